@@ -1,0 +1,64 @@
+"""2-D convolution with implicit-GEMM HIP kernels on the GPU path.
+
+SURVEY.md §2b N4: ResNet-18/50's 3x3 and 1x1 convs in bf16 as
+implicit-GEMM MFMA kernels (fwd, dgrad, wgrad) in csrc/conv_bf16.hip.
+CPU path = F.conv2d (oracle).
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from . import native_enabled, require_ext
+
+
+def _use_native(*ts: torch.Tensor) -> bool:
+    return ts[0].is_cuda and native_enabled() and all(
+        t.dtype == torch.bfloat16 for t in ts
+    )
+
+
+class _Conv2dFn(torch.autograd.Function):
+    generate_vmap_rule = True
+
+    @staticmethod
+    def forward(x, w, stride: int, padding: int):
+        if _use_native(x, w):
+            return require_ext().conv2d_fwd(
+                x.contiguous(), w.contiguous(), stride, padding
+            )
+        return F.conv2d(x, w, None, stride, padding)
+
+    @staticmethod
+    def setup_context(ctx, inputs, output):
+        x, w, stride, padding = inputs
+        ctx.save_for_backward(x, w)
+        ctx.stride, ctx.padding = stride, padding
+
+    @staticmethod
+    def backward(ctx, gy):
+        x, w = ctx.saved_tensors
+        s, p = ctx.stride, ctx.padding
+        gy = gy.contiguous()
+        if _use_native(gy, w, x):
+            m = require_ext()
+            gx = m.conv2d_dgrad(gy, w.contiguous(), x.shape[2], x.shape[3],
+                                s, p)
+            gw = m.conv2d_wgrad(gy, x.contiguous(), w.shape[2], w.shape[3],
+                                s, p)
+        else:
+            gx = torch.nn.grad.conv2d_input(x.shape, w, gy, s, p)
+            gw = torch.nn.grad.conv2d_weight(x, w.shape, gy, s, p)
+        return gx, gw, None, None
+
+
+def colearn_conv2d(
+    x: torch.Tensor, w: torch.Tensor, stride: int = 1, padding: int = 0
+) -> torch.Tensor:
+    if torch.is_autocast_enabled("cuda") and x.is_cuda:
+        dt = torch.get_autocast_dtype("cuda")
+        x, w = x.to(dt), w.to(dt)
+    return _Conv2dFn.apply(x, w, stride, padding)

@@ -1,0 +1,175 @@
+"""BatchNorm2d and LayerNorm with fused HIP kernels on the GPU path.
+
+SURVEY.md §2b N6 (BatchNorm, ResNet path) and N7 (LayerNorm, BERT path).
+Both are memory-bound: the HIP kernels (csrc/norms.hip) do one fused sweep
+per pass with vectorized bf16 loads (G13) and wave-level reductions; stats
+accumulate in fp32.  CPU path = stock PyTorch math (oracle).
+
+Autocast note: like PyTorch's own norms, statistics are computed in fp32;
+inputs may be bf16, parameters stay fp32.
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+
+from . import native_enabled, require_ext
+
+
+def _native(x: torch.Tensor) -> bool:
+    return x.is_cuda and native_enabled() and x.dtype == torch.bfloat16
+
+
+class _BatchNormFn(torch.autograd.Function):
+    generate_vmap_rule = True
+
+    @staticmethod
+    def forward(x, weight, bias, running_mean, running_var,
+                training: bool, momentum: float, eps: float):
+        if _native(x):
+            m = require_ext()
+            y, save_mean, save_invstd = m.batchnorm_fwd(
+                x.contiguous(), weight, bias, running_mean, running_var,
+                training, momentum, eps,
+            )
+            return y, save_mean, save_invstd
+        if training:
+            dims = (0, 2, 3)
+            xf = x.float()
+            mean = xf.mean(dim=dims)
+            var = xf.var(dim=dims, unbiased=False)
+            n = x.numel() / x.shape[1]
+            with torch.no_grad():
+                running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
+                running_var.mul_(1 - momentum).add_(
+                    var * n / max(n - 1, 1), alpha=momentum
+                )
+            invstd = (var + eps).rsqrt()
+        else:
+            # clone: an input returned as an output may not be saved for
+            # backward under the setup_context protocol
+            mean = running_mean.clone()
+            invstd = (running_var + eps).rsqrt()
+            xf = x.float()
+        y = ((xf - mean[None, :, None, None])
+             * invstd[None, :, None, None]
+             * weight[None, :, None, None]
+             + bias[None, :, None, None]).to(x.dtype)
+        return y, mean, invstd
+
+    @staticmethod
+    def setup_context(ctx, inputs, output):
+        x, weight = inputs[0], inputs[1]
+        _, save_mean, save_invstd = output
+        ctx.save_for_backward(x, weight, save_mean, save_invstd)
+        ctx.training = inputs[5]
+        ctx.mark_non_differentiable(output[1], output[2])
+
+    @staticmethod
+    def backward(ctx, gy, _gmean, _ginvstd):
+        x, weight, mean, invstd = ctx.saved_tensors
+        gy = gy.contiguous()
+        if _native(x):
+            gx, gw, gb = require_ext().batchnorm_bwd(
+                gy, x.contiguous(), weight, mean, invstd, ctx.training
+            )
+        else:
+            xf, gyf = x.float(), gy.float()
+            xhat = (xf - mean[None, :, None, None]) * invstd[None, :, None, None]
+            n = x.numel() / x.shape[1]
+            gb = gyf.sum(dim=(0, 2, 3))
+            gw = (gyf * xhat).sum(dim=(0, 2, 3))
+            if ctx.training:
+                gx = (weight * invstd)[None, :, None, None] * (
+                    gyf - gb[None, :, None, None] / n
+                    - xhat * gw[None, :, None, None] / n
+                )
+            else:
+                gx = (weight * invstd)[None, :, None, None] * gyf
+            gx = gx.to(x.dtype)
+        return gx, gw, gb, None, None, None, None, None
+
+
+class ColearnBatchNorm2d(nn.Module):
+    def __init__(self, num_features: int, eps: float = 1e-5,
+                 momentum: float = 0.1) -> None:
+        super().__init__()
+        self.num_features = num_features
+        self.eps = eps
+        self.momentum = momentum
+        self.weight = nn.Parameter(torch.ones(num_features))
+        self.bias = nn.Parameter(torch.zeros(num_features))
+        self.register_buffer("running_mean", torch.zeros(num_features))
+        self.register_buffer("running_var", torch.ones(num_features))
+        self.register_buffer("num_batches_tracked",
+                             torch.tensor(0, dtype=torch.long))
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if self.training:
+            self.num_batches_tracked += 1
+        y, _, _ = _BatchNormFn.apply(
+            x, self.weight, self.bias, self.running_mean, self.running_var,
+            self.training, self.momentum, self.eps,
+        )
+        return y
+
+
+class _LayerNormFn(torch.autograd.Function):
+    generate_vmap_rule = True
+
+    @staticmethod
+    def forward(x, weight, bias, eps: float):
+        if _native(x):
+            y, mean, invstd = require_ext().layernorm_fwd(
+                x.contiguous(), weight, bias, eps
+            )
+            return y, mean, invstd
+        xf = x.float()
+        mean = xf.mean(dim=-1)
+        var = xf.var(dim=-1, unbiased=False)
+        invstd = (var + eps).rsqrt()
+        y = ((xf - mean[..., None]) * invstd[..., None] * weight
+             + bias).to(x.dtype)
+        return y, mean, invstd
+
+    @staticmethod
+    def setup_context(ctx, inputs, output):
+        x, weight = inputs[0], inputs[1]
+        _, mean, invstd = output
+        ctx.save_for_backward(x, weight, mean, invstd)
+        ctx.mark_non_differentiable(output[1], output[2])
+
+    @staticmethod
+    def backward(ctx, gy, _gmean, _ginvstd):
+        x, weight, mean, invstd = ctx.saved_tensors
+        gy = gy.contiguous()
+        if _native(x):
+            gx, gw, gb = require_ext().layernorm_bwd(
+                gy, x.contiguous(), weight, mean, invstd
+            )
+        else:
+            xf, gyf = x.float(), gy.float()
+            xhat = (xf - mean[..., None]) * invstd[..., None]
+            h = x.shape[-1]
+            gw = (gyf * xhat).reshape(-1, h).sum(dim=0)
+            gb = gyf.reshape(-1, h).sum(dim=0)
+            wg = gyf * weight
+            gx = invstd[..., None] * (
+                wg - wg.mean(dim=-1, keepdim=True)
+                - xhat * (wg * xhat).mean(dim=-1, keepdim=True)
+            )
+            gx = gx.to(x.dtype)
+        return gx, gw, gb, None
+
+
+class ColearnLayerNorm(nn.Module):
+    def __init__(self, hidden: int, eps: float = 1e-12) -> None:
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(hidden))
+        self.bias = nn.Parameter(torch.zeros(hidden))
+        self.eps = eps
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        y, _, _ = _LayerNormFn.apply(x, self.weight, self.bias, self.eps)
+        return y

@@ -1,0 +1,91 @@
+"""Flat parameter/gradient storage: the memory layout of the whole engine.
+
+MI355X-first design decision (no analog in the reference, which ships whole
+state dicts over websockets): every model trains out of ONE contiguous fp32
+parameter buffer and ONE contiguous fp32 gradient buffer.
+
+* ``nn.Parameter`` data are re-pointed to views of the flat buffer, so the
+  forward/backward graph reads/writes it directly,
+* ``p.grad`` is pre-bound to the matching view of the flat grad buffer, so
+  autograd accumulates into it in place (no gather pass),
+* the fused SGD/Adam HIP kernel (ops.fused_optim) is ONE launch over the
+  two buffers,
+* client deltas, FedProx reference weights, DP noise and the bucketed RCCL
+  all-reduce all operate on the same flat layout with zero repacking,
+* with 288 GB HBM3E per GPU the whole-model flat buffers (ResNet-50:
+  ~100 MB fp32; BERT-base: ~440 MB) are trivially resident; we keep THREE
+  copies (live weights, round-start global weights, delta accumulator).
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Tuple
+
+import torch
+import torch.nn as nn
+
+
+class FlatParams:
+    """Flatten a model's parameters into one fp32 buffer (in place)."""
+
+    def __init__(self, model: nn.Module, device: torch.device) -> None:
+        self.model = model
+        self.device = device
+        params: List[Tuple[str, nn.Parameter]] = [
+            (n, p) for n, p in model.named_parameters()
+        ]
+        self.names = [n for n, _ in params]
+        self.shapes = [p.shape for _, p in params]
+        self.numels = [p.numel() for _, p in params]
+        total = sum(self.numels)
+        self.numel = total
+        self.flat_w = torch.empty(total, dtype=torch.float32, device=device)
+        self.flat_g = torch.zeros(total, dtype=torch.float32, device=device)
+        # copy initial values then alias the parameters into the flat buffer
+        offset = 0
+        self._views: Dict[str, torch.Tensor] = {}
+        for (name, p), n in zip(params, self.numels):
+            view = self.flat_w[offset : offset + n].view(p.shape)
+            view.copy_(p.detach().to(device=device, dtype=torch.float32))
+            p.data = view
+            gview = self.flat_g[offset : offset + n].view(p.shape)
+            p.grad = gview
+            self._views[name] = view
+            offset += n
+
+    # ------------------------------------------------------------------
+    def grad_views_rebind(self) -> None:
+        """Re-bind p.grad views (autograd can replace .grad if zeroed to
+        None somewhere); call before backward if unsure."""
+        offset = 0
+        for p, n in zip(self.model.parameters(), self.numels):
+            if p.grad is None or p.grad.data_ptr() != self.flat_g[offset:offset + n].data_ptr():
+                p.grad = self.flat_g[offset : offset + n].view(p.shape)
+            offset += n
+
+    def view_of(self, name: str) -> torch.Tensor:
+        return self._views[name]
+
+    def clone_weights(self) -> torch.Tensor:
+        return self.flat_w.detach().clone()
+
+    def load_flat(self, flat: torch.Tensor) -> None:
+        self.flat_w.copy_(flat)
+
+    def state_dict_from_flat(self, flat: torch.Tensor) -> Dict[str, torch.Tensor]:
+        out = {}
+        offset = 0
+        for name, shape, n in zip(self.names, self.shapes, self.numels):
+            out[name] = flat[offset : offset + n].view(shape).clone()
+            offset += n
+        return out
+
+    def flat_from_state_dict(self, sd: Dict[str, torch.Tensor]) -> torch.Tensor:
+        flat = torch.empty(self.numel, dtype=torch.float32, device=self.device)
+        offset = 0
+        for name, n in zip(self.names, self.numels):
+            flat[offset : offset + n].copy_(
+                sd[name].reshape(-1).to(device=self.device, dtype=torch.float32)
+            )
+            offset += n
+        return flat

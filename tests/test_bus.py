@@ -1,0 +1,45 @@
+from colearn_federated_learning_amd.federated.bus import MessageBus, Topics
+
+
+def test_publish_subscribe():
+    bus = MessageBus()
+    got = []
+    bus.subscribe("a/b", lambda t, p: got.append((t, p)))
+    bus.publish("a/b", 1)
+    bus.publish("a/c", 2)
+    assert got == [("a/b", 1)]
+
+
+def test_wildcard_plus():
+    bus = MessageBus()
+    got = []
+    bus.subscribe("colearn/role/+", lambda t, p: got.append(t))
+    bus.publish("colearn/role/3", {"role": "trainer"})
+    bus.publish("colearn/role/3/extra", {})
+    assert got == ["colearn/role/3"]
+
+
+def test_wildcard_hash():
+    bus = MessageBus()
+    got = []
+    bus.subscribe("colearn/#", lambda t, p: got.append(t))
+    bus.publish("colearn/availability", {})
+    bus.publish("colearn/role/1", {})
+    bus.publish("other/x", {})
+    assert got == ["colearn/availability", "colearn/role/1"]
+
+
+def test_retained_delivered_on_subscribe():
+    bus = MessageBus()
+    bus.publish(Topics.AVAILABILITY, {"client_id": 1}, retain=True)
+    got = []
+    bus.subscribe(Topics.AVAILABILITY, lambda t, p: got.append(p))
+    assert got == [{"client_id": 1}]
+
+
+def test_collect():
+    bus = MessageBus()
+    bus.publish("colearn/availability", {"client_id": 0}, retain=True)
+    bus.publish("colearn/metrics/0", {"x": 1}, retain=True)
+    avail = bus.collect("colearn/availability")
+    assert list(avail.values()) == [{"client_id": 0}]

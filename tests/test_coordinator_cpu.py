@@ -1,0 +1,77 @@
+"""Driver config #1 end-to-end (CPU plumbing) + control-plane semantics."""
+
+import torch
+
+from colearn_federated_learning_amd import (
+    FederatedServer,
+    MessageBus,
+    driver_config,
+    run,
+)
+from colearn_federated_learning_amd.federated.bus import Topics
+
+
+def test_config1_end_to_end_loss_decreases():
+    cfg = driver_config(1).override(rounds=4, num_evaluators=0)
+    hist = run(cfg)
+    assert len(hist) == 4
+    assert hist[-1]["mean_client_loss"] < hist[0]["mean_client_loss"]
+
+
+def test_evaluation_accuracy_improves():
+    cfg = driver_config(1).override(rounds=4, num_clients=3,
+                                    num_evaluators=1)
+    hist = run(cfg)
+    assert hist[-1]["eval_accuracy"] > 0.5  # well above 10% chance
+
+
+def test_control_plane_messages():
+    cfg = driver_config(1).override(rounds=1, num_clients=2)
+    bus = MessageBus()
+    events = []
+    bus.subscribe("colearn/#", lambda t, p: events.append((t, p)))
+    server = FederatedServer(cfg, bus=bus)
+    server.train()
+    topics = [t for t, _ in events]
+    # availability announced by both clients
+    assert topics.count(Topics.AVAILABILITY) == 2
+    # roles assigned
+    assert any(t.startswith("colearn/role/") for t in topics)
+    # round + session lifecycle
+    assert Topics.ROUND in topics
+    sessions = [p for t, p in events if t == Topics.SESSION]
+    assert sessions[0]["state"] == "start"
+    assert sessions[-1]["state"] == "stop"
+    # per-client results reported
+    assert any(t.startswith("colearn/result/") for t in topics)
+
+
+def test_metrics_jsonl(tmp_path):
+    p = tmp_path / "metrics.jsonl"
+    cfg = driver_config(1).override(rounds=2, metrics_path=str(p))
+    run(cfg)
+    import json
+
+    lines = [json.loads(line) for line in p.read_text().splitlines()]
+    assert len(lines) == 2
+    assert {"round", "samples_per_sec", "mean_client_loss",
+            "time_train_s", "time_aggregate_s"} <= set(lines[0])
+
+
+def test_clients_per_round_subset():
+    cfg = driver_config(1).override(rounds=2, num_clients=4,
+                                    clients_per_round=2)
+    bus = MessageBus()
+    server = FederatedServer(cfg, bus=bus)
+    rec = server.run_round(0)
+    assert len(rec["client_losses"]) == 2
+
+
+def test_fedprox_changes_update():
+    base = driver_config(1).override(rounds=1)
+    prox = driver_config(1).override(rounds=1, train__fedprox_mu=0.5)
+    s1 = FederatedServer(base)
+    s2 = FederatedServer(prox)
+    s1.run_round(0)
+    s2.run_round(0)
+    assert not torch.allclose(s1.global_flat, s2.global_flat)
