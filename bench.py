@@ -1,0 +1,118 @@
+"""Benchmark harness (driver contract).
+
+Flagship: the BASELINE.json headline config — ResNet-18 CIFAR-10-shape
+FedAvg, 8 IID clients, bf16, synthetic data, random-init weights.  One
+bench "step" is one federated ROUND: every one of the 8 clients runs
+local SGD over its shard (local_epochs=1) on its owning GPU, then the
+round aggregates via the weighted RCCL all-reduce over xGMI.  The metric
+is whole-node samples/sec: (samples trained per round x rounds) / elapsed.
+
+Scaling is STRONG: the 8 clients are fixed (that is the named config) and
+are multiplexed over however many GPUs the driver gives us — 8 clients on
+1 GPU at N=1 down to 1 client per GPU at N=8.
+
+Launch:  python bench.py --gpus N --steps K --warmup W
+(for N>1 the driver launches via torch.distributed.run, one rank per GPU.)
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=5)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--config", type=int, default=2,
+                    help="BASELINE.json driver config index (default: "
+                         "flagship ResNet-18 FedAvg)")
+    ap.add_argument("--samples-per-client", type=int, default=2048)
+    args = ap.parse_args()
+
+    from colearn_federated_learning_amd.config import driver_config
+    from colearn_federated_learning_amd.federated.coordinator import (
+        FederatedServer,
+    )
+    from colearn_federated_learning_amd.utils.dist import init_distributed
+
+    cfg = driver_config(args.config)
+    on_gpu = torch.cuda.is_available()
+    cfg = cfg.override(
+        rounds=args.warmup + args.steps,
+        data__samples_per_client=args.samples_per_client,
+        train__dtype="bfloat16" if on_gpu else "float32",
+        num_evaluators=0,
+        checkpoint_dir=None,
+        metrics_path=None,
+    )
+    rank, world, device = init_distributed(cfg.agg.backend)
+    server = FederatedServer(cfg, rank=rank, world=world, device=device)
+
+    def barrier_sync():
+        if dist.is_initialized():
+            dist.barrier()
+        if device.type == "cuda":
+            torch.cuda.synchronize(device)
+
+    for r in range(args.warmup):
+        server.run_round(r)
+    barrier_sync()
+    t0 = time.perf_counter()
+    total_samples = 0.0
+    for r in range(args.warmup, args.warmup + args.steps):
+        rec = server.run_round(r)
+        total_samples += rec["n_samples"]
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX elapsed over ranks (the job is as slow as its slowest rank)
+    if dist.is_initialized():
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        if dist.get_backend() != "gloo":
+            t = t.to(device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank == 0:
+        value = total_samples / elapsed
+        result = {
+            "metric": "samples/sec (whole node), ResNet-18 FedAvg 8 clients",
+            "value": value,
+            "unit": "samples/sec",
+            "n_gpus": args.gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "bf16" if on_gpu else "float32",
+            "data": "synthetic",
+            "config": {
+                "model": cfg.model.name,
+                "global_batch": cfg.train.batch_size * cfg.num_clients,
+                "local_batch": cfg.train.batch_size,
+                "num_clients": cfg.num_clients,
+                "samples_per_client": cfg.data.samples_per_client,
+                "image": f"{cfg.model.in_channels}x{cfg.model.image_size}"
+                         f"x{cfg.model.image_size}",
+                "parallelism": f"federated dp{args.gpus}",
+                "local_epochs": cfg.train.local_epochs,
+            },
+        }
+        print(json.dumps(result))
+
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

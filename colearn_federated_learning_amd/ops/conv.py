@@ -27,9 +27,8 @@ class _Conv2dFn(torch.autograd.Function):
     @staticmethod
     def forward(x, w, stride: int, padding: int):
         if _use_native(x, w):
-            return require_ext().conv2d_fwd(
-                x.contiguous(), w.contiguous(), stride, padding
-            )
+            # layout normalization (channels-last) happens in the binding
+            return require_ext().conv2d_fwd(x, w, stride, padding)
         return F.conv2d(x, w, None, stride, padding)
 
     @staticmethod
@@ -42,13 +41,10 @@ class _Conv2dFn(torch.autograd.Function):
     def backward(ctx, gy):
         x, w = ctx.saved_tensors
         s, p = ctx.stride, ctx.padding
-        gy = gy.contiguous()
         if _use_native(gy, w, x):
             m = require_ext()
-            gx = m.conv2d_dgrad(gy, w.contiguous(), x.shape[2], x.shape[3],
-                                s, p)
-            gw = m.conv2d_wgrad(gy, x.contiguous(), w.shape[2], w.shape[3],
-                                s, p)
+            gx = m.conv2d_dgrad(gy, w, x.shape[2], x.shape[3], s, p)
+            gw = m.conv2d_wgrad(gy, x, w.shape[2], w.shape[3], s, p)
         else:
             gx = torch.nn.grad.conv2d_input(x.shape, w, gy, s, p)
             gw = torch.nn.grad.conv2d_weight(x, w.shape, gy, s, p)

@@ -1,0 +1,60 @@
+// Host-visible launcher API shared by bindings.cpp (compiled with g++) and
+// the .hip translation units (compiled with hipcc).  bf16 pointers cross
+// this boundary as void*.
+#pragma once
+
+#include <hip/hip_runtime_api.h>
+
+#include <cstdint>
+
+namespace colearn {
+
+struct ConvShape {
+  int N, H, W, C;  // input
+  int K, R, S;     // out channels, filter
+  int P, Q;        // output spatial
+  int stride, pad;
+};
+
+void launch_fused_sgd(float* w, const float* g, float* m, const float* ref,
+                      int64_t n, float lr, float momentum, float wd,
+                      float mu, bool has_mom, bool has_prox,
+                      hipStream_t stream);
+void launch_fused_adam(float* w, const float* g, float* m, float* v,
+                       const float* ref, int64_t n, float lr, float b1,
+                       float b2, float eps, float wd, float mu, float bc1,
+                       float bc2, bool has_prox, hipStream_t stream);
+void launch_sq_norm_per_row(const float* g, float* sq, int64_t rows,
+                            int64_t cols, hipStream_t stream);
+void launch_scale_rows_mean(const float* g, const float* scale, float* out,
+                            int64_t rows, int64_t cols, hipStream_t stream);
+void launch_topk(const float* v, int64_t n, int64_t k, int* idx, float* val,
+                 unsigned* workspace, hipStream_t stream);
+void launch_scatter_add_pairs(float* out, const int* idx, const float* val,
+                              int64_t k, hipStream_t stream);
+void launch_layernorm_fwd(const void* x, const float* w, const float* b,
+                          void* y, float* mean, float* invstd, int64_t rows,
+                          int H, float eps, hipStream_t stream);
+void launch_layernorm_bwd(const void* gy, const void* x, const float* w,
+                          const float* mean, const float* invstd, void* gx,
+                          float* gw, float* gb, int64_t rows, int H,
+                          hipStream_t stream);
+void launch_batchnorm_fwd(const void* x, const float* w, const float* b,
+                          float* running_mean, float* running_var, void* y,
+                          float* mean, float* invstd, float* sums, int64_t R,
+                          int C, float momentum, float eps, bool training,
+                          hipStream_t stream);
+void launch_batchnorm_bwd(const void* gy, const void* x, const float* w,
+                          const float* mean, const float* invstd, void* gx,
+                          float* acc, int64_t R, int C, bool training,
+                          hipStream_t stream);
+void launch_gemm_nt(const void* A, const void* B, void* C, int64_t batch,
+                    int64_t M, int64_t N, int64_t K, hipStream_t stream);
+void launch_conv_fwd(const void* x, const void* w, void* y, ConvShape cs,
+                     hipStream_t stream);
+void launch_conv_dgrad(const void* gy, const void* wt, void* gx,
+                       ConvShape cs, hipStream_t stream);
+void launch_conv_wgrad(const void* gy, const void* x, float* gwf, void* gw,
+                       ConvShape cs, hipStream_t stream);
+
+}  // namespace colearn

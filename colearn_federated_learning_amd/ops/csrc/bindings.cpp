@@ -1,0 +1,379 @@
+// Python bindings for the colearn gfx950 kernels.
+//
+// Layout policy (MI355X-first, no CUDA-compat):
+//  * GEMM family: bf16, row-major; NN/TN flavors route through the NT
+//    kernel with an explicit transpose pass (v1 — see gemm_bf16.hip).
+//  * conv family: channels-last (NHWC) activations; weights re-packed
+//    K-major per call ([CO][R][S][CI] fwd, [CI][R][S][CO] dgrad).
+//  * norms: LayerNorm on [.., H] bf16; BatchNorm on channels-last bf16.
+//  * optimizers / DP / top-k: flat fp32 buffers.
+
+#include <torch/extension.h>
+
+#include <c10/hip/HIPGuard.h>
+#include <c10/hip/HIPStream.h>
+
+#include "api.h"
+
+#define HIP_CHECK_LAST()                                          \
+  do {                                                            \
+    hipError_t e_ = hipGetLastError();                            \
+    TORCH_CHECK(e_ == hipSuccess, "HIP kernel launch failed: ",   \
+                hipGetErrorString(e_));                           \
+  } while (0)
+
+namespace colearn {
+
+
+namespace {
+
+hipStream_t stream() { return c10::hip::getCurrentHIPStream().stream(); }
+
+void check_f32(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kFloat32, name,
+              " must be a float32 GPU tensor");
+}
+void check_bf16(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kBFloat16, name,
+              " must be a bfloat16 GPU tensor");
+}
+
+// ---- optimizers -----------------------------------------------------------
+void fused_sgd(torch::Tensor w, torch::Tensor g, torch::Tensor m,
+               torch::Tensor ref, double lr, double momentum, double wd,
+               double mu) {
+  check_f32(w, "w");
+  check_f32(g, "g");
+  const c10::hip::HIPGuard guard(w.device());
+  const bool has_mom = m.numel() > 0;
+  const bool has_prox = ref.numel() > 0;
+  launch_fused_sgd(w.data_ptr<float>(), g.data_ptr<float>(),
+                   has_mom ? m.data_ptr<float>() : nullptr,
+                   has_prox ? ref.data_ptr<float>() : nullptr, w.numel(),
+                   static_cast<float>(lr), static_cast<float>(momentum),
+                   static_cast<float>(wd), static_cast<float>(mu), has_mom,
+                   has_prox, stream());
+  HIP_CHECK_LAST();
+}
+
+void fused_adam(torch::Tensor w, torch::Tensor g, torch::Tensor m,
+                torch::Tensor v, torch::Tensor ref, double lr, double b1,
+                double b2, double eps, double wd, double mu, double bc1,
+                double bc2) {
+  check_f32(w, "w");
+  const c10::hip::HIPGuard guard(w.device());
+  const bool has_prox = ref.numel() > 0;
+  launch_fused_adam(
+      w.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
+      v.data_ptr<float>(), has_prox ? ref.data_ptr<float>() : nullptr,
+      w.numel(), static_cast<float>(lr), static_cast<float>(b1),
+      static_cast<float>(b2), static_cast<float>(eps),
+      static_cast<float>(wd), static_cast<float>(mu),
+      static_cast<float>(bc1), static_cast<float>(bc2), has_prox, stream());
+  HIP_CHECK_LAST();
+}
+
+// ---- DP / top-k -----------------------------------------------------------
+void sq_norm_per_row(torch::Tensor g, torch::Tensor sq) {
+  check_f32(g, "g");
+  check_f32(sq, "sq");
+  const c10::hip::HIPGuard guard(g.device());
+  auto gc = g.contiguous();
+  launch_sq_norm_per_row(gc.data_ptr<float>(), sq.data_ptr<float>(),
+                         gc.size(0), gc.size(1), stream());
+  HIP_CHECK_LAST();
+}
+
+void scale_rows_mean(torch::Tensor g, torch::Tensor scale,
+                     torch::Tensor out) {
+  check_f32(g, "g");
+  const c10::hip::HIPGuard guard(g.device());
+  auto gc = g.contiguous();
+  launch_scale_rows_mean(gc.data_ptr<float>(), scale.data_ptr<float>(),
+                         out.data_ptr<float>(), gc.size(0), gc.size(1),
+                         stream());
+  HIP_CHECK_LAST();
+}
+
+std::tuple<torch::Tensor, torch::Tensor> topk_compress(torch::Tensor flat,
+                                                       int64_t k) {
+  check_f32(flat, "flat");
+  const c10::hip::HIPGuard guard(flat.device());
+  auto fc = flat.contiguous();
+  k = std::min<int64_t>(k, fc.numel());
+  auto idx = torch::empty({k}, fc.options().dtype(torch::kInt32));
+  auto val = torch::empty({k}, fc.options());
+  auto ws = torch::zeros({520}, fc.options().dtype(torch::kInt32));
+  launch_topk(fc.data_ptr<float>(), fc.numel(), k, idx.data_ptr<int>(),
+              val.data_ptr<float>(),
+              reinterpret_cast<unsigned*>(ws.data_ptr<int>()), stream());
+  HIP_CHECK_LAST();
+  return {idx, val};
+}
+
+void scatter_add_pairs(torch::Tensor out, torch::Tensor idx,
+                       torch::Tensor val) {
+  check_f32(out, "out");
+  const c10::hip::HIPGuard guard(out.device());
+  launch_scatter_add_pairs(out.data_ptr<float>(), idx.data_ptr<int>(),
+                           val.data_ptr<float>(), idx.numel(), stream());
+  HIP_CHECK_LAST();
+}
+
+// ---- LayerNorm ------------------------------------------------------------
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> layernorm_fwd(
+    torch::Tensor x, torch::Tensor w, torch::Tensor b, double eps) {
+  check_bf16(x, "x");
+  const c10::hip::HIPGuard guard(x.device());
+  auto xc = x.contiguous();
+  const int H = static_cast<int>(x.size(-1));
+  const int64_t rows = x.numel() / H;
+  auto y = torch::empty_like(xc);
+  auto mean = torch::empty({rows}, x.options().dtype(torch::kFloat32));
+  auto invstd = torch::empty_like(mean);
+  launch_layernorm_fwd(
+      xc.data_ptr(), w.data_ptr<float>(),
+      b.data_ptr<float>(), y.data_ptr(),
+      mean.data_ptr<float>(), invstd.data_ptr<float>(), rows, H,
+      static_cast<float>(eps), stream());
+  HIP_CHECK_LAST();
+  auto stat_shape = x.sizes().vec();
+  stat_shape.pop_back();
+  return {y, mean.view(stat_shape), invstd.view(stat_shape)};
+}
+
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> layernorm_bwd(
+    torch::Tensor gy, torch::Tensor x, torch::Tensor w, torch::Tensor mean,
+    torch::Tensor invstd) {
+  check_bf16(gy, "gy");
+  const c10::hip::HIPGuard guard(x.device());
+  auto gc = gy.contiguous(), xc = x.contiguous();
+  auto mc = mean.contiguous(), ic = invstd.contiguous();
+  const int H = static_cast<int>(x.size(-1));
+  const int64_t rows = x.numel() / H;
+  auto gx = torch::empty_like(xc);
+  auto gw = torch::zeros({H}, x.options().dtype(torch::kFloat32));
+  auto gb = torch::zeros_like(gw);
+  launch_layernorm_bwd(
+      gc.data_ptr(),
+      xc.data_ptr(), w.data_ptr<float>(),
+      mc.data_ptr<float>(), ic.data_ptr<float>(),
+      gx.data_ptr(), gw.data_ptr<float>(),
+      gb.data_ptr<float>(), rows, H, stream());
+  HIP_CHECK_LAST();
+  return {gx, gw, gb};
+}
+
+// ---- BatchNorm (channels-last) -------------------------------------------
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
+    torch::Tensor x, torch::Tensor w, torch::Tensor b,
+    torch::Tensor running_mean, torch::Tensor running_var, bool training,
+    double momentum, double eps) {
+  check_bf16(x, "x");
+  TORCH_CHECK(x.dim() == 4, "batchnorm expects NCHW tensor");
+  const int C = static_cast<int>(x.size(1));
+  TORCH_CHECK(C % 8 == 0, "native batchnorm needs C % 8 == 0, got ", C);
+  const c10::hip::HIPGuard guard(x.device());
+  auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
+  const int64_t R = x.numel() / C;
+  auto y = torch::empty_like(xc);
+  auto mean = torch::empty({C}, x.options().dtype(torch::kFloat32));
+  auto invstd = torch::empty_like(mean);
+  auto sums = torch::zeros({2 * C}, mean.options());
+  launch_batchnorm_fwd(
+      xc.data_ptr(), w.data_ptr<float>(),
+      b.data_ptr<float>(), running_mean.data_ptr<float>(),
+      running_var.data_ptr<float>(), y.data_ptr(),
+      mean.data_ptr<float>(), invstd.data_ptr<float>(),
+      sums.data_ptr<float>(), R, C, static_cast<float>(momentum),
+      static_cast<float>(eps), training, stream());
+  HIP_CHECK_LAST();
+  return {y, mean, invstd};
+}
+
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
+    torch::Tensor gy, torch::Tensor x, torch::Tensor w, torch::Tensor mean,
+    torch::Tensor invstd, bool training) {
+  check_bf16(gy, "gy");
+  const int C = static_cast<int>(x.size(1));
+  const c10::hip::HIPGuard guard(x.device());
+  auto gc = gy.contiguous(torch::MemoryFormat::ChannelsLast);
+  auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
+  const int64_t R = x.numel() / C;
+  auto gx = torch::empty_like(xc);
+  auto acc = torch::zeros({2 * C}, x.options().dtype(torch::kFloat32));
+  launch_batchnorm_bwd(gc.data_ptr(),
+                       xc.data_ptr(),
+                       w.data_ptr<float>(), mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(),
+                       gx.data_ptr(),
+                       acc.data_ptr<float>(), R, C, training, stream());
+  HIP_CHECK_LAST();
+  auto gb = acc.narrow(0, 0, C).clone();
+  auto gw = acc.narrow(0, C, C).clone();
+  return {gx, gw, gb};
+}
+
+// ---- GEMM family ----------------------------------------------------------
+torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b) {
+  check_bf16(a, "a");
+  check_bf16(b, "b");
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(1),
+              "gemm_nt shape mismatch");
+  const c10::hip::HIPGuard guard(a.device());
+  auto ac = a.contiguous(), bc = b.contiguous();
+  const int64_t M = a.size(0), N = b.size(0), K = a.size(1);
+  auto c = torch::empty({M, N}, a.options());
+  launch_gemm_nt(ac.data_ptr(), bc.data_ptr(), c.data_ptr(), 1, M, N, K,
+                 stream());
+  HIP_CHECK_LAST();
+  return c;
+}
+
+torch::Tensor gemm_nn(torch::Tensor a, torch::Tensor b) {
+  return gemm_nt(a, b.transpose(0, 1).contiguous());
+}
+
+torch::Tensor gemm_tn(torch::Tensor a, torch::Tensor b) {
+  return gemm_nt(a.transpose(0, 1).contiguous(),
+                 b.transpose(0, 1).contiguous());
+}
+
+torch::Tensor bmm_nt(torch::Tensor a, torch::Tensor b) {
+  check_bf16(a, "a");
+  TORCH_CHECK(a.dim() == 3 && b.dim() == 3, "bmm expects 3-D");
+  const c10::hip::HIPGuard guard(a.device());
+  auto ac = a.contiguous(), bc = b.contiguous();
+  const int64_t B = a.size(0), M = a.size(1), N = b.size(1), K = a.size(2);
+  auto c = torch::empty({B, M, N}, a.options());
+  launch_gemm_nt(ac.data_ptr(), bc.data_ptr(), c.data_ptr(), B, M, N, K,
+                 stream());
+  HIP_CHECK_LAST();
+  return c;
+}
+
+torch::Tensor bmm_nn(torch::Tensor a, torch::Tensor b) {
+  return bmm_nt(a, b.transpose(1, 2).contiguous());
+}
+
+torch::Tensor bmm_tn(torch::Tensor a, torch::Tensor b) {
+  return bmm_nt(a.transpose(1, 2).contiguous(),
+                b.transpose(1, 2).contiguous());
+}
+
+// ---- conv family ----------------------------------------------------------
+ConvShape make_shape(const torch::Tensor& x, const torch::Tensor& w,
+                     int64_t stride, int64_t pad) {
+  ConvShape cs;
+  cs.N = static_cast<int>(x.size(0));
+  cs.C = static_cast<int>(x.size(1));
+  cs.H = static_cast<int>(x.size(2));
+  cs.W = static_cast<int>(x.size(3));
+  cs.K = static_cast<int>(w.size(0));
+  cs.R = static_cast<int>(w.size(2));
+  cs.S = static_cast<int>(w.size(3));
+  cs.stride = static_cast<int>(stride);
+  cs.pad = static_cast<int>(pad);
+  cs.P = (cs.H + 2 * cs.pad - cs.R) / cs.stride + 1;
+  cs.Q = (cs.W + 2 * cs.pad - cs.S) / cs.stride + 1;
+  return cs;
+}
+
+torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
+                         int64_t pad) {
+  check_bf16(x, "x");
+  check_bf16(w, "w");
+  const c10::hip::HIPGuard guard(x.device());
+  ConvShape cs = make_shape(x, w, stride, pad);
+  auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
+  // repack weight K-major: [CO][R][S][CI]
+  auto wp = w.permute({0, 2, 3, 1}).contiguous();
+  auto y = torch::empty({cs.N, cs.K, cs.P, cs.Q},
+                        x.options().memory_format(
+                            torch::MemoryFormat::ChannelsLast));
+  launch_conv_fwd(xc.data_ptr(), wp.data_ptr(), y.data_ptr(), cs, stream());
+  HIP_CHECK_LAST();
+  return y;
+}
+
+torch::Tensor conv2d_dgrad(torch::Tensor gy, torch::Tensor w, int64_t H,
+                           int64_t W, int64_t stride, int64_t pad) {
+  check_bf16(gy, "gy");
+  const c10::hip::HIPGuard guard(gy.device());
+  ConvShape cs;
+  cs.N = static_cast<int>(gy.size(0));
+  cs.K = static_cast<int>(gy.size(1));
+  cs.P = static_cast<int>(gy.size(2));
+  cs.Q = static_cast<int>(gy.size(3));
+  cs.C = static_cast<int>(w.size(1));
+  cs.R = static_cast<int>(w.size(2));
+  cs.S = static_cast<int>(w.size(3));
+  cs.H = static_cast<int>(H);
+  cs.W = static_cast<int>(W);
+  cs.stride = static_cast<int>(stride);
+  cs.pad = static_cast<int>(pad);
+  auto gc = gy.contiguous(torch::MemoryFormat::ChannelsLast);
+  // wT K-major: [CI][R][S][CO]
+  auto wt = w.permute({1, 2, 3, 0}).contiguous();
+  auto gx = torch::empty({cs.N, cs.C, cs.H, cs.W},
+                         gy.options().memory_format(
+                             torch::MemoryFormat::ChannelsLast));
+  launch_conv_dgrad(gc.data_ptr(), wt.data_ptr(), gx.data_ptr(), cs,
+                    stream());
+  HIP_CHECK_LAST();
+  return gx;
+}
+
+torch::Tensor conv2d_wgrad(torch::Tensor gy, torch::Tensor x, int64_t R,
+                           int64_t S, int64_t stride, int64_t pad) {
+  check_bf16(gy, "gy");
+  const c10::hip::HIPGuard guard(gy.device());
+  ConvShape cs;
+  cs.N = static_cast<int>(x.size(0));
+  cs.C = static_cast<int>(x.size(1));
+  cs.H = static_cast<int>(x.size(2));
+  cs.W = static_cast<int>(x.size(3));
+  cs.K = static_cast<int>(gy.size(1));
+  cs.P = static_cast<int>(gy.size(2));
+  cs.Q = static_cast<int>(gy.size(3));
+  cs.R = static_cast<int>(R);
+  cs.S = static_cast<int>(S);
+  cs.stride = static_cast<int>(stride);
+  cs.pad = static_cast<int>(pad);
+  auto gc = gy.contiguous(torch::MemoryFormat::ChannelsLast);
+  auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
+  const int64_t rsc = static_cast<int64_t>(cs.R) * cs.S * cs.C;
+  auto gwf = torch::zeros({cs.K, rsc},
+                          x.options().dtype(torch::kFloat32));
+  auto gw_rsci = torch::empty({cs.K, cs.R, cs.S, cs.C}, x.options());
+  launch_conv_wgrad(gc.data_ptr(), xc.data_ptr(), gwf.data_ptr<float>(),
+                    gw_rsci.data_ptr(), cs, stream());
+  HIP_CHECK_LAST();
+  return gw_rsci.permute({0, 3, 1, 2}).contiguous();
+}
+
+}  // namespace
+}  // namespace colearn
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  using namespace colearn;
+  m.def("fused_sgd", &fused_sgd);
+  m.def("fused_adam", &fused_adam);
+  m.def("sq_norm_per_row", &sq_norm_per_row);
+  m.def("scale_rows_mean", &scale_rows_mean);
+  m.def("topk_compress", &topk_compress);
+  m.def("scatter_add_pairs", &scatter_add_pairs);
+  m.def("layernorm_fwd", &layernorm_fwd);
+  m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("batchnorm_fwd", &batchnorm_fwd);
+  m.def("batchnorm_bwd", &batchnorm_bwd);
+  m.def("gemm_nt", &gemm_nt);
+  m.def("gemm_nn", &gemm_nn);
+  m.def("gemm_tn", &gemm_tn);
+  m.def("bmm_nt", &bmm_nt);
+  m.def("bmm_nn", &bmm_nn);
+  m.def("bmm_tn", &bmm_tn);
+  m.def("conv2d_fwd", &conv2d_fwd);
+  m.def("conv2d_dgrad", &conv2d_dgrad);
+  m.def("conv2d_wgrad", &conv2d_wgrad);
+}

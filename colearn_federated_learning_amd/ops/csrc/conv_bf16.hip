@@ -1,0 +1,334 @@
+// Implicit-GEMM 2-D convolution for gfx950, channels-last bf16
+// (ResNet 3x3 & 1x1 convs, fwd + dgrad + wgrad).
+//
+// All three passes are MFMA tile GEMMs that GATHER their A operand from
+// the NHWC image instead of materializing im2col:
+//   fwd   : y[npq, co]  = x_gather[npq, rsci] @ w[co, rsci]^T      (NT)
+//   dgrad : gx[nhw, ci] = gy_gather[nhw, rsco] @ wT[ci, rsco]^T    (NT)
+//   wgrad : gw[co, rsci] = gy[npq, co]^T @ x_gather[npq, rsci]     (TN,
+//           split-K over npq with fp32 atomic accumulation)
+// Weights arrive pre-permuted K-major from the wrapper ([CO][R][S][CI]
+// for fwd, [CI][R][S][CO] for dgrad) so their staging is identical to the
+// dense GEMM's.  Channel gathers vectorize 8-wide whenever C % 8 == 0
+// (every ResNet conv except the 3-channel stem, which takes the scalar
+// path).
+
+#include "api.h"
+#include "tile_gemm.h"
+
+namespace colearn {
+
+// ---- forward A-gather: row m = (n,p,q), k = (r,s,ci) ---------------------
+DEV_INLINE void conv_fwd_gather(const short* __restrict__ x,
+                                const ConvShape cs, int64_t m, int64_t k,
+                                short8v_t& out) {
+  out = mfma_frag_zero();
+  const int64_t K_dim = static_cast<int64_t>(cs.R) * cs.S * cs.C;
+  if (k >= K_dim) return;
+  const int q = static_cast<int>(m % cs.Q);
+  const int p = static_cast<int>((m / cs.Q) % cs.P);
+  const int n = static_cast<int>(m / (static_cast<int64_t>(cs.P) * cs.Q));
+  if ((cs.C & 7) == 0) {
+    const int ci = static_cast<int>(k % cs.C);
+    const int s = static_cast<int>((k / cs.C) % cs.S);
+    const int r = static_cast<int>(k / (static_cast<int64_t>(cs.C) * cs.S));
+    const int ih = p * cs.stride - cs.pad + r;
+    const int iw = q * cs.stride - cs.pad + s;
+    if (ih < 0 || ih >= cs.H || iw < 0 || iw >= cs.W) return;
+    out = *reinterpret_cast<const short8v_t*>(
+        x + ((static_cast<int64_t>(n) * cs.H + ih) * cs.W + iw) * cs.C + ci);
+  } else {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int64_t kk = k + j;
+      if (kk >= K_dim) break;
+      const int ci = static_cast<int>(kk % cs.C);
+      const int s = static_cast<int>((kk / cs.C) % cs.S);
+      const int r =
+          static_cast<int>(kk / (static_cast<int64_t>(cs.C) * cs.S));
+      const int ih = p * cs.stride - cs.pad + r;
+      const int iw = q * cs.stride - cs.pad + s;
+      if (ih < 0 || ih >= cs.H || iw < 0 || iw >= cs.W) continue;
+      out[j] = x[((static_cast<int64_t>(n) * cs.H + ih) * cs.W + iw) * cs.C +
+                 ci];
+    }
+  }
+}
+
+// ---- dgrad A-gather: row m = (n,h,w), k = (r,s,co) -----------------------
+DEV_INLINE void conv_dgrad_gather(const short* __restrict__ gy,
+                                  const ConvShape cs, int64_t m, int64_t k,
+                                  short8v_t& out) {
+  out = mfma_frag_zero();
+  const int64_t K_dim = static_cast<int64_t>(cs.R) * cs.S * cs.K;
+  if (k >= K_dim) return;
+  const int w = static_cast<int>(m % cs.W);
+  const int h = static_cast<int>((m / cs.W) % cs.H);
+  const int n = static_cast<int>(m / (static_cast<int64_t>(cs.H) * cs.W));
+  if ((cs.K & 7) == 0) {
+    const int co = static_cast<int>(k % cs.K);
+    const int s = static_cast<int>((k / cs.K) % cs.S);
+    const int r = static_cast<int>(k / (static_cast<int64_t>(cs.K) * cs.S));
+    const int ph = h + cs.pad - r;
+    const int pw = w + cs.pad - s;
+    if (ph < 0 || pw < 0 || ph % cs.stride || pw % cs.stride) return;
+    const int po = ph / cs.stride, qo = pw / cs.stride;
+    if (po >= cs.P || qo >= cs.Q) return;
+    out = *reinterpret_cast<const short8v_t*>(
+        gy + ((static_cast<int64_t>(n) * cs.P + po) * cs.Q + qo) * cs.K + co);
+  } else {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int64_t kk = k + j;
+      if (kk >= K_dim) break;
+      const int co = static_cast<int>(kk % cs.K);
+      const int s = static_cast<int>((kk / cs.K) % cs.S);
+      const int r =
+          static_cast<int>(kk / (static_cast<int64_t>(cs.K) * cs.S));
+      const int ph = h + cs.pad - r;
+      const int pw = w + cs.pad - s;
+      if (ph < 0 || pw < 0 || ph % cs.stride || pw % cs.stride) continue;
+      const int po = ph / cs.stride, qo = pw / cs.stride;
+      if (po >= cs.P || qo >= cs.Q) continue;
+      out[j] = gy[((static_cast<int64_t>(n) * cs.P + po) * cs.Q + qo) * cs.K +
+                  co];
+    }
+  }
+}
+
+// ---- the shared implicit-GEMM NT kernel (fwd / dgrad via GATHER flag) ----
+template <int GATHER>  // 0 = conv fwd (gather x), 1 = dgrad (gather gy)
+__global__ __launch_bounds__(kBlockThreads) void conv_nt_kernel(
+    const short* __restrict__ Asrc, const short* __restrict__ Bw,
+    bf16* __restrict__ C, ConvShape cs, int64_t M, int64_t Nd, int64_t Kd,
+    int nbm, int nbn) {
+  __shared__ short lds[2][2][BM * BK];
+  const int bid = xcd_swizzle(blockIdx.x, nbm * nbn);
+  const int64_t bm = static_cast<int64_t>(bid / nbn) * BM;
+  const int64_t bn = static_cast<int64_t>(bid % nbn) * BN;
+  const int wid = threadIdx.x / kWave;
+  const int wm = (wid >> 1) * 64, wn = (wid & 1) * 64;
+
+  f32x4_t acc[4][4];
+#pragma unroll
+  for (int m = 0; m < 4; ++m)
+#pragma unroll
+    for (int n = 0; n < 4; ++n) acc[m][n] = {0.f, 0.f, 0.f, 0.f};
+
+  const int row = threadIdx.x >> 1;
+  const int half = threadIdx.x & 1;
+  const int ntiles = static_cast<int>((Kd + BK - 1) / BK);
+
+  short8v_t a0, a1, b0, b1;
+  auto load_tile = [&](int t) {
+    const int64_t k = static_cast<int64_t>(t) * BK + half * 16;
+    const int64_t am = bm + row;
+    if (am < M) {
+      if constexpr (GATHER == 0) {
+        conv_fwd_gather(Asrc, cs, am, k, a0);
+        conv_fwd_gather(Asrc, cs, am, k + 8, a1);
+      } else {
+        conv_dgrad_gather(Asrc, cs, am, k, a0);
+        conv_dgrad_gather(Asrc, cs, am, k + 8, a1);
+      }
+    } else {
+      a0 = a1 = mfma_frag_zero();
+    }
+    b0 = b1 = mfma_frag_zero();
+    const int64_t bn_row = bn + row;
+    if (bn_row < Nd) {
+      const short* p = Bw + bn_row * Kd + k;
+      if (k + 16 <= Kd) {
+        b0 = *reinterpret_cast<const short8v_t*>(p);
+        b1 = *reinterpret_cast<const short8v_t*>(p + 8);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 16; ++j)
+          if (k + j < Kd) {
+            if (j < 8) b0[j] = p[j]; else b1[j - 8] = p[j];
+          }
+      }
+    }
+  };
+  auto write_tile = [&](int buf) {
+    *reinterpret_cast<short8v_t*>(reinterpret_cast<char*>(lds[buf][0]) +
+                                  lds_slot_byte(row, half * 2)) = a0;
+    *reinterpret_cast<short8v_t*>(reinterpret_cast<char*>(lds[buf][0]) +
+                                  lds_slot_byte(row, half * 2 + 1)) = a1;
+    *reinterpret_cast<short8v_t*>(reinterpret_cast<char*>(lds[buf][1]) +
+                                  lds_slot_byte(row, half * 2)) = b0;
+    *reinterpret_cast<short8v_t*>(reinterpret_cast<char*>(lds[buf][1]) +
+                                  lds_slot_byte(row, half * 2 + 1)) = b1;
+  };
+
+  load_tile(0);
+  write_tile(0);
+  __syncthreads();
+  int cur = 0;
+  for (int t = 0; t < ntiles; ++t) {
+    if (t + 1 < ntiles) load_tile(t + 1);
+    short8v_t af[4], bfr[4];
+#pragma unroll
+    for (int m = 0; m < 4; ++m)
+      af[m] = frag_from_lds(lds[cur][0], wm + m * 16);
+#pragma unroll
+    for (int n = 0; n < 4; ++n)
+      bfr[n] = frag_from_lds(lds[cur][1], wn + n * 16);
+#pragma unroll
+    for (int m = 0; m < 4; ++m)
+#pragma unroll
+      for (int n = 0; n < 4; ++n)
+        acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[m], bfr[n], acc[m][n], 0, 0, 0);
+    if (t + 1 < ntiles) write_tile(cur ^ 1);
+    __syncthreads();
+    cur ^= 1;
+  }
+  store_tile_bf16(reinterpret_cast<const float*>(acc), C, Nd, M, Nd, bm, bn,
+                  wm, wn);
+}
+
+// ==== wgrad ===============================================================
+// gw[co, rsci] += sum over an npq chunk; 64x64 tile per block, 4 waves in
+// 2x2, BK=32 over npq; both operands transposed-staged into K-major LDS
+// (scatter b16 writes — v1 correctness; tr_b16 reads are the planned
+// optimization).  fp32 atomicAdd epilogue combines the split-K blocks.
+__global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
+    const short* __restrict__ gy, const short* __restrict__ x,
+    float* __restrict__ gwf, ConvShape cs, int64_t NPQ, int64_t RSC,
+    int nbm, int nbn, int splitk) {
+  constexpr int TM = 64, TN = 64;
+  __shared__ short Gs[TM][BK + 2];  // [co][k] (+2 pad: scatter writes)
+  __shared__ short Xs[TN][BK + 2];  // [rsci][k]
+  const int tile = blockIdx.x;
+  const int co0 = (tile / nbn) * TM;
+  const int rc0 = (tile % nbn) * TN;
+  const int64_t chunk = (NPQ + splitk - 1) / splitk;
+  const int64_t k_begin = static_cast<int64_t>(blockIdx.y) * chunk;
+  const int64_t k_end = min(k_begin + chunk, NPQ);
+
+  const int wid = threadIdx.x / kWave;
+  const int wm = (wid >> 1) * 32, wn = (wid & 1) * 32;
+  f32x4_t acc[2][2];
+#pragma unroll
+  for (int m = 0; m < 2; ++m)
+#pragma unroll
+    for (int n = 0; n < 2; ++n) acc[m][n] = {0.f, 0.f, 0.f, 0.f};
+
+  // staging decomposition: 256 threads, 8 elems each of G and X per step.
+  const int kk = threadIdx.x / 8;        // 0..31 (npq within step)
+  const int e8 = (threadIdx.x % 8) * 8;  // 8-element column block
+
+  for (int64_t k0 = k_begin; k0 < k_end; k0 += BK) {
+    const int64_t npq = k0 + kk;
+    short8v_t gv = mfma_frag_zero();
+    short8v_t xv = mfma_frag_zero();
+    if (npq < k_end) {
+      // G: gy[npq][co0+e8 .. +8]
+      if (co0 + e8 < cs.K) {
+        const short* p = gy + npq * cs.K + co0 + e8;
+        if ((cs.K & 7) == 0 && co0 + e8 + 8 <= cs.K)
+          gv = *reinterpret_cast<const short8v_t*>(p);
+        else
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            if (co0 + e8 + j < cs.K) gv[j] = p[j];
+      }
+      // X: x_gather[npq][rc0+e8 .. +8]
+      conv_fwd_gather(x, cs, npq, rc0 + e8, xv);
+    }
+    __syncthreads();  // previous iteration's reads complete
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      Gs[e8 + j][kk] = gv[j];
+      Xs[e8 + j][kk] = xv[j];
+    }
+    __syncthreads();
+    // fragments: A rows = co (l&15), k = (l>>4)*8 + j from Gs[co][k]
+    const int lane = threadIdx.x % kWave;
+    const int fr = lane & 15, kg = (lane >> 4) * 8;
+    short8v_t af[2], bfr[2];
+#pragma unroll
+    for (int m = 0; m < 2; ++m) {
+      const short* p = &Gs[wm + m * 16 + fr][kg];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) af[m][j] = p[j];
+    }
+#pragma unroll
+    for (int n = 0; n < 2; ++n) {
+      const short* p = &Xs[wn + n * 16 + fr][kg];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) bfr[n][j] = p[j];
+    }
+#pragma unroll
+    for (int m = 0; m < 2; ++m)
+#pragma unroll
+      for (int n = 0; n < 2; ++n)
+        acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[m], bfr[n], acc[m][n], 0, 0, 0);
+  }
+  // epilogue: atomicAdd fp32 (C/D map: col=lane&15, row=(lane>>4)*4+j)
+  const int lane = threadIdx.x % kWave;
+  const int cif = lane & 15, rg = lane >> 4;
+#pragma unroll
+  for (int m = 0; m < 2; ++m)
+#pragma unroll
+    for (int n = 0; n < 2; ++n)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int co = co0 + wm + m * 16 + rg * 4 + j;
+        const int rc = rc0 + wn + n * 16 + cif;
+        if (co < cs.K && rc < RSC)
+          atomicAdd(gwf + static_cast<int64_t>(co) * RSC + rc,
+                    acc[m][n][j]);
+      }
+}
+
+__global__ void f32_to_bf16_kernel(const float* __restrict__ in,
+                                   bf16* __restrict__ out, int64_t n) {
+  int64_t i = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+  for (; i < n; i += grid_stride()) out[i] = f2b(in[i]);
+}
+
+// ---- launchers ------------------------------------------------------------
+void launch_conv_fwd(const void* x, const void* w, void* y, ConvShape cs,
+                     hipStream_t stream) {
+  const int64_t M = static_cast<int64_t>(cs.N) * cs.P * cs.Q;
+  const int64_t Kd = static_cast<int64_t>(cs.R) * cs.S * cs.C;
+  const int nbm = static_cast<int>((M + BM - 1) / BM);
+  const int nbn = (cs.K + BN - 1) / BN;
+  conv_nt_kernel<0><<<dim3(nbm * nbn), kBlockThreads, 0, stream>>>(
+      static_cast<const short*>(x), static_cast<const short*>(w),
+      static_cast<bf16*>(y), cs, M, cs.K, Kd, nbm, nbn);
+}
+
+void launch_conv_dgrad(const void* gy, const void* wt, void* gx,
+                       ConvShape cs, hipStream_t stream) {
+  const int64_t M = static_cast<int64_t>(cs.N) * cs.H * cs.W;
+  const int64_t Kd = static_cast<int64_t>(cs.R) * cs.S * cs.K;
+  const int nbm = static_cast<int>((M + BM - 1) / BM);
+  const int nbn = (cs.C + BN - 1) / BN;
+  conv_nt_kernel<1><<<dim3(nbm * nbn), kBlockThreads, 0, stream>>>(
+      static_cast<const short*>(gy), static_cast<const short*>(wt),
+      static_cast<bf16*>(gx), cs, M, cs.C, Kd, nbm, nbn);
+}
+
+void launch_conv_wgrad(const void* gy, const void* x, float* gwf, void* gw,
+                       ConvShape cs, hipStream_t stream) {
+  const int64_t NPQ = static_cast<int64_t>(cs.N) * cs.P * cs.Q;
+  const int64_t RSC = static_cast<int64_t>(cs.R) * cs.S * cs.C;
+  const int nbm = (cs.K + 63) / 64;
+  const int nbn = static_cast<int>((RSC + 63) / 64);
+  int64_t want = 512 / (nbm * nbn);
+  if (want < 1) want = 1;
+  int64_t cap = (NPQ + BK - 1) / BK;
+  const int splitk = static_cast<int>(want < cap ? want : cap);
+  conv_wgrad_kernel<<<dim3(nbm * nbn, splitk), kBlockThreads, 0, stream>>>(
+      static_cast<const short*>(gy), static_cast<const short*>(x), gwf, cs,
+      NPQ, RSC, nbm, nbn, splitk);
+  const int64_t n = static_cast<int64_t>(cs.K) * RSC;
+  f32_to_bf16_kernel<<<elemwise_grid(n), 256, 0, stream>>>(
+      gwf, static_cast<bf16*>(gw), n);
+}
+
+}  // namespace colearn

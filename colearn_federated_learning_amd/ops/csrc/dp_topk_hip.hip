@@ -1,0 +1,228 @@
+#include "hip/hip_runtime.h"
+// DP-FedAvg + top-k sparsification kernels.
+//
+// sq_norm_per_row / scale_rows_mean: the fused per-sample clip pipeline —
+// per-sample grad norms (one reduction sweep over [B, D]), then
+// clip-scale + batch-mean in a second sweep.  Both are pure-bandwidth
+// float4 kernels.
+//
+// topk_compress: 16-bit radix-select over the |v| bit pattern (two 256-bin
+// histogram rounds + ticketed compaction) — no sort, O(3) sweeps of the
+// data, exact k output.  scatter_add_pairs: the decompression side.
+
+#include "api.h"
+#include "common.h"
+
+namespace colearn {
+
+// ---- per-sample squared norms: sq[b] += sum_d g[b,d]^2 --------------------
+__global__ void sq_norm_per_row_kernel(const float* __restrict__ g,
+                                       float* __restrict__ sq, int64_t rows,
+                                       int64_t cols) {
+  __shared__ float scratch[4];
+  const int row = blockIdx.x;
+  if (row >= rows) return;
+  const float* grow = g + row * cols;
+  float acc = 0.f;
+  int64_t i = (static_cast<int64_t>(blockIdx.y) * blockDim.x + threadIdx.x) * 4;
+  const int64_t stride = static_cast<int64_t>(gridDim.y) * blockDim.x * 4;
+  for (; i + 3 < cols; i += stride) {
+    const float4 v = *reinterpret_cast<const float4*>(grow + i);
+    acc += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+  }
+  if (blockIdx.y == 0 && threadIdx.x < (cols & 3)) {
+    const float v = grow[(cols & ~int64_t(3)) + threadIdx.x];
+    acc += v * v;
+  }
+  acc = block_allreduce_sum<256>(acc, scratch);
+  if (threadIdx.x == 0) atomicAdd(sq + row, acc);
+}
+
+// ---- out[d] = (1/rows) * sum_b scale[b] * g[b,d] --------------------------
+__global__ void scale_rows_mean_kernel(const float* __restrict__ g,
+                                       const float* __restrict__ scale,
+                                       float* __restrict__ out, int64_t rows,
+                                       int64_t cols) {
+  int64_t d = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) * 4;
+  const float inv = 1.f / static_cast<float>(rows);
+  const int64_t stride = grid_stride() * 4;
+  for (; d + 3 < cols; d += stride) {
+    float acc[4] = {0.f, 0.f, 0.f, 0.f};
+    for (int64_t b = 0; b < rows; ++b) {
+      const float s = scale[b];
+      const float4 v = *reinterpret_cast<const float4*>(g + b * cols + d);
+      acc[0] += s * v.x;
+      acc[1] += s * v.y;
+      acc[2] += s * v.z;
+      acc[3] += s * v.w;
+    }
+    *reinterpret_cast<float4*>(out + d) = make_float4(
+        acc[0] * inv, acc[1] * inv, acc[2] * inv, acc[3] * inv);
+  }
+  if (blockIdx.x == 0 && threadIdx.x < (cols & 3)) {
+    const int64_t j = (cols & ~int64_t(3)) + threadIdx.x;
+    float acc = 0.f;
+    for (int64_t b = 0; b < rows; ++b) acc += scale[b] * g[b * cols + j];
+    out[j] = acc * inv;
+  }
+}
+
+// ---- top-k radix select ---------------------------------------------------
+// key(v) = top 16 bits of |v|'s f32 bit pattern (monotone in |v|).
+
+DEV_INLINE unsigned absbits(float v) {
+  union {
+    float f;
+    unsigned u;
+  } c;
+  c.f = v;
+  return c.u & 0x7FFFFFFFu;
+}
+
+__global__ void topk_hist1_kernel(const float* __restrict__ v, int64_t n,
+                                  unsigned* __restrict__ hist /*[256]*/) {
+  __shared__ unsigned lh[256];
+  if (threadIdx.x < 256) lh[threadIdx.x] = 0;
+  __syncthreads();
+  int64_t i = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+  for (; i < n; i += grid_stride())
+    atomicAdd(&lh[absbits(v[i]) >> 24], 1u);
+  __syncthreads();
+  if (threadIdx.x < 256 && lh[threadIdx.x])
+    atomicAdd(hist + threadIdx.x, lh[threadIdx.x]);
+}
+
+// pick the bin where the k-th largest falls; single block of 256.
+// out: sel[0]=bin, sel[1]=count strictly above bin, sel[2]=k remaining in bin
+__global__ void topk_select_kernel(const unsigned* __restrict__ hist,
+                                   int64_t k, unsigned* __restrict__ sel) {
+  __shared__ unsigned h[256];
+  h[threadIdx.x] = hist[threadIdx.x];
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    unsigned above = 0;
+    int bin = 0;
+    for (int b = 255; b >= 0; --b) {
+      if (above + h[b] >= static_cast<unsigned>(k)) {
+        bin = b;
+        break;
+      }
+      above += h[b];
+    }
+    sel[0] = static_cast<unsigned>(bin);
+    sel[1] = above;
+    sel[2] = static_cast<unsigned>(k) - above;
+  }
+}
+
+__global__ void topk_hist2_kernel(const float* __restrict__ v, int64_t n,
+                                  const unsigned* __restrict__ sel1,
+                                  unsigned* __restrict__ hist2 /*[256]*/) {
+  __shared__ unsigned lh[256];
+  if (threadIdx.x < 256) lh[threadIdx.x] = 0;
+  __syncthreads();
+  const unsigned bin = sel1[0];
+  int64_t i = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+  for (; i < n; i += grid_stride()) {
+    const unsigned key = absbits(v[i]);
+    if ((key >> 24) == bin) atomicAdd(&lh[(key >> 16) & 0xFFu], 1u);
+  }
+  __syncthreads();
+  if (threadIdx.x < 256 && lh[threadIdx.x])
+    atomicAdd(hist2 + threadIdx.x, lh[threadIdx.x]);
+}
+
+// threshold = (bin1<<24)|(bin2<<16); counts strictly above threshold plus a
+// ticket budget for ties at the threshold make the output exactly k.
+// out2: [0]=threshold, [1]=need_eq budget at threshold
+__global__ void topk_select2_kernel(const unsigned* __restrict__ sel1,
+                                    const unsigned* __restrict__ hist2,
+                                    unsigned* __restrict__ out2) {
+  __shared__ unsigned h[256];
+  h[threadIdx.x] = hist2[threadIdx.x];
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    const unsigned rem = sel1[2];  // entries of k still needed inside bin1
+    unsigned above = 0;
+    int bin = 0;
+    for (int b = 255; b >= 0; --b) {
+      if (above + h[b] >= rem) {
+        bin = b;
+        break;
+      }
+      above += h[b];
+    }
+    out2[0] = (sel1[0] << 24) | (static_cast<unsigned>(bin) << 16);
+    out2[1] = rem - above;  // tickets for keys exactly at the threshold
+    // counters used by the compaction pass
+    out2[2] = 0;  // ticket counter (== threshold keys taken)
+    out2[3] = 0;  // output cursor
+  }
+}
+
+__global__ void topk_compact_kernel(const float* __restrict__ v, int64_t n,
+                                    unsigned* __restrict__ ctl /*out2*/,
+                                    int* __restrict__ idx,
+                                    float* __restrict__ val, int64_t k) {
+  const unsigned thresh = ctl[0];
+  const unsigned need_eq = ctl[1];
+  int64_t i = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+  for (; i < n; i += grid_stride()) {
+    const unsigned key = absbits(v[i]) & 0xFFFF0000u;
+    if (key < thresh) continue;
+    if (key == thresh) {
+      if (atomicAdd(&ctl[2], 1u) >= need_eq) continue;  // tie budget spent
+    }
+    const unsigned pos = atomicAdd(&ctl[3], 1u);
+    if (pos < static_cast<unsigned>(k)) {
+      idx[pos] = static_cast<int>(i);
+      val[pos] = v[i];
+    }
+  }
+}
+
+__global__ void scatter_add_pairs_kernel(float* __restrict__ out,
+                                         const int* __restrict__ idx,
+                                         const float* __restrict__ val,
+                                         int64_t k) {
+  int64_t i = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+  for (; i < k; i += grid_stride()) atomicAdd(out + idx[i], val[i]);
+}
+
+// ---- launchers ------------------------------------------------------------
+void launch_sq_norm_per_row(const float* g, float* sq, int64_t rows,
+                            int64_t cols, hipStream_t stream) {
+  const int chunks =
+      max(1, min(16, static_cast<int>((cols + 262143) / 262144)));
+  dim3 grid(static_cast<unsigned>(rows), chunks);
+ hipLaunchKernelGGL(( sq_norm_per_row_kernel), dim3(grid), dim3(256), 0, stream, g, sq, rows, cols);
+}
+
+void launch_scale_rows_mean(const float* g, const float* scale, float* out,
+                            int64_t rows, int64_t cols, hipStream_t stream) {
+ hipLaunchKernelGGL(( scale_rows_mean_kernel), dim3(elemwise_grid((cols + 3) / 4)), dim3(256), 0, stream, 
+      g, scale, out, rows, cols);
+}
+
+void launch_topk(const float* v, int64_t n, int64_t k, int* idx, float* val,
+                 unsigned* workspace /* >= 520 unsigned, zeroed */,
+                 hipStream_t stream) {
+  unsigned* hist1 = workspace;        // 256
+  unsigned* hist2 = workspace + 256;  // 256
+  unsigned* sel1 = workspace + 512;   // 4
+  unsigned* ctl = workspace + 516;    // 4
+  const int grid = elemwise_grid(n);
+ hipLaunchKernelGGL(( topk_hist1_kernel), dim3(grid), dim3(256), 0, stream, v, n, hist1);
+ hipLaunchKernelGGL(( topk_select_kernel), dim3(1), dim3(256), 0, stream, hist1, k, sel1);
+ hipLaunchKernelGGL(( topk_hist2_kernel), dim3(grid), dim3(256), 0, stream, v, n, sel1, hist2);
+ hipLaunchKernelGGL(( topk_select2_kernel), dim3(1), dim3(256), 0, stream, sel1, hist2, ctl);
+ hipLaunchKernelGGL(( topk_compact_kernel), dim3(grid), dim3(256), 0, stream, v, n, ctl, idx, val, k);
+}
+
+void launch_scatter_add_pairs(float* out, const int* idx, const float* val,
+                              int64_t k, hipStream_t stream) {
+ hipLaunchKernelGGL(( scatter_add_pairs_kernel), dim3(elemwise_grid(k)), dim3(256), 0, stream, out, idx,
+                                                                 val, k);
+}
+
+}  // namespace colearn

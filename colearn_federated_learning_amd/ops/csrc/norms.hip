@@ -1,0 +1,335 @@
+// Fused LayerNorm (BERT path, N7) and BatchNorm (ResNet path, N6) kernels.
+//
+// Memory-bound ops: the design targets the HBM roofline with vectorized
+// bf16x8 access (guide G13 — scalar bf16 is 2x slower) and single-sweep
+// fusion per pass.  BatchNorm operates on CHANNELS-LAST (NHWC) data — the
+// layout the whole ResNet GPU path uses — so per-channel statistics
+// accumulate with perfectly coalesced lanes (consecutive lanes read
+// consecutive channels).  Stats and parameters are fp32.
+//
+// LayerNorm: one wave per row (row = token), wave shuffle reductions, no
+// LDS round-trip for the row stats.
+
+#include "api.h"
+#include "common.h"
+
+namespace colearn {
+
+// ==== LayerNorm ============================================================
+// x: [rows, H] bf16, w/b: [H] fp32, y: bf16, mean/invstd: [rows] fp32
+__global__ void layernorm_fwd_kernel(const bf16* __restrict__ x,
+                                     const float* __restrict__ w,
+                                     const float* __restrict__ b,
+                                     bf16* __restrict__ y,
+                                     float* __restrict__ mean,
+                                     float* __restrict__ invstd, int64_t rows,
+                                     int H, float eps) {
+  const int wid = threadIdx.x / kWave;
+  const int lane = threadIdx.x % kWave;
+  const int waves = blockDim.x / kWave;
+  const float invH = 1.f / H;
+  for (int64_t row = static_cast<int64_t>(blockIdx.x) * waves + wid;
+       row < rows; row += static_cast<int64_t>(gridDim.x) * waves) {
+    const short* xr = reinterpret_cast<const short*>(x + row * H);
+    float s = 0.f, ss = 0.f;
+    for (int h = lane * 8; h + 7 < H; h += kWave * 8) {
+      const short8 v = *reinterpret_cast<const short8*>(xr + h);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float f = bits2f(v.x[j]);
+        s += f;
+        ss += f * f;
+      }
+    }
+    for (int h = (H & ~7) + lane; h < H; h += kWave) {  // H % 8 tail
+      const float f = bits2f(xr[h]);
+      s += f;
+      ss += f * f;
+    }
+    s = wave_allreduce_sum(s) * invH;
+    ss = wave_allreduce_sum(ss) * invH;
+    const float istd = rsqrtf(fmaxf(ss - s * s, 0.f) + eps);
+    if (lane == 0) {
+      mean[row] = s;
+      invstd[row] = istd;
+    }
+    short* yr = reinterpret_cast<short*>(y + row * H);
+    for (int h = lane * 8; h + 7 < H; h += kWave * 8) {
+      const short8 v = *reinterpret_cast<const short8*>(xr + h);
+      const float4 w0 = *reinterpret_cast<const float4*>(w + h);
+      const float4 w1 = *reinterpret_cast<const float4*>(w + h + 4);
+      const float4 b0 = *reinterpret_cast<const float4*>(b + h);
+      const float4 b1 = *reinterpret_cast<const float4*>(b + h + 4);
+      const float wv[8] = {w0.x, w0.y, w0.z, w0.w, w1.x, w1.y, w1.z, w1.w};
+      const float bv[8] = {b0.x, b0.y, b0.z, b0.w, b1.x, b1.y, b1.z, b1.w};
+      short8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o.x[j] = f2bits((bits2f(v.x[j]) - s) * istd * wv[j] + bv[j]);
+      *reinterpret_cast<short8*>(yr + h) = o;
+    }
+    for (int h = (H & ~7) + lane; h < H; h += kWave)
+      yr[h] = f2bits((bits2f(xr[h]) - s) * istd * w[h] + b[h]);
+  }
+}
+
+// per-row part of the backward: gx.  (gw/gb accumulate in a column kernel.)
+__global__ void layernorm_bwd_gx_kernel(
+    const bf16* __restrict__ gy, const bf16* __restrict__ x,
+    const float* __restrict__ w, const float* __restrict__ mean,
+    const float* __restrict__ invstd, bf16* __restrict__ gx, int64_t rows,
+    int H) {
+  const int wid = threadIdx.x / kWave;
+  const int lane = threadIdx.x % kWave;
+  const int waves = blockDim.x / kWave;
+  const float invH = 1.f / H;
+  for (int64_t row = static_cast<int64_t>(blockIdx.x) * waves + wid;
+       row < rows; row += static_cast<int64_t>(gridDim.x) * waves) {
+    const short* gr = reinterpret_cast<const short*>(gy + row * H);
+    const short* xr = reinterpret_cast<const short*>(x + row * H);
+    const float mu = mean[row], istd = invstd[row];
+    float s1 = 0.f, s2 = 0.f;  // sum(wg), sum(wg * xhat)
+    for (int h = lane; h < H; h += kWave) {
+      const float wg = bits2f(gr[h]) * w[h];
+      const float xhat = (bits2f(xr[h]) - mu) * istd;
+      s1 += wg;
+      s2 += wg * xhat;
+    }
+    s1 = wave_allreduce_sum(s1) * invH;
+    s2 = wave_allreduce_sum(s2) * invH;
+    short* or_ = reinterpret_cast<short*>(gx + row * H);
+    for (int h = lane; h < H; h += kWave) {
+      const float wg = bits2f(gr[h]) * w[h];
+      const float xhat = (bits2f(xr[h]) - mu) * istd;
+      or_[h] = f2bits(istd * (wg - s1 - xhat * s2));
+    }
+  }
+}
+
+// column reduction: gw[h] = sum_rows gy*xhat, gb[h] = sum_rows gy.
+// threads along h (coalesced), grid.y chunks the rows; fp32 atomics combine.
+__global__ void layernorm_bwd_gwb_kernel(
+    const bf16* __restrict__ gy, const bf16* __restrict__ x,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    float* __restrict__ gw, float* __restrict__ gb, int64_t rows, int H) {
+  const int h = blockIdx.x * blockDim.x + threadIdx.x;
+  if (h >= H) return;
+  float aw = 0.f, ab = 0.f;
+  for (int64_t row = blockIdx.y; row < rows; row += gridDim.y) {
+    const float g = bits2f(reinterpret_cast<const short*>(gy)[row * H + h]);
+    const float xv = bits2f(reinterpret_cast<const short*>(x)[row * H + h]);
+    aw += g * (xv - mean[row]) * invstd[row];
+    ab += g;
+  }
+  atomicAdd(gw + h, aw);
+  atomicAdd(gb + h, ab);
+}
+
+// ==== BatchNorm (channels-last) ===========================================
+// x viewed as [R, C] with R = N*H*W; stats per channel c.
+// Pass 1: per-channel sum + sumsq via LDS-binned accumulation.
+__global__ void bn_reduce_kernel(const bf16* __restrict__ x, int64_t R, int C,
+                                 float* __restrict__ sums /*[2*C]*/) {
+  extern __shared__ float lbin[];  // [2*C]: sum, sumsq
+  for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) lbin[i] = 0.f;
+  __syncthreads();
+  const int64_t n = R * C;
+  int64_t i = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) * 8;
+  const int64_t stride = grid_stride() * 8;
+  for (; i + 7 < n; i += stride) {
+    const short8 v = *reinterpret_cast<const short8*>(
+        reinterpret_cast<const short*>(x) + i);
+    const int c0 = static_cast<int>(i % C);  // C % 8 == 0: no row crossing
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float f = bits2f(v.x[j]);
+      atomicAdd(&lbin[c0 + j], f);
+      atomicAdd(&lbin[C + c0 + j], f * f);
+    }
+  }
+  __syncthreads();
+  for (int i2 = threadIdx.x; i2 < 2 * C; i2 += blockDim.x)
+    if (lbin[i2] != 0.f) atomicAdd(sums + i2, lbin[i2]);
+}
+
+// Pass 2: finalize stats + update running stats (tiny launch over C).
+__global__ void bn_stats_kernel(const float* __restrict__ sums, int C,
+                                int64_t R, float* __restrict__ mean,
+                                float* __restrict__ invstd,
+                                float* __restrict__ running_mean,
+                                float* __restrict__ running_var,
+                                float momentum, float eps, bool training) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  if (training) {
+    const float m = sums[c] / R;
+    const float var = fmaxf(sums[C + c] / R - m * m, 0.f);
+    mean[c] = m;
+    invstd[c] = rsqrtf(var + eps);
+    const float unbiased = var * R / fmaxf(static_cast<float>(R - 1), 1.f);
+    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
+    running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+  } else {
+    mean[c] = running_mean[c];
+    invstd[c] = rsqrtf(running_var[c] + eps);
+  }
+}
+
+// Pass 3: y = (x - mean)*invstd*w + b, elementwise over [R, C].
+__global__ void bn_normalize_kernel(const bf16* __restrict__ x,
+                                    const float* __restrict__ mean,
+                                    const float* __restrict__ invstd,
+                                    const float* __restrict__ w,
+                                    const float* __restrict__ b,
+                                    bf16* __restrict__ y, int64_t R, int C) {
+  const int64_t n = R * C;
+  int64_t i = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) * 8;
+  const int64_t stride = grid_stride() * 8;
+  for (; i + 7 < n; i += stride) {
+    const short8 v = *reinterpret_cast<const short8*>(
+        reinterpret_cast<const short*>(x) + i);
+    const int c0 = static_cast<int>(i % C);
+    short8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = c0 + j;
+      o.x[j] = f2bits((bits2f(v.x[j]) - mean[c]) * invstd[c] * w[c] + b[c]);
+    }
+    *reinterpret_cast<short8*>(reinterpret_cast<short*>(y) + i) = o;
+  }
+}
+
+// Backward pass 1: gb[c] = sum gy, gw[c] = sum gy*xhat (LDS-binned).
+__global__ void bn_bwd_reduce_kernel(const bf16* __restrict__ gy,
+                                     const bf16* __restrict__ x,
+                                     const float* __restrict__ mean,
+                                     const float* __restrict__ invstd,
+                                     int64_t R, int C,
+                                     float* __restrict__ acc /*[2*C]*/) {
+  extern __shared__ float lbin[];  // [2*C]: gb, gw
+  for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) lbin[i] = 0.f;
+  __syncthreads();
+  const int64_t n = R * C;
+  int64_t i = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) * 8;
+  const int64_t stride = grid_stride() * 8;
+  for (; i + 7 < n; i += stride) {
+    const short8 g = *reinterpret_cast<const short8*>(
+        reinterpret_cast<const short*>(gy) + i);
+    const short8 v = *reinterpret_cast<const short8*>(
+        reinterpret_cast<const short*>(x) + i);
+    const int c0 = static_cast<int>(i % C);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = c0 + j;
+      const float gf = bits2f(g.x[j]);
+      atomicAdd(&lbin[c], gf);
+      atomicAdd(&lbin[C + c], gf * (bits2f(v.x[j]) - mean[c]) * invstd[c]);
+    }
+  }
+  __syncthreads();
+  for (int i2 = threadIdx.x; i2 < 2 * C; i2 += blockDim.x)
+    if (lbin[i2] != 0.f) atomicAdd(acc + i2, lbin[i2]);
+}
+
+// Backward pass 2: gx = w*invstd*(gy - gb/n - xhat*gw/n)   (training)
+//                  gx = w*invstd*gy                        (eval)
+__global__ void bn_bwd_gx_kernel(const bf16* __restrict__ gy,
+                                 const bf16* __restrict__ x,
+                                 const float* __restrict__ mean,
+                                 const float* __restrict__ invstd,
+                                 const float* __restrict__ w,
+                                 const float* __restrict__ acc /*[2*C]*/,
+                                 bf16* __restrict__ gx, int64_t R, int C,
+                                 bool training) {
+  const int64_t n = R * C;
+  const float invn = 1.f / R;
+  int64_t i = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) * 8;
+  const int64_t stride = grid_stride() * 8;
+  for (; i + 7 < n; i += stride) {
+    const short8 g = *reinterpret_cast<const short8*>(
+        reinterpret_cast<const short*>(gy) + i);
+    const short8 v = *reinterpret_cast<const short8*>(
+        reinterpret_cast<const short*>(x) + i);
+    const int c0 = static_cast<int>(i % C);
+    short8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = c0 + j;
+      const float gf = bits2f(g.x[j]);
+      float r;
+      if (training) {
+        const float xhat = (bits2f(v.x[j]) - mean[c]) * invstd[c];
+        r = w[c] * invstd[c] *
+            (gf - acc[c] * invn - xhat * acc[C + c] * invn);
+      } else {
+        r = w[c] * invstd[c] * gf;
+      }
+      o.x[j] = f2bits(r);
+    }
+    *reinterpret_cast<short8*>(reinterpret_cast<short*>(gx) + i) = o;
+  }
+}
+
+// ---- launchers ------------------------------------------------------------
+void launch_layernorm_fwd(const void* x, const float* w, const float* b,
+                          void* y, float* mean, float* invstd, int64_t rows,
+                          int H, float eps, hipStream_t stream) {
+  const int waves = 4;  // 256-thread block, one wave per row
+  int64_t g = (rows + waves - 1) / waves;
+  const int grid = static_cast<int>(g < 2048 ? g : 2048);
+  layernorm_fwd_kernel<<<grid, waves * kWave, 0, stream>>>(
+      static_cast<const bf16*>(x), w, b, static_cast<bf16*>(y), mean,
+      invstd, rows, H, eps);
+}
+
+void launch_layernorm_bwd(const void* gy_, const void* x_, const float* w,
+                          const float* mean, const float* invstd, void* gx_,
+                          float* gw, float* gb, int64_t rows, int H,
+                          hipStream_t stream) {
+  const bf16* gy = static_cast<const bf16*>(gy_);
+  const bf16* x = static_cast<const bf16*>(x_);
+  bf16* gx = static_cast<bf16*>(gx_);
+  const int waves = 4;
+  int64_t g = (rows + waves - 1) / waves;
+  const int grid = static_cast<int>(g < 2048 ? g : 2048);
+  layernorm_bwd_gx_kernel<<<grid, waves * kWave, 0, stream>>>(
+      gy, x, w, mean, invstd, gx, rows, H);
+  dim3 g2((H + 255) / 256, 64);
+  layernorm_bwd_gwb_kernel<<<g2, 256, 0, stream>>>(gy, x, mean, invstd, gw,
+                                                   gb, rows, H);
+}
+
+void launch_batchnorm_fwd(const void* x_, const float* w, const float* b,
+                          float* running_mean, float* running_var, void* y_,
+                          float* mean, float* invstd, float* sums, int64_t R,
+                          int C, float momentum, float eps, bool training,
+                          hipStream_t stream) {
+  const bf16* x = static_cast<const bf16*>(x_);
+  bf16* y = static_cast<bf16*>(y_);
+  const size_t lds = 2 * C * sizeof(float);
+  if (training) {
+    bn_reduce_kernel<<<elemwise_grid(R * C / 8), 256, lds, stream>>>(x, R, C,
+                                                                     sums);
+  }
+  bn_stats_kernel<<<(C + 255) / 256, 256, 0, stream>>>(
+      sums, C, R, mean, invstd, running_mean, running_var, momentum, eps,
+      training);
+  bn_normalize_kernel<<<elemwise_grid(R * C / 8), 256, 0, stream>>>(
+      x, mean, invstd, w, b, y, R, C);
+}
+
+void launch_batchnorm_bwd(const void* gy_, const void* x_, const float* w,
+                          const float* mean, const float* invstd, void* gx_,
+                          float* acc, int64_t R, int C, bool training,
+                          hipStream_t stream) {
+  const bf16* gy = static_cast<const bf16*>(gy_);
+  const bf16* x = static_cast<const bf16*>(x_);
+  bf16* gx = static_cast<bf16*>(gx_);
+  const size_t lds = 2 * C * sizeof(float);
+  bn_bwd_reduce_kernel<<<elemwise_grid(R * C / 8), 256, lds, stream>>>(
+      gy, x, mean, invstd, R, C, acc);
+  bn_bwd_gx_kernel<<<elemwise_grid(R * C / 8), 256, 0, stream>>>(
+      gy, x, mean, invstd, w, acc, gx, R, C, training);
+}
+
+}  // namespace colearn

@@ -1,0 +1,82 @@
+// Shared MFMA tile-GEMM core for gfx950 (guide §5 anatomy).
+//
+// Geometry: 256-thread block = 4 waves in a 2x2 grid; block tile 128x128;
+// per-wave output 64x64 = 4x4 fragments of v_mfma_f32_16x16x32_bf16;
+// K-step BK=32 (one MFMA K per step).  LDS tiles are K-major
+// ([row][k], 64 B rows) with a 16-B-slot XOR swizzle (slot ^= (row>>2)&3)
+// that makes the ds_read_b128 fragment reads bank-conflict-free
+// (guide §6 G4: row stride 64 B = 16 banks, 4 rows wrap the bank space,
+// so the swizzle must key on row>>2).
+//
+// Fragment maps (guide §3, 16x16x32 bf16):
+//   A: lane l holds A[i = l&15][k = (l>>4)*8 + j], j = 0..7
+//   B: lane l holds B[k = (l>>4)*8 + j][col = l&15]
+//   C/D: col = lane&15, row = (lane>>4)*4 + reg
+#pragma once
+
+#include "common.h"
+
+namespace colearn {
+
+typedef __attribute__((ext_vector_type(8))) short short8v_t;
+typedef __attribute__((ext_vector_type(4))) float f32x4_t;
+
+constexpr int BM = 128;
+constexpr int BN = 128;
+constexpr int BK = 32;
+constexpr int kBlockThreads = 256;
+
+// LDS tile: [128 rows][32 k] bf16 = 8 KiB; 4 16-B slots per row.
+constexpr int kSlotsPerRow = BK * 2 / 16;  // 4
+
+DEV_INLINE int lds_slot_byte(int row, int slot) {
+  return row * (BK * 2) + ((slot ^ ((row >> 2) & 3)) * 16);
+}
+
+// read one A/B fragment (8 bf16) for this lane from a K-major LDS tile.
+DEV_INLINE short8v_t frag_from_lds(const short* tile, int row_base) {
+  const int lane = threadIdx.x % kWave;
+  const int row = row_base + (lane & 15);
+  const int slot = lane >> 4;  // k-group: k = slot*8 + j
+  return *reinterpret_cast<const short8v_t*>(
+      reinterpret_cast<const char*>(tile) + lds_slot_byte(row, slot));
+}
+
+DEV_INLINE short8v_t mfma_frag_zero() {
+  short8v_t z = {0, 0, 0, 0, 0, 0, 0, 0};
+  return z;
+}
+
+// Epilogue: write acc[4][4] (each f32x4_t) of one wave into C (bf16,
+// row-major, ldc) with bounds checks.  Wave origin (wm, wn) in the block
+// tile; block origin (bm, bn) in the matrix.
+DEV_INLINE void store_tile_bf16(float const* acc_flat /* [4][4][4] */,
+                                bf16* __restrict__ c, int64_t ldc, int64_t M,
+                                int64_t N, int64_t bm, int64_t bn, int wm,
+                                int wn) {
+  const int lane = threadIdx.x % kWave;
+  const int col_in_frag = lane & 15;
+  const int row_grp = lane >> 4;  // rows row_grp*4 + j
+#pragma unroll
+  for (int m = 0; m < 4; ++m) {
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int64_t row = bm + wm + m * 16 + row_grp * 4 + j;
+        const int64_t col = bn + wn + n * 16 + col_in_frag;
+        if (row < M && col < N)
+          c[row * ldc + col] = f2b(acc_flat[(m * 4 + n) * 4 + j]);
+      }
+    }
+  }
+}
+
+// XCD-aware bijective block swizzle (guide T1): contiguous chunks per XCD.
+DEV_INLINE int xcd_swizzle(int bid, int nblocks) {
+  const int q = nblocks / kNumXCD, r = nblocks % kNumXCD;
+  const int xcd = bid % kNumXCD, idx = bid / kNumXCD;
+  return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+}
+
+}  // namespace colearn

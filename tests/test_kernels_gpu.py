@@ -1,0 +1,315 @@
+"""Numerics tests for the gfx950 HIP kernels vs plain-PyTorch fp32 oracles.
+
+Every kernel compares against the CPU/fp32 reference implementation on the
+SAME bf16-rounded inputs, so tolerances only cover output rounding +
+accumulation-order differences.  Asymmetric random operands are used
+everywhere (guide §5.4 rule 16: symmetric inputs miss operand/output
+transposes).
+"""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda"
+
+
+def _ext():
+    from colearn_federated_learning_amd.ops import require_ext
+
+    return require_ext()
+
+
+# ---- fused optimizers ------------------------------------------------------
+def test_fused_sgd_matches_cpu():
+    from colearn_federated_learning_amd.ops.fused_optim import FlatSGD
+
+    torch.manual_seed(0)
+    n = 1_000_003  # odd size exercises the scalar tail
+    w0 = torch.randn(n)
+    grads = [torch.randn(n) for _ in range(3)]
+    ref_w = w0.clone()
+    ref_g = torch.zeros(n)
+    ref = FlatSGD(ref_w, ref_g, lr=0.1, momentum=0.9, weight_decay=0.01)
+    gw = w0.clone().to(DEV)
+    gg = torch.zeros(n, device=DEV)
+    gpu = FlatSGD(gw, gg, lr=0.1, momentum=0.9, weight_decay=0.01)
+    for g in grads:
+        ref_g.copy_(g)
+        ref.step()
+        gg.copy_(g.to(DEV))
+        gpu.step()
+    torch.testing.assert_close(gw.cpu(), ref_w, rtol=1e-6, atol=1e-6)
+
+
+def test_fused_sgd_fedprox():
+    from colearn_federated_learning_amd.ops.fused_optim import FlatSGD
+
+    torch.manual_seed(1)
+    n = 4097
+    w0 = torch.randn(n)
+    refw = torch.randn(n)
+    g = torch.randn(n)
+    cpu_w = w0.clone()
+    cpu = FlatSGD(cpu_w, g.clone(), lr=0.05, momentum=0.9, fedprox_mu=0.01,
+                  flat_ref=refw)
+    cpu.step()
+    gw = w0.clone().to(DEV)
+    gpu = FlatSGD(gw, g.clone().to(DEV), lr=0.05, momentum=0.9,
+                  fedprox_mu=0.01, flat_ref=refw.to(DEV))
+    gpu.step()
+    torch.testing.assert_close(gw.cpu(), cpu_w, rtol=1e-6, atol=1e-6)
+
+
+def test_fused_adam_matches_cpu():
+    from colearn_federated_learning_amd.ops.fused_optim import FlatAdam
+
+    torch.manual_seed(2)
+    n = 123_457
+    w0 = torch.randn(n)
+    grads = [torch.randn(n) for _ in range(3)]
+    cpu_w = w0.clone()
+    cpu_g = torch.zeros(n)
+    cpu = FlatAdam(cpu_w, cpu_g, lr=1e-3, weight_decay=0.01)
+    gw = w0.clone().to(DEV)
+    gg = torch.zeros(n, device=DEV)
+    gpu = FlatAdam(gw, gg, lr=1e-3, weight_decay=0.01)
+    for g in grads:
+        cpu_g.copy_(g)
+        cpu.step()
+        gg.copy_(g.to(DEV))
+        gpu.step()
+    torch.testing.assert_close(gw.cpu(), cpu_w, rtol=1e-5, atol=1e-6)
+
+
+# ---- DP kernels ------------------------------------------------------------
+def test_dp_clip_and_reduce_gpu():
+    from colearn_federated_learning_amd.ops.dp import clip_and_reduce
+
+    torch.manual_seed(3)
+    b = 32
+    gs_cpu = [torch.randn(b, 1000), torch.randn(b, 17, 13)]
+    out_cpu = clip_and_reduce(gs_cpu, 0.7)
+    out_gpu = clip_and_reduce([g.to(DEV) for g in gs_cpu], 0.7)
+    for c, g in zip(out_cpu, out_gpu):
+        torch.testing.assert_close(g.cpu(), c, rtol=1e-4, atol=1e-6)
+
+
+def test_topk_gpu():
+    from colearn_federated_learning_amd.ops.topk import (
+        topk_compress,
+        topk_decompress,
+    )
+
+    torch.manual_seed(4)
+    v = torch.randn(1_000_000, device=DEV)
+    k = 5000
+    idx, val = topk_compress(v, k)
+    assert idx.shape == (k,)
+    # the selected |values| dominate the unselected ones
+    thresh = v.abs().topk(k).values.min()
+    assert (val.abs() >= thresh - 1e-5).all()
+    torch.testing.assert_close(val, v[idx.long()])
+    dense = topk_decompress(idx, val, v.numel())
+    torch.testing.assert_close(dense[idx.long()].cpu(), val.cpu())
+    assert (dense != 0).sum() == k
+
+
+# ---- norms -----------------------------------------------------------------
+def test_layernorm_gpu_vs_fp32():
+    from colearn_federated_learning_amd.ops.norms import ColearnLayerNorm
+
+    torch.manual_seed(5)
+    h, rows = 768, 512
+    x = torch.randn(rows, h).bfloat16()
+    ln = ColearnLayerNorm(h, eps=1e-5)
+    # CPU fp32 oracle on the same bf16-rounded input
+    x_cpu = x.float().requires_grad_(True)
+    ref = torch.nn.LayerNorm(h, eps=1e-5)
+    y_ref = ref(x_cpu)
+    gy = torch.randn(rows, h)
+    y_ref.backward(gy)
+
+    ln_gpu = ColearnLayerNorm(h, eps=1e-5).to(DEV)
+    xg = x.to(DEV).requires_grad_(True)
+    y = ln_gpu(xg)
+    y.backward(gy.bfloat16().to(DEV))
+    torch.testing.assert_close(y.float().cpu(), y_ref.detach(), rtol=2e-2,
+                               atol=2e-2)
+    torch.testing.assert_close(xg.grad.float().cpu(), x_cpu.grad, rtol=5e-2,
+                               atol=2e-2)
+    torch.testing.assert_close(ln_gpu.weight.grad.cpu(), ref.weight.grad,
+                               rtol=2e-2, atol=0.5)
+    torch.testing.assert_close(ln_gpu.bias.grad.cpu(), ref.bias.grad,
+                               rtol=2e-2, atol=0.5)
+
+
+def test_batchnorm_gpu_vs_fp32():
+    from colearn_federated_learning_amd.ops.norms import ColearnBatchNorm2d
+
+    torch.manual_seed(6)
+    n, c, h, w = 16, 64, 14, 14
+    x = torch.randn(n, c, h, w).bfloat16()
+    gy = torch.randn(n, c, h, w)
+
+    ref = torch.nn.BatchNorm2d(c)
+    x_cpu = x.float().requires_grad_(True)
+    y_ref = ref(x_cpu)
+    y_ref.backward(gy)
+
+    bn = ColearnBatchNorm2d(c).to(DEV)
+    xg = x.to(DEV).requires_grad_(True)
+    y = bn(xg)
+    y.backward(gy.bfloat16().to(DEV))
+    torch.testing.assert_close(y.float().cpu(), y_ref.detach(), rtol=3e-2,
+                               atol=3e-2)
+    torch.testing.assert_close(bn.running_mean.cpu(), ref.running_mean,
+                               rtol=1e-2, atol=1e-3)
+    torch.testing.assert_close(bn.running_var.cpu(), ref.running_var,
+                               rtol=1e-2, atol=1e-3)
+    torch.testing.assert_close(xg.grad.float().cpu(), x_cpu.grad, rtol=5e-2,
+                               atol=3e-2)
+    torch.testing.assert_close(bn.weight.grad.cpu(), ref.weight.grad,
+                               rtol=3e-2, atol=0.5)
+    torch.testing.assert_close(bn.bias.grad.cpu(), ref.bias.grad, rtol=3e-2,
+                               atol=0.5)
+
+
+# ---- GEMM family -----------------------------------------------------------
+@pytest.mark.parametrize("m,n,k", [(128, 128, 128), (256, 512, 768),
+                                   (100, 30, 52), (4096, 768, 768)])
+def test_gemm_nt(m, n, k):
+    torch.manual_seed(7)
+    a = torch.randn(m, k).bfloat16()
+    b = torch.randn(n, k).bfloat16()
+    ref = a.float() @ b.float().T
+    out = _ext().gemm_nt(a.to(DEV), b.to(DEV))
+    tol = 0.1 * max(1.0, (k / 512) ** 0.5)
+    torch.testing.assert_close(out.float().cpu(), ref, rtol=2e-2, atol=tol)
+
+
+def test_gemm_identity_asymmetric():
+    """A = I with an ASYMMETRIC B catches swapped output maps (§5.4 r16)."""
+    m = 64
+    a = torch.eye(m).bfloat16()
+    b = torch.arange(m * m, dtype=torch.float32).reshape(m, m)
+    b = (b / (m * m)).bfloat16()
+    out = _ext().gemm_nt(a.to(DEV), b.to(DEV))  # I @ B^T = B^T
+    torch.testing.assert_close(out.float().cpu(), b.float().T, rtol=1e-3,
+                               atol=1e-3)
+
+
+def test_gemm_nn_tn():
+    torch.manual_seed(8)
+    m, n, k = 160, 96, 224
+    a = torch.randn(m, k).bfloat16()
+    b = torch.randn(k, n).bfloat16()
+    ref = a.float() @ b.float()
+    out = _ext().gemm_nn(a.to(DEV), b.to(DEV))
+    torch.testing.assert_close(out.float().cpu(), ref, rtol=2e-2, atol=0.1)
+    at = torch.randn(k, m).bfloat16()
+    ref2 = at.float().T @ b.float()
+    out2 = _ext().gemm_tn(at.to(DEV), b.to(DEV))
+    torch.testing.assert_close(out2.float().cpu(), ref2, rtol=2e-2, atol=0.1)
+
+
+def test_bmm():
+    torch.manual_seed(9)
+    B, m, n, k = 6, 128, 128, 64
+    a = torch.randn(B, m, k).bfloat16()
+    b = torch.randn(B, n, k).bfloat16()
+    ref = torch.matmul(a.float(), b.float().transpose(1, 2))
+    out = _ext().bmm_nt(a.to(DEV), b.to(DEV))
+    torch.testing.assert_close(out.float().cpu(), ref, rtol=2e-2, atol=0.1)
+    b2 = torch.randn(B, k, n).bfloat16()
+    ref2 = torch.matmul(a.float(), b2.float())
+    out2 = _ext().bmm_nn(a.to(DEV), b2.to(DEV))
+    torch.testing.assert_close(out2.float().cpu(), ref2, rtol=2e-2, atol=0.1)
+
+
+# ---- conv family -----------------------------------------------------------
+@pytest.mark.parametrize(
+    "n,c,hw,co,ks,stride,pad",
+    [
+        (4, 64, 16, 64, 3, 1, 1),     # resnet body 3x3
+        (4, 64, 16, 128, 1, 2, 0),    # downsample 1x1/2
+        (4, 128, 8, 128, 3, 2, 1),    # strided 3x3
+        (2, 3, 32, 64, 3, 1, 1),      # stem (C=3 scalar path)
+    ],
+)
+def test_conv_fwd(n, c, hw, co, ks, stride, pad):
+    torch.manual_seed(10)
+    x = torch.randn(n, c, hw, hw).bfloat16()
+    w = (torch.randn(co, c, ks, ks) * 0.1).bfloat16()
+    ref = torch.nn.functional.conv2d(x.float(), w.float(), None, stride, pad)
+    out = _ext().conv2d_fwd(x.to(DEV), w.to(DEV), stride, pad)
+    torch.testing.assert_close(out.float().cpu(), ref, rtol=3e-2, atol=0.1)
+
+
+@pytest.mark.parametrize(
+    "n,c,hw,co,ks,stride,pad",
+    [
+        (4, 64, 16, 64, 3, 1, 1),
+        (4, 64, 16, 128, 1, 2, 0),
+        (2, 128, 8, 128, 3, 2, 1),
+    ],
+)
+def test_conv_bwd(n, c, hw, co, ks, stride, pad):
+    torch.manual_seed(11)
+    x = torch.randn(n, c, hw, hw).bfloat16()
+    w = (torch.randn(co, c, ks, ks) * 0.1).bfloat16()
+    po = (hw + 2 * pad - ks) // stride + 1
+    gy = torch.randn(n, co, po, po).bfloat16()
+    gx_ref = torch.nn.grad.conv2d_input(
+        x.shape, w.float(), gy.float(), stride, pad
+    )
+    gw_ref = torch.nn.grad.conv2d_weight(
+        x.float(), w.shape, gy.float(), stride, pad
+    )
+    m = _ext()
+    gx = m.conv2d_dgrad(gy.to(DEV), w.to(DEV), hw, hw, stride, pad)
+    gw = m.conv2d_wgrad(gy.to(DEV), x.to(DEV), ks, ks, stride, pad)
+    torch.testing.assert_close(gx.float().cpu(), gx_ref, rtol=3e-2, atol=0.1)
+    npq = n * po * po
+    tol = 0.15 * max(1.0, (npq / 1024) ** 0.5)
+    torch.testing.assert_close(gw.float().cpu(), gw_ref, rtol=3e-2, atol=tol)
+
+
+# ---- end-to-end ------------------------------------------------------------
+def test_resnet18_step_native_vs_cpu():
+    """One bf16 training step on GPU (native kernels) vs fp32 CPU oracle:
+    losses must agree to bf16 tolerance."""
+    from colearn_federated_learning_amd.config import DataConfig, ModelConfig
+    from colearn_federated_learning_amd.models import build_model
+
+    torch.manual_seed(12)
+    mcpu = build_model(ModelConfig(name="resnet18"), DataConfig())
+    mgpu = build_model(ModelConfig(name="resnet18"), DataConfig())
+    mgpu.load_state_dict(mcpu.state_dict())
+    mgpu = mgpu.to(DEV)
+    x = torch.randn(8, 3, 32, 32)
+    y = torch.randint(0, 10, (8,))
+    loss_cpu = mcpu.loss(x, y)
+    with torch.autocast("cuda", torch.bfloat16):
+        loss_gpu = mgpu.loss(x.to(DEV), y.to(DEV))
+    assert abs(loss_cpu.item() - loss_gpu.item()) < 0.1
+    loss_gpu.backward()
+    gnorm = sum(
+        float(p.grad.norm()) for p in mgpu.parameters() if p.grad is not None
+    )
+    assert gnorm > 0 and torch.isfinite(torch.tensor(gnorm))
+
+
+def test_native_extension_required_on_gpu(monkeypatch):
+    """The GPU path must fail loudly without the extension, never fall back."""
+    import colearn_federated_learning_amd.ops as ops
+
+    monkeypatch.setattr(ops, "_EXT", None)
+    monkeypatch.setattr(ops, "_EXT_TRIED", True)
+    from colearn_federated_learning_amd.ops.gemm import colearn_linear
+
+    x = torch.randn(8, 16).bfloat16().to(DEV)
+    w = torch.randn(4, 16).bfloat16().to(DEV)
+    with pytest.raises(RuntimeError, match="extension"):
+        colearn_linear(x, w)
