@@ -10,8 +10,8 @@
 
 #include <torch/extension.h>
 
-#include <c10/hip/HIPGuard.h>
-#include <c10/hip/HIPStream.h>
+#include <ATen/hip/impl/HIPGuardImplMasqueradingAsCUDA.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 
 #include "api.h"
 
@@ -27,7 +27,9 @@ namespace colearn {
 
 namespace {
 
-hipStream_t stream() { return c10::hip::getCurrentHIPStream().stream(); }
+hipStream_t stream() {
+  return c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+}
 
 void check_f32(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kFloat32, name,
@@ -44,7 +46,7 @@ void fused_sgd(torch::Tensor w, torch::Tensor g, torch::Tensor m,
                double mu) {
   check_f32(w, "w");
   check_f32(g, "g");
-  const c10::hip::HIPGuard guard(w.device());
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(w.device());
   const bool has_mom = m.numel() > 0;
   const bool has_prox = ref.numel() > 0;
   launch_fused_sgd(w.data_ptr<float>(), g.data_ptr<float>(),
@@ -61,7 +63,7 @@ void fused_adam(torch::Tensor w, torch::Tensor g, torch::Tensor m,
                 double b2, double eps, double wd, double mu, double bc1,
                 double bc2) {
   check_f32(w, "w");
-  const c10::hip::HIPGuard guard(w.device());
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(w.device());
   const bool has_prox = ref.numel() > 0;
   launch_fused_adam(
       w.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
@@ -77,7 +79,7 @@ void fused_adam(torch::Tensor w, torch::Tensor g, torch::Tensor m,
 void sq_norm_per_row(torch::Tensor g, torch::Tensor sq) {
   check_f32(g, "g");
   check_f32(sq, "sq");
-  const c10::hip::HIPGuard guard(g.device());
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(g.device());
   auto gc = g.contiguous();
   launch_sq_norm_per_row(gc.data_ptr<float>(), sq.data_ptr<float>(),
                          gc.size(0), gc.size(1), stream());
@@ -87,7 +89,7 @@ void sq_norm_per_row(torch::Tensor g, torch::Tensor sq) {
 void scale_rows_mean(torch::Tensor g, torch::Tensor scale,
                      torch::Tensor out) {
   check_f32(g, "g");
-  const c10::hip::HIPGuard guard(g.device());
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(g.device());
   auto gc = g.contiguous();
   launch_scale_rows_mean(gc.data_ptr<float>(), scale.data_ptr<float>(),
                          out.data_ptr<float>(), gc.size(0), gc.size(1),
@@ -98,7 +100,7 @@ void scale_rows_mean(torch::Tensor g, torch::Tensor scale,
 std::tuple<torch::Tensor, torch::Tensor> topk_compress(torch::Tensor flat,
                                                        int64_t k) {
   check_f32(flat, "flat");
-  const c10::hip::HIPGuard guard(flat.device());
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(flat.device());
   auto fc = flat.contiguous();
   k = std::min<int64_t>(k, fc.numel());
   auto idx = torch::empty({k}, fc.options().dtype(torch::kInt32));
@@ -114,7 +116,7 @@ std::tuple<torch::Tensor, torch::Tensor> topk_compress(torch::Tensor flat,
 void scatter_add_pairs(torch::Tensor out, torch::Tensor idx,
                        torch::Tensor val) {
   check_f32(out, "out");
-  const c10::hip::HIPGuard guard(out.device());
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(out.device());
   launch_scatter_add_pairs(out.data_ptr<float>(), idx.data_ptr<int>(),
                            val.data_ptr<float>(), idx.numel(), stream());
   HIP_CHECK_LAST();
@@ -124,7 +126,7 @@ void scatter_add_pairs(torch::Tensor out, torch::Tensor idx,
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> layernorm_fwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor b, double eps) {
   check_bf16(x, "x");
-  const c10::hip::HIPGuard guard(x.device());
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   auto xc = x.contiguous();
   const int H = static_cast<int>(x.size(-1));
   const int64_t rows = x.numel() / H;
@@ -146,7 +148,7 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> layernorm_bwd(
     torch::Tensor gy, torch::Tensor x, torch::Tensor w, torch::Tensor mean,
     torch::Tensor invstd) {
   check_bf16(gy, "gy");
-  const c10::hip::HIPGuard guard(x.device());
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   auto gc = gy.contiguous(), xc = x.contiguous();
   auto mc = mean.contiguous(), ic = invstd.contiguous();
   const int H = static_cast<int>(x.size(-1));
@@ -173,7 +175,7 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
   TORCH_CHECK(x.dim() == 4, "batchnorm expects NCHW tensor");
   const int C = static_cast<int>(x.size(1));
   TORCH_CHECK(C % 8 == 0, "native batchnorm needs C % 8 == 0, got ", C);
-  const c10::hip::HIPGuard guard(x.device());
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
   const int64_t R = x.numel() / C;
   auto y = torch::empty_like(xc);
@@ -196,7 +198,7 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
     torch::Tensor invstd, bool training) {
   check_bf16(gy, "gy");
   const int C = static_cast<int>(x.size(1));
-  const c10::hip::HIPGuard guard(x.device());
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   auto gc = gy.contiguous(torch::MemoryFormat::ChannelsLast);
   auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
   const int64_t R = x.numel() / C;
@@ -220,7 +222,7 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b) {
   check_bf16(b, "b");
   TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(1),
               "gemm_nt shape mismatch");
-  const c10::hip::HIPGuard guard(a.device());
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(a.device());
   auto ac = a.contiguous(), bc = b.contiguous();
   const int64_t M = a.size(0), N = b.size(0), K = a.size(1);
   auto c = torch::empty({M, N}, a.options());
@@ -242,7 +244,7 @@ torch::Tensor gemm_tn(torch::Tensor a, torch::Tensor b) {
 torch::Tensor bmm_nt(torch::Tensor a, torch::Tensor b) {
   check_bf16(a, "a");
   TORCH_CHECK(a.dim() == 3 && b.dim() == 3, "bmm expects 3-D");
-  const c10::hip::HIPGuard guard(a.device());
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(a.device());
   auto ac = a.contiguous(), bc = b.contiguous();
   const int64_t B = a.size(0), M = a.size(1), N = b.size(1), K = a.size(2);
   auto c = torch::empty({B, M, N}, a.options());
@@ -283,7 +285,7 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
                          int64_t pad) {
   check_bf16(x, "x");
   check_bf16(w, "w");
-  const c10::hip::HIPGuard guard(x.device());
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   ConvShape cs = make_shape(x, w, stride, pad);
   auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
   // repack weight K-major: [CO][R][S][CI]
@@ -299,7 +301,7 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
 torch::Tensor conv2d_dgrad(torch::Tensor gy, torch::Tensor w, int64_t H,
                            int64_t W, int64_t stride, int64_t pad) {
   check_bf16(gy, "gy");
-  const c10::hip::HIPGuard guard(gy.device());
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(gy.device());
   ConvShape cs;
   cs.N = static_cast<int>(gy.size(0));
   cs.K = static_cast<int>(gy.size(1));
@@ -327,7 +329,7 @@ torch::Tensor conv2d_dgrad(torch::Tensor gy, torch::Tensor w, int64_t H,
 torch::Tensor conv2d_wgrad(torch::Tensor gy, torch::Tensor x, int64_t R,
                            int64_t S, int64_t stride, int64_t pad) {
   check_bf16(gy, "gy");
-  const c10::hip::HIPGuard guard(gy.device());
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(gy.device());
   ConvShape cs;
   cs.N = static_cast<int>(x.size(0));
   cs.C = static_cast<int>(x.size(1));
