@@ -29,7 +29,8 @@ void launch_sq_norm_per_row(const float* g, float* sq, int64_t rows,
 void launch_scale_rows_mean(const float* g, const float* scale, float* out,
                             int64_t rows, int64_t cols, hipStream_t stream);
 void launch_topk(const float* v, int64_t n, int64_t k, int* idx, float* val,
-                 unsigned* workspace, hipStream_t stream);
+                 unsigned* workspace, int* cand, int cand_cap,
+                 hipStream_t stream);
 void launch_scatter_add_pairs(float* out, const int* idx, const float* val,
                               int64_t k, hipStream_t stream);
 void launch_layernorm_fwd(const void* x, const float* w, const float* b,

@@ -106,9 +106,12 @@ std::tuple<torch::Tensor, torch::Tensor> topk_compress(torch::Tensor flat,
   auto idx = torch::empty({k}, fc.options().dtype(torch::kInt32));
   auto val = torch::empty({k}, fc.options());
   auto ws = torch::zeros({520}, fc.options().dtype(torch::kInt32));
+  const int cand_cap = static_cast<int>(k) + 4096;
+  auto cand = torch::empty({cand_cap}, fc.options().dtype(torch::kInt32));
   launch_topk(fc.data_ptr<float>(), fc.numel(), k, idx.data_ptr<int>(),
               val.data_ptr<float>(),
-              reinterpret_cast<unsigned*>(ws.data_ptr<int>()), stream());
+              reinterpret_cast<unsigned*>(ws.data_ptr<int>()),
+              cand.data_ptr<int>(), cand_cap, stream());
   HIP_CHECK_LAST();
   return {idx, val};
 }

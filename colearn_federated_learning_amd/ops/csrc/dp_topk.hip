@@ -159,23 +159,59 @@ __global__ void topk_select2_kernel(const unsigned* __restrict__ sel1,
   }
 }
 
+// entries strictly above the 16-bit threshold go straight to the output;
+// entries AT the threshold become candidates for the exact final pass.
 __global__ void topk_compact_kernel(const float* __restrict__ v, int64_t n,
                                     unsigned* __restrict__ ctl /*out2*/,
                                     int* __restrict__ idx,
-                                    float* __restrict__ val, int64_t k) {
+                                    float* __restrict__ val, int64_t k,
+                                    int* __restrict__ cand, int cand_cap) {
   const unsigned thresh = ctl[0];
-  const unsigned need_eq = ctl[1];
   int64_t i = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
   for (; i < n; i += grid_stride()) {
     const unsigned key = absbits(v[i]) & 0xFFFF0000u;
     if (key < thresh) continue;
     if (key == thresh) {
-      if (atomicAdd(&ctl[2], 1u) >= need_eq) continue;  // tie budget spent
+      const unsigned c = atomicAdd(&ctl[2], 1u);
+      if (c < static_cast<unsigned>(cand_cap)) cand[c] = static_cast<int>(i);
+      continue;
     }
     const unsigned pos = atomicAdd(&ctl[3], 1u);
     if (pos < static_cast<unsigned>(k)) {
       idx[pos] = static_cast<int>(i);
       val[pos] = v[i];
+    }
+  }
+}
+
+// exact tie resolution: among the candidates (all sharing the threshold's
+// 16-bit key), rank by (|v| desc, index asc) and keep the need_eq best.
+// One block; O(c^2/threads) comparisons, c is the candidate count (small:
+// one 16-bit bucket of the distribution).
+__global__ void topk_ties_kernel(const float* __restrict__ v,
+                                 unsigned* __restrict__ ctl,
+                                 const int* __restrict__ cand, int cand_cap,
+                                 int* __restrict__ idx,
+                                 float* __restrict__ val, int64_t k) {
+  const unsigned need_eq = ctl[1];
+  if (need_eq == 0) return;
+  int c = static_cast<int>(ctl[2]);
+  if (c > cand_cap) c = cand_cap;
+  for (int a = threadIdx.x; a < c; a += blockDim.x) {
+    const int ia = cand[a];
+    const float va = fabsf(v[ia]);
+    unsigned rank = 0;
+    for (int b2 = 0; b2 < c; ++b2) {
+      const int ib = cand[b2];
+      const float vb = fabsf(v[ib]);
+      rank += (vb > va) || (vb == va && ib < ia);
+    }
+    if (rank < need_eq) {
+      const unsigned pos = atomicAdd(&ctl[3], 1u);
+      if (pos < static_cast<unsigned>(k)) {
+        idx[pos] = ia;
+        val[pos] = v[ia];
+      }
     }
   }
 }
@@ -205,7 +241,7 @@ void launch_scale_rows_mean(const float* g, const float* scale, float* out,
 
 void launch_topk(const float* v, int64_t n, int64_t k, int* idx, float* val,
                  unsigned* workspace /* >= 520 unsigned, zeroed */,
-                 hipStream_t stream) {
+                 int* cand, int cand_cap, hipStream_t stream) {
   unsigned* hist1 = workspace;        // 256
   unsigned* hist2 = workspace + 256;  // 256
   unsigned* sel1 = workspace + 512;   // 4
@@ -215,7 +251,10 @@ void launch_topk(const float* v, int64_t n, int64_t k, int* idx, float* val,
   topk_select_kernel<<<1, 256, 0, stream>>>(hist1, k, sel1);
   topk_hist2_kernel<<<grid, 256, 0, stream>>>(v, n, sel1, hist2);
   topk_select2_kernel<<<1, 256, 0, stream>>>(sel1, hist2, ctl);
-  topk_compact_kernel<<<grid, 256, 0, stream>>>(v, n, ctl, idx, val, k);
+  topk_compact_kernel<<<grid, 256, 0, stream>>>(v, n, ctl, idx, val, k,
+                                                cand, cand_cap);
+  topk_ties_kernel<<<1, 256, 0, stream>>>(v, ctl, cand, cand_cap, idx, val,
+                                          k);
 }
 
 void launch_scatter_add_pairs(float* out, const int* idx, const float* val,
