@@ -93,7 +93,10 @@ class FederatedClient:
         dev = self.device
         use_bf16 = tc.dtype == "bfloat16" and dev.type == "cuda"
         gen = self._gen(round_idx, "batches")
-        total_loss, total_n, steps = 0.0, 0, 0
+        # accumulate the loss ON DEVICE: a float() per step would force a
+        # GPU sync per training step and serialize the whole pipeline.
+        loss_acc = torch.zeros((), device=dev)
+        total_n, steps = 0, 0
         devices = [dev] if dev.type == "cuda" else []
         with torch.random.fork_rng(devices=devices):
             torch.manual_seed(
@@ -117,10 +120,10 @@ class FederatedClient:
                         loss.backward()
                     optimizer.step()
                     bs = x.shape[0]
-                    total_loss += float(loss.detach()) * bs
+                    loss_acc += loss.detach() * bs
                     total_n += bs
                     steps += 1
-        avg = total_loss / max(total_n, 1)
+        avg = float(loss_acc) / max(total_n, 1)
         res = FitResult(n_samples=total_n, loss=avg)
         if self.bus is not None:
             self.bus.publish(

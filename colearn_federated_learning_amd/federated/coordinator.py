@@ -81,6 +81,11 @@ class FederatedServer:
             if cid % world != rank:
                 continue
             shard = make_client_dataset(cfg.data, cfg.model, cid, shards[cid])
+            if self.device.type == "cuda":
+                # 288 GB HBM3E per GPU: client shards live device-resident
+                # for the whole session instead of streaming from host.
+                shard.x = shard.x.to(self.device)
+                shard.y = shard.y.to(self.device)
             c = FederatedClient(cid, cfg, shard, self.device, bus=self.bus)
             c.announce()
             self.clients[cid] = c
