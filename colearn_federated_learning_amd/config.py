@@ -52,6 +52,8 @@ class TrainingConfig:
     # FedProx proximal term strength (0 => plain FedAvg local objective)
     fedprox_mu: float = 0.0
     max_local_steps: Optional[int] = None  # cap steps/epoch (bench + tests)
+    # capture the client train step in a hipGraph on the GPU (steprunner.py)
+    use_graphs: bool = True
 
 
 @dataclass

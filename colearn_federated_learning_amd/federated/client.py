@@ -78,8 +78,9 @@ class FederatedClient:
     def fit(
         self,
         model: torch.nn.Module,
-        optimizer: torch.optim.Optimizer,
+        optimizer,
         round_idx: int,
+        runner=None,
     ) -> FitResult:
         """Local training for ``local_epochs`` over this client's shard.
 
@@ -108,17 +109,23 @@ class FederatedClient:
                         break
                     x = x.to(dev, non_blocking=True)
                     y = y.to(dev, non_blocking=True)
-                    optimizer.zero_grad()
-                    if self.cfg.dp.enabled:
+                    if runner is not None and runner.matches(x):
+                        # hipGraph fast path: fwd+bwd+optimizer in ONE
+                        # graph replay (steprunner.py)
+                        loss = runner.step(x, y)
+                    elif self.cfg.dp.enabled:
+                        optimizer.zero_grad()
                         loss = _per_sample_clipped_backward(
                             model, x, y, self.cfg.dp.clip_norm, use_bf16
                         )
+                        optimizer.step()
                     else:
+                        optimizer.zero_grad()
                         with torch.autocast("cuda", torch.bfloat16,
                                             enabled=use_bf16):
                             loss = model.loss(x, y)
                         loss.backward()
-                    optimizer.step()
+                        optimizer.step()
                     bs = x.shape[0]
                     loss_acc += loss.detach() * bs
                     total_n += bs

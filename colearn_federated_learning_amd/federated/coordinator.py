@@ -105,6 +105,19 @@ class FederatedServer:
             cfg.metrics_path if rank == 0 else None
         )
         self.timer = PhaseTimer(self.device)
+        # hipGraph-captured client step (see steprunner.py); one capture
+        # shared by every virtual client on this rank.
+        sample_shape = (
+            (cfg.model.in_channels, cfg.model.image_size,
+             cfg.model.image_size)
+            if cfg.data.dataset != "text_synth" else None
+        )
+        from .steprunner import make_step_runner
+
+        self.step_runner = make_step_runner(
+            self.model, self.optimizer, self.flat, cfg, self.device,
+            sample_shape,
+        )
         if cfg.resume and cfg.checkpoint_dir:
             self._resume()
 
@@ -118,9 +131,11 @@ class FederatedServer:
             k: v.to(self.device) for k, v in ck["model_state_dict"].items()
         }
         # restore only trainable parameters into the flat buffer; buffers
-        # (BN running stats) load through the module.
+        # (BN running stats) load through the module.  copy_ in place: the
+        # optimizer's FedProx reference and any captured hipGraph alias
+        # this exact tensor.
         self.model.load_state_dict(state, strict=False)
-        self.global_flat = self.flat.flat_from_state_dict(state)
+        self.global_flat.copy_(self.flat.flat_from_state_dict(state))
         self.flat.load_flat(self.global_flat)
         self.sampler.load_state_dict(ck["sampler_state"])
         self.start_round = ck["round"] + 1
@@ -172,7 +187,8 @@ class FederatedServer:
                 t_fit = time.perf_counter()
                 self.flat.load_flat(self.global_flat)
                 self.optimizer.reset()
-                res = client.fit(self.model, self.optimizer, round_idx)
+                res = client.fit(self.model, self.optimizer, round_idx,
+                                 runner=self.step_runner)
                 if (cfg.straggler_timeout_s is not None
                         and time.perf_counter() - t_fit
                         > cfg.straggler_timeout_s):

@@ -36,10 +36,27 @@ class _Conv2d(nn.Module):
         return colearn_conv2d(x, self.weight, self.stride, self.padding)
 
 
-def _norm(planes: int, group_norm: bool) -> nn.Module:
+class _GNAct(nn.Module):
+    """GroupNorm with the same (x, residual) interface as the fused BN —
+    the DP-mode substitute (eager add + relu)."""
+
+    def __init__(self, planes: int, relu: bool) -> None:
+        super().__init__()
+        self.gn = nn.GroupNorm(min(32, planes), planes)
+        self.do_relu = relu
+
+    def forward(self, x: torch.Tensor, residual=None) -> torch.Tensor:
+        y = self.gn(x)
+        if residual is not None:
+            y = y + residual
+        return F.relu(y) if self.do_relu else y
+
+
+def _norm(planes: int, group_norm: bool, relu: bool = False) -> nn.Module:
     if group_norm:
-        return nn.GroupNorm(min(32, planes), planes)
-    return ColearnBatchNorm2d(planes)
+        return _GNAct(planes, relu)
+    # relu/residual fuse into the BN kernel sweeps on gfx950
+    return ColearnBatchNorm2d(planes, relu=relu)
 
 
 class BasicBlock(nn.Module):
@@ -49,9 +66,9 @@ class BasicBlock(nn.Module):
                  group_norm: bool) -> None:
         super().__init__()
         self.conv1 = _Conv2d(inplanes, planes, 3, stride, 1)
-        self.bn1 = _norm(planes, group_norm)
+        self.bn1 = _norm(planes, group_norm, relu=True)
         self.conv2 = _Conv2d(planes, planes, 3, 1, 1)
-        self.bn2 = _norm(planes, group_norm)
+        self.bn2 = _norm(planes, group_norm, relu=True)  # fused +idt, relu
         self.downsample: Optional[nn.Module] = None
         if stride != 1 or inplanes != planes:
             self.downsample = nn.Sequential(
@@ -61,9 +78,8 @@ class BasicBlock(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         idt = x if self.downsample is None else self.downsample(x)
-        out = F.relu(self.bn1(self.conv1(x)))
-        out = self.bn2(self.conv2(out))
-        return F.relu(out + idt)
+        out = self.bn1(self.conv1(x))
+        return self.bn2(self.conv2(out), residual=idt)
 
 
 class Bottleneck(nn.Module):
@@ -74,11 +90,11 @@ class Bottleneck(nn.Module):
         super().__init__()
         width = planes
         self.conv1 = _Conv2d(inplanes, width, 1, 1, 0)
-        self.bn1 = _norm(width, group_norm)
+        self.bn1 = _norm(width, group_norm, relu=True)
         self.conv2 = _Conv2d(width, width, 3, stride, 1)
-        self.bn2 = _norm(width, group_norm)
+        self.bn2 = _norm(width, group_norm, relu=True)
         self.conv3 = _Conv2d(width, planes * 4, 1, 1, 0)
-        self.bn3 = _norm(planes * 4, group_norm)
+        self.bn3 = _norm(planes * 4, group_norm, relu=True)  # fused +idt
         self.downsample: Optional[nn.Module] = None
         if stride != 1 or inplanes != planes * 4:
             self.downsample = nn.Sequential(
@@ -88,10 +104,9 @@ class Bottleneck(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         idt = x if self.downsample is None else self.downsample(x)
-        out = F.relu(self.bn1(self.conv1(x)))
-        out = F.relu(self.bn2(self.conv2(out)))
-        out = self.bn3(self.conv3(out))
-        return F.relu(out + idt)
+        out = self.bn1(self.conv1(x))
+        out = self.bn2(self.conv2(out))
+        return self.bn3(self.conv3(out), residual=idt)
 
 
 class ResNet(ColearnModel):
@@ -108,14 +123,14 @@ class ResNet(ColearnModel):
         self.inplanes = 64
         if small_input:
             self.stem = nn.Sequential(
-                _Conv2d(in_channels, 64, 3, 1, 1), _norm(64, group_norm),
-                nn.ReLU(inplace=True),
+                _Conv2d(in_channels, 64, 3, 1, 1),
+                _norm(64, group_norm, relu=True),
             )
             self.pool = nn.Identity()
         else:
             self.stem = nn.Sequential(
-                _Conv2d(in_channels, 64, 7, 2, 3), _norm(64, group_norm),
-                nn.ReLU(inplace=True),
+                _Conv2d(in_channels, 64, 7, 2, 3),
+                _norm(64, group_norm, relu=True),
             )
             self.pool = nn.MaxPool2d(3, 2, 1)
         self.layer1 = self._make_layer(block, 64, layers[0], 1, group_norm)

@@ -59,16 +59,17 @@ void fused_sgd(torch::Tensor w, torch::Tensor g, torch::Tensor m,
 }
 
 void fused_adam(torch::Tensor w, torch::Tensor g, torch::Tensor m,
-                torch::Tensor v, torch::Tensor ref, double lr, double b1,
-                double b2, double eps, double wd, double mu, double bc1,
-                double bc2) {
+                torch::Tensor v, torch::Tensor ref, torch::Tensor bc,
+                double lr, double b1, double b2, double eps, double wd,
+                double mu, double bc1, double bc2) {
   check_f32(w, "w");
   const c10::hip::HIPGuardMasqueradingAsCUDA guard(w.device());
   const bool has_prox = ref.numel() > 0;
   launch_fused_adam(
       w.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
       v.data_ptr<float>(), has_prox ? ref.data_ptr<float>() : nullptr,
-      w.numel(), static_cast<float>(lr), static_cast<float>(b1),
+      bc.numel() > 0 ? bc.data_ptr<float>() : nullptr, w.numel(),
+      static_cast<float>(lr), static_cast<float>(b1),
       static_cast<float>(b2), static_cast<float>(eps),
       static_cast<float>(wd), static_cast<float>(mu),
       static_cast<float>(bc1), static_cast<float>(bc2), has_prox, stream());
@@ -172,51 +173,60 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> layernorm_bwd(
 // ---- BatchNorm (channels-last) -------------------------------------------
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor b,
-    torch::Tensor running_mean, torch::Tensor running_var, bool training,
-    double momentum, double eps) {
+    torch::Tensor running_mean, torch::Tensor running_var,
+    torch::Tensor residual, bool training, double momentum, double eps,
+    bool relu) {
   check_bf16(x, "x");
   TORCH_CHECK(x.dim() == 4, "batchnorm expects NCHW tensor");
   const int C = static_cast<int>(x.size(1));
   TORCH_CHECK(C % 8 == 0, "native batchnorm needs C % 8 == 0, got ", C);
   const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
+  const bool has_res = residual.numel() > 0;
+  torch::Tensor rc;
+  if (has_res) rc = residual.contiguous(torch::MemoryFormat::ChannelsLast);
   const int64_t R = x.numel() / C;
   auto y = torch::empty_like(xc);
   auto mean = torch::empty({C}, x.options().dtype(torch::kFloat32));
   auto invstd = torch::empty_like(mean);
   auto sums = torch::zeros({2 * C}, mean.options());
   launch_batchnorm_fwd(
-      xc.data_ptr(), w.data_ptr<float>(),
-      b.data_ptr<float>(), running_mean.data_ptr<float>(),
-      running_var.data_ptr<float>(), y.data_ptr(),
-      mean.data_ptr<float>(), invstd.data_ptr<float>(),
+      xc.data_ptr(), has_res ? rc.data_ptr() : nullptr,
+      w.data_ptr<float>(), b.data_ptr<float>(),
+      running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
+      y.data_ptr(), mean.data_ptr<float>(), invstd.data_ptr<float>(),
       sums.data_ptr<float>(), R, C, static_cast<float>(momentum),
-      static_cast<float>(eps), training, stream());
+      static_cast<float>(eps), training, relu, stream());
   HIP_CHECK_LAST();
   return {y, mean, invstd};
 }
 
-std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
-    torch::Tensor gy, torch::Tensor x, torch::Tensor w, torch::Tensor mean,
-    torch::Tensor invstd, bool training) {
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
+batchnorm_bwd(torch::Tensor gy, torch::Tensor y, torch::Tensor x,
+              torch::Tensor w, torch::Tensor mean, torch::Tensor invstd,
+              bool training, bool relu) {
   check_bf16(gy, "gy");
   const int C = static_cast<int>(x.size(1));
   const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   auto gc = gy.contiguous(torch::MemoryFormat::ChannelsLast);
   auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
+  torch::Tensor yc;
+  if (relu) yc = y.contiguous(torch::MemoryFormat::ChannelsLast);
   const int64_t R = x.numel() / C;
   auto gx = torch::empty_like(xc);
+  torch::Tensor gym;  // masked gy == the residual grad
+  if (relu) gym = torch::empty_like(gc);
   auto acc = torch::zeros({2 * C}, x.options().dtype(torch::kFloat32));
-  launch_batchnorm_bwd(gc.data_ptr(),
-                       xc.data_ptr(),
-                       w.data_ptr<float>(), mean.data_ptr<float>(),
-                       invstd.data_ptr<float>(),
-                       gx.data_ptr(),
-                       acc.data_ptr<float>(), R, C, training, stream());
+  launch_batchnorm_bwd(gc.data_ptr(), relu ? yc.data_ptr() : nullptr,
+                       xc.data_ptr(), w.data_ptr<float>(),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       gx.data_ptr(), relu ? gym.data_ptr() : nullptr,
+                       acc.data_ptr<float>(), R, C, training, relu,
+                       stream());
   HIP_CHECK_LAST();
   auto gb = acc.narrow(0, 0, C).clone();
   auto gw = acc.narrow(0, C, C).clone();
-  return {gx, gw, gb};
+  return {gx, gw, gb, relu ? gym : gc};
 }
 
 // ---- GEMM family ----------------------------------------------------------

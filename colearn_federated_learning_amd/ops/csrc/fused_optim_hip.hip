@@ -69,13 +69,22 @@ __global__ void fused_adam_kernel(float* __restrict__ w,
                                   const float* __restrict__ g,
                                   float* __restrict__ m,
                                   float* __restrict__ v,
-                                  const float* __restrict__ ref, int64_t n,
-                                  float lr, float b1, float b2, float eps,
-                                  float wd, float mu, float bc1, float bc2) {
+                                  const float* __restrict__ ref,
+                                  const float* __restrict__ bc /*[2] or null*/,
+                                  int64_t n, float lr, float b1, float b2,
+                                  float eps, float wd, float mu, float bc1,
+                                  float bc2) {
+  // device-side bias correction: under hipGraph replay the scalar args are
+  // frozen at capture, so a live [bc1, bc2] tensor overrides them.
+  if (bc != nullptr) {
+    bc1 = bc[0];
+    bc2 = bc[1];
+  }
   int64_t i = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) * 4;
   const int64_t stride = grid_stride() * 4;
   const float step = lr / bc1;
   const float inv_bc2 = 1.f / bc2;
+  (void)0;
   for (; i + 3 < n; i += stride) {
     float4 wv = *reinterpret_cast<float4*>(w + i);
     const float4 gv = *reinterpret_cast<const float4*>(g + i);
@@ -136,17 +145,18 @@ void launch_fused_sgd(float* w, const float* g, float* m, const float* ref,
 }
 
 void launch_fused_adam(float* w, const float* g, float* m, float* v,
-                       const float* ref, int64_t n, float lr, float b1,
-                       float b2, float eps, float wd, float mu, float bc1,
-                       float bc2, bool has_prox, hipStream_t stream) {
+                       const float* ref, const float* bc, int64_t n,
+                       float lr, float b1, float b2, float eps, float wd,
+                       float mu, float bc1, float bc2, bool has_prox,
+                       hipStream_t stream) {
   const int block = 256;
   const int grid = elemwise_grid((n + 3) / 4, block);
   if (has_prox)
    hipLaunchKernelGGL(( fused_adam_kernel<true>), dim3(grid), dim3(block), 0, stream, 
-        w, g, m, v, ref, n, lr, b1, b2, eps, wd, mu, bc1, bc2);
+        w, g, m, v, ref, bc, n, lr, b1, b2, eps, wd, mu, bc1, bc2);
   else
    hipLaunchKernelGGL(( fused_adam_kernel<false>), dim3(grid), dim3(block), 0, stream, 
-        w, g, m, v, ref, n, lr, b1, b2, eps, wd, mu, bc1, bc2);
+        w, g, m, v, ref, bc, n, lr, b1, b2, eps, wd, mu, bc1, bc2);
 }
 
 }  // namespace colearn

@@ -125,26 +125,55 @@ __global__ void layernorm_bwd_gwb_kernel(
   atomicAdd(gb + h, ab);
 }
 
-// ==== BatchNorm (channels-last) ===========================================
-// x viewed as [R, C] with R = N*H*W; stats per channel c.
-// Pass 1: per-channel sum + sumsq via LDS-binned accumulation.
-__global__ void bn_reduce_kernel(const bf16* __restrict__ x, int64_t R, int C,
-                                 float* __restrict__ sums /*[2*C]*/) {
+// ==== BatchNorm (channels-last) with fused residual-add + ReLU ===========
+// x viewed as [R, C], R = N*H*W; stats per channel.  With 256-thread
+// blocks x 8 elems/thread, a thread's channel block is FIXED whenever
+// (256*8) % C == 0 (every ResNet channel count) -- partial sums then
+// accumulate in REGISTERS and LDS atomics fire once per block instead of
+// once per element (the v1 LDS-atomic path measured 31 us on 8 MB; this
+// is the fix).  The residual add + ReLU of the ResNet block fuse into the
+// normalize/backward sweeps, eliminating the separate elementwise
+// kernels and one full activation-tensor round trip per block.
+
+template <bool FIXED_C>
+__global__ void bn_reduce_kernel(const bf16* __restrict__ x, int64_t R,
+                                 int C, float* __restrict__ sums /*[2C]*/) {
   extern __shared__ float lbin[];  // [2*C]: sum, sumsq
   for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) lbin[i] = 0.f;
   __syncthreads();
   const int64_t n = R * C;
   int64_t i = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) * 8;
   const int64_t stride = grid_stride() * 8;
-  for (; i + 7 < n; i += stride) {
-    const short8 v = *reinterpret_cast<const short8*>(
-        reinterpret_cast<const short*>(x) + i);
-    const int c0 = static_cast<int>(i % C);  // C % 8 == 0: no row crossing
+  if constexpr (FIXED_C) {
+    float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    float ss[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    const int c0 = static_cast<int>(i % C);  // invariant: stride % C == 0
+    for (; i + 7 < n; i += stride) {
+      const short8 v = *reinterpret_cast<const short8*>(
+          reinterpret_cast<const short*>(x) + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float f = bits2f(v.x[j]);
+        s[j] += f;
+        ss[j] += f * f;
+      }
+    }
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      const float f = bits2f(v.x[j]);
-      atomicAdd(&lbin[c0 + j], f);
-      atomicAdd(&lbin[C + c0 + j], f * f);
+      atomicAdd(&lbin[c0 + j], s[j]);
+      atomicAdd(&lbin[C + c0 + j], ss[j]);
+    }
+  } else {
+    for (; i + 7 < n; i += stride) {
+      const short8 v = *reinterpret_cast<const short8*>(
+          reinterpret_cast<const short*>(x) + i);
+      const int c0 = static_cast<int>(i % C);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float f = bits2f(v.x[j]);
+        atomicAdd(&lbin[c0 + j], f);
+        atomicAdd(&lbin[C + c0 + j], f * f);
+      }
     }
   }
   __syncthreads();
@@ -175,8 +204,10 @@ __global__ void bn_stats_kernel(const float* __restrict__ sums, int C,
   }
 }
 
-// Pass 3: y = (x - mean)*invstd*w + b, elementwise over [R, C].
+// Pass 3: y = act((x - mean)*invstd*w + b [+ res]), elementwise [R, C].
+template <bool RES, bool RELU>
 __global__ void bn_normalize_kernel(const bf16* __restrict__ x,
+                                    const bf16* __restrict__ res,
                                     const float* __restrict__ mean,
                                     const float* __restrict__ invstd,
                                     const float* __restrict__ w,
@@ -188,42 +219,76 @@ __global__ void bn_normalize_kernel(const bf16* __restrict__ x,
   for (; i + 7 < n; i += stride) {
     const short8 v = *reinterpret_cast<const short8*>(
         reinterpret_cast<const short*>(x) + i);
+    short8 rv;
+    if constexpr (RES)
+      rv = *reinterpret_cast<const short8*>(
+          reinterpret_cast<const short*>(res) + i);
     const int c0 = static_cast<int>(i % C);
     short8 o;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       const int c = c0 + j;
-      o.x[j] = f2bits((bits2f(v.x[j]) - mean[c]) * invstd[c] * w[c] + b[c]);
+      float r = (bits2f(v.x[j]) - mean[c]) * invstd[c] * w[c] + b[c];
+      if constexpr (RES) r += bits2f(rv.x[j]);
+      if constexpr (RELU) r = fmaxf(r, 0.f);
+      o.x[j] = f2bits(r);
     }
     *reinterpret_cast<short8*>(reinterpret_cast<short*>(y) + i) = o;
   }
 }
 
-// Backward pass 1: gb[c] = sum gy, gw[c] = sum gy*xhat (LDS-binned).
-__global__ void bn_bwd_reduce_kernel(const bf16* __restrict__ gy,
-                                     const bf16* __restrict__ x,
-                                     const float* __restrict__ mean,
-                                     const float* __restrict__ invstd,
-                                     int64_t R, int C,
-                                     float* __restrict__ acc /*[2*C]*/) {
-  extern __shared__ float lbin[];  // [2*C]: gb, gw
+// Backward pass 1: gym = gy * relu_mask(y); gb[c] = sum gym,
+// gw[c] = sum gym*xhat; optionally writes gym out (it IS the residual
+// grad, and pass 2 consumes it).
+template <bool FIXED_C, bool RELU>
+__global__ void bn_bwd_reduce_kernel(
+    const bf16* __restrict__ gy, const bf16* __restrict__ y,
+    const bf16* __restrict__ x, const float* __restrict__ mean,
+    const float* __restrict__ invstd, int64_t R, int C,
+    float* __restrict__ acc /*[2C]*/, bf16* __restrict__ gym_out) {
+  extern __shared__ float lbin[];
   for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) lbin[i] = 0.f;
   __syncthreads();
   const int64_t n = R * C;
   int64_t i = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) * 8;
   const int64_t stride = grid_stride() * 8;
+  float ab[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  float aw[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  const int c0_fixed = static_cast<int>(i % C);
   for (; i + 7 < n; i += stride) {
     const short8 g = *reinterpret_cast<const short8*>(
         reinterpret_cast<const short*>(gy) + i);
     const short8 v = *reinterpret_cast<const short8*>(
         reinterpret_cast<const short*>(x) + i);
-    const int c0 = static_cast<int>(i % C);
+    short8 yv;
+    if constexpr (RELU)
+      yv = *reinterpret_cast<const short8*>(
+          reinterpret_cast<const short*>(y) + i);
+    const int c0 = FIXED_C ? c0_fixed : static_cast<int>(i % C);
+    short8 gm;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       const int c = c0 + j;
-      const float gf = bits2f(g.x[j]);
-      atomicAdd(&lbin[c], gf);
-      atomicAdd(&lbin[C + c], gf * (bits2f(v.x[j]) - mean[c]) * invstd[c]);
+      float gf = bits2f(g.x[j]);
+      if constexpr (RELU) gf = bits2f(yv.x[j]) > 0.f ? gf : 0.f;
+      gm.x[j] = f2bits(gf);
+      const float xh = (bits2f(v.x[j]) - mean[c]) * invstd[c];
+      if constexpr (FIXED_C) {
+        ab[j] += gf;
+        aw[j] += gf * xh;
+      } else {
+        atomicAdd(&lbin[c], gf);
+        atomicAdd(&lbin[C + c], gf * xh);
+      }
+    }
+    if constexpr (RELU)
+      *reinterpret_cast<short8*>(reinterpret_cast<short*>(gym_out) + i) = gm;
+  }
+  if constexpr (FIXED_C) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      atomicAdd(&lbin[c0_fixed + j], ab[j]);
+      atomicAdd(&lbin[C + c0_fixed + j], aw[j]);
     }
   }
   __syncthreads();
@@ -231,9 +296,9 @@ __global__ void bn_bwd_reduce_kernel(const bf16* __restrict__ gy,
     if (lbin[i2] != 0.f) atomicAdd(acc + i2, lbin[i2]);
 }
 
-// Backward pass 2: gx = w*invstd*(gy - gb/n - xhat*gw/n)   (training)
-//                  gx = w*invstd*gy                        (eval)
-__global__ void bn_bwd_gx_kernel(const bf16* __restrict__ gy,
+// Backward pass 2: gx = w*invstd*(gym - gb/n - xhat*gw/n)   (training)
+//                  gx = w*invstd*gym                        (eval)
+__global__ void bn_bwd_gx_kernel(const bf16* __restrict__ gym,
                                  const bf16* __restrict__ x,
                                  const float* __restrict__ mean,
                                  const float* __restrict__ invstd,
@@ -247,7 +312,7 @@ __global__ void bn_bwd_gx_kernel(const bf16* __restrict__ gy,
   const int64_t stride = grid_stride() * 8;
   for (; i + 7 < n; i += stride) {
     const short8 g = *reinterpret_cast<const short8*>(
-        reinterpret_cast<const short*>(gy) + i);
+        reinterpret_cast<const short*>(gym) + i);
     const short8 v = *reinterpret_cast<const short8*>(
         reinterpret_cast<const short*>(x) + i);
     const int c0 = static_cast<int>(i % C);
@@ -299,37 +364,75 @@ void launch_layernorm_bwd(const void* gy_, const void* x_, const float* w,
                                                    gb, rows, H);
 }
 
-void launch_batchnorm_fwd(const void* x_, const float* w, const float* b,
-                          float* running_mean, float* running_var, void* y_,
-                          float* mean, float* invstd, float* sums, int64_t R,
-                          int C, float momentum, float eps, bool training,
-                          hipStream_t stream) {
+void launch_batchnorm_fwd(const void* x_, const void* res_, const float* w,
+                          const float* b, float* running_mean,
+                          float* running_var, void* y_, float* mean,
+                          float* invstd, float* sums, int64_t R, int C,
+                          float momentum, float eps, bool training,
+                          bool relu, hipStream_t stream) {
   const bf16* x = static_cast<const bf16*>(x_);
+  const bf16* res = static_cast<const bf16*>(res_);
   bf16* y = static_cast<bf16*>(y_);
   const size_t lds = 2 * C * sizeof(float);
+  const bool fixed = (256 * 8) % C == 0;
+  const int grid = elemwise_grid(R * C / 8);
   if (training) {
-    bn_reduce_kernel<<<elemwise_grid(R * C / 8), 256, lds, stream>>>(x, R, C,
-                                                                     sums);
+    if (fixed)
+      bn_reduce_kernel<true><<<grid, 256, lds, stream>>>(x, R, C, sums);
+    else
+      bn_reduce_kernel<false><<<grid, 256, lds, stream>>>(x, R, C, sums);
   }
   bn_stats_kernel<<<(C + 255) / 256, 256, 0, stream>>>(
       sums, C, R, mean, invstd, running_mean, running_var, momentum, eps,
       training);
-  bn_normalize_kernel<<<elemwise_grid(R * C / 8), 256, 0, stream>>>(
-      x, mean, invstd, w, b, y, R, C);
+  if (res != nullptr) {
+    if (relu)
+      bn_normalize_kernel<true, true><<<grid, 256, 0, stream>>>(
+          x, res, mean, invstd, w, b, y, R, C);
+    else
+      bn_normalize_kernel<true, false><<<grid, 256, 0, stream>>>(
+          x, res, mean, invstd, w, b, y, R, C);
+  } else {
+    if (relu)
+      bn_normalize_kernel<false, true><<<grid, 256, 0, stream>>>(
+          x, nullptr, mean, invstd, w, b, y, R, C);
+    else
+      bn_normalize_kernel<false, false><<<grid, 256, 0, stream>>>(
+          x, nullptr, mean, invstd, w, b, y, R, C);
+  }
 }
 
-void launch_batchnorm_bwd(const void* gy_, const void* x_, const float* w,
-                          const float* mean, const float* invstd, void* gx_,
+void launch_batchnorm_bwd(const void* gy_, const void* y_, const void* x_,
+                          const float* w, const float* mean,
+                          const float* invstd, void* gx_, void* gym_,
                           float* acc, int64_t R, int C, bool training,
-                          hipStream_t stream) {
+                          bool relu, hipStream_t stream) {
   const bf16* gy = static_cast<const bf16*>(gy_);
+  const bf16* y = static_cast<const bf16*>(y_);
   const bf16* x = static_cast<const bf16*>(x_);
   bf16* gx = static_cast<bf16*>(gx_);
+  bf16* gym = static_cast<bf16*>(gym_);
   const size_t lds = 2 * C * sizeof(float);
-  bn_bwd_reduce_kernel<<<elemwise_grid(R * C / 8), 256, lds, stream>>>(
-      gy, x, mean, invstd, R, C, acc);
-  bn_bwd_gx_kernel<<<elemwise_grid(R * C / 8), 256, 0, stream>>>(
-      gy, x, mean, invstd, w, acc, gx, R, C, training);
+  const bool fixed = (256 * 8) % C == 0;
+  const int grid = elemwise_grid(R * C / 8);
+  if (relu) {
+    if (fixed)
+      bn_bwd_reduce_kernel<true, true><<<grid, 256, lds, stream>>>(
+          gy, y, x, mean, invstd, R, C, acc, gym);
+    else
+      bn_bwd_reduce_kernel<false, true><<<grid, 256, lds, stream>>>(
+          gy, y, x, mean, invstd, R, C, acc, gym);
+  } else {
+    if (fixed)
+      bn_bwd_reduce_kernel<true, false><<<grid, 256, lds, stream>>>(
+          gy, y, x, mean, invstd, R, C, acc, nullptr);
+    else
+      bn_bwd_reduce_kernel<false, false><<<grid, 256, lds, stream>>>(
+          gy, y, x, mean, invstd, R, C, acc, nullptr);
+  }
+  const bf16* gym_in = relu ? gym : gy;
+  bn_bwd_gx_kernel<<<grid, 256, 0, stream>>>(gym_in, x, mean, invstd, w,
+                                             acc, gx, R, C, training);
 }
 
 }  // namespace colearn

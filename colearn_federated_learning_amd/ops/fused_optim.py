@@ -56,17 +56,26 @@ class FlatSGD:
         if self.m is not None:
             self.m.zero_()
 
+    def host_prepare(self) -> None:
+        pass  # SGD's update is stateless on the host: graph-safe as-is
+
+    @torch.no_grad()
+    def device_step(self) -> None:
+        mu = self.fedprox_mu
+        require_ext().fused_sgd(
+            self.w, self.g,
+            self.m if self.m is not None
+            else torch.empty(0, device=self.w.device),
+            self.ref if (mu != 0.0 and self.ref is not None)
+            else torch.empty(0, device=self.w.device),
+            self.lr, self.momentum, self.weight_decay, mu,
+        )
+
     @torch.no_grad()
     def step(self) -> None:
         mu = self.fedprox_mu
         if self.w.is_cuda and native_enabled():
-            require_ext().fused_sgd(
-                self.w, self.g,
-                self.m if self.m is not None else torch.empty(0, device=self.w.device),
-                self.ref if (mu != 0.0 and self.ref is not None)
-                else torch.empty(0, device=self.w.device),
-                self.lr, self.momentum, self.weight_decay, mu,
-            )
+            self.device_step()
             return
         g = self.g
         if self.weight_decay != 0.0:
@@ -107,27 +116,53 @@ class FlatAdam:
         self.m = torch.zeros_like(flat_params)
         self.v = torch.zeros_like(flat_params)
         self.t = 0
+        # live bias-correction buffer: under hipGraph replay the kernel's
+        # scalar args are frozen, so [bc1, bc2] travels through device
+        # memory, refreshed by host_prepare() before each step/replay.
+        self._bc = (
+            torch.ones(2, device=flat_params.device)
+            if flat_params.is_cuda else None
+        )
 
     def reset(self) -> None:
         self.m.zero_()
         self.v.zero_()
         self.t = 0
 
+    def host_prepare(self) -> None:
+        """Host-side half of step(): advance t, refresh the device bc
+        buffer.  Must run OUTSIDE any graph capture/replay."""
+        self.t += 1
+        if self._bc is not None:
+            bc = torch.tensor(
+                [1.0 - self.beta1 ** self.t, 1.0 - self.beta2 ** self.t],
+                dtype=torch.float32,
+            )
+            self._bc.copy_(bc, non_blocking=True)
+
+    @torch.no_grad()
+    def device_step(self) -> None:
+        """Device-side half (graph-capturable)."""
+        mu = self.fedprox_mu
+        require_ext().fused_adam(
+            self.w, self.g, self.m, self.v,
+            self.ref if (mu != 0.0 and self.ref is not None)
+            else torch.empty(0, device=self.w.device),
+            self._bc,
+            self.lr, self.beta1, self.beta2, self.eps,
+            self.weight_decay, mu, 1.0, 1.0,
+        )
+
     @torch.no_grad()
     def step(self) -> None:
+        if self.w.is_cuda and native_enabled():
+            self.host_prepare()
+            self.device_step()
+            return
         self.t += 1
         bc1 = 1.0 - self.beta1 ** self.t
         bc2 = 1.0 - self.beta2 ** self.t
         mu = self.fedprox_mu
-        if self.w.is_cuda and native_enabled():
-            require_ext().fused_adam(
-                self.w, self.g, self.m, self.v,
-                self.ref if (mu != 0.0 and self.ref is not None)
-                else torch.empty(0, device=self.w.device),
-                self.lr, self.beta1, self.beta2, self.eps,
-                self.weight_decay, mu, bc1, bc2,
-            )
-            return
         g = self.g
         if self.weight_decay != 0.0:
             g = g.add(self.w, alpha=self.weight_decay)

@@ -22,16 +22,21 @@ def _native(x: torch.Tensor) -> bool:
 
 
 class _BatchNormFn(torch.autograd.Function):
+    """BatchNorm with optional fused residual-add + ReLU (the ResNet block
+    epilogue runs inside the BN sweeps on gfx950 — see csrc/norms.hip)."""
+
     generate_vmap_rule = True
 
     @staticmethod
-    def forward(x, weight, bias, running_mean, running_var,
-                training: bool, momentum: float, eps: float):
+    def forward(x, weight, bias, running_mean, running_var, residual,
+                training: bool, momentum: float, eps: float, relu: bool):
         if _native(x):
             m = require_ext()
             y, save_mean, save_invstd = m.batchnorm_fwd(
-                x.contiguous(), weight, bias, running_mean, running_var,
-                training, momentum, eps,
+                x, weight, bias, running_mean, running_var,
+                residual if residual is not None
+                else torch.empty(0, device=x.device, dtype=x.dtype),
+                training, momentum, eps, relu,
             )
             return y, save_mean, save_invstd
         if training:
@@ -52,30 +57,37 @@ class _BatchNormFn(torch.autograd.Function):
             mean = running_mean.clone()
             invstd = (running_var + eps).rsqrt()
             xf = x.float()
-        y = ((xf - mean[None, :, None, None])
-             * invstd[None, :, None, None]
-             * weight[None, :, None, None]
-             + bias[None, :, None, None]).to(x.dtype)
-        return y, mean, invstd
+        y = (xf - mean[None, :, None, None]) * invstd[None, :, None, None] \
+            * weight[None, :, None, None] + bias[None, :, None, None]
+        if residual is not None:
+            y = y + residual.float()
+        if relu:
+            y = y.relu()
+        return y.to(x.dtype), mean, invstd
 
     @staticmethod
     def setup_context(ctx, inputs, output):
         x, weight = inputs[0], inputs[1]
-        _, save_mean, save_invstd = output
-        ctx.save_for_backward(x, weight, save_mean, save_invstd)
-        ctx.training = inputs[5]
+        y, save_mean, save_invstd = output
+        ctx.save_for_backward(x, weight, save_mean, save_invstd, y)
+        ctx.training = inputs[6]
+        ctx.relu = inputs[9]
+        ctx.has_res = inputs[5] is not None
         ctx.mark_non_differentiable(output[1], output[2])
 
     @staticmethod
     def backward(ctx, gy, _gmean, _ginvstd):
-        x, weight, mean, invstd = ctx.saved_tensors
-        gy = gy.contiguous()
+        x, weight, mean, invstd, y = ctx.saved_tensors
         if _native(x):
-            gx, gw, gb = require_ext().batchnorm_bwd(
-                gy, x.contiguous(), weight, mean, invstd, ctx.training
+            gx, gw, gb, gres = require_ext().batchnorm_bwd(
+                gy, y, x, weight, mean, invstd, ctx.training, ctx.relu
             )
         else:
-            xf, gyf = x.float(), gy.float()
+            gyf = gy.float()
+            if ctx.relu:
+                gyf = gyf * (y > 0).float()
+            gres = gyf.to(x.dtype)
+            xf = x.float()
             xhat = (xf - mean[None, :, None, None]) * invstd[None, :, None, None]
             n = x.numel() / x.shape[1]
             gb = gyf.sum(dim=(0, 2, 3))
@@ -88,16 +100,20 @@ class _BatchNormFn(torch.autograd.Function):
             else:
                 gx = (weight * invstd)[None, :, None, None] * gyf
             gx = gx.to(x.dtype)
-        return gx, gw, gb, None, None, None, None, None
+        return (gx, gw, gb, None, None,
+                gres if ctx.has_res else None, None, None, None, None)
 
 
 class ColearnBatchNorm2d(nn.Module):
+    """BatchNorm2d with an optional fused ``+ residual`` / ReLU epilogue."""
+
     def __init__(self, num_features: int, eps: float = 1e-5,
-                 momentum: float = 0.1) -> None:
+                 momentum: float = 0.1, relu: bool = False) -> None:
         super().__init__()
         self.num_features = num_features
         self.eps = eps
         self.momentum = momentum
+        self.relu = relu
         self.weight = nn.Parameter(torch.ones(num_features))
         self.bias = nn.Parameter(torch.zeros(num_features))
         self.register_buffer("running_mean", torch.zeros(num_features))
@@ -105,12 +121,13 @@ class ColearnBatchNorm2d(nn.Module):
         self.register_buffer("num_batches_tracked",
                              torch.tensor(0, dtype=torch.long))
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, residual=None) -> torch.Tensor:
         if self.training:
-            self.num_batches_tracked += 1
+            with torch.no_grad():
+                self.num_batches_tracked += 1
         y, _, _ = _BatchNormFn.apply(
             x, self.weight, self.bias, self.running_mean, self.running_var,
-            self.training, self.momentum, self.eps,
+            residual, self.training, self.momentum, self.eps, self.relu,
         )
         return y
 
