@@ -37,9 +37,10 @@ void launch_scatter_add_pairs(float* out, const int* idx, const float* val,
 void launch_layernorm_fwd(const void* x, const float* w, const float* b,
                           void* y, float* mean, float* invstd, int64_t rows,
                           int H, float eps, hipStream_t stream);
+int norms_reduce_grid(int64_t n8);
 void launch_layernorm_bwd(const void* gy, const void* x, const float* w,
                           const float* mean, const float* invstd, void* gx,
-                          float* gw, float* gb, int64_t rows, int H,
+                          float* gw_gb, float* partials, int64_t rows, int H,
                           hipStream_t stream);
 void launch_batchnorm_fwd(const void* x, const void* res, const float* w,
                           const float* b, float* running_mean,
@@ -58,6 +59,7 @@ void launch_conv_fwd(const void* x, const void* w, void* y, ConvShape cs,
                      hipStream_t stream);
 void launch_conv_dgrad(const void* gy, const void* wt, void* gx,
                        ConvShape cs, hipStream_t stream);
+int conv_wgrad_splitk(ConvShape cs);
 void launch_conv_wgrad(const void* gy, const void* x, float* gwf, void* gw,
                        ConvShape cs, hipStream_t stream);
 

@@ -158,16 +158,17 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> layernorm_bwd(
   const int H = static_cast<int>(x.size(-1));
   const int64_t rows = x.numel() / H;
   auto gx = torch::empty_like(xc);
-  auto gw = torch::zeros({H}, x.options().dtype(torch::kFloat32));
-  auto gb = torch::zeros_like(gw);
+  auto gw_gb = torch::empty({2 * H}, x.options().dtype(torch::kFloat32));
+  auto partials = torch::empty({64 * 2 * H},
+                               x.options().dtype(torch::kFloat32));
   launch_layernorm_bwd(
-      gc.data_ptr(),
-      xc.data_ptr(), w.data_ptr<float>(),
-      mc.data_ptr<float>(), ic.data_ptr<float>(),
-      gx.data_ptr(), gw.data_ptr<float>(),
-      gb.data_ptr<float>(), rows, H, stream());
+      gc.data_ptr(), xc.data_ptr(), w.data_ptr<float>(),
+      mc.data_ptr<float>(), ic.data_ptr<float>(), gx.data_ptr(),
+      gw_gb.data_ptr<float>(), partials.data_ptr<float>(), rows, H,
+      stream());
   HIP_CHECK_LAST();
-  return {gx, gw, gb};
+  return {gx, gw_gb.narrow(0, 0, H).clone(),
+          gw_gb.narrow(0, H, H).clone()};
 }
 
 // ---- BatchNorm (channels-last) -------------------------------------------
@@ -189,7 +190,9 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
   auto y = torch::empty_like(xc);
   auto mean = torch::empty({C}, x.options().dtype(torch::kFloat32));
   auto invstd = torch::empty_like(mean);
-  auto sums = torch::zeros({2 * C}, mean.options());
+  const int rgrid = norms_reduce_grid(R * C / 8);
+  auto sums = torch::empty(
+      {static_cast<int64_t>(rgrid + 1) * 2 * C}, mean.options());
   launch_batchnorm_fwd(
       xc.data_ptr(), has_res ? rc.data_ptr() : nullptr,
       w.data_ptr<float>(), b.data_ptr<float>(),
@@ -216,7 +219,9 @@ batchnorm_bwd(torch::Tensor gy, torch::Tensor y, torch::Tensor x,
   auto gx = torch::empty_like(xc);
   torch::Tensor gym;  // masked gy == the residual grad
   if (relu) gym = torch::empty_like(gc);
-  auto acc = torch::zeros({2 * C}, x.options().dtype(torch::kFloat32));
+  const int rgrid = norms_reduce_grid(R * C / 8);
+  auto acc = torch::empty({static_cast<int64_t>(rgrid + 1) * 2 * C},
+                          x.options().dtype(torch::kFloat32));
   launch_batchnorm_bwd(gc.data_ptr(), relu ? yc.data_ptr() : nullptr,
                        xc.data_ptr(), w.data_ptr<float>(),
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
@@ -224,8 +229,9 @@ batchnorm_bwd(torch::Tensor gy, torch::Tensor y, torch::Tensor x,
                        acc.data_ptr<float>(), R, C, training, relu,
                        stream());
   HIP_CHECK_LAST();
-  auto gb = acc.narrow(0, 0, C).clone();
-  auto gw = acc.narrow(0, C, C).clone();
+  const int64_t off = static_cast<int64_t>(rgrid) * 2 * C;
+  auto gb = acc.narrow(0, off, C).clone();
+  auto gw = acc.narrow(0, off + C, C).clone();
   return {gx, gw, gb, relu ? gym : gc};
 }
 
@@ -358,7 +364,8 @@ torch::Tensor conv2d_wgrad(torch::Tensor gy, torch::Tensor x, int64_t R,
   auto gc = gy.contiguous(torch::MemoryFormat::ChannelsLast);
   auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
   const int64_t rsc = static_cast<int64_t>(cs.R) * cs.S * cs.C;
-  auto gwf = torch::zeros({cs.K, rsc},
+  const int splitk = conv_wgrad_splitk(cs);
+  auto gwf = torch::empty({splitk, cs.K, rsc},
                           x.options().dtype(torch::kFloat32));
   auto gw_rsci = torch::empty({cs.K, cs.R, cs.S, cs.C}, x.options());
   launch_conv_wgrad(gc.data_ptr(), xc.data_ptr(), gwf.data_ptr<float>(),

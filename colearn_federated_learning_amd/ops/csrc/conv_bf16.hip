@@ -267,7 +267,9 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
         acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af[m], bfr[n], acc[m][n], 0, 0, 0);
   }
-  // epilogue: atomicAdd fp32 (C/D map: col=lane&15, row=(lane>>4)*4+j)
+  // epilogue: plain stores into this block's split-K slab (deterministic;
+  // the combine kernel sums slabs in fixed slice order).
+  float* slab = gwf + static_cast<int64_t>(blockIdx.y) * cs.K * RSC;
   const int lane = threadIdx.x % kWave;
   const int cif = lane & 15, rg = lane >> 4;
 #pragma unroll
@@ -279,9 +281,20 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
         const int co = co0 + wm + m * 16 + rg * 4 + j;
         const int rc = rc0 + wn + n * 16 + cif;
         if (co < cs.K && rc < RSC)
-          atomicAdd(gwf + static_cast<int64_t>(co) * RSC + rc,
-                    acc[m][n][j]);
+          slab[static_cast<int64_t>(co) * RSC + rc] = acc[m][n][j];
       }
+}
+
+// sum the split-K slabs in slice order and emit bf16.
+__global__ void wgrad_combine_kernel(const float* __restrict__ gwf,
+                                     bf16* __restrict__ gw, int64_t n,
+                                     int splitk) {
+  int64_t i = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+  for (; i < n; i += grid_stride()) {
+    float acc = 0.f;
+    for (int sk = 0; sk < splitk; ++sk) acc += gwf[sk * n + i];
+    gw[i] = f2b(acc);
+  }
 }
 
 __global__ void f32_to_bf16_kernel(const float* __restrict__ in,
@@ -319,16 +332,24 @@ void launch_conv_wgrad(const void* gy, const void* x, float* gwf, void* gw,
   const int64_t RSC = static_cast<int64_t>(cs.R) * cs.S * cs.C;
   const int nbm = (cs.K + 63) / 64;
   const int nbn = static_cast<int>((RSC + 63) / 64);
-  int64_t want = 512 / (nbm * nbn);
-  if (want < 1) want = 1;
-  int64_t cap = (NPQ + BK - 1) / BK;
-  const int splitk = static_cast<int>(want < cap ? want : cap);
+  const int splitk = conv_wgrad_splitk(cs);
   conv_wgrad_kernel<<<dim3(nbm * nbn, splitk), kBlockThreads, 0, stream>>>(
       static_cast<const short*>(gy), static_cast<const short*>(x), gwf, cs,
       NPQ, RSC, nbm, nbn, splitk);
   const int64_t n = static_cast<int64_t>(cs.K) * RSC;
-  f32_to_bf16_kernel<<<elemwise_grid(n), 256, 0, stream>>>(
-      gwf, static_cast<bf16*>(gw), n);
+  wgrad_combine_kernel<<<elemwise_grid(n), 256, 0, stream>>>(
+      gwf, static_cast<bf16*>(gw), n, splitk);
+}
+
+int conv_wgrad_splitk(ConvShape cs) {
+  const int64_t NPQ = static_cast<int64_t>(cs.N) * cs.P * cs.Q;
+  const int64_t RSC = static_cast<int64_t>(cs.R) * cs.S * cs.C;
+  const int nbm = (cs.K + 63) / 64;
+  const int nbn = static_cast<int>((RSC + 63) / 64);
+  int64_t want = 512 / (nbm * nbn);
+  if (want < 1) want = 1;
+  int64_t cap = (NPQ + BK - 1) / BK;
+  return static_cast<int>(want < cap ? want : cap);
 }
 
 }  // namespace colearn

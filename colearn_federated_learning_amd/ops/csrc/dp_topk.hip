@@ -227,9 +227,9 @@ __global__ void scatter_add_pairs_kernel(float* __restrict__ out,
 // ---- launchers ------------------------------------------------------------
 void launch_sq_norm_per_row(const float* g, float* sq, int64_t rows,
                             int64_t cols, hipStream_t stream) {
-  const int chunks =
-      max(1, min(16, static_cast<int>((cols + 262143) / 262144)));
-  dim3 grid(static_cast<unsigned>(rows), chunks);
+  // one block per row: the per-row sum is a single deterministic
+  // block-reduction (atomicAdd across chunk-blocks would not be).
+  dim3 grid(static_cast<unsigned>(rows), 1);
   sq_norm_per_row_kernel<<<grid, 256, 0, stream>>>(g, sq, rows, cols);
 }
 

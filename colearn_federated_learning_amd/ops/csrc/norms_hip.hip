@@ -16,6 +16,8 @@
 
 namespace colearn {
 
+int norms_reduce_grid(int64_t n8) { return elemwise_grid(n8); }
+
 // ==== LayerNorm ============================================================
 // x: [rows, H] bf16, w/b: [H] fp32, y: bf16, mean/invstd: [rows] fp32
 __global__ void layernorm_fwd_kernel(const bf16* __restrict__ x,
@@ -112,7 +114,7 @@ __global__ void layernorm_bwd_gx_kernel(
 __global__ void layernorm_bwd_gwb_kernel(
     const bf16* __restrict__ gy, const bf16* __restrict__ x,
     const float* __restrict__ mean, const float* __restrict__ invstd,
-    float* __restrict__ gw, float* __restrict__ gb, int64_t rows, int H) {
+    float* __restrict__ partials /*[gridDim.y][2H]*/, int64_t rows, int H) {
   const int h = blockIdx.x * blockDim.x + threadIdx.x;
   if (h >= H) return;
   float aw = 0.f, ab = 0.f;
@@ -122,8 +124,9 @@ __global__ void layernorm_bwd_gwb_kernel(
     aw += g * (xv - mean[row]) * invstd[row];
     ab += g;
   }
-  atomicAdd(gw + h, aw);
-  atomicAdd(gb + h, ab);
+  float* prow = partials + static_cast<int64_t>(blockIdx.y) * 2 * H;
+  prow[h] = aw;
+  prow[H + h] = ab;
 }
 
 // ==== BatchNorm (channels-last) with fused residual-add + ReLU ===========
@@ -178,8 +181,21 @@ __global__ void bn_reduce_kernel(const bf16* __restrict__ x, int64_t R,
     }
   }
   __syncthreads();
+  // deterministic: per-block partial row, combined in fixed order later
   for (int i2 = threadIdx.x; i2 < 2 * C; i2 += blockDim.x)
-    if (lbin[i2] != 0.f) atomicAdd(sums + i2, lbin[i2]);
+    sums[static_cast<int64_t>(blockIdx.x) * 2 * C + i2] = lbin[i2];
+}
+
+// combine per-block partial rows in block order (deterministic).
+__global__ void partial_combine_kernel(const float* __restrict__ partials,
+                                       float* __restrict__ out, int width,
+                                       int nblk) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= width) return;
+  float acc = 0.f;
+  for (int b = 0; b < nblk; ++b)
+    acc += partials[static_cast<int64_t>(b) * width + i];
+  out[i] = acc;
 }
 
 // Pass 2: finalize stats + update running stats (tiny launch over C).
@@ -294,7 +310,7 @@ __global__ void bn_bwd_reduce_kernel(
   }
   __syncthreads();
   for (int i2 = threadIdx.x; i2 < 2 * C; i2 += blockDim.x)
-    if (lbin[i2] != 0.f) atomicAdd(acc + i2, lbin[i2]);
+    acc[static_cast<int64_t>(blockIdx.x) * 2 * C + i2] = lbin[i2];
 }
 
 // Backward pass 2: gx = w*invstd*(gym - gb/n - xhat*gw/n)   (training)
@@ -350,8 +366,8 @@ void launch_layernorm_fwd(const void* x, const float* w, const float* b,
 
 void launch_layernorm_bwd(const void* gy_, const void* x_, const float* w,
                           const float* mean, const float* invstd, void* gx_,
-                          float* gw, float* gb, int64_t rows, int H,
-                          hipStream_t stream) {
+                          float* gw_gb /*[2H] out*/, float* partials,
+                          int64_t rows, int H, hipStream_t stream) {
   const bf16* gy = static_cast<const bf16*>(gy_);
   const bf16* x = static_cast<const bf16*>(x_);
   bf16* gx = static_cast<bf16*>(gx_);
@@ -360,9 +376,12 @@ void launch_layernorm_bwd(const void* gy_, const void* x_, const float* w,
   const int grid = static_cast<int>(g < 2048 ? g : 2048);
  hipLaunchKernelGGL(( layernorm_bwd_gx_kernel), dim3(grid), dim3(waves * kWave), 0, stream, 
       gy, x, w, mean, invstd, gx, rows, H);
-  dim3 g2((H + 255) / 256, 64);
- hipLaunchKernelGGL(( layernorm_bwd_gwb_kernel), dim3(g2), dim3(256), 0, stream, gy, x, mean, invstd, gw,
-                                                   gb, rows, H);
+  const int nchunk = 64;
+  dim3 g2((H + 255) / 256, nchunk);
+ hipLaunchKernelGGL(( layernorm_bwd_gwb_kernel), dim3(g2), dim3(256), 0, stream, gy, x, mean, invstd,
+                                                   partials, rows, H);
+ hipLaunchKernelGGL(( partial_combine_kernel), dim3((2 * H + 255) / 256), dim3(256), 0, stream, 
+      partials, gw_gb, 2 * H, nchunk);
 }
 
 void launch_batchnorm_fwd(const void* x_, const void* res_, const float* w,
@@ -378,10 +397,16 @@ void launch_batchnorm_fwd(const void* x_, const void* res_, const float* w,
   const bool fixed = (256 * 8) % C == 0;
   const int grid = elemwise_grid(R * C / 8);
   if (training) {
+    // sums layout: [grid][2C] partials, combined into the last 2C floats
+    float* partials = sums;
+    float* combined = sums + static_cast<int64_t>(grid) * 2 * C;
     if (fixed)
-     hipLaunchKernelGGL(( bn_reduce_kernel<true>), dim3(grid), dim3(256), lds, stream, x, R, C, sums);
+     hipLaunchKernelGGL(( bn_reduce_kernel<true>), dim3(grid), dim3(256), lds, stream, x, R, C, partials);
     else
-     hipLaunchKernelGGL(( bn_reduce_kernel<false>), dim3(grid), dim3(256), lds, stream, x, R, C, sums);
+     hipLaunchKernelGGL(( bn_reduce_kernel<false>), dim3(grid), dim3(256), lds, stream, x, R, C, partials);
+   hipLaunchKernelGGL(( partial_combine_kernel), dim3((2 * C + 255) / 256), dim3(256), 0, stream, 
+        partials, combined, 2 * C, grid);
+    sums = combined;
   }
  hipLaunchKernelGGL(( bn_stats_kernel), dim3((C + 255) / 256), dim3(256), 0, stream, 
       sums, C, R, mean, invstd, running_mean, running_var, momentum, eps,
@@ -416,6 +441,9 @@ void launch_batchnorm_bwd(const void* gy_, const void* y_, const void* x_,
   const size_t lds = 2 * C * sizeof(float);
   const bool fixed = (256 * 8) % C == 0;
   const int grid = elemwise_grid(R * C / 8);
+  float* partials = acc;
+  float* combined = acc + static_cast<int64_t>(grid) * 2 * C;
+  acc = partials;
   if (relu) {
     if (fixed)
      hipLaunchKernelGGL(( bn_bwd_reduce_kernel<true, true>), dim3(grid), dim3(256), lds, stream, 
@@ -431,9 +459,11 @@ void launch_batchnorm_bwd(const void* gy_, const void* y_, const void* x_,
      hipLaunchKernelGGL(( bn_bwd_reduce_kernel<false, false>), dim3(grid), dim3(256), lds, stream, 
           gy, y, x, mean, invstd, R, C, acc, nullptr);
   }
+ hipLaunchKernelGGL(( partial_combine_kernel), dim3((2 * C + 255) / 256), dim3(256), 0, stream, 
+      partials, combined, 2 * C, grid);
   const bf16* gym_in = relu ? gym : gy;
  hipLaunchKernelGGL(( bn_bwd_gx_kernel), dim3(grid), dim3(256), 0, stream, gym_in, x, mean, invstd, w,
-                                             acc, gx, R, C, training);
+                                             combined, gx, R, C, training);
 }
 
 }  // namespace colearn
