@@ -15,7 +15,11 @@
 
 namespace colearn {
 
-int norms_reduce_grid(int64_t n8) { return elemwise_grid(n8); }
+// Reduction grids are CAPPED at 256 blocks: each block then carries real
+// register-accumulation work (large grids degenerate to one iteration per
+// thread, where the per-block LDS-atomic flush dominates), and the
+// fixed-order combine stays cheap.
+int norms_reduce_grid(int64_t n8) { return elemwise_grid(n8, 256, 256); }
 
 // ==== LayerNorm ============================================================
 // x: [rows, H] bf16, w/b: [H] fp32, y: bf16, mean/invstd: [rows] fp32
@@ -185,16 +189,19 @@ __global__ void bn_reduce_kernel(const bf16* __restrict__ x, int64_t R,
     sums[static_cast<int64_t>(blockIdx.x) * 2 * C + i2] = lbin[i2];
 }
 
-// combine per-block partial rows in block order (deterministic).
+// combine per-block partial rows (deterministic: fixed lane->entry map +
+// fixed shuffle tree).  One wave per output element.
 __global__ void partial_combine_kernel(const float* __restrict__ partials,
                                        float* __restrict__ out, int width,
                                        int nblk) {
-  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  const int lane = threadIdx.x % kWave;
+  const int i = blockIdx.x * (blockDim.x / kWave) + threadIdx.x / kWave;
   if (i >= width) return;
   float acc = 0.f;
-  for (int b = 0; b < nblk; ++b)
+  for (int b = lane; b < nblk; b += kWave)
     acc += partials[static_cast<int64_t>(b) * width + i];
-  out[i] = acc;
+  acc = wave_allreduce_sum(acc);
+  if (lane == 0) out[i] = acc;
 }
 
 // Pass 2: finalize stats + update running stats (tiny launch over C).
@@ -379,7 +386,7 @@ void launch_layernorm_bwd(const void* gy_, const void* x_, const float* w,
   dim3 g2((H + 255) / 256, nchunk);
   layernorm_bwd_gwb_kernel<<<g2, 256, 0, stream>>>(gy, x, mean, invstd,
                                                    partials, rows, H);
-  partial_combine_kernel<<<(2 * H + 255) / 256, 256, 0, stream>>>(
+  partial_combine_kernel<<<(2 * H + 3) / 4, 256, 0, stream>>>(
       partials, gw_gb, 2 * H, nchunk);
 }
 
@@ -395,16 +402,18 @@ void launch_batchnorm_fwd(const void* x_, const void* res_, const float* w,
   const size_t lds = 2 * C * sizeof(float);
   const bool fixed = (256 * 8) % C == 0;
   const int grid = elemwise_grid(R * C / 8);
+  const int rgrid = norms_reduce_grid(R * C / 8);
   if (training) {
-    // sums layout: [grid][2C] partials, combined into the last 2C floats
+    // sums layout: [rgrid][2C] partials, combined into the last 2C floats
     float* partials = sums;
-    float* combined = sums + static_cast<int64_t>(grid) * 2 * C;
+    float* combined = sums + static_cast<int64_t>(rgrid) * 2 * C;
     if (fixed)
-      bn_reduce_kernel<true><<<grid, 256, lds, stream>>>(x, R, C, partials);
+      bn_reduce_kernel<true><<<rgrid, 256, lds, stream>>>(x, R, C, partials);
     else
-      bn_reduce_kernel<false><<<grid, 256, lds, stream>>>(x, R, C, partials);
-    partial_combine_kernel<<<(2 * C + 255) / 256, 256, 0, stream>>>(
-        partials, combined, 2 * C, grid);
+      bn_reduce_kernel<false><<<rgrid, 256, lds, stream>>>(x, R, C,
+                                                           partials);
+    partial_combine_kernel<<<(2 * C + 3) / 4, 256, 0, stream>>>(
+        partials, combined, 2 * C, rgrid);
     sums = combined;
   }
   bn_stats_kernel<<<(C + 255) / 256, 256, 0, stream>>>(
@@ -440,26 +449,27 @@ void launch_batchnorm_bwd(const void* gy_, const void* y_, const void* x_,
   const size_t lds = 2 * C * sizeof(float);
   const bool fixed = (256 * 8) % C == 0;
   const int grid = elemwise_grid(R * C / 8);
+  const int rgrid = norms_reduce_grid(R * C / 8);
   float* partials = acc;
-  float* combined = acc + static_cast<int64_t>(grid) * 2 * C;
+  float* combined = acc + static_cast<int64_t>(rgrid) * 2 * C;
   acc = partials;
   if (relu) {
     if (fixed)
-      bn_bwd_reduce_kernel<true, true><<<grid, 256, lds, stream>>>(
+      bn_bwd_reduce_kernel<true, true><<<rgrid, 256, lds, stream>>>(
           gy, y, x, mean, invstd, R, C, acc, gym);
     else
-      bn_bwd_reduce_kernel<false, true><<<grid, 256, lds, stream>>>(
+      bn_bwd_reduce_kernel<false, true><<<rgrid, 256, lds, stream>>>(
           gy, y, x, mean, invstd, R, C, acc, gym);
   } else {
     if (fixed)
-      bn_bwd_reduce_kernel<true, false><<<grid, 256, lds, stream>>>(
+      bn_bwd_reduce_kernel<true, false><<<rgrid, 256, lds, stream>>>(
           gy, y, x, mean, invstd, R, C, acc, nullptr);
     else
-      bn_bwd_reduce_kernel<false, false><<<grid, 256, lds, stream>>>(
+      bn_bwd_reduce_kernel<false, false><<<rgrid, 256, lds, stream>>>(
           gy, y, x, mean, invstd, R, C, acc, nullptr);
   }
-  partial_combine_kernel<<<(2 * C + 255) / 256, 256, 0, stream>>>(
-      partials, combined, 2 * C, grid);
+  partial_combine_kernel<<<(2 * C + 3) / 4, 256, 0, stream>>>(
+      partials, combined, 2 * C, rgrid);
   const bf16* gym_in = relu ? gym : gy;
   bn_bwd_gx_kernel<<<grid, 256, 0, stream>>>(gym_in, x, mean, invstd, w,
                                              combined, gx, R, C, training);

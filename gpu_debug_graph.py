@@ -1,4 +1,4 @@
-"""Single-step graphed vs eager comparison (debug)."""
+"""Determinism + graph-vs-eager localization (debug)."""
 import torch
 from colearn_federated_learning_amd.config import driver_config
 from colearn_federated_learning_amd.federated.coordinator import FederatedServer
@@ -6,23 +6,21 @@ from colearn_federated_learning_amd.federated.coordinator import FederatedServer
 dev = torch.device("cuda", 0)
 base = driver_config(2).override(rounds=1, num_clients=1,
                                  data__samples_per_client=64,
-                                 train__batch_size=32,
-                                 train__max_local_steps=1)
-sg = FederatedServer(base, device=dev)
-se = FederatedServer(base.override(train__use_graphs=False), device=dev)
-print("init equal:", torch.equal(sg.global_flat, se.global_flat))
+                                 train__batch_size=32)
 
-# run one fit each with identical start
-c_g = list(sg.clients.values())[0]
-c_e = list(se.clients.values())[0]
-sg.flat.load_flat(sg.global_flat); sg.optimizer.reset()
-se.flat.load_flat(se.global_flat); se.optimizer.reset()
-r_g = c_g.fit(sg.model, sg.optimizer, 0, runner=sg.step_runner)
-r_e = c_e.fit(se.model, se.optimizer, 0, runner=None)
-torch.cuda.synchronize()
-print("loss graphed:", r_g.loss, "eager:", r_e.loss)
-dw = (sg.flat.flat_w - se.flat.flat_w)
-dg = (sg.flat.flat_g - se.flat.flat_g)
-print("w diff max:", float(dw.abs().max()), "g diff max:", float(dg.abs().max()))
-print("g norm graphed:", float(sg.flat.flat_g.norm()), "eager:", float(se.flat.flat_g.norm()))
-print("w norm graphed:", float(sg.flat.flat_w.norm()), "eager:", float(se.flat.flat_w.norm()))
+def run_fit(cfg):
+    s = FederatedServer(cfg, device=dev)
+    c = list(s.clients.values())[0]
+    s.flat.load_flat(s.global_flat); s.optimizer.reset()
+    r = c.fit(s.model, s.optimizer, 0, runner=s.step_runner)
+    torch.cuda.synchronize()
+    return s.flat.flat_w.clone(), s.flat.flat_g.clone(), r.loss
+
+w1, g1, l1 = run_fit(base)                                   # graphed
+w2, g2, l2 = run_fit(base)                                   # graphed again
+w3, g3, l3 = run_fit(base.override(train__use_graphs=False)) # eager
+w4, g4, l4 = run_fit(base.override(train__use_graphs=False)) # eager again
+print("graphed repeat identical:", torch.equal(w1, w2), l1, l2)
+print("eager repeat identical:", torch.equal(w3, w4), l3, l4)
+print("graph vs eager: wdiff", float((w1-w3).abs().max()),
+      "gdiff", float((g1-g3).abs().max()), "loss", l1, l3)
