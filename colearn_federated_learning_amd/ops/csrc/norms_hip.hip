@@ -143,19 +143,22 @@ __global__ void layernorm_bwd_gwb_kernel(
 // normalize/backward sweeps, eliminating the separate elementwise
 // kernels and one full activation-tensor round trip per block.
 
+// In-block combine is fully deterministic on the FIXED_C path: each
+// thread's 8-channel partials land in a private LDS slot and one thread
+// per channel sums its group's slots in fixed order (no atomics anywhere
+// in the chain).  The !FIXED_C fallback (C not dividing 2048 — never hit
+// by the shipped models) keeps LDS atomics and is only
+// tolerance-deterministic.
 template <bool FIXED_C>
 __global__ void bn_reduce_kernel(const bf16* __restrict__ x, int64_t R,
-                                 int C, float* __restrict__ sums /*[2C]*/) {
-  extern __shared__ float lbin[];  // [2*C]: sum, sumsq
-  for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) lbin[i] = 0.f;
-  __syncthreads();
+                                 int C, float* __restrict__ sums) {
+  extern __shared__ float lbin[];  // FIXED: [256][16]; else [2C]
   const int64_t n = R * C;
   int64_t i = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) * 8;
   const int64_t stride = grid_stride() * 8;
   if constexpr (FIXED_C) {
     float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
     float ss[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-    const int c0 = static_cast<int>(i % C);  // invariant: stride % C == 0
     for (; i + 7 < n; i += stride) {
       const short8 v = *reinterpret_cast<const short8*>(
           reinterpret_cast<const short*>(x) + i);
@@ -168,10 +171,27 @@ __global__ void bn_reduce_kernel(const bf16* __restrict__ x, int64_t R,
     }
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      atomicAdd(&lbin[c0 + j], s[j]);
-      atomicAdd(&lbin[C + c0 + j], ss[j]);
+      lbin[threadIdx.x * 16 + j] = s[j];
+      lbin[threadIdx.x * 16 + 8 + j] = ss[j];
+    }
+    __syncthreads();
+    // channel c is covered by threads tid % (C/8) == c/8, element c%8
+    const int nb = C / 8;           // channel blocks
+    const int gsz = blockDim.x / nb;  // threads per channel block
+    for (int c = threadIdx.x; c < C; c += blockDim.x) {
+      const int cb = c / 8, j = c % 8;
+      float a = 0.f, b = 0.f;
+      for (int k = 0; k < gsz; ++k) {
+        a += lbin[(cb + k * nb) * 16 + j];
+        b += lbin[(cb + k * nb) * 16 + 8 + j];
+      }
+      float* out = sums + static_cast<int64_t>(blockIdx.x) * 2 * C;
+      out[c] = a;
+      out[C + c] = b;
     }
   } else {
+    for (int i2 = threadIdx.x; i2 < 2 * C; i2 += blockDim.x) lbin[i2] = 0.f;
+    __syncthreads();
     for (; i + 7 < n; i += stride) {
       const short8 v = *reinterpret_cast<const short8*>(
           reinterpret_cast<const short*>(x) + i);
@@ -183,11 +203,10 @@ __global__ void bn_reduce_kernel(const bf16* __restrict__ x, int64_t R,
         atomicAdd(&lbin[C + c0 + j], f * f);
       }
     }
+    __syncthreads();
+    for (int i2 = threadIdx.x; i2 < 2 * C; i2 += blockDim.x)
+      sums[static_cast<int64_t>(blockIdx.x) * 2 * C + i2] = lbin[i2];
   }
-  __syncthreads();
-  // deterministic: per-block partial row, combined in fixed order later
-  for (int i2 = threadIdx.x; i2 < 2 * C; i2 += blockDim.x)
-    sums[static_cast<int64_t>(blockIdx.x) * 2 * C + i2] = lbin[i2];
 }
 
 // combine per-block partial rows (deterministic: fixed lane->entry map +
@@ -270,9 +289,11 @@ __global__ void bn_bwd_reduce_kernel(
     const bf16* __restrict__ x, const float* __restrict__ mean,
     const float* __restrict__ invstd, int64_t R, int C,
     float* __restrict__ acc /*[2C]*/, bf16* __restrict__ gym_out) {
-  extern __shared__ float lbin[];
-  for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) lbin[i] = 0.f;
-  __syncthreads();
+  extern __shared__ float lbin[];  // FIXED: [256][16]; else [2C]
+  if constexpr (!FIXED_C) {
+    for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) lbin[i] = 0.f;
+    __syncthreads();
+  }
   const int64_t n = R * C;
   int64_t i = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) * 8;
   const int64_t stride = grid_stride() * 8;
@@ -400,8 +421,9 @@ void launch_batchnorm_fwd(const void* x_, const void* res_, const float* w,
   const bf16* x = static_cast<const bf16*>(x_);
   const bf16* res = static_cast<const bf16*>(res_);
   bf16* y = static_cast<bf16*>(y_);
-  const size_t lds = 2 * C * sizeof(float);
   const bool fixed = (256 * 8) % C == 0;
+  const size_t lds =
+      (fixed ? 256 * 16 : 2 * C) * sizeof(float);
   const int grid = elemwise_grid(R * C / 8);
   const int rgrid = norms_reduce_grid(R * C / 8);
   if (training) {
@@ -447,8 +469,9 @@ void launch_batchnorm_bwd(const void* gy_, const void* y_, const void* x_,
   const bf16* x = static_cast<const bf16*>(x_);
   bf16* gx = static_cast<bf16*>(gx_);
   bf16* gym = static_cast<bf16*>(gym_);
-  const size_t lds = 2 * C * sizeof(float);
   const bool fixed = (256 * 8) % C == 0;
+  const size_t lds =
+      (fixed ? 256 * 16 : 2 * C) * sizeof(float);
   const int grid = elemwise_grid(R * C / 8);
   const int rgrid = norms_reduce_grid(R * C / 8);
   float* partials = acc;
