@@ -330,15 +330,32 @@ __global__ void bn_bwd_reduce_kernel(
       *reinterpret_cast<short8*>(reinterpret_cast<short*>(gym_out) + i) = gm;
   }
   if constexpr (FIXED_C) {
+    // fixed-order in-block combine (no atomics => deterministic)
+    (void)c0_fixed;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      atomicAdd(&lbin[c0_fixed + j], ab[j]);
-      atomicAdd(&lbin[C + c0_fixed + j], aw[j]);
+      lbin[threadIdx.x * 16 + j] = ab[j];
+      lbin[threadIdx.x * 16 + 8 + j] = aw[j];
     }
+    __syncthreads();
+    const int nb = C / 8;
+    const int gsz = blockDim.x / nb;
+    for (int c = threadIdx.x; c < C; c += blockDim.x) {
+      const int cb = c / 8, j = c % 8;
+      float a = 0.f, b2 = 0.f;
+      for (int k = 0; k < gsz; ++k) {
+        a += lbin[(cb + k * nb) * 16 + j];
+        b2 += lbin[(cb + k * nb) * 16 + 8 + j];
+      }
+      float* out = acc + static_cast<int64_t>(blockIdx.x) * 2 * C;
+      out[c] = a;
+      out[C + c] = b2;
+    }
+  } else {
+    __syncthreads();
+    for (int i2 = threadIdx.x; i2 < 2 * C; i2 += blockDim.x)
+      acc[static_cast<int64_t>(blockIdx.x) * 2 * C + i2] = lbin[i2];
   }
-  __syncthreads();
-  for (int i2 = threadIdx.x; i2 < 2 * C; i2 += blockDim.x)
-    acc[static_cast<int64_t>(blockIdx.x) * 2 * C + i2] = lbin[i2];
 }
 
 // Backward pass 2: gx = w*invstd*(gym - gb/n - xhat*gw/n)   (training)
