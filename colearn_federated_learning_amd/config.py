@@ -54,6 +54,8 @@ class TrainingConfig:
     max_local_steps: Optional[int] = None  # cap steps/epoch (bench + tests)
     # capture the client train step in a hipGraph on the GPU (steprunner.py)
     use_graphs: bool = True
+    # stacked-client execution (federated/stacked.py); sequential if False
+    use_stacked: bool = True
 
 
 @dataclass

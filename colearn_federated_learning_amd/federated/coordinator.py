@@ -33,7 +33,7 @@ from ..ops.dp import PrivacyAccountant, add_dp_noise
 from ..ops.fused_optim import make_flat_optimizer
 from ..ops.topk import topk_compress, topk_decompress
 from ..parallel.aggregator import Aggregator
-from ..parallel.flat import FlatParams
+from ..parallel.flat import FlatBuffers, FlatParams
 from .bus import MessageBus, Topics
 from .checkpoint import load_checkpoint, save_checkpoint
 from .client import FederatedClient
@@ -112,12 +112,28 @@ class FederatedServer:
              cfg.model.image_size)
             if cfg.data.dataset != "text_synth" else None
         )
-        from .steprunner import make_step_runner
+        # FedAvg aggregates full model state: float buffers (BN running
+        # stats) are averaged alongside the weight deltas.
+        self.flatbuf = FlatBuffers(self.model)
+        self.global_buf = self.flatbuf.gather()
+        self.buf_acc = torch.zeros_like(self.global_buf)
+        # stacked-client engine: all local clients train simultaneously
+        # with per-client weights (federated/stacked.py) when applicable
+        from .stacked import StackedEngine, stackable
 
-        self.step_runner = make_step_runner(
-            self.model, self.optimizer, self.flat, cfg, self.device,
-            sample_shape,
+        local_clients = [self.clients[c] for c in sorted(self.clients)]
+        self.stacked = (
+            StackedEngine(cfg, local_clients, self.device)
+            if stackable(cfg, local_clients) else None
         )
+        self.step_runner = None
+        if self.stacked is None:
+            from .steprunner import make_step_runner
+
+            self.step_runner = make_step_runner(
+                self.model, self.optimizer, self.flat, cfg, self.device,
+                sample_shape,
+            )
         if cfg.resume and cfg.checkpoint_dir:
             self._resume()
 
@@ -138,6 +154,7 @@ class FederatedServer:
         self.global_flat.copy_(self.flat.flat_from_state_dict(state))
         self.flat.load_flat(self.global_flat)
         self.sampler.load_state_dict(ck["sampler_state"])
+        self.global_buf = self.flatbuf.gather()
         self.start_round = ck["round"] + 1
         if self.accountant is not None and "dp_steps" in ck:
             self.accountant.steps = ck["dp_steps"]
@@ -173,36 +190,59 @@ class FederatedServer:
 
         # --- local training of this rank's sampled clients -------------
         self.delta_acc.zero_()
+        self.buf_acc.zero_()
         local_weight = 0.0
         local_losses: Dict[int, float] = {}
         n_samples_round = 0
         topk_ratio = cfg.agg.topk_ratio
+
+        def fold_client(cid, delta, bufs, loss, n_samples):
+            nonlocal local_weight, n_samples_round
+            w = 1.0 if cfg.dp.enabled else float(n_samples)
+            if topk_ratio > 0:
+                k = max(int(delta.numel() * topk_ratio), 1)
+                idx, val = topk_compress(delta.contiguous(), k)
+                delta = topk_decompress(idx, val, delta.numel())
+            self.delta_acc.add_(delta, alpha=w)
+            if bufs is not None and bufs.numel():
+                self.buf_acc.add_(bufs.to(self.buf_acc.device), alpha=w)
+            local_weight += w
+            local_losses[cid] = loss
+            n_samples_round += n_samples
+
         with self.timer.phase("train"):
-            for cid in plan.trainers:
-                client = self.clients.get(cid)
-                if client is None:
-                    continue
-                if self._dropped(round_idx, cid):
-                    continue
-                t_fit = time.perf_counter()
-                self.flat.load_flat(self.global_flat)
-                self.optimizer.reset()
-                res = client.fit(self.model, self.optimizer, round_idx,
-                                 runner=self.step_runner)
-                if (cfg.straggler_timeout_s is not None
-                        and time.perf_counter() - t_fit
-                        > cfg.straggler_timeout_s):
-                    continue  # straggler: excluded, aggregation reweights
-                w = 1.0 if cfg.dp.enabled else float(res.n_samples)
-                delta = self.flat.flat_w - self.global_flat
-                if topk_ratio > 0:
-                    k = max(int(delta.numel() * topk_ratio), 1)
-                    idx, val = topk_compress(delta, k)
-                    delta = topk_decompress(idx, val, delta.numel())
-                self.delta_acc.add_(delta, alpha=w)
-                local_weight += w
-                local_losses[cid] = res.loss
-                n_samples_round += res.n_samples
+            if self.stacked is not None:
+                sampled = set(plan.trainers)
+                deltas, bufs, losses, n_per_client = \
+                    self.stacked.train_round(
+                        self.global_flat, self.global_buf, round_idx
+                    )
+                for i, cid in enumerate(self.stacked.client_ids):
+                    if cid not in sampled or self._dropped(round_idx, cid):
+                        continue
+                    fold_client(cid, deltas[i],
+                                bufs[i] if bufs.numel() else None,
+                                losses[cid], n_per_client)
+            else:
+                for cid in plan.trainers:
+                    client = self.clients.get(cid)
+                    if client is None:
+                        continue
+                    if self._dropped(round_idx, cid):
+                        continue
+                    t_fit = time.perf_counter()
+                    self.flat.load_flat(self.global_flat)
+                    self.flatbuf.load(self.global_buf)
+                    self.optimizer.reset()
+                    res = client.fit(self.model, self.optimizer, round_idx,
+                                     runner=self.step_runner)
+                    if (cfg.straggler_timeout_s is not None
+                            and time.perf_counter() - t_fit
+                            > cfg.straggler_timeout_s):
+                        continue  # straggler excluded; reweighted below
+                    fold_client(cid, self.flat.flat_w - self.global_flat,
+                                self.flatbuf.gather(), res.loss,
+                                res.n_samples)
 
         # --- aggregation: weighted all-reduce over xGMI -----------------
         with self.timer.phase("aggregate"):
@@ -217,9 +257,13 @@ class FederatedServer:
                              cfg.dp.noise_multiplier, generator=g)
                 if self.accountant is not None:
                     self.accountant.step()
+            self.aggregator.allreduce_tensor(self.buf_acc)
             if total_weight > 0:
                 self.global_flat.add_(self.delta_acc, alpha=1.0 / total_weight)
+                if self.buf_acc.numel():
+                    self.global_buf.copy_(self.buf_acc / total_weight)
             self.flat.load_flat(self.global_flat)
+            self.flatbuf.load(self.global_buf)
 
         # --- evaluation of the NEW global model -------------------------
         eval_metrics: Dict[str, float] = {}

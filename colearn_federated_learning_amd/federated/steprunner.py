@@ -61,9 +61,12 @@ class GraphedStepRunner:
 
         def one_step():
             flat.flat_g.zero_()
+            loss_fn = getattr(model, "loss_terms", None)
             with torch.autocast("cuda", torch.bfloat16, enabled=use_bf16):
-                loss = model.loss(self.static_x, self.static_y)
-            loss.backward()
+                loss = (loss_fn(self.static_x, self.static_y)
+                        if loss_fn is not None
+                        else model.loss(self.static_x, self.static_y))
+            (loss.sum() if loss.dim() else loss).backward()
             optimizer.device_step()
             return loss
 

@@ -44,7 +44,7 @@ class _Conv2dFn(torch.autograd.Function):
         if _use_native(gy, w, x):
             m = require_ext()
             gx = m.conv2d_dgrad(gy, w, x.shape[2], x.shape[3], s, p)
-            gw = m.conv2d_wgrad(gy, x, w.shape[2], w.shape[3], s, p)
+            gw = m.conv2d_wgrad(gy, x, w.shape[2], w.shape[3], s, p, 1)
         else:
             gx = torch.nn.grad.conv2d_input(x.shape, w, gy, s, p)
             gw = torch.nn.grad.conv2d_weight(x, w.shape, gy, s, p)

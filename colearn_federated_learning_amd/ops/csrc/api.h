@@ -17,14 +17,14 @@ struct ConvShape {
 };
 
 void launch_fused_sgd(float* w, const float* g, float* m, const float* ref,
-                      int64_t n, float lr, float momentum, float wd,
-                      float mu, bool has_mom, bool has_prox,
-                      hipStream_t stream);
+                      int64_t n, int64_t ref_period, float lr,
+                      float momentum, float wd, float mu, bool has_mom,
+                      bool has_prox, hipStream_t stream);
 void launch_fused_adam(float* w, const float* g, float* m, float* v,
                        const float* ref, const float* bc, int64_t n,
-                       float lr, float b1, float b2, float eps, float wd,
-                       float mu, float bc1, float bc2, bool has_prox,
-                       hipStream_t stream);
+                       int64_t ref_period, float lr, float b1, float b2,
+                       float eps, float wd, float mu, float bc1, float bc2,
+                       bool has_prox, hipStream_t stream);
 void launch_sq_norm_per_row(const float* g, float* sq, int64_t rows,
                             int64_t cols, hipStream_t stream);
 void launch_scale_rows_mean(const float* g, const float* scale, float* out,
@@ -46,21 +46,21 @@ void launch_batchnorm_fwd(const void* x, const void* res, const float* w,
                           const float* b, float* running_mean,
                           float* running_var, void* y, float* mean,
                           float* invstd, float* sums, int64_t R, int C,
-                          float momentum, float eps, bool training,
+                          int G, float momentum, float eps, bool training,
                           bool relu, hipStream_t stream);
 void launch_batchnorm_bwd(const void* gy, const void* y, const void* x,
                           const float* w, const float* mean,
                           const float* invstd, void* gx, void* gym,
-                          float* acc, int64_t R, int C, bool training,
-                          bool relu, hipStream_t stream);
+                          float* acc, int64_t R, int C, int G,
+                          bool training, bool relu, hipStream_t stream);
 void launch_gemm_nt(const void* A, const void* B, void* C, int64_t batch,
                     int64_t M, int64_t N, int64_t K, hipStream_t stream);
 void launch_conv_fwd(const void* x, const void* w, void* y, ConvShape cs,
-                     hipStream_t stream);
+                     int G, hipStream_t stream);
 void launch_conv_dgrad(const void* gy, const void* wt, void* gx,
-                       ConvShape cs, hipStream_t stream);
-int conv_wgrad_splitk(ConvShape cs);
+                       ConvShape cs, int G, hipStream_t stream);
+int conv_wgrad_splitk(ConvShape cs, int G);
 void launch_conv_wgrad(const void* gy, const void* x, float* gwf, void* gw,
-                       ConvShape cs, hipStream_t stream);
+                       ConvShape cs, int G, hipStream_t stream);
 
 }  // namespace colearn

@@ -49,12 +49,18 @@ void fused_sgd(torch::Tensor w, torch::Tensor g, torch::Tensor m,
   const c10::hip::HIPGuardMasqueradingAsCUDA guard(w.device());
   const bool has_mom = m.numel() > 0;
   const bool has_prox = ref.numel() > 0;
+  const int64_t period =
+      (has_prox && ref.numel() != w.numel()) ? ref.numel() : 0;
+  if (period) {
+    TORCH_CHECK(w.numel() % period == 0 && period % 4 == 0,
+                "stacked FedProx ref must divide w and be 4-aligned");
+  }
   launch_fused_sgd(w.data_ptr<float>(), g.data_ptr<float>(),
                    has_mom ? m.data_ptr<float>() : nullptr,
                    has_prox ? ref.data_ptr<float>() : nullptr, w.numel(),
-                   static_cast<float>(lr), static_cast<float>(momentum),
-                   static_cast<float>(wd), static_cast<float>(mu), has_mom,
-                   has_prox, stream());
+                   period, static_cast<float>(lr),
+                   static_cast<float>(momentum), static_cast<float>(wd),
+                   static_cast<float>(mu), has_mom, has_prox, stream());
   HIP_CHECK_LAST();
 }
 
@@ -65,10 +71,12 @@ void fused_adam(torch::Tensor w, torch::Tensor g, torch::Tensor m,
   check_f32(w, "w");
   const c10::hip::HIPGuardMasqueradingAsCUDA guard(w.device());
   const bool has_prox = ref.numel() > 0;
+  const int64_t period =
+      (has_prox && ref.numel() != w.numel()) ? ref.numel() : 0;
   launch_fused_adam(
       w.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
       v.data_ptr<float>(), has_prox ? ref.data_ptr<float>() : nullptr,
-      bc.numel() > 0 ? bc.data_ptr<float>() : nullptr, w.numel(),
+      bc.numel() > 0 ? bc.data_ptr<float>() : nullptr, w.numel(), period,
       static_cast<float>(lr), static_cast<float>(b1),
       static_cast<float>(b2), static_cast<float>(eps),
       static_cast<float>(wd), static_cast<float>(mu),
@@ -176,30 +184,33 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor b,
     torch::Tensor running_mean, torch::Tensor running_var,
     torch::Tensor residual, bool training, double momentum, double eps,
-    bool relu) {
+    bool relu, int64_t G) {
   check_bf16(x, "x");
   TORCH_CHECK(x.dim() == 4, "batchnorm expects NCHW tensor");
   const int C = static_cast<int>(x.size(1));
   TORCH_CHECK(C % 8 == 0, "native batchnorm needs C % 8 == 0, got ", C);
+  TORCH_CHECK(x.size(0) % G == 0, "stacked batch must divide by G");
+  TORCH_CHECK(w.numel() == G * C, "stacked BN weight must be [G*C]");
   const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
   const bool has_res = residual.numel() > 0;
   torch::Tensor rc;
   if (has_res) rc = residual.contiguous(torch::MemoryFormat::ChannelsLast);
-  const int64_t R = x.numel() / C;
+  const int64_t R = x.numel() / C / G;  // rows per client
   auto y = torch::empty_like(xc);
-  auto mean = torch::empty({C}, x.options().dtype(torch::kFloat32));
+  auto mean = torch::empty({G * C}, x.options().dtype(torch::kFloat32));
   auto invstd = torch::empty_like(mean);
   const int rgrid = norms_reduce_grid(R * C / 8);
   auto sums = torch::empty(
-      {static_cast<int64_t>(rgrid + 1) * 2 * C}, mean.options());
+      {G * static_cast<int64_t>(rgrid + 1) * 2 * C}, mean.options());
   launch_batchnorm_fwd(
       xc.data_ptr(), has_res ? rc.data_ptr() : nullptr,
       w.data_ptr<float>(), b.data_ptr<float>(),
       running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
       y.data_ptr(), mean.data_ptr<float>(), invstd.data_ptr<float>(),
-      sums.data_ptr<float>(), R, C, static_cast<float>(momentum),
-      static_cast<float>(eps), training, relu, stream());
+      sums.data_ptr<float>(), R, C, static_cast<int>(G),
+      static_cast<float>(momentum), static_cast<float>(eps), training,
+      relu, stream());
   HIP_CHECK_LAST();
   return {y, mean, invstd};
 }
@@ -207,7 +218,7 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
 batchnorm_bwd(torch::Tensor gy, torch::Tensor y, torch::Tensor x,
               torch::Tensor w, torch::Tensor mean, torch::Tensor invstd,
-              bool training, bool relu) {
+              bool training, bool relu, int64_t G) {
   check_bf16(gy, "gy");
   const int C = static_cast<int>(x.size(1));
   const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
@@ -215,23 +226,25 @@ batchnorm_bwd(torch::Tensor gy, torch::Tensor y, torch::Tensor x,
   auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
   torch::Tensor yc;
   if (relu) yc = y.contiguous(torch::MemoryFormat::ChannelsLast);
-  const int64_t R = x.numel() / C;
+  const int64_t R = x.numel() / C / G;
   auto gx = torch::empty_like(xc);
   torch::Tensor gym;  // masked gy == the residual grad
   if (relu) gym = torch::empty_like(gc);
   const int rgrid = norms_reduce_grid(R * C / 8);
-  auto acc = torch::empty({static_cast<int64_t>(rgrid + 1) * 2 * C},
+  auto acc = torch::empty({G * static_cast<int64_t>(rgrid + 1) * 2 * C},
                           x.options().dtype(torch::kFloat32));
   launch_batchnorm_bwd(gc.data_ptr(), relu ? yc.data_ptr() : nullptr,
                        xc.data_ptr(), w.data_ptr<float>(),
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
                        gx.data_ptr(), relu ? gym.data_ptr() : nullptr,
-                       acc.data_ptr<float>(), R, C, training, relu,
-                       stream());
+                       acc.data_ptr<float>(), R, C, static_cast<int>(G),
+                       training, relu, stream());
   HIP_CHECK_LAST();
-  const int64_t off = static_cast<int64_t>(rgrid) * 2 * C;
-  auto gb = acc.narrow(0, off, C).clone();
-  auto gw = acc.narrow(0, off + C, C).clone();
+  const int64_t off = static_cast<int64_t>(G) * rgrid * 2 * C;
+  // combined region is [G][2C]: gb = even C-halves, gw = odd C-halves
+  auto comb = acc.narrow(0, off, G * 2 * C).view({G, 2, C});
+  auto gb = comb.select(1, 0).reshape({G * C}).clone();
+  auto gw = comb.select(1, 1).reshape({G * C}).clone();
   return {gx, gw, gb, relu ? gym : gc};
 }
 
@@ -283,16 +296,19 @@ torch::Tensor bmm_tn(torch::Tensor a, torch::Tensor b) {
 }
 
 // ---- conv family ----------------------------------------------------------
+// weight 4-D [CO,CI,R,S] = single model; 5-D [G,CO,CI,R,S] = stacked
+// clients (x batch is then G*B, client-major).
 ConvShape make_shape(const torch::Tensor& x, const torch::Tensor& w,
-                     int64_t stride, int64_t pad) {
+                     int64_t stride, int64_t pad, int64_t G) {
   ConvShape cs;
-  cs.N = static_cast<int>(x.size(0));
+  cs.N = static_cast<int>(x.size(0) / G);  // batch per client
   cs.C = static_cast<int>(x.size(1));
   cs.H = static_cast<int>(x.size(2));
   cs.W = static_cast<int>(x.size(3));
-  cs.K = static_cast<int>(w.size(0));
-  cs.R = static_cast<int>(w.size(2));
-  cs.S = static_cast<int>(w.size(3));
+  const int wd = w.dim();
+  cs.K = static_cast<int>(w.size(wd - 4));
+  cs.R = static_cast<int>(w.size(wd - 2));
+  cs.S = static_cast<int>(w.size(wd - 1));
   cs.stride = static_cast<int>(stride);
   cs.pad = static_cast<int>(pad);
   cs.P = (cs.H + 2 * cs.pad - cs.R) / cs.stride + 1;
@@ -305,14 +321,18 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
   check_bf16(x, "x");
   check_bf16(w, "w");
   const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
-  ConvShape cs = make_shape(x, w, stride, pad);
+  const int64_t G = w.dim() == 5 ? w.size(0) : 1;
+  ConvShape cs = make_shape(x, w, stride, pad, G);
   auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
-  // repack weight K-major: [CO][R][S][CI]
-  auto wp = w.permute({0, 2, 3, 1}).contiguous();
-  auto y = torch::empty({cs.N, cs.K, cs.P, cs.Q},
+  // repack weight K-major: [(G)][CO][R][S][CI]
+  auto wp = (w.dim() == 5 ? w.permute({0, 1, 3, 4, 2})
+                          : w.permute({0, 2, 3, 1}))
+                .contiguous();
+  auto y = torch::empty({x.size(0), cs.K, cs.P, cs.Q},
                         x.options().memory_format(
                             torch::MemoryFormat::ChannelsLast));
-  launch_conv_fwd(xc.data_ptr(), wp.data_ptr(), y.data_ptr(), cs, stream());
+  launch_conv_fwd(xc.data_ptr(), wp.data_ptr(), y.data_ptr(), cs,
+                  static_cast<int>(G), stream());
   HIP_CHECK_LAST();
   return y;
 }
@@ -321,36 +341,41 @@ torch::Tensor conv2d_dgrad(torch::Tensor gy, torch::Tensor w, int64_t H,
                            int64_t W, int64_t stride, int64_t pad) {
   check_bf16(gy, "gy");
   const c10::hip::HIPGuardMasqueradingAsCUDA guard(gy.device());
+  const int64_t G = w.dim() == 5 ? w.size(0) : 1;
+  const int wd = w.dim();
   ConvShape cs;
-  cs.N = static_cast<int>(gy.size(0));
+  cs.N = static_cast<int>(gy.size(0) / G);
   cs.K = static_cast<int>(gy.size(1));
   cs.P = static_cast<int>(gy.size(2));
   cs.Q = static_cast<int>(gy.size(3));
-  cs.C = static_cast<int>(w.size(1));
-  cs.R = static_cast<int>(w.size(2));
-  cs.S = static_cast<int>(w.size(3));
+  cs.C = static_cast<int>(w.size(wd - 3));
+  cs.R = static_cast<int>(w.size(wd - 2));
+  cs.S = static_cast<int>(w.size(wd - 1));
   cs.H = static_cast<int>(H);
   cs.W = static_cast<int>(W);
   cs.stride = static_cast<int>(stride);
   cs.pad = static_cast<int>(pad);
   auto gc = gy.contiguous(torch::MemoryFormat::ChannelsLast);
-  // wT K-major: [CI][R][S][CO]
-  auto wt = w.permute({1, 2, 3, 0}).contiguous();
-  auto gx = torch::empty({cs.N, cs.C, cs.H, cs.W},
+  // wT K-major: [(G)][CI][R][S][CO]
+  auto wt = (w.dim() == 5 ? w.permute({0, 2, 3, 4, 1})
+                          : w.permute({1, 2, 3, 0}))
+                .contiguous();
+  auto gx = torch::empty({gy.size(0), cs.C, cs.H, cs.W},
                          gy.options().memory_format(
                              torch::MemoryFormat::ChannelsLast));
   launch_conv_dgrad(gc.data_ptr(), wt.data_ptr(), gx.data_ptr(), cs,
-                    stream());
+                    static_cast<int>(G), stream());
   HIP_CHECK_LAST();
   return gx;
 }
 
 torch::Tensor conv2d_wgrad(torch::Tensor gy, torch::Tensor x, int64_t R,
-                           int64_t S, int64_t stride, int64_t pad) {
+                           int64_t S, int64_t stride, int64_t pad,
+                           int64_t G) {
   check_bf16(gy, "gy");
   const c10::hip::HIPGuardMasqueradingAsCUDA guard(gy.device());
   ConvShape cs;
-  cs.N = static_cast<int>(x.size(0));
+  cs.N = static_cast<int>(x.size(0) / G);
   cs.C = static_cast<int>(x.size(1));
   cs.H = static_cast<int>(x.size(2));
   cs.W = static_cast<int>(x.size(3));
@@ -364,14 +389,15 @@ torch::Tensor conv2d_wgrad(torch::Tensor gy, torch::Tensor x, int64_t R,
   auto gc = gy.contiguous(torch::MemoryFormat::ChannelsLast);
   auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
   const int64_t rsc = static_cast<int64_t>(cs.R) * cs.S * cs.C;
-  const int splitk = conv_wgrad_splitk(cs);
-  auto gwf = torch::empty({splitk, cs.K, rsc},
+  const int splitk = conv_wgrad_splitk(cs, static_cast<int>(G));
+  auto gwf = torch::empty({G * splitk, cs.K, rsc},
                           x.options().dtype(torch::kFloat32));
-  auto gw_rsci = torch::empty({cs.K, cs.R, cs.S, cs.C}, x.options());
+  auto gw_rsci = torch::empty({G, cs.K, cs.R, cs.S, cs.C}, x.options());
   launch_conv_wgrad(gc.data_ptr(), xc.data_ptr(), gwf.data_ptr<float>(),
-                    gw_rsci.data_ptr(), cs, stream());
+                    gw_rsci.data_ptr(), cs, static_cast<int>(G), stream());
   HIP_CHECK_LAST();
-  return gw_rsci.permute({0, 3, 1, 2}).contiguous();
+  auto out = gw_rsci.permute({0, 1, 4, 2, 3}).contiguous();
+  return G == 1 ? out.squeeze(0) : out;
 }
 
 }  // namespace

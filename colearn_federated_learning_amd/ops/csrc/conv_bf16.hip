@@ -1,17 +1,25 @@
 // Implicit-GEMM 2-D convolution for gfx950, channels-last bf16
-// (ResNet 3x3 & 1x1 convs, fwd + dgrad + wgrad).
+// (ResNet 3x3 & 1x1 convs, fwd + dgrad + wgrad), with
+// STACKED-CLIENT execution.
 //
 // All three passes are MFMA tile GEMMs that GATHER their A operand from
 // the NHWC image instead of materializing im2col:
 //   fwd   : y[npq, co]  = x_gather[npq, rsci] @ w[co, rsci]^T      (NT)
 //   dgrad : gx[nhw, ci] = gy_gather[nhw, rsco] @ wT[ci, rsco]^T    (NT)
 //   wgrad : gw[co, rsci] = gy[npq, co]^T @ x_gather[npq, rsci]     (TN,
-//           split-K over npq with fp32 atomic accumulation)
-// Weights arrive pre-permuted K-major from the wrapper ([CO][R][S][CI]
-// for fwd, [CI][R][S][CO] for dgrad) so their staging is identical to the
-// dense GEMM's.  Channel gathers vectorize 8-wide whenever C % 8 == 0
-// (every ResNet conv except the 3-channel stem, which takes the scalar
-// path).
+//           deterministic split-K slabs over npq)
+//
+// Stacking (the FL multiplexing answer on a 256-CU chip): G virtual
+// clients train simultaneously with per-client weights w[G][...] and
+// batches stacked along M; grid.z = client.  Each block offsets its
+// operand pointers by its client index, so tile structure is unchanged
+// while the launch gets G x the workgroups — this is what fills the chip
+// for the small deep-layer shapes (measured: ResNet-18 layer4 tiles
+// alone = 32 workgroups = 12% occupancy at G=1).
+//
+// Weights arrive pre-permuted K-major from the wrapper ([G][CO][R][S][CI]
+// fwd, [G][CI][R][S][CO] dgrad).  Channel gathers vectorize 8-wide when
+// C % 8 == 0 (every ResNet conv except the 3-channel stem).
 
 #include "api.h"
 #include "tile_gemm.h"
@@ -56,6 +64,9 @@ DEV_INLINE void conv_fwd_gather(const short* __restrict__ x,
 }
 
 // ---- dgrad A-gather: row m = (n,h,w), k = (r,s,co) -----------------------
+// STRIDE1: compile-time fast path without the divisibility checks (most
+// ResNet convs are stride 1).
+template <bool STRIDE1>
 DEV_INLINE void conv_dgrad_gather(const short* __restrict__ gy,
                                   const ConvShape cs, int64_t m, int64_t k,
                                   short8v_t& out) {
@@ -71,8 +82,16 @@ DEV_INLINE void conv_dgrad_gather(const short* __restrict__ gy,
     const int r = static_cast<int>(k / (static_cast<int64_t>(cs.K) * cs.S));
     const int ph = h + cs.pad - r;
     const int pw = w + cs.pad - s;
-    if (ph < 0 || pw < 0 || ph % cs.stride || pw % cs.stride) return;
-    const int po = ph / cs.stride, qo = pw / cs.stride;
+    if (ph < 0 || pw < 0) return;
+    int po, qo;
+    if constexpr (STRIDE1) {
+      po = ph;
+      qo = pw;
+    } else {
+      if (ph % cs.stride || pw % cs.stride) return;
+      po = ph / cs.stride;
+      qo = pw / cs.stride;
+    }
     if (po >= cs.P || qo >= cs.Q) return;
     out = *reinterpret_cast<const short8v_t*>(
         gy + ((static_cast<int64_t>(n) * cs.P + po) * cs.Q + qo) * cs.K + co);
@@ -96,16 +115,23 @@ DEV_INLINE void conv_dgrad_gather(const short* __restrict__ gy,
   }
 }
 
-// ---- the shared implicit-GEMM NT kernel (fwd / dgrad via GATHER flag) ----
-template <int GATHER>  // 0 = conv fwd (gather x), 1 = dgrad (gather gy)
+// ---- the shared implicit-GEMM NT kernel ----------------------------------
+// GATHER: 0 = conv fwd (gather x), 1 = dgrad stride1, 2 = dgrad strided
+template <int GATHER>
 __global__ __launch_bounds__(kBlockThreads) void conv_nt_kernel(
     const short* __restrict__ Asrc, const short* __restrict__ Bw,
-    bf16* __restrict__ C, ConvShape cs, int64_t M, int64_t Nd, int64_t Kd,
+    bf16* __restrict__ C, ConvShape cs, int64_t Mc /* rows per client */,
+    int64_t Nd, int64_t Kd, int64_t a_stride /* A elems per client */,
     int nbm, int nbn) {
   __shared__ short lds[2][2][BM * BK];
   const int bid = xcd_swizzle(blockIdx.x, nbm * nbn);
-  const int64_t bm = static_cast<int64_t>(bid / nbn) * BM;
+  const int64_t bm_local = static_cast<int64_t>(bid / nbn) * BM;
   const int64_t bn = static_cast<int64_t>(bid % nbn) * BN;
+  const int g = blockIdx.z;  // stacked client index
+  const short* A = Asrc + static_cast<int64_t>(g) * a_stride;
+  const short* Bg = Bw + static_cast<int64_t>(g) * Nd * Kd;
+  bf16* Cg = C + static_cast<int64_t>(g) * Mc * Nd;
+
   const int wid = threadIdx.x / kWave;
   const int wm = (wid >> 1) * 64, wn = (wid & 1) * 64;
 
@@ -122,14 +148,17 @@ __global__ __launch_bounds__(kBlockThreads) void conv_nt_kernel(
   short8v_t a0, a1, b0, b1;
   auto load_tile = [&](int t) {
     const int64_t k = static_cast<int64_t>(t) * BK + half * 16;
-    const int64_t am = bm + row;
-    if (am < M) {
+    const int64_t am = bm_local + row;
+    if (am < Mc) {
       if constexpr (GATHER == 0) {
-        conv_fwd_gather(Asrc, cs, am, k, a0);
-        conv_fwd_gather(Asrc, cs, am, k + 8, a1);
+        conv_fwd_gather(A, cs, am, k, a0);
+        conv_fwd_gather(A, cs, am, k + 8, a1);
+      } else if constexpr (GATHER == 1) {
+        conv_dgrad_gather<true>(A, cs, am, k, a0);
+        conv_dgrad_gather<true>(A, cs, am, k + 8, a1);
       } else {
-        conv_dgrad_gather(Asrc, cs, am, k, a0);
-        conv_dgrad_gather(Asrc, cs, am, k + 8, a1);
+        conv_dgrad_gather<false>(A, cs, am, k, a0);
+        conv_dgrad_gather<false>(A, cs, am, k + 8, a1);
       }
     } else {
       a0 = a1 = mfma_frag_zero();
@@ -137,7 +166,7 @@ __global__ __launch_bounds__(kBlockThreads) void conv_nt_kernel(
     b0 = b1 = mfma_frag_zero();
     const int64_t bn_row = bn + row;
     if (bn_row < Nd) {
-      const short* p = Bw + bn_row * Kd + k;
+      const short* p = Bg + bn_row * Kd + k;
       if (k + 16 <= Kd) {
         b0 = *reinterpret_cast<const short8v_t*>(p);
         b1 = *reinterpret_cast<const short8v_t*>(p + 8);
@@ -145,7 +174,10 @@ __global__ __launch_bounds__(kBlockThreads) void conv_nt_kernel(
 #pragma unroll
         for (int j = 0; j < 16; ++j)
           if (k + j < Kd) {
-            if (j < 8) b0[j] = p[j]; else b1[j - 8] = p[j];
+            if (j < 8)
+              b0[j] = p[j];
+            else
+              b1[j - 8] = p[j];
           }
       }
     }
@@ -184,22 +216,26 @@ __global__ __launch_bounds__(kBlockThreads) void conv_nt_kernel(
     __syncthreads();
     cur ^= 1;
   }
-  store_tile_bf16(reinterpret_cast<const float*>(acc), C, Nd, M, Nd, bm, bn,
-                  wm, wn);
+  store_tile_bf16(reinterpret_cast<const float*>(acc), Cg, Nd, Mc, Nd,
+                  bm_local, bn, wm, wn);
 }
 
 // ==== wgrad ===============================================================
-// gw[co, rsci] += sum over an npq chunk; 64x64 tile per block, 4 waves in
-// 2x2, BK=32 over npq; both operands transposed-staged into K-major LDS
-// (scatter b16 writes — v1 correctness; tr_b16 reads are the planned
-// optimization).  fp32 atomicAdd epilogue combines the split-K blocks.
+// gw[g, co, rsci]: 64x64 tile per block, BK=32 over npq; deterministic
+// split-K slabs per client.  grid = (tiles, splitk, G).
 __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
     const short* __restrict__ gy, const short* __restrict__ x,
     float* __restrict__ gwf, ConvShape cs, int64_t NPQ, int64_t RSC,
     int nbm, int nbn, int splitk) {
   constexpr int TM = 64, TN = 64;
-  __shared__ short Gs[TM][BK + 2];  // [co][k] (+2 pad: scatter writes)
-  __shared__ short Xs[TN][BK + 2];  // [rsci][k]
+  __shared__ short Gs[TM][BK + 2];
+  __shared__ short Xs[TN][BK + 2];
+  const int g = blockIdx.z;
+  const short* gyg = gy + static_cast<int64_t>(g) * NPQ * cs.K;
+  const short* xg =
+      x + static_cast<int64_t>(g) * cs.N * cs.H * cs.W * cs.C;
+  float* slab = gwf + (static_cast<int64_t>(g) * splitk + blockIdx.y) *
+                          cs.K * RSC;
   const int tile = blockIdx.x;
   const int co0 = (tile / nbn) * TM;
   const int rc0 = (tile % nbn) * TN;
@@ -215,18 +251,16 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
 #pragma unroll
     for (int n = 0; n < 2; ++n) acc[m][n] = {0.f, 0.f, 0.f, 0.f};
 
-  // staging decomposition: 256 threads, 8 elems each of G and X per step.
-  const int kk = threadIdx.x / 8;        // 0..31 (npq within step)
-  const int e8 = (threadIdx.x % 8) * 8;  // 8-element column block
+  const int kk = threadIdx.x / 8;
+  const int e8 = (threadIdx.x % 8) * 8;
 
   for (int64_t k0 = k_begin; k0 < k_end; k0 += BK) {
     const int64_t npq = k0 + kk;
     short8v_t gv = mfma_frag_zero();
     short8v_t xv = mfma_frag_zero();
     if (npq < k_end) {
-      // G: gy[npq][co0+e8 .. +8]
       if (co0 + e8 < cs.K) {
-        const short* p = gy + npq * cs.K + co0 + e8;
+        const short* p = gyg + npq * cs.K + co0 + e8;
         if ((cs.K & 7) == 0 && co0 + e8 + 8 <= cs.K)
           gv = *reinterpret_cast<const short8v_t*>(p);
         else
@@ -234,17 +268,15 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
           for (int j = 0; j < 8; ++j)
             if (co0 + e8 + j < cs.K) gv[j] = p[j];
       }
-      // X: x_gather[npq][rc0+e8 .. +8]
-      conv_fwd_gather(x, cs, npq, rc0 + e8, xv);
+      conv_fwd_gather(xg, cs, npq, rc0 + e8, xv);
     }
-    __syncthreads();  // previous iteration's reads complete
+    __syncthreads();
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       Gs[e8 + j][kk] = gv[j];
       Xs[e8 + j][kk] = xv[j];
     }
     __syncthreads();
-    // fragments: A rows = co (l&15), k = (l>>4)*8 + j from Gs[co][k]
     const int lane = threadIdx.x % kWave;
     const int fr = lane & 15, kg = (lane >> 4) * 8;
     short8v_t af[2], bfr[2];
@@ -267,9 +299,7 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
         acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af[m], bfr[n], acc[m][n], 0, 0, 0);
   }
-  // epilogue: plain stores into this block's split-K slab (deterministic;
-  // the combine kernel sums slabs in fixed slice order).
-  float* slab = gwf + static_cast<int64_t>(blockIdx.y) * cs.K * RSC;
+  // deterministic slab epilogue
   const int lane = threadIdx.x % kWave;
   const int cif = lane & 15, rg = lane >> 4;
 #pragma unroll
@@ -285,71 +315,79 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
       }
 }
 
-// sum the split-K slabs in slice order and emit bf16.
+// sum the split-K slabs of each client in slice order and emit bf16.
 __global__ void wgrad_combine_kernel(const float* __restrict__ gwf,
-                                     bf16* __restrict__ gw, int64_t n,
+                                     bf16* __restrict__ gw, int64_t n_per_g,
                                      int splitk) {
+  const int g = blockIdx.y;
+  const float* base = gwf + static_cast<int64_t>(g) * splitk * n_per_g;
+  bf16* out = gw + static_cast<int64_t>(g) * n_per_g;
   int64_t i = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
-  for (; i < n; i += grid_stride()) {
+  const int64_t stride = static_cast<int64_t>(blockDim.x) * gridDim.x;
+  for (; i < n_per_g; i += stride) {
     float acc = 0.f;
-    for (int sk = 0; sk < splitk; ++sk) acc += gwf[sk * n + i];
-    gw[i] = f2b(acc);
+    for (int sk = 0; sk < splitk; ++sk) acc += base[sk * n_per_g + i];
+    out[i] = f2b(acc);
   }
-}
-
-__global__ void f32_to_bf16_kernel(const float* __restrict__ in,
-                                   bf16* __restrict__ out, int64_t n) {
-  int64_t i = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
-  for (; i < n; i += grid_stride()) out[i] = f2b(in[i]);
 }
 
 // ---- launchers ------------------------------------------------------------
 void launch_conv_fwd(const void* x, const void* w, void* y, ConvShape cs,
-                     hipStream_t stream) {
-  const int64_t M = static_cast<int64_t>(cs.N) * cs.P * cs.Q;
+                     int G, hipStream_t stream) {
+  const int64_t Mc = static_cast<int64_t>(cs.N) * cs.P * cs.Q;
   const int64_t Kd = static_cast<int64_t>(cs.R) * cs.S * cs.C;
-  const int nbm = static_cast<int>((M + BM - 1) / BM);
+  const int64_t a_stride = static_cast<int64_t>(cs.N) * cs.H * cs.W * cs.C;
+  const int nbm = static_cast<int>((Mc + BM - 1) / BM);
   const int nbn = (cs.K + BN - 1) / BN;
-  conv_nt_kernel<0><<<dim3(nbm * nbn), kBlockThreads, 0, stream>>>(
+  conv_nt_kernel<0><<<dim3(nbm * nbn, 1, G), kBlockThreads, 0, stream>>>(
       static_cast<const short*>(x), static_cast<const short*>(w),
-      static_cast<bf16*>(y), cs, M, cs.K, Kd, nbm, nbn);
+      static_cast<bf16*>(y), cs, Mc, cs.K, Kd, a_stride, nbm, nbn);
 }
 
 void launch_conv_dgrad(const void* gy, const void* wt, void* gx,
-                       ConvShape cs, hipStream_t stream) {
-  const int64_t M = static_cast<int64_t>(cs.N) * cs.H * cs.W;
+                       ConvShape cs, int G, hipStream_t stream) {
+  const int64_t Mc = static_cast<int64_t>(cs.N) * cs.H * cs.W;
   const int64_t Kd = static_cast<int64_t>(cs.R) * cs.S * cs.K;
-  const int nbm = static_cast<int>((M + BM - 1) / BM);
+  const int64_t a_stride = static_cast<int64_t>(cs.N) * cs.P * cs.Q * cs.K;
+  const int nbm = static_cast<int>((Mc + BM - 1) / BM);
   const int nbn = (cs.C + BN - 1) / BN;
-  conv_nt_kernel<1><<<dim3(nbm * nbn), kBlockThreads, 0, stream>>>(
-      static_cast<const short*>(gy), static_cast<const short*>(wt),
-      static_cast<bf16*>(gx), cs, M, cs.C, Kd, nbm, nbn);
+  if (cs.stride == 1)
+    conv_nt_kernel<1><<<dim3(nbm * nbn, 1, G), kBlockThreads, 0, stream>>>(
+        static_cast<const short*>(gy), static_cast<const short*>(wt),
+        static_cast<bf16*>(gx), cs, Mc, cs.C, Kd, a_stride, nbm, nbn);
+  else
+    conv_nt_kernel<2><<<dim3(nbm * nbn, 1, G), kBlockThreads, 0, stream>>>(
+        static_cast<const short*>(gy), static_cast<const short*>(wt),
+        static_cast<bf16*>(gx), cs, Mc, cs.C, Kd, a_stride, nbm, nbn);
 }
 
-void launch_conv_wgrad(const void* gy, const void* x, float* gwf, void* gw,
-                       ConvShape cs, hipStream_t stream) {
+int conv_wgrad_splitk(ConvShape cs, int G) {
   const int64_t NPQ = static_cast<int64_t>(cs.N) * cs.P * cs.Q;
   const int64_t RSC = static_cast<int64_t>(cs.R) * cs.S * cs.C;
   const int nbm = (cs.K + 63) / 64;
   const int nbn = static_cast<int>((RSC + 63) / 64);
-  const int splitk = conv_wgrad_splitk(cs);
-  conv_wgrad_kernel<<<dim3(nbm * nbn, splitk), kBlockThreads, 0, stream>>>(
-      static_cast<const short*>(gy), static_cast<const short*>(x), gwf, cs,
-      NPQ, RSC, nbm, nbn, splitk);
-  const int64_t n = static_cast<int64_t>(cs.K) * RSC;
-  wgrad_combine_kernel<<<elemwise_grid(n), 256, 0, stream>>>(
-      gwf, static_cast<bf16*>(gw), n, splitk);
-}
-
-int conv_wgrad_splitk(ConvShape cs) {
-  const int64_t NPQ = static_cast<int64_t>(cs.N) * cs.P * cs.Q;
-  const int64_t RSC = static_cast<int64_t>(cs.R) * cs.S * cs.C;
-  const int nbm = (cs.K + 63) / 64;
-  const int nbn = static_cast<int>((RSC + 63) / 64);
-  int64_t want = 512 / (nbm * nbn);
+  int64_t want = 512 / (static_cast<int64_t>(nbm) * nbn * G);
   if (want < 1) want = 1;
   int64_t cap = (NPQ + BK - 1) / BK;
   return static_cast<int>(want < cap ? want : cap);
+}
+
+void launch_conv_wgrad(const void* gy, const void* x, float* gwf, void* gw,
+                       ConvShape cs, int G, hipStream_t stream) {
+  const int64_t NPQ = static_cast<int64_t>(cs.N) * cs.P * cs.Q;
+  const int64_t RSC = static_cast<int64_t>(cs.R) * cs.S * cs.C;
+  const int nbm = (cs.K + 63) / 64;
+  const int nbn = static_cast<int>((RSC + 63) / 64);
+  const int splitk = conv_wgrad_splitk(cs, G);
+  conv_wgrad_kernel<<<dim3(nbm * nbn, splitk, G), kBlockThreads, 0,
+                      stream>>>(static_cast<const short*>(gy),
+                                static_cast<const short*>(x), gwf, cs, NPQ,
+                                RSC, nbm, nbn, splitk);
+  const int64_t n_per_g = static_cast<int64_t>(cs.K) * RSC;
+  int cgrid_x = elemwise_grid(n_per_g, 256, G > 0 ? 2048 / G : 2048);
+  if (cgrid_x < 1) cgrid_x = 1;
+  wgrad_combine_kernel<<<dim3(cgrid_x, G), 256, 0, stream>>>(
+      gwf, static_cast<bf16*>(gw), n_per_g, splitk);
 }
 
 }  // namespace colearn

@@ -13,13 +13,16 @@ namespace colearn {
 
 // w, g, m are flat fp32; HAS_MOM / HAS_PROX compile-time to keep the
 // inner loop branch-free (guide G5).
+// ref_period > 0: FedProx reference wraps every ref_period elements —
+// the stacked-client case, where w is [G][P] but the global reference is
+// one [P] vector (P padded to a multiple of 4 by parallel.flat).
 template <bool HAS_MOM, bool HAS_PROX>
 __global__ void fused_sgd_kernel(float* __restrict__ w,
                                  const float* __restrict__ g,
                                  float* __restrict__ m,
                                  const float* __restrict__ ref, int64_t n,
-                                 float lr, float momentum, float wd,
-                                 float mu) {
+                                 int64_t ref_period, float lr,
+                                 float momentum, float wd, float mu) {
   int64_t i = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) * 4;
   const int64_t stride = grid_stride() * 4;
   for (; i + 3 < n; i += stride) {
@@ -30,7 +33,8 @@ __global__ void fused_sgd_kernel(float* __restrict__ w,
 #pragma unroll
     for (int k = 0; k < 4; ++k) gx[k] += wd * wx[k];
     if constexpr (HAS_PROX) {
-      const float4 rv = *reinterpret_cast<const float4*>(ref + i);
+      const int64_t ri = ref_period > 0 ? (i % ref_period) : i;
+      const float4 rv = *reinterpret_cast<const float4*>(ref + ri);
       const float rx[4] = {rv.x, rv.y, rv.z, rv.w};
 #pragma unroll
       for (int k = 0; k < 4; ++k) gx[k] += mu * (wx[k] - rx[k]);
@@ -55,7 +59,8 @@ __global__ void fused_sgd_kernel(float* __restrict__ w,
   if (blockIdx.x == 0 && threadIdx.x < (n & 3)) {
     const int64_t j = (n & ~int64_t(3)) + threadIdx.x;
     float gj = g[j] + wd * w[j];
-    if constexpr (HAS_PROX) gj += mu * (w[j] - ref[j]);
+    if constexpr (HAS_PROX)
+      gj += mu * (w[j] - ref[ref_period > 0 ? j % ref_period : j]);
     if constexpr (HAS_MOM) {
       m[j] = momentum * m[j] + gj;
       gj = m[j];
@@ -71,9 +76,9 @@ __global__ void fused_adam_kernel(float* __restrict__ w,
                                   float* __restrict__ v,
                                   const float* __restrict__ ref,
                                   const float* __restrict__ bc /*[2] or null*/,
-                                  int64_t n, float lr, float b1, float b2,
-                                  float eps, float wd, float mu, float bc1,
-                                  float bc2) {
+                                  int64_t n, int64_t ref_period, float lr,
+                                  float b1, float b2, float eps, float wd,
+                                  float mu, float bc1, float bc2) {
   // device-side bias correction: under hipGraph replay the scalar args are
   // frozen at capture, so a live [bc1, bc2] tensor overrides them.
   if (bc != nullptr) {
@@ -96,7 +101,8 @@ __global__ void fused_adam_kernel(float* __restrict__ w,
     float vx[4] = {vv.x, vv.y, vv.z, vv.w};
     float rx[4] = {0.f, 0.f, 0.f, 0.f};
     if constexpr (HAS_PROX) {
-      const float4 rv = *reinterpret_cast<const float4*>(ref + i);
+      const int64_t ri = ref_period > 0 ? (i % ref_period) : i;
+      const float4 rv = *reinterpret_cast<const float4*>(ref + ri);
       rx[0] = rv.x; rx[1] = rv.y; rx[2] = rv.z; rx[3] = rv.w;
     }
 #pragma unroll
@@ -118,7 +124,8 @@ __global__ void fused_adam_kernel(float* __restrict__ w,
   if (blockIdx.x == 0 && threadIdx.x < (n & 3)) {
     const int64_t j = (n & ~int64_t(3)) + threadIdx.x;
     float gj = g[j] + wd * w[j];
-    if constexpr (HAS_PROX) gj += mu * (w[j] - ref[j]);
+    if constexpr (HAS_PROX)
+      gj += mu * (w[j] - ref[ref_period > 0 ? j % ref_period : j]);
     m[j] = b1 * m[j] + (1.f - b1) * gj;
     v[j] = b2 * v[j] + (1.f - b2) * gj * gj;
     w[j] -= step * m[j] / (sqrtf(v[j] * inv_bc2) + eps);
@@ -126,37 +133,40 @@ __global__ void fused_adam_kernel(float* __restrict__ w,
 }
 
 void launch_fused_sgd(float* w, const float* g, float* m, const float* ref,
-                      int64_t n, float lr, float momentum, float wd, float mu,
-                      bool has_mom, bool has_prox, hipStream_t stream) {
+                      int64_t n, int64_t ref_period, float lr,
+                      float momentum, float wd, float mu, bool has_mom,
+                      bool has_prox, hipStream_t stream) {
   const int block = 256;
   const int grid = elemwise_grid((n + 3) / 4, block);
   if (has_mom && has_prox)
    hipLaunchKernelGGL(( fused_sgd_kernel<true, true>), dim3(grid), dim3(block), 0, stream, 
-        w, g, m, ref, n, lr, momentum, wd, mu);
+        w, g, m, ref, n, ref_period, lr, momentum, wd, mu);
   else if (has_mom)
    hipLaunchKernelGGL(( fused_sgd_kernel<true, false>), dim3(grid), dim3(block), 0, stream, 
-        w, g, m, ref, n, lr, momentum, wd, mu);
+        w, g, m, ref, n, ref_period, lr, momentum, wd, mu);
   else if (has_prox)
    hipLaunchKernelGGL(( fused_sgd_kernel<false, true>), dim3(grid), dim3(block), 0, stream, 
-        w, g, m, ref, n, lr, momentum, wd, mu);
+        w, g, m, ref, n, ref_period, lr, momentum, wd, mu);
   else
    hipLaunchKernelGGL(( fused_sgd_kernel<false, false>), dim3(grid), dim3(block), 0, stream, 
-        w, g, m, ref, n, lr, momentum, wd, mu);
+        w, g, m, ref, n, ref_period, lr, momentum, wd, mu);
 }
 
 void launch_fused_adam(float* w, const float* g, float* m, float* v,
                        const float* ref, const float* bc, int64_t n,
-                       float lr, float b1, float b2, float eps, float wd,
-                       float mu, float bc1, float bc2, bool has_prox,
-                       hipStream_t stream) {
+                       int64_t ref_period, float lr, float b1, float b2,
+                       float eps, float wd, float mu, float bc1, float bc2,
+                       bool has_prox, hipStream_t stream) {
   const int block = 256;
   const int grid = elemwise_grid((n + 3) / 4, block);
   if (has_prox)
    hipLaunchKernelGGL(( fused_adam_kernel<true>), dim3(grid), dim3(block), 0, stream, 
-        w, g, m, v, ref, bc, n, lr, b1, b2, eps, wd, mu, bc1, bc2);
+        w, g, m, v, ref, bc, n, ref_period, lr, b1, b2, eps, wd, mu, bc1,
+        bc2);
   else
    hipLaunchKernelGGL(( fused_adam_kernel<false>), dim3(grid), dim3(block), 0, stream, 
-        w, g, m, v, ref, bc, n, lr, b1, b2, eps, wd, mu, bc1, bc2);
+        w, g, m, v, ref, bc, n, ref_period, lr, b1, b2, eps, wd, mu, bc1,
+        bc2);
 }
 
 }  // namespace colearn

@@ -153,6 +153,9 @@ __global__ void bn_reduce_kernel(const bf16* __restrict__ x, int64_t R,
                                  int C, float* __restrict__ sums) {
   extern __shared__ float lbin[];  // FIXED: [256][16]; else [2C]
   const int64_t n = R * C;
+  // stacked clients: grid.z picks the client's activation + partial slab
+  x += static_cast<int64_t>(blockIdx.z) * n;
+  sums += static_cast<int64_t>(blockIdx.z) * gridDim.x * 2 * C;
   int64_t i = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) * 8;
   const int64_t stride = grid_stride() * 8;
   if constexpr (FIXED_C) {
@@ -213,6 +216,8 @@ __global__ void bn_reduce_kernel(const bf16* __restrict__ x, int64_t R,
 __global__ void partial_combine_kernel(const float* __restrict__ partials,
                                        float* __restrict__ out, int width,
                                        int nblk) {
+  partials += static_cast<int64_t>(blockIdx.y) * nblk * width;
+  out += static_cast<int64_t>(blockIdx.y) * width;
   const int lane = threadIdx.x % kWave;
   const int i = blockIdx.x * (blockDim.x / kWave) + threadIdx.x / kWave;
   if (i >= width) return;
@@ -256,6 +261,15 @@ __global__ void bn_normalize_kernel(const bf16* __restrict__ x,
                                     const float* __restrict__ b,
                                     bf16* __restrict__ y, int64_t R, int C) {
   const int64_t n = R * C;
+  const int64_t goff = static_cast<int64_t>(blockIdx.z) * n;
+  const int gc = blockIdx.z * C;
+  x += goff;
+  if (RES) res += goff;
+  y += goff;
+  mean += gc;
+  invstd += gc;
+  w += gc;
+  b += gc;
   int64_t i = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) * 8;
   const int64_t stride = grid_stride() * 8;
   for (; i + 7 < n; i += stride) {
@@ -289,6 +303,16 @@ __global__ void bn_bwd_reduce_kernel(
     const float* __restrict__ invstd, int64_t R, int C,
     float* __restrict__ acc /*[2C]*/, bf16* __restrict__ gym_out) {
   extern __shared__ float lbin[];  // FIXED: [256][16]; else [2C]
+  {
+    const int64_t goff = static_cast<int64_t>(blockIdx.z) * R * C;
+    gy += goff;
+    x += goff;
+    if (y != nullptr) y += goff;
+    if (gym_out != nullptr) gym_out += goff;
+    mean += blockIdx.z * C;
+    invstd += blockIdx.z * C;
+    acc += static_cast<int64_t>(blockIdx.z) * gridDim.x * 2 * C;
+  }
   if constexpr (!FIXED_C) {
     for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) lbin[i] = 0.f;
     __syncthreads();
@@ -368,6 +392,14 @@ __global__ void bn_bwd_gx_kernel(const bf16* __restrict__ gym,
                                  bf16* __restrict__ gx, int64_t R, int C,
                                  bool training) {
   const int64_t n = R * C;
+  const int64_t goff = static_cast<int64_t>(blockIdx.z) * n;
+  gym += goff;
+  x += goff;
+  gx += goff;
+  mean += blockIdx.z * C;
+  invstd += blockIdx.z * C;
+  w += blockIdx.z * C;
+  acc += static_cast<int64_t>(blockIdx.z) * 2 * C;
   const float invn = 1.f / R;
   int64_t i = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) * 8;
   const int64_t stride = grid_stride() * 8;
@@ -432,7 +464,7 @@ void launch_batchnorm_fwd(const void* x_, const void* res_, const float* w,
                           const float* b, float* running_mean,
                           float* running_var, void* y_, float* mean,
                           float* invstd, float* sums, int64_t R, int C,
-                          float momentum, float eps, bool training,
+                          int G, float momentum, float eps, bool training,
                           bool relu, hipStream_t stream) {
   const bf16* x = static_cast<const bf16*>(x_);
   const bf16* res = static_cast<const bf16*>(res_);
@@ -443,34 +475,38 @@ void launch_batchnorm_fwd(const void* x_, const void* res_, const float* w,
   const int grid = elemwise_grid(R * C / 8);
   const int rgrid = norms_reduce_grid(R * C / 8);
   if (training) {
-    // sums layout: [rgrid][2C] partials, combined into the last 2C floats
+    // sums layout: [G][rgrid][2C] partials + [G][2C] combined at the end
     float* partials = sums;
-    float* combined = sums + static_cast<int64_t>(rgrid) * 2 * C;
+    float* combined =
+        sums + static_cast<int64_t>(G) * rgrid * 2 * C;
     if (fixed)
-      bn_reduce_kernel<true><<<rgrid, 256, lds, stream>>>(x, R, C, partials);
+      bn_reduce_kernel<true><<<dim3(rgrid, 1, G), 256, lds, stream>>>(
+          x, R, C, partials);
     else
-      bn_reduce_kernel<false><<<rgrid, 256, lds, stream>>>(x, R, C,
-                                                           partials);
-    partial_combine_kernel<<<(2 * C + 3) / 4, 256, 0, stream>>>(
+      bn_reduce_kernel<false><<<dim3(rgrid, 1, G), 256, lds, stream>>>(
+          x, R, C, partials);
+    partial_combine_kernel<<<dim3((2 * C + 3) / 4, G), 256, 0, stream>>>(
         partials, combined, 2 * C, rgrid);
     sums = combined;
   }
-  bn_stats_kernel<<<(C + 255) / 256, 256, 0, stream>>>(
-      sums, C, R, mean, invstd, running_mean, running_var, momentum, eps,
-      training);
+  // stats are elementwise over G*C channel instances
+  bn_stats_kernel<<<(G * C + 255) / 256, 256, 0, stream>>>(
+      sums, G * C, R, mean, invstd, running_mean, running_var, momentum,
+      eps, training);
+  dim3 ngrid(grid, 1, G);
   if (res != nullptr) {
     if (relu)
-      bn_normalize_kernel<true, true><<<grid, 256, 0, stream>>>(
+      bn_normalize_kernel<true, true><<<ngrid, 256, 0, stream>>>(
           x, res, mean, invstd, w, b, y, R, C);
     else
-      bn_normalize_kernel<true, false><<<grid, 256, 0, stream>>>(
+      bn_normalize_kernel<true, false><<<ngrid, 256, 0, stream>>>(
           x, res, mean, invstd, w, b, y, R, C);
   } else {
     if (relu)
-      bn_normalize_kernel<false, true><<<grid, 256, 0, stream>>>(
+      bn_normalize_kernel<false, true><<<ngrid, 256, 0, stream>>>(
           x, nullptr, mean, invstd, w, b, y, R, C);
     else
-      bn_normalize_kernel<false, false><<<grid, 256, 0, stream>>>(
+      bn_normalize_kernel<false, false><<<ngrid, 256, 0, stream>>>(
           x, nullptr, mean, invstd, w, b, y, R, C);
   }
 }
@@ -478,8 +514,8 @@ void launch_batchnorm_fwd(const void* x_, const void* res_, const float* w,
 void launch_batchnorm_bwd(const void* gy_, const void* y_, const void* x_,
                           const float* w, const float* mean,
                           const float* invstd, void* gx_, void* gym_,
-                          float* acc, int64_t R, int C, bool training,
-                          bool relu, hipStream_t stream) {
+                          float* acc, int64_t R, int C, int G,
+                          bool training, bool relu, hipStream_t stream) {
   const bf16* gy = static_cast<const bf16*>(gy_);
   const bf16* y = static_cast<const bf16*>(y_);
   const bf16* x = static_cast<const bf16*>(x_);
@@ -491,28 +527,29 @@ void launch_batchnorm_bwd(const void* gy_, const void* y_, const void* x_,
   const int grid = elemwise_grid(R * C / 8);
   const int rgrid = norms_reduce_grid(R * C / 8);
   float* partials = acc;
-  float* combined = acc + static_cast<int64_t>(rgrid) * 2 * C;
+  float* combined = acc + static_cast<int64_t>(G) * rgrid * 2 * C;
   acc = partials;
+  dim3 rg(rgrid, 1, G);
   if (relu) {
     if (fixed)
-      bn_bwd_reduce_kernel<true, true><<<rgrid, 256, lds, stream>>>(
+      bn_bwd_reduce_kernel<true, true><<<rg, 256, lds, stream>>>(
           gy, y, x, mean, invstd, R, C, acc, gym);
     else
-      bn_bwd_reduce_kernel<false, true><<<rgrid, 256, lds, stream>>>(
+      bn_bwd_reduce_kernel<false, true><<<rg, 256, lds, stream>>>(
           gy, y, x, mean, invstd, R, C, acc, gym);
   } else {
     if (fixed)
-      bn_bwd_reduce_kernel<true, false><<<rgrid, 256, lds, stream>>>(
+      bn_bwd_reduce_kernel<true, false><<<rg, 256, lds, stream>>>(
           gy, y, x, mean, invstd, R, C, acc, nullptr);
     else
-      bn_bwd_reduce_kernel<false, false><<<rgrid, 256, lds, stream>>>(
+      bn_bwd_reduce_kernel<false, false><<<rg, 256, lds, stream>>>(
           gy, y, x, mean, invstd, R, C, acc, nullptr);
   }
-  partial_combine_kernel<<<(2 * C + 3) / 4, 256, 0, stream>>>(
+  partial_combine_kernel<<<dim3((2 * C + 3) / 4, G), 256, 0, stream>>>(
       partials, combined, 2 * C, rgrid);
   const bf16* gym_in = relu ? gym : gy;
-  bn_bwd_gx_kernel<<<grid, 256, 0, stream>>>(gym_in, x, mean, invstd, w,
-                                             combined, gx, R, C, training);
+  bn_bwd_gx_kernel<<<dim3(grid, 1, G), 256, 0, stream>>>(
+      gym_in, x, mean, invstd, w, combined, gx, R, C, training);
 }
 
 }  // namespace colearn

@@ -29,16 +29,19 @@ class _BatchNormFn(torch.autograd.Function):
 
     @staticmethod
     def forward(x, weight, bias, running_mean, running_var, residual,
-                training: bool, momentum: float, eps: float, relu: bool):
+                training: bool, momentum: float, eps: float, relu: bool,
+                groups: int = 1):
         if _native(x):
             m = require_ext()
             y, save_mean, save_invstd = m.batchnorm_fwd(
                 x, weight, bias, running_mean, running_var,
                 residual if residual is not None
                 else torch.empty(0, device=x.device, dtype=x.dtype),
-                training, momentum, eps, relu,
+                training, momentum, eps, relu, groups,
             )
             return y, save_mean, save_invstd
+        assert groups == 1, "stacked BN has no CPU fast path here"
+
         if training:
             dims = (0, 2, 3)
             xf = x.float()
@@ -72,6 +75,7 @@ class _BatchNormFn(torch.autograd.Function):
         ctx.save_for_backward(x, weight, save_mean, save_invstd, y)
         ctx.training = inputs[6]
         ctx.relu = inputs[9]
+        ctx.groups = inputs[10] if len(inputs) > 10 else 1
         ctx.has_res = inputs[5] is not None
         ctx.mark_non_differentiable(output[1], output[2])
 
@@ -80,7 +84,8 @@ class _BatchNormFn(torch.autograd.Function):
         x, weight, mean, invstd, y = ctx.saved_tensors
         if _native(x):
             gx, gw, gb, gres = require_ext().batchnorm_bwd(
-                gy, y, x, weight, mean, invstd, ctx.training, ctx.relu
+                gy, y, x, weight, mean, invstd, ctx.training, ctx.relu,
+                ctx.groups,
             )
         else:
             gyf = gy.float()
@@ -101,7 +106,8 @@ class _BatchNormFn(torch.autograd.Function):
                 gx = (weight * invstd)[None, :, None, None] * gyf
             gx = gx.to(x.dtype)
         return (gx, gw, gb, None, None,
-                gres if ctx.has_res else None, None, None, None, None)
+                gres if ctx.has_res else None, None, None, None, None,
+                None)
 
 
 class ColearnBatchNorm2d(nn.Module):
@@ -127,7 +133,7 @@ class ColearnBatchNorm2d(nn.Module):
                 self.num_batches_tracked += 1
         y, _, _ = _BatchNormFn.apply(
             x, self.weight, self.bias, self.running_mean, self.running_var,
-            residual, self.training, self.momentum, self.eps, self.relu,
+            residual, self.training, self.momentum, self.eps, self.relu, 1,
         )
         return y
 

@@ -142,6 +142,10 @@ class Aggregator:
         if self.world > 1:
             dist.broadcast(flat, src=src)
 
+    def allreduce_tensor(self, t: torch.Tensor) -> None:
+        if self.world > 1 and t.numel():
+            dist.all_reduce(t, op=dist.ReduceOp.SUM)
+
     def allreduce_scalar(self, value: float, op=None) -> float:
         if self.world == 1:
             return value

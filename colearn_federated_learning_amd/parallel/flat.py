@@ -89,3 +89,38 @@ class FlatParams:
             )
             offset += n
         return flat
+
+
+class FlatBuffers:
+    """Flat view over a model's float buffers (BN running stats).
+
+    FedAvg aggregates the WHOLE model state, buffers included — the
+    coordinator averages these alongside the weight deltas so evaluation
+    statistics don't depend on client visit order."""
+
+    def __init__(self, model: nn.Module) -> None:
+        self.items = [
+            (n, b) for n, b in model.named_buffers()
+            if b.dtype.is_floating_point
+        ]
+        self.numels = [b.numel() for _, b in self.items]
+        self.numel = sum(self.numels)
+
+    @torch.no_grad()
+    def gather(self, device=None) -> torch.Tensor:
+        if not self.items:
+            return torch.empty(0)
+        dev = device or self.items[0][1].device
+        out = torch.empty(self.numel, dtype=torch.float32, device=dev)
+        off = 0
+        for (_, b), n in zip(self.items, self.numels):
+            out[off : off + n] = b.reshape(-1).float()
+            off += n
+        return out
+
+    @torch.no_grad()
+    def load(self, flat: torch.Tensor) -> None:
+        off = 0
+        for (_, b), n in zip(self.items, self.numels):
+            b.copy_(flat[off : off + n].view(b.shape).to(b.dtype))
+            off += n
