@@ -9,11 +9,13 @@ def _cfg(**over):
     # lr small enough that training is stable: the stacked and sequential
     # paths have different fp reduction orders, and a diverging run
     # amplifies that noise chaotically.
-    return driver_config(2).override(
+    base = dict(
         rounds=2, num_clients=4, data__samples_per_client=32,
         train__batch_size=8, train__dtype="float32", num_evaluators=1,
-        train__lr=0.01, **over,
+        train__lr=0.01,
     )
+    base.update(over)
+    return driver_config(2).override(**base)
 
 
 def test_stacked_matches_sequential_resnet():
