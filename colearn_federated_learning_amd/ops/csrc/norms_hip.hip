@@ -229,26 +229,30 @@ __global__ void partial_combine_kernel(const float* __restrict__ partials,
   if (lane == 0) out[i] = acc;
 }
 
-// Pass 2: finalize stats + update running stats (tiny launch over C).
+// Pass 2: finalize stats + update running stats over all G*C channel
+// instances.  The combined sums are laid out [G][2C]: sum at g*2C + c,
+// sumsq at g*2C + C + c.
 __global__ void bn_stats_kernel(const float* __restrict__ sums, int C,
-                                int64_t R, float* __restrict__ mean,
+                                int G, int64_t R, float* __restrict__ mean,
                                 float* __restrict__ invstd,
                                 float* __restrict__ running_mean,
                                 float* __restrict__ running_var,
                                 float momentum, float eps, bool training) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= G * C) return;
   if (training) {
-    const float m = sums[c] / R;
-    const float var = fmaxf(sums[C + c] / R - m * m, 0.f);
-    mean[c] = m;
-    invstd[c] = rsqrtf(var + eps);
+    const int g = i / C, c = i % C;
+    const float m = sums[static_cast<int64_t>(g) * 2 * C + c] / R;
+    const float var = fmaxf(
+        sums[static_cast<int64_t>(g) * 2 * C + C + c] / R - m * m, 0.f);
+    mean[i] = m;
+    invstd[i] = rsqrtf(var + eps);
     const float unbiased = var * R / fmaxf(static_cast<float>(R - 1), 1.f);
-    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
-    running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+    running_mean[i] = (1.f - momentum) * running_mean[i] + momentum * m;
+    running_var[i] = (1.f - momentum) * running_var[i] + momentum * unbiased;
   } else {
-    mean[c] = running_mean[c];
-    invstd[c] = rsqrtf(running_var[c] + eps);
+    mean[i] = running_mean[i];
+    invstd[i] = rsqrtf(running_var[i] + eps);
   }
 }
 
@@ -492,7 +496,7 @@ void launch_batchnorm_fwd(const void* x_, const void* res_, const float* w,
   }
   // stats are elementwise over G*C channel instances
  hipLaunchKernelGGL(( bn_stats_kernel), dim3((G * C + 255) / 256), dim3(256), 0, stream, 
-      sums, G * C, R, mean, invstd, running_mean, running_var, momentum,
+      sums, C, G, R, mean, invstd, running_mean, running_var, momentum,
       eps, training);
   dim3 ngrid(grid, 1, G);
   if (res != nullptr) {
