@@ -269,7 +269,7 @@ def test_conv_bwd(n, c, hw, co, ks, stride, pad):
     )
     m = _ext()
     gx = m.conv2d_dgrad(gy.to(DEV), w.to(DEV), hw, hw, stride, pad)
-    gw = m.conv2d_wgrad(gy.to(DEV), x.to(DEV), ks, ks, stride, pad)
+    gw = m.conv2d_wgrad(gy.to(DEV), x.to(DEV), ks, ks, stride, pad, 1)
     torch.testing.assert_close(gx.float().cpu(), gx_ref, rtol=3e-2, atol=0.1)
     npq = n * po * po
     tol = 0.15 * max(1.0, (npq / 1024) ** 0.5)
@@ -326,7 +326,7 @@ def test_graphed_fit_matches_eager():
 
     base = driver_config(2).override(
         rounds=1, num_clients=2, data__samples_per_client=128,
-        train__batch_size=32,
+        train__batch_size=32, train__use_stacked=False,
     )
     dev = torch.device("cuda", 0)
     s_graph = FederatedServer(base, device=dev)
@@ -424,3 +424,26 @@ def test_stacked_gpu_deterministic():
         s.run_round(0)
         outs.append(s.global_flat.clone())
     assert torch.equal(outs[0], outs[1])
+
+
+def test_stacked_graphed_matches_eager():
+    """Stacked engine with hipGraph replay == stacked eager, bitwise."""
+    from colearn_federated_learning_amd.config import driver_config
+    from colearn_federated_learning_amd.federated.coordinator import (
+        FederatedServer,
+    )
+
+    base = dict(rounds=1, num_clients=4, data__samples_per_client=64,
+                train__batch_size=32)
+    dev = torch.device("cuda", 0)
+    s_g = FederatedServer(driver_config(2).override(**base), device=dev)
+    assert s_g.stacked is not None and s_g.stacked.runner is not None
+    s_e = FederatedServer(
+        driver_config(2).override(train__use_graphs=False, **base),
+        device=dev,
+    )
+    assert s_e.stacked is not None and s_e.stacked.runner is None
+    s_g.run_round(0)
+    s_e.run_round(0)
+    torch.testing.assert_close(s_g.global_flat, s_e.global_flat,
+                               rtol=1e-6, atol=1e-7)
