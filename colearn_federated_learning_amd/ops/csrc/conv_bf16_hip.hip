@@ -118,29 +118,36 @@ DEV_INLINE void conv_dgrad_gather(const short* __restrict__ gy,
 
 // ---- the shared implicit-GEMM NT kernel ----------------------------------
 // GATHER: 0 = conv fwd (gather x), 1 = dgrad stride1, 2 = dgrad strided
-template <int GATHER>
+// TN: output-channel tile width — 64 for the CO=64 layers (a 128-wide
+// tile would waste half its MFMAs there), 128 otherwise.
+template <int GATHER, int TN>
 __global__ __launch_bounds__(kBlockThreads) void conv_nt_kernel(
     const short* __restrict__ Asrc, const short* __restrict__ Bw,
     bf16* __restrict__ C, ConvShape cs, int64_t Mc /* rows per client */,
     int64_t Nd, int64_t Kd, int64_t a_stride /* A elems per client */,
     int nbm, int nbn) {
-  __shared__ short lds[2][2][BM * BK];
+  __shared__ short lds_a[2][BM * BK];
+  __shared__ short lds_b[2][TN * BK];
   const int bid = xcd_swizzle(blockIdx.x, nbm * nbn);
   const int64_t bm_local = static_cast<int64_t>(bid / nbn) * BM;
-  const int64_t bn = static_cast<int64_t>(bid % nbn) * BN;
+  const int64_t bn = static_cast<int64_t>(bid % nbn) * TN;
   const int g = blockIdx.z;  // stacked client index
   const short* A = Asrc + static_cast<int64_t>(g) * a_stride;
   const short* Bg = Bw + static_cast<int64_t>(g) * Nd * Kd;
   bf16* Cg = C + static_cast<int64_t>(g) * Mc * Nd;
 
   const int wid = threadIdx.x / kWave;
-  const int wm = (wid >> 1) * 64, wn = (wid & 1) * 64;
+  // TN=128: 2x2 waves of 64x64; TN=64: 4x1 waves of 32x64
+  constexpr int MF = TN == 128 ? 4 : 2;  // 16-row fragments per wave (M)
+  constexpr int NF = 4;                  // 16-col fragments per wave (N)
+  const int wm = TN == 128 ? (wid >> 1) * 64 : wid * 32;
+  const int wn = TN == 128 ? (wid & 1) * 64 : 0;
 
-  f32x4_t acc[4][4];
+  f32x4_t acc[MF][NF];
 #pragma unroll
-  for (int m = 0; m < 4; ++m)
+  for (int m = 0; m < MF; ++m)
 #pragma unroll
-    for (int n = 0; n < 4; ++n) acc[m][n] = {0.f, 0.f, 0.f, 0.f};
+    for (int n = 0; n < NF; ++n) acc[m][n] = {0.f, 0.f, 0.f, 0.f};
 
   const int row = threadIdx.x >> 1;
   const int half = threadIdx.x & 1;
@@ -166,7 +173,7 @@ __global__ __launch_bounds__(kBlockThreads) void conv_nt_kernel(
     }
     b0 = b1 = mfma_frag_zero();
     const int64_t bn_row = bn + row;
-    if (bn_row < Nd) {
+    if (row < TN && bn_row < Nd) {
       const short* p = Bg + bn_row * Kd + k;
       if (k + 16 <= Kd) {
         b0 = *reinterpret_cast<const short8v_t*>(p);
@@ -184,14 +191,16 @@ __global__ __launch_bounds__(kBlockThreads) void conv_nt_kernel(
     }
   };
   auto write_tile = [&](int buf) {
-    *reinterpret_cast<short8v_t*>(reinterpret_cast<char*>(lds[buf][0]) +
+    *reinterpret_cast<short8v_t*>(reinterpret_cast<char*>(lds_a[buf]) +
                                   lds_slot_byte(row, half * 2)) = a0;
-    *reinterpret_cast<short8v_t*>(reinterpret_cast<char*>(lds[buf][0]) +
+    *reinterpret_cast<short8v_t*>(reinterpret_cast<char*>(lds_a[buf]) +
                                   lds_slot_byte(row, half * 2 + 1)) = a1;
-    *reinterpret_cast<short8v_t*>(reinterpret_cast<char*>(lds[buf][1]) +
-                                  lds_slot_byte(row, half * 2)) = b0;
-    *reinterpret_cast<short8v_t*>(reinterpret_cast<char*>(lds[buf][1]) +
-                                  lds_slot_byte(row, half * 2 + 1)) = b1;
+    if (row < TN) {
+      *reinterpret_cast<short8v_t*>(reinterpret_cast<char*>(lds_b[buf]) +
+                                    lds_slot_byte(row, half * 2)) = b0;
+      *reinterpret_cast<short8v_t*>(reinterpret_cast<char*>(lds_b[buf]) +
+                                    lds_slot_byte(row, half * 2 + 1)) = b1;
+    }
   };
 
   load_tile(0);
@@ -200,35 +209,51 @@ __global__ __launch_bounds__(kBlockThreads) void conv_nt_kernel(
   int cur = 0;
   for (int t = 0; t < ntiles; ++t) {
     if (t + 1 < ntiles) load_tile(t + 1);
-    short8v_t af[4], bfr[4];
+    short8v_t af[MF], bfr[NF];
 #pragma unroll
-    for (int m = 0; m < 4; ++m)
-      af[m] = frag_from_lds(lds[cur][0], wm + m * 16);
+    for (int m = 0; m < MF; ++m)
+      af[m] = frag_from_lds(lds_a[cur], wm + m * 16);
 #pragma unroll
-    for (int n = 0; n < 4; ++n)
-      bfr[n] = frag_from_lds(lds[cur][1], wn + n * 16);
+    for (int n = 0; n < NF; ++n)
+      bfr[n] = frag_from_lds(lds_b[cur], wn + n * 16);
 #pragma unroll
-    for (int m = 0; m < 4; ++m)
+    for (int m = 0; m < MF; ++m)
 #pragma unroll
-      for (int n = 0; n < 4; ++n)
+      for (int n = 0; n < NF; ++n)
         acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af[m], bfr[n], acc[m][n], 0, 0, 0);
     if (t + 1 < ntiles) write_tile(cur ^ 1);
     __syncthreads();
     cur ^= 1;
   }
-  store_tile_bf16(reinterpret_cast<const float*>(acc), Cg, Nd, Mc, Nd,
-                  bm_local, bn, wm, wn);
+  // bounds-checked epilogue (C/D map: col=lane&15, row=(lane>>4)*4+reg)
+  const int lane = threadIdx.x % kWave;
+  const int col_in_frag = lane & 15;
+  const int row_grp = lane >> 4;
+#pragma unroll
+  for (int m = 0; m < MF; ++m)
+#pragma unroll
+    for (int n = 0; n < NF; ++n)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int64_t r = bm_local + wm + m * 16 + row_grp * 4 + j;
+        const int64_t c = bn + wn + n * 16 + col_in_frag;
+        if (r < Mc && c < Nd)
+          Cg[r * Nd + c] = f2b(acc[m][n][j]);
+      }
 }
 
 // ==== wgrad ===============================================================
-// gw[g, co, rsci]: 64x64 tile per block, BK=32 over npq; deterministic
-// split-K slabs per client.  grid = (tiles, splitk, G).
+// gw[g, co, rsci]: 128x128 tile per block (4 waves, 4x4 fragments each,
+// 128 FLOP per staged LDS element — 2x the 64x64 v1), BK=32 over npq;
+// deterministic split-K slabs per client.  grid = (tiles, splitk, G).
+// Both operands transposed-scatter into K-major LDS (b16 writes; the
+// tr_b16 hardware-transpose read is the planned next step).
 __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
     const short* __restrict__ gy, const short* __restrict__ x,
     float* __restrict__ gwf, ConvShape cs, int64_t NPQ, int64_t RSC,
     int nbm, int nbn, int splitk) {
-  constexpr int TM = 64, TN = 64;
+  constexpr int TM = 128, TN = 128;
   __shared__ short Gs[TM][BK + 2];
   __shared__ short Xs[TN][BK + 2];
   const int g = blockIdx.z;
@@ -245,58 +270,69 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
   const int64_t k_end = min(k_begin + chunk, NPQ);
 
   const int wid = threadIdx.x / kWave;
-  const int wm = (wid >> 1) * 32, wn = (wid & 1) * 32;
-  f32x4_t acc[2][2];
+  const int wm = (wid >> 1) * 64, wn = (wid & 1) * 64;
+  f32x4_t acc[4][4];
 #pragma unroll
-  for (int m = 0; m < 2; ++m)
+  for (int m = 0; m < 4; ++m)
 #pragma unroll
-    for (int n = 0; n < 2; ++n) acc[m][n] = {0.f, 0.f, 0.f, 0.f};
+    for (int n = 0; n < 4; ++n) acc[m][n] = {0.f, 0.f, 0.f, 0.f};
 
-  const int kk = threadIdx.x / 8;
-  const int e8 = (threadIdx.x % 8) * 8;
+  // staging: thread t covers npq row kk = t>>3, element block (t&7)*16
+  const int kk = threadIdx.x >> 3;         // 0..31
+  const int e16 = (threadIdx.x & 7) * 16;  // 16-wide column block
 
   for (int64_t k0 = k_begin; k0 < k_end; k0 += BK) {
     const int64_t npq = k0 + kk;
-    short8v_t gv = mfma_frag_zero();
-    short8v_t xv = mfma_frag_zero();
+    short8v_t gv0 = mfma_frag_zero(), gv1 = mfma_frag_zero();
+    short8v_t xv0 = mfma_frag_zero(), xv1 = mfma_frag_zero();
     if (npq < k_end) {
-      if (co0 + e8 < cs.K) {
-        const short* p = gyg + npq * cs.K + co0 + e8;
-        if ((cs.K & 7) == 0 && co0 + e8 + 8 <= cs.K)
-          gv = *reinterpret_cast<const short8v_t*>(p);
-        else
+      if (co0 + e16 < cs.K) {
+        const short* p = gyg + npq * cs.K + co0 + e16;
+        if ((cs.K & 7) == 0 && co0 + e16 + 16 <= cs.K) {
+          gv0 = *reinterpret_cast<const short8v_t*>(p);
+          gv1 = *reinterpret_cast<const short8v_t*>(p + 8);
+        } else {
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            if (co0 + e8 + j < cs.K) gv[j] = p[j];
+          for (int j = 0; j < 16; ++j)
+            if (co0 + e16 + j < cs.K) {
+              if (j < 8)
+                gv0[j] = p[j];
+              else
+                gv1[j - 8] = p[j];
+            }
+        }
       }
-      conv_fwd_gather(xg, cs, npq, rc0 + e8, xv);
+      conv_fwd_gather(xg, cs, npq, rc0 + e16, xv0);
+      conv_fwd_gather(xg, cs, npq, rc0 + e16 + 8, xv1);
     }
     __syncthreads();
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      Gs[e8 + j][kk] = gv[j];
-      Xs[e8 + j][kk] = xv[j];
+      Gs[e16 + j][kk] = gv0[j];
+      Gs[e16 + 8 + j][kk] = gv1[j];
+      Xs[e16 + j][kk] = xv0[j];
+      Xs[e16 + 8 + j][kk] = xv1[j];
     }
     __syncthreads();
     const int lane = threadIdx.x % kWave;
     const int fr = lane & 15, kg = (lane >> 4) * 8;
-    short8v_t af[2], bfr[2];
+    short8v_t af[4], bfr[4];
 #pragma unroll
-    for (int m = 0; m < 2; ++m) {
+    for (int m = 0; m < 4; ++m) {
       const short* p = &Gs[wm + m * 16 + fr][kg];
 #pragma unroll
       for (int j = 0; j < 8; ++j) af[m][j] = p[j];
     }
 #pragma unroll
-    for (int n = 0; n < 2; ++n) {
+    for (int n = 0; n < 4; ++n) {
       const short* p = &Xs[wn + n * 16 + fr][kg];
 #pragma unroll
       for (int j = 0; j < 8; ++j) bfr[n][j] = p[j];
     }
 #pragma unroll
-    for (int m = 0; m < 2; ++m)
+    for (int m = 0; m < 4; ++m)
 #pragma unroll
-      for (int n = 0; n < 2; ++n)
+      for (int n = 0; n < 4; ++n)
         acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af[m], bfr[n], acc[m][n], 0, 0, 0);
   }
@@ -304,9 +340,9 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
   const int lane = threadIdx.x % kWave;
   const int cif = lane & 15, rg = lane >> 4;
 #pragma unroll
-  for (int m = 0; m < 2; ++m)
+  for (int m = 0; m < 4; ++m)
 #pragma unroll
-    for (int n = 0; n < 2; ++n)
+    for (int n = 0; n < 4; ++n)
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
         const int co = co0 + wm + m * 16 + rg * 4 + j;
@@ -339,10 +375,19 @@ void launch_conv_fwd(const void* x, const void* w, void* y, ConvShape cs,
   const int64_t Kd = static_cast<int64_t>(cs.R) * cs.S * cs.C;
   const int64_t a_stride = static_cast<int64_t>(cs.N) * cs.H * cs.W * cs.C;
   const int nbm = static_cast<int>((Mc + BM - 1) / BM);
-  const int nbn = (cs.K + BN - 1) / BN;
- hipLaunchKernelGGL(( conv_nt_kernel<0>), dim3(dim3(nbm * nbn, 1, G)), dim3(kBlockThreads), 0, stream, 
-      static_cast<const short*>(x), static_cast<const short*>(w),
-      static_cast<bf16*>(y), cs, Mc, cs.K, Kd, a_stride, nbm, nbn);
+  if (cs.K <= 64) {
+    const int nbn = (cs.K + 63) / 64;
+   hipLaunchKernelGGL(( conv_nt_kernel<0, 64>), dim3(dim3(nbm * nbn, 1, G)), dim3(kBlockThreads), 0,
+                            stream, 
+        static_cast<const short*>(x), static_cast<const short*>(w),
+        static_cast<bf16*>(y), cs, Mc, cs.K, Kd, a_stride, nbm, nbn);
+  } else {
+    const int nbn = (cs.K + BN - 1) / BN;
+   hipLaunchKernelGGL(( conv_nt_kernel<0, 128>), dim3(dim3(nbm * nbn, 1, G)), dim3(kBlockThreads), 0,
+                             stream, 
+        static_cast<const short*>(x), static_cast<const short*>(w),
+        static_cast<bf16*>(y), cs, Mc, cs.K, Kd, a_stride, nbm, nbn);
+  }
 }
 
 void launch_conv_dgrad(const void* gy, const void* wt, void* gx,
@@ -351,22 +396,36 @@ void launch_conv_dgrad(const void* gy, const void* wt, void* gx,
   const int64_t Kd = static_cast<int64_t>(cs.R) * cs.S * cs.K;
   const int64_t a_stride = static_cast<int64_t>(cs.N) * cs.P * cs.Q * cs.K;
   const int nbm = static_cast<int>((Mc + BM - 1) / BM);
-  const int nbn = (cs.C + BN - 1) / BN;
-  if (cs.stride == 1)
-   hipLaunchKernelGGL(( conv_nt_kernel<1>), dim3(dim3(nbm * nbn, 1, G)), dim3(kBlockThreads), 0, stream, 
-        static_cast<const short*>(gy), static_cast<const short*>(wt),
-        static_cast<bf16*>(gx), cs, Mc, cs.C, Kd, a_stride, nbm, nbn);
-  else
-   hipLaunchKernelGGL(( conv_nt_kernel<2>), dim3(dim3(nbm * nbn, 1, G)), dim3(kBlockThreads), 0, stream, 
-        static_cast<const short*>(gy), static_cast<const short*>(wt),
-        static_cast<bf16*>(gx), cs, Mc, cs.C, Kd, a_stride, nbm, nbn);
+  if (cs.C <= 64) {
+    const int nbn = (cs.C + 63) / 64;
+    dim3 grid(nbm * nbn, 1, G);
+    if (cs.stride == 1)
+     hipLaunchKernelGGL(( conv_nt_kernel<1, 64>), dim3(grid), dim3(kBlockThreads), 0, stream, 
+          static_cast<const short*>(gy), static_cast<const short*>(wt),
+          static_cast<bf16*>(gx), cs, Mc, cs.C, Kd, a_stride, nbm, nbn);
+    else
+     hipLaunchKernelGGL(( conv_nt_kernel<2, 64>), dim3(grid), dim3(kBlockThreads), 0, stream, 
+          static_cast<const short*>(gy), static_cast<const short*>(wt),
+          static_cast<bf16*>(gx), cs, Mc, cs.C, Kd, a_stride, nbm, nbn);
+  } else {
+    const int nbn = (cs.C + BN - 1) / BN;
+    dim3 grid(nbm * nbn, 1, G);
+    if (cs.stride == 1)
+     hipLaunchKernelGGL(( conv_nt_kernel<1, 128>), dim3(grid), dim3(kBlockThreads), 0, stream, 
+          static_cast<const short*>(gy), static_cast<const short*>(wt),
+          static_cast<bf16*>(gx), cs, Mc, cs.C, Kd, a_stride, nbm, nbn);
+    else
+     hipLaunchKernelGGL(( conv_nt_kernel<2, 128>), dim3(grid), dim3(kBlockThreads), 0, stream, 
+          static_cast<const short*>(gy), static_cast<const short*>(wt),
+          static_cast<bf16*>(gx), cs, Mc, cs.C, Kd, a_stride, nbm, nbn);
+  }
 }
 
 int conv_wgrad_splitk(ConvShape cs, int G) {
   const int64_t NPQ = static_cast<int64_t>(cs.N) * cs.P * cs.Q;
   const int64_t RSC = static_cast<int64_t>(cs.R) * cs.S * cs.C;
-  const int nbm = (cs.K + 63) / 64;
-  const int nbn = static_cast<int>((RSC + 63) / 64);
+  const int nbm = (cs.K + 127) / 128;
+  const int nbn = static_cast<int>((RSC + 127) / 128);
   int64_t want = 512 / (static_cast<int64_t>(nbm) * nbn * G);
   if (want < 1) want = 1;
   int64_t cap = (NPQ + BK - 1) / BK;
@@ -377,8 +436,8 @@ void launch_conv_wgrad(const void* gy, const void* x, float* gwf, void* gw,
                        ConvShape cs, int G, hipStream_t stream) {
   const int64_t NPQ = static_cast<int64_t>(cs.N) * cs.P * cs.Q;
   const int64_t RSC = static_cast<int64_t>(cs.R) * cs.S * cs.C;
-  const int nbm = (cs.K + 63) / 64;
-  const int nbn = static_cast<int>((RSC + 63) / 64);
+  const int nbm = (cs.K + 127) / 128;
+  const int nbn = static_cast<int>((RSC + 127) / 128);
   const int splitk = conv_wgrad_splitk(cs, G);
  hipLaunchKernelGGL(( conv_wgrad_kernel), dim3(dim3(nbm * nbn, splitk, G)), dim3(kBlockThreads), 0,
                       stream, static_cast<const short*>(gy),

@@ -277,24 +277,57 @@ __global__ void bn_normalize_kernel(const bf16* __restrict__ x,
   b += gc;
   int64_t i = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) * 8;
   const int64_t stride = grid_stride() * 8;
-  for (; i + 7 < n; i += stride) {
-    const short8 v = *reinterpret_cast<const short8*>(
-        reinterpret_cast<const short*>(x) + i);
-    short8 rv;
-    if constexpr (RES)
-      rv = *reinterpret_cast<const short8*>(
-          reinterpret_cast<const short*>(res) + i);
+  // stride % C == 0 for every ResNet channel count => per-thread channel
+  // block is loop-invariant: hoist the expensive 64-bit modulo + the
+  // 4 channel-parameter loads out of the sweep.
+  const bool fixed = stride % C == 0;
+  if (fixed) {
     const int c0 = static_cast<int>(i % C);
-    short8 o;
+    float mv[8], iv[8], wv[8], bv[8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      const int c = c0 + j;
-      float r = (bits2f(v.x[j]) - mean[c]) * invstd[c] * w[c] + b[c];
-      if constexpr (RES) r += bits2f(rv.x[j]);
-      if constexpr (RELU) r = fmaxf(r, 0.f);
-      o.x[j] = f2bits(r);
+      mv[j] = mean[c0 + j];
+      iv[j] = invstd[c0 + j];
+      wv[j] = w[c0 + j];
+      bv[j] = b[c0 + j];
     }
-    *reinterpret_cast<short8*>(reinterpret_cast<short*>(y) + i) = o;
+    for (; i + 7 < n; i += stride) {
+      const short8 v = *reinterpret_cast<const short8*>(
+          reinterpret_cast<const short*>(x) + i);
+      short8 rv;
+      if constexpr (RES)
+        rv = *reinterpret_cast<const short8*>(
+            reinterpret_cast<const short*>(res) + i);
+      short8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float r = (bits2f(v.x[j]) - mv[j]) * iv[j] * wv[j] + bv[j];
+        if constexpr (RES) r += bits2f(rv.x[j]);
+        if constexpr (RELU) r = fmaxf(r, 0.f);
+        o.x[j] = f2bits(r);
+      }
+      *reinterpret_cast<short8*>(reinterpret_cast<short*>(y) + i) = o;
+    }
+  } else {
+    for (; i + 7 < n; i += stride) {
+      const short8 v = *reinterpret_cast<const short8*>(
+          reinterpret_cast<const short*>(x) + i);
+      short8 rv;
+      if constexpr (RES)
+        rv = *reinterpret_cast<const short8*>(
+            reinterpret_cast<const short*>(res) + i);
+      const int c0 = static_cast<int>(i % C);
+      short8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int c = c0 + j;
+        float r = (bits2f(v.x[j]) - mean[c]) * invstd[c] * w[c] + b[c];
+        if constexpr (RES) r += bits2f(rv.x[j]);
+        if constexpr (RELU) r = fmaxf(r, 0.f);
+        o.x[j] = f2bits(r);
+      }
+      *reinterpret_cast<short8*>(reinterpret_cast<short*>(y) + i) = o;
+    }
   }
 }
 
@@ -408,6 +441,39 @@ __global__ void bn_bwd_gx_kernel(const bf16* __restrict__ gym,
   const float invn = 1.f / R;
   int64_t i = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) * 8;
   const int64_t stride = grid_stride() * 8;
+  if (stride % C == 0) {
+    // loop-invariant channels: precompute the per-channel coefficients
+    // gx = a[j]*gym + (bcoef[j] + d[j]*x)   (training folds the acc terms)
+    const int c0 = static_cast<int>(i % C);
+    float a[8], bco[8], d[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = c0 + j;
+      const float wi = w[c] * invstd[c];
+      a[j] = wi;
+      if (training) {
+        const float gwn = acc[C + c] * invn * invstd[c];
+        bco[j] = -wi * (acc[c] * invn - gwn * mean[c]);
+        d[j] = -wi * gwn;
+      } else {
+        bco[j] = 0.f;
+        d[j] = 0.f;
+      }
+    }
+    for (; i + 7 < n; i += stride) {
+      const short8 g = *reinterpret_cast<const short8*>(
+          reinterpret_cast<const short*>(gym) + i);
+      const short8 v = *reinterpret_cast<const short8*>(
+          reinterpret_cast<const short*>(x) + i);
+      short8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o.x[j] = f2bits(a[j] * bits2f(g.x[j]) + bco[j] +
+                        d[j] * bits2f(v.x[j]));
+      *reinterpret_cast<short8*>(reinterpret_cast<short*>(gx) + i) = o;
+    }
+    return;
+  }
   for (; i + 7 < n; i += stride) {
     const short8 g = *reinterpret_cast<const short8*>(
         reinterpret_cast<const short*>(gym) + i);
