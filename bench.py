@@ -81,10 +81,17 @@ def main() -> None:
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
+    metric_names = {
+        1: "samples/sec (whole node), logreg FedAvg 2 clients",
+        2: "samples/sec (whole node), ResNet-18 FedAvg 8 clients",
+        3: "samples/sec (whole node), ResNet-50 FedProx 8 clients",
+        4: "samples/sec (whole node), BERT-base MLM FedAvg 8 clients",
+        5: "samples/sec (whole node), DP-FedAvg+top-k ResNet-18 64 clients",
+    }
     if rank == 0:
         value = total_samples / elapsed
         result = {
-            "metric": "samples/sec (whole node), ResNet-18 FedAvg 8 clients",
+            "metric": metric_names[args.config],
             "value": value,
             "unit": "samples/sec",
             "n_gpus": args.gpus,
@@ -98,6 +105,7 @@ def main() -> None:
             "data": "synthetic",
             "config": {
                 "model": cfg.model.name,
+                "driver_config": args.config,
                 "global_batch": cfg.train.batch_size * cfg.num_clients,
                 "local_batch": cfg.train.batch_size,
                 "num_clients": cfg.num_clients,
