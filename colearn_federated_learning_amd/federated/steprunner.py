@@ -71,7 +71,10 @@ class GraphedStepRunner:
             return loss
 
         model.train()
-        self.static_x.normal_()
+        if self.static_x.dtype.is_floating_point:
+            self.static_x.normal_()
+        else:
+            self.static_x.random_(0, 100)  # token ids (warmup only)
         if y_dtype == torch.long:
             self.static_y.random_(0, 2)
         # warmup on a side stream (required before capture)

@@ -57,3 +57,17 @@ def native_enabled() -> bool:
     """Native kernels are used for GPU tensors unless explicitly disabled
     (COLEARN_DISABLE_NATIVE=1 is for A/B benchmarking only)."""
     return os.environ.get("COLEARN_DISABLE_NATIVE", "0") != "1"
+
+
+def is_vmapped(*tensors) -> bool:
+    """functorch-batched tensors (the DP per-sample path) cannot enter the
+    native kernels — the composite fallback implementations are the vmap
+    rules."""
+    import torch
+
+    try:
+        return any(
+            torch._C._functorch.is_batchedtensor(t) for t in tensors
+        )
+    except Exception:
+        return False

@@ -16,13 +16,12 @@ from typing import Optional
 import torch
 import torch.nn.functional as F
 
-from . import native_enabled, require_ext
+from . import is_vmapped, native_enabled, require_ext
 
 
 def _use_native(*ts: torch.Tensor) -> bool:
-    return ts[0].is_cuda and native_enabled() and all(
-        t.dtype == torch.bfloat16 for t in ts
-    )
+    return (ts[0].is_cuda and native_enabled() and not is_vmapped(*ts)
+            and all(t.dtype == torch.bfloat16 for t in ts))
 
 
 class _LinearFn(torch.autograd.Function):

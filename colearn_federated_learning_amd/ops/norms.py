@@ -14,11 +14,12 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
-from . import native_enabled, require_ext
+from . import is_vmapped, native_enabled, require_ext
 
 
 def _native(x: torch.Tensor) -> bool:
-    return x.is_cuda and native_enabled() and x.dtype == torch.bfloat16
+    return (x.is_cuda and native_enabled() and not is_vmapped(x)
+            and x.dtype == torch.bfloat16)
 
 
 class _BatchNormFn(torch.autograd.Function):
