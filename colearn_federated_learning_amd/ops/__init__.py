@@ -60,14 +60,16 @@ def native_enabled() -> bool:
 
 
 def is_vmapped(*tensors) -> bool:
-    """functorch-batched tensors (the DP per-sample path) cannot enter the
-    native kernels — the composite fallback implementations are the vmap
-    rules."""
+    """True while any functorch transform (vmap/grad — the DP per-sample
+    path) is active, or when a tensor is functorch-wrapped: native kernels
+    can't see through the wrappers, so the composite fallbacks serve as
+    the vmap rules."""
     import torch
 
     try:
-        return any(
-            torch._C._functorch.is_batchedtensor(t) for t in tensors
-        )
+        fc = torch._C._functorch
+        if fc.peek_interpreter_stack() is not None:
+            return True
+        return any(fc.is_batchedtensor(t) for t in tensors)
     except Exception:
         return False
