@@ -35,6 +35,8 @@ def main() -> None:
                     help="BASELINE.json driver config index (default: "
                          "flagship ResNet-18 FedAvg)")
     ap.add_argument("--samples-per-client", type=int, default=2048)
+    ap.add_argument("--set", action="append", default=[],
+                    help="config overrides, dotted key=value")
     args = ap.parse_args()
 
     from colearn_federated_learning_amd.config import driver_config
@@ -42,6 +44,8 @@ def main() -> None:
         FederatedServer,
     )
     from colearn_federated_learning_amd.utils.dist import init_distributed
+
+    from colearn_federated_learning_amd.config import parse_overrides
 
     cfg = driver_config(args.config)
     on_gpu = torch.cuda.is_available()
@@ -53,6 +57,8 @@ def main() -> None:
         checkpoint_dir=None,
         metrics_path=None,
     )
+    if args.set:
+        cfg = cfg.override(**parse_overrides(args.set))
     rank, world, device = init_distributed(cfg.agg.backend)
     server = FederatedServer(cfg, rank=rank, world=world, device=device)
 
