@@ -10,9 +10,14 @@ host work collapses to two buffer copies + one graph launch.
 
 State contract: the graph reads/writes only fixed-storage tensors — the
 flat parameter/grad buffers (parallel.flat), optimizer state, BN running
-stats, and the static input buffers.  Warmup + capture mutate training
-state, so the runner snapshots and restores it; construction is therefore
-side-effect-free.  RNG-consuming ops (dropout) use PyTorch's graph-safe
+stats, and the static input buffers.  Warmup + capture MUTATE that state;
+the engine reloads every piece of it from the global model at the start
+of each round (load_flat / buffer broadcast / optimizer.reset), so
+construction needs no restore.  (A restore was tried and removed: ANY
+default-stream copy_ issued between hipGraph capture and the first replay
+intermittently faults the replay on ROCm 7.2 when the graph contains
+sort/embedding kernels — the BERT path.  Keeping the post-capture window
+clean avoids it.)  RNG-consuming ops (dropout) use PyTorch's graph-safe
 philox machinery.
 """
 
@@ -46,19 +51,6 @@ class GraphedStepRunner:
                                     device=device)
         self.static_y = torch.zeros(y_shape, dtype=y_dtype, device=device)
 
-        # ---- snapshot every tensor the warmup will mutate ----------------
-        snap_w = flat.flat_w.detach().clone()
-        snap_g = flat.flat_g.detach().clone()
-        snap_opt = {
-            k: v.detach().clone()
-            for k, v in vars(optimizer).items()
-            if isinstance(v, torch.Tensor)
-        }
-        snap_buf = {
-            k: v.detach().clone() for k, v in model.named_buffers()
-        }
-        snap_t = getattr(optimizer, "t", None)
-
         def one_step():
             flat.flat_g.zero_()
             loss_fn = getattr(model, "loss_terms", None)
@@ -90,18 +82,7 @@ class GraphedStepRunner:
         optimizer.host_prepare()
         with torch.cuda.graph(self.graph):
             self.static_loss = one_step()
-
-        # ---- restore ----------------------------------------------------
-        with torch.no_grad():
-            flat.flat_w.copy_(snap_w)
-            flat.flat_g.copy_(snap_g)
-            for k, v in snap_opt.items():
-                getattr(optimizer, k).copy_(v)
-            bufs = dict(model.named_buffers())
-            for k, v in snap_buf.items():
-                bufs[k].copy_(v)
-        if snap_t is not None:
-            optimizer.t = snap_t
+        optimizer.t = 0
         torch.cuda.synchronize(device)
 
     def matches(self, x: torch.Tensor) -> bool:
