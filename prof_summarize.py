@@ -1,0 +1,49 @@
+"""Summarize a rocprofv3 rocpd SQLite db into a small text report.
+
+Usage: python prof_summarize.py <dir-with-*_results.db> <out.txt>
+Run ON the GPU box so only the summary travels back through gpurun_out.
+"""
+
+import glob
+import sqlite3
+import sys
+
+
+def main() -> None:
+    dbs = glob.glob(sys.argv[1] + "/**/*_results.db", recursive=True)
+    out = sys.argv[2]
+    with open(out, "w") as f:
+        for path in dbs:
+            db = sqlite3.connect(path)
+            cur = db.cursor()
+            tables = [
+                r[0] for r in cur.execute(
+                    "SELECT name FROM sqlite_master WHERE type='table'"
+                )
+            ]
+            disp = [t for t in tables if t.startswith("rocpd_kernel_dispatch")]
+            if not disp:
+                continue
+            u = disp[0].replace("rocpd_kernel_dispatch_", "")
+            f.write(f"== {path} ==\n")
+            q = (
+                f"SELECT s.display_name, COUNT(*), SUM(d.end-d.start)/1e6,"
+                f" AVG(d.end-d.start)/1e3, MAX(d.grid_size_x)"
+                f" FROM rocpd_kernel_dispatch_{u} d"
+                f" JOIN rocpd_info_kernel_symbol_{u} s ON d.kernel_id=s.id"
+                f" GROUP BY s.display_name ORDER BY 3 DESC LIMIT 30"
+            )
+            for name, n, ms, us, gx in cur.execute(q):
+                f.write(f"{ms:9.2f} ms n={n:6d} avg={us:8.1f}us "
+                        f"maxgrid={gx:9d} {name[:90]}\n")
+            tot = cur.execute(
+                f"SELECT SUM(end-start)/1e6 FROM rocpd_kernel_dispatch_{u}"
+            ).fetchone()[0]
+            lo, hi = cur.execute(
+                f"SELECT MIN(start), MAX(end) FROM rocpd_kernel_dispatch_{u}"
+            ).fetchone()
+            f.write(f"TOTAL {tot:.1f} ms kernels; span {(hi-lo)/1e6:.1f} ms\n\n")
+
+
+if __name__ == "__main__":
+    main()
