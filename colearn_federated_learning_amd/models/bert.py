@@ -19,7 +19,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..ops.gemm import colearn_linear, colearn_matmul
+from ..ops.gemm import colearn_linear, colearn_matmul, colearn_matmul_nt
 from ..ops.norms import ColearnLayerNorm
 from .base import ColearnModel
 
@@ -63,7 +63,7 @@ class BertSelfAttention(nn.Module):
             qkv[:, :, i].transpose(1, 2).reshape(b * self.h, s, self.dh)
             for i in range(3)
         )
-        scores = colearn_matmul(q, k.transpose(-1, -2)) / math.sqrt(self.dh)
+        scores = colearn_matmul_nt(q, k) / math.sqrt(self.dh)
         probs = self.drop(F.softmax(scores.float(), dim=-1).to(v.dtype))
         ctx = colearn_matmul(probs, v)
         ctx = ctx.view(b, self.h, s, self.dh).transpose(1, 2).reshape(b, s, d)

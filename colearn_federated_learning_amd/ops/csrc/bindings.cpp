@@ -249,50 +249,54 @@ batchnorm_bwd(torch::Tensor gy, torch::Tensor y, torch::Tensor x,
 }
 
 // ---- GEMM family ----------------------------------------------------------
-torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b) {
+// all four operand layouts hit one kernel with transposed LDS staging —
+// no host transpose passes.
+torch::Tensor gemm2(torch::Tensor a, torch::Tensor b, bool a_km, bool b_km,
+                    int64_t M, int64_t N, int64_t K) {
   check_bf16(a, "a");
   check_bf16(b, "b");
-  TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(1),
-              "gemm_nt shape mismatch");
   const c10::hip::HIPGuardMasqueradingAsCUDA guard(a.device());
   auto ac = a.contiguous(), bc = b.contiguous();
-  const int64_t M = a.size(0), N = b.size(0), K = a.size(1);
-  auto c = torch::empty({M, N}, a.options());
-  launch_gemm_nt(ac.data_ptr(), bc.data_ptr(), c.data_ptr(), 1, M, N, K,
-                 stream());
+  const int64_t batch = a.dim() == 3 ? a.size(0) : 1;
+  auto c = batch > 1 ? torch::empty({batch, M, N}, a.options())
+                     : torch::empty({M, N}, a.options());
+  launch_gemm(ac.data_ptr(), bc.data_ptr(), c.data_ptr(), batch, M, N, K,
+              a_km, b_km, stream());
   HIP_CHECK_LAST();
   return c;
+}
+
+torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b) {
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(1),
+              "gemm_nt shape mismatch");
+  return gemm2(a, b, true, true, a.size(0), b.size(0), a.size(1));
 }
 
 torch::Tensor gemm_nn(torch::Tensor a, torch::Tensor b) {
-  return gemm_nt(a, b.transpose(0, 1).contiguous());
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(0),
+              "gemm_nn shape mismatch");
+  return gemm2(a, b, true, false, a.size(0), b.size(1), a.size(1));
 }
 
 torch::Tensor gemm_tn(torch::Tensor a, torch::Tensor b) {
-  return gemm_nt(a.transpose(0, 1).contiguous(),
-                 b.transpose(0, 1).contiguous());
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(0) == b.size(0),
+              "gemm_tn shape mismatch");
+  return gemm2(a, b, false, false, a.size(1), b.size(1), a.size(0));
 }
 
 torch::Tensor bmm_nt(torch::Tensor a, torch::Tensor b) {
-  check_bf16(a, "a");
   TORCH_CHECK(a.dim() == 3 && b.dim() == 3, "bmm expects 3-D");
-  const c10::hip::HIPGuardMasqueradingAsCUDA guard(a.device());
-  auto ac = a.contiguous(), bc = b.contiguous();
-  const int64_t B = a.size(0), M = a.size(1), N = b.size(1), K = a.size(2);
-  auto c = torch::empty({B, M, N}, a.options());
-  launch_gemm_nt(ac.data_ptr(), bc.data_ptr(), c.data_ptr(), B, M, N, K,
-                 stream());
-  HIP_CHECK_LAST();
-  return c;
+  return gemm2(a, b, true, true, a.size(1), b.size(1), a.size(2));
 }
 
 torch::Tensor bmm_nn(torch::Tensor a, torch::Tensor b) {
-  return bmm_nt(a, b.transpose(1, 2).contiguous());
+  TORCH_CHECK(a.dim() == 3 && b.dim() == 3, "bmm expects 3-D");
+  return gemm2(a, b, true, false, a.size(1), b.size(2), a.size(2));
 }
 
 torch::Tensor bmm_tn(torch::Tensor a, torch::Tensor b) {
-  return bmm_nt(a.transpose(1, 2).contiguous(),
-                b.transpose(1, 2).contiguous());
+  TORCH_CHECK(a.dim() == 3 && b.dim() == 3, "bmm expects 3-D");
+  return gemm2(a, b, false, false, a.size(2), b.size(2), a.size(1));
 }
 
 // ---- conv family ----------------------------------------------------------

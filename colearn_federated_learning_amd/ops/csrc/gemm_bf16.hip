@@ -1,83 +1,38 @@
-// bf16 MFMA GEMM, NT layout: C[M,N] = A[M,K] @ B[N,K]^T  (SURVEY.md N5).
+// bf16 MFMA GEMM family for gfx950: one kernel template
+// covering all four operand layouts, batched.
 //
-// The one hot GEMM kernel of the engine: linear forward consumes it
-// directly; NN/TN flavors and the attention BMMs route through it with a
-// cheap transpose pass (v1 — dedicated transposed-staging kernels are a
-// later optimization once rocprof shows the transpose passes matter).
+//   C[M,N] = A . B  with A in {[M,K] (K-major), [K,M]} and
+//                        B in {[N,K] (K-major), [K,N]}
 //
-// Structure: guide §5 "minimum 2-phase" schedule with register staging
-// (T14): issue next tile's global loads BEFORE the MFMA phase so HBM
-// latency hides under compute, write them to the other LDS buffer after,
-// ONE barrier per K-step, double-buffered LDS, XCD-aware block swizzle.
-// Batched via blockIdx.z (strides in elements).
+// Both operands stage into canonical K-major LDS tiles.  K-major sources
+// stage with 16-B vector loads; K-strided sources load contiguous along
+// their leading dim and scatter-transpose into LDS (b16 writes).  This is
+// what removes the host-side `.transpose().contiguous()` passes that v1
+// used for NN/TN/bmm flavors — measured at 63% of BERT's GEMM time.
+//
+// Structure: guide §5 "minimum 2-phase" schedule, register staging with
+// early load issue (T14), ONE barrier per K-step, double-buffered LDS,
+// XOR-swizzled tiles (tile_gemm.h), XCD-aware block swizzle.
 
 #include "api.h"
 #include "tile_gemm.h"
 
 namespace colearn {
 
-// each thread stages 32 B of A and 32 B of B per K-step:
-// thread t: row = t>>1, half = t&1 covers k [half*16, half*16+16).
-struct StageRegs {
-  short8v_t a0, a1, b0, b1;
-};
-
-DEV_INLINE void stage_load(const short* __restrict__ A,
-                           const short* __restrict__ B, int64_t M, int64_t N,
-                           int64_t K, int64_t lda, int64_t ldb, int64_t bm,
-                           int64_t bn, int64_t k0, StageRegs& r) {
-  const int row = threadIdx.x >> 1;
-  const int half = threadIdx.x & 1;
-  const int64_t k = k0 + half * 16;
-  const int64_t am = bm + row, bn_row = bn + row;
-  r.a0 = r.a1 = r.b0 = r.b1 = mfma_frag_zero();
-  if (am < M) {
-    const short* p = A + am * lda + k;
-    if (k + 16 <= K) {
-      r.a0 = *reinterpret_cast<const short8v_t*>(p);
-      r.a1 = *reinterpret_cast<const short8v_t*>(p + 8);
-    } else {
-#pragma unroll
-      for (int j = 0; j < 16; ++j)
-        if (k + j < K) {
-          if (j < 8) r.a0[j] = p[j]; else r.a1[j - 8] = p[j];
-        }
-    }
-  }
-  if (bn_row < N) {
-    const short* p = B + bn_row * ldb + k;
-    if (k + 16 <= K) {
-      r.b0 = *reinterpret_cast<const short8v_t*>(p);
-      r.b1 = *reinterpret_cast<const short8v_t*>(p + 8);
-    } else {
-#pragma unroll
-      for (int j = 0; j < 16; ++j)
-        if (k + j < K) {
-          if (j < 8) r.b0[j] = p[j]; else r.b1[j - 8] = p[j];
-        }
-    }
-  }
+// byte offset of element (row, k) in a swizzled K-major LDS tile
+DEV_INLINE int lds_elem_byte(int row, int k) {
+  return row * (BK * 2) + (((k >> 3) ^ ((row >> 2) & 3)) * 16) +
+         (k & 7) * 2;
 }
 
-DEV_INLINE void stage_write(short* As, short* Bs, const StageRegs& r) {
-  const int row = threadIdx.x >> 1;
-  const int half = threadIdx.x & 1;
-  *reinterpret_cast<short8v_t*>(reinterpret_cast<char*>(As) +
-                                lds_slot_byte(row, half * 2)) = r.a0;
-  *reinterpret_cast<short8v_t*>(reinterpret_cast<char*>(As) +
-                                lds_slot_byte(row, half * 2 + 1)) = r.a1;
-  *reinterpret_cast<short8v_t*>(reinterpret_cast<char*>(Bs) +
-                                lds_slot_byte(row, half * 2)) = r.b0;
-  *reinterpret_cast<short8v_t*>(reinterpret_cast<char*>(Bs) +
-                                lds_slot_byte(row, half * 2 + 1)) = r.b1;
-}
-
-__global__ __launch_bounds__(kBlockThreads) void gemm_nt_kernel(
+// A_KM: A is [M,K] (k contiguous); else A is [K,M] (m contiguous).
+// B_KM: B is [N,K] (k contiguous); else B is [K,N] (n contiguous).
+template <bool A_KM, bool B_KM>
+__global__ __launch_bounds__(kBlockThreads) void gemm_kernel(
     const short* __restrict__ A, const short* __restrict__ B,
-    bf16* __restrict__ C, int64_t M, int64_t N, int64_t K, int64_t lda,
-    int64_t ldb, int64_t ldc, int64_t sA, int64_t sB, int64_t sC,
-    int nbm, int nbn) {
-  __shared__ short lds[2][2][BM * BK];  // [dbuf][A|B][tile]
+    bf16* __restrict__ C, int64_t M, int64_t N, int64_t K, int64_t sA,
+    int64_t sB, int64_t sC, int nbm, int nbn) {
+  __shared__ short lds[2][2][BM * BK];
   const int bid = xcd_swizzle(blockIdx.x, nbm * nbn);
   const int64_t bm = static_cast<int64_t>(bid / nbn) * BM;
   const int64_t bn = static_cast<int64_t>(bid % nbn) * BN;
@@ -95,46 +50,152 @@ __global__ __launch_bounds__(kBlockThreads) void gemm_nt_kernel(
     for (int n = 0; n < 4; ++n) acc[m][n] = {0.f, 0.f, 0.f, 0.f};
 
   const int ntiles = static_cast<int>((K + BK - 1) / BK);
-  StageRegs regs;
-  stage_load(A, B, M, N, K, lda, ldb, bm, bn, 0, regs);
-  stage_write(lds[0][0], lds[0][1], regs);
-  __syncthreads();
+  // K-major staging decomposition: thread t = (row = t>>1, khalf = t&1)
+  const int row = threadIdx.x >> 1;
+  const int half = threadIdx.x & 1;
+  // K-strided staging decomposition: thread t = (kk = t>>3, col16 = t&7)
+  const int kk = threadIdx.x >> 3;
+  const int c16 = (threadIdx.x & 7) * 16;
 
+  short8v_t ra0, ra1, rb0, rb1;
+
+  auto load_km = [&](const short* __restrict__ P, int64_t rows,
+                     int64_t k, int64_t r, short8v_t& v0, short8v_t& v1) {
+    v0 = v1 = mfma_frag_zero();
+    if (r < rows) {
+      const short* p = P + r * K + k;
+      if (k + 16 <= K) {
+        v0 = *reinterpret_cast<const short8v_t*>(p);
+        v1 = *reinterpret_cast<const short8v_t*>(p + 8);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 16; ++j)
+          if (k + j < K) {
+            if (j < 8)
+              v0[j] = p[j];
+            else
+              v1[j - 8] = p[j];
+          }
+      }
+    }
+  };
+  auto load_ks = [&](const short* __restrict__ P, int64_t cols,
+                     int64_t k, int64_t c, short8v_t& v0, short8v_t& v1) {
+    // P is [K, cols]; load 16 contiguous columns at row k
+    v0 = v1 = mfma_frag_zero();
+    if (k < K && c < cols) {
+      const short* p = P + k * cols + c;
+      if (c + 16 <= cols) {
+        v0 = *reinterpret_cast<const short8v_t*>(p);
+        v1 = *reinterpret_cast<const short8v_t*>(p + 8);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 16; ++j)
+          if (c + j < cols) {
+            if (j < 8)
+              v0[j] = p[j];
+            else
+              v1[j - 8] = p[j];
+          }
+      }
+    }
+  };
+
+  auto load_tile = [&](int t) {
+    const int64_t k0 = static_cast<int64_t>(t) * BK;
+    if constexpr (A_KM)
+      load_km(A, M, k0 + half * 16, bm + row, ra0, ra1);
+    else
+      load_ks(A, M, k0 + kk, bm + c16, ra0, ra1);
+    if constexpr (B_KM)
+      load_km(B, N, k0 + half * 16, bn + row, rb0, rb1);
+    else
+      load_ks(B, N, k0 + kk, bn + c16, rb0, rb1);
+  };
+  auto write_tile = [&](int buf) {
+    char* As = reinterpret_cast<char*>(lds[buf][0]);
+    char* Bs = reinterpret_cast<char*>(lds[buf][1]);
+    if constexpr (A_KM) {
+      *reinterpret_cast<short8v_t*>(As + lds_slot_byte(row, half * 2)) = ra0;
+      *reinterpret_cast<short8v_t*>(
+          As + lds_slot_byte(row, half * 2 + 1)) = ra1;
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        *reinterpret_cast<short*>(As + lds_elem_byte(c16 + j, kk)) = ra0[j];
+        *reinterpret_cast<short*>(As + lds_elem_byte(c16 + 8 + j, kk)) =
+            ra1[j];
+      }
+    }
+    if constexpr (B_KM) {
+      *reinterpret_cast<short8v_t*>(Bs + lds_slot_byte(row, half * 2)) = rb0;
+      *reinterpret_cast<short8v_t*>(
+          Bs + lds_slot_byte(row, half * 2 + 1)) = rb1;
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        *reinterpret_cast<short*>(Bs + lds_elem_byte(c16 + j, kk)) = rb0[j];
+        *reinterpret_cast<short*>(Bs + lds_elem_byte(c16 + 8 + j, kk)) =
+            rb1[j];
+      }
+    }
+  };
+
+  load_tile(0);
+  write_tile(0);
+  __syncthreads();
   int cur = 0;
   for (int t = 0; t < ntiles; ++t) {
-    if (t + 1 < ntiles)  // issue next tile's loads before compute (T14)
-      stage_load(A, B, M, N, K, lda, ldb, bm, bn,
-                 static_cast<int64_t>(t + 1) * BK, regs);
-    const short* As = lds[cur][0];
-    const short* Bs = lds[cur][1];
-    short8v_t af[4], bf[4];
+    if (t + 1 < ntiles) load_tile(t + 1);
+    short8v_t af[4], bfr[4];
 #pragma unroll
-    for (int m = 0; m < 4; ++m) af[m] = frag_from_lds(As, wm + m * 16);
+    for (int m = 0; m < 4; ++m)
+      af[m] = frag_from_lds(lds[cur][0], wm + m * 16);
 #pragma unroll
-    for (int n = 0; n < 4; ++n) bf[n] = frag_from_lds(Bs, wn + n * 16);
+    for (int n = 0; n < 4; ++n)
+      bfr[n] = frag_from_lds(lds[cur][1], wn + n * 16);
 #pragma unroll
     for (int m = 0; m < 4; ++m)
 #pragma unroll
       for (int n = 0; n < 4; ++n)
         acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            af[m], bf[n], acc[m][n], 0, 0, 0);
-    if (t + 1 < ntiles) stage_write(lds[cur ^ 1][0], lds[cur ^ 1][1], regs);
+            af[m], bfr[n], acc[m][n], 0, 0, 0);
+    if (t + 1 < ntiles) write_tile(cur ^ 1);
     __syncthreads();
     cur ^= 1;
   }
-  store_tile_bf16(reinterpret_cast<const float*>(acc), C, ldc, M, N, bm, bn,
+  store_tile_bf16(reinterpret_cast<const float*>(acc), C, N, M, N, bm, bn,
                   wm, wn);
 }
 
-void launch_gemm_nt(const void* A, const void* B, void* C, int64_t batch,
-                    int64_t M, int64_t N, int64_t K, hipStream_t stream) {
+void launch_gemm(const void* A, const void* B, void* C, int64_t batch,
+                 int64_t M, int64_t N, int64_t K, bool a_km, bool b_km,
+                 hipStream_t stream) {
   const int nbm = static_cast<int>((M + BM - 1) / BM);
   const int nbn = static_cast<int>((N + BN - 1) / BN);
   dim3 grid(nbm * nbn, 1, static_cast<unsigned>(batch));
-  gemm_nt_kernel<<<grid, kBlockThreads, 0, stream>>>(
-      static_cast<const short*>(A), static_cast<const short*>(B),
-      static_cast<bf16*>(C), M, N, K, K, K, N, M * K, N * K, M * N, nbm,
-      nbn);
+  const short* a = static_cast<const short*>(A);
+  const short* b = static_cast<const short*>(B);
+  bf16* c = static_cast<bf16*>(C);
+  const int64_t sA = M * K, sB = N * K, sC = M * N;
+  if (a_km && b_km)
+    gemm_kernel<true, true><<<grid, kBlockThreads, 0, stream>>>(
+        a, b, c, M, N, K, sA, sB, sC, nbm, nbn);
+  else if (a_km && !b_km)
+    gemm_kernel<true, false><<<grid, kBlockThreads, 0, stream>>>(
+        a, b, c, M, N, K, sA, sB, sC, nbm, nbn);
+  else if (!a_km && b_km)
+    gemm_kernel<false, true><<<grid, kBlockThreads, 0, stream>>>(
+        a, b, c, M, N, K, sA, sB, sC, nbm, nbn);
+  else
+    gemm_kernel<false, false><<<grid, kBlockThreads, 0, stream>>>(
+        a, b, c, M, N, K, sA, sB, sC, nbm, nbn);
+}
+
+// legacy entry (conv path compatibility)
+void launch_gemm_nt(const void* A, const void* B, void* C, int64_t batch,
+                    int64_t M, int64_t N, int64_t K, hipStream_t stream) {
+  launch_gemm(A, B, C, batch, M, N, K, true, true, stream);
 }
 
 }  // namespace colearn

@@ -110,3 +110,41 @@ def colearn_matmul(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
         dt = torch.get_autocast_dtype("cuda")
         a, b = a.to(dt), b.to(dt)
     return _MatmulFn.apply(a, b)
+
+
+class _MatmulNTFn(torch.autograd.Function):
+    """a @ b^T for 3-D operands with both K-major (attention scores:
+    q [B,S,dh] x k [B,S,dh] -> [B,S,S]) — no transpose copies anywhere."""
+
+    generate_vmap_rule = True
+
+    @staticmethod
+    def forward(a: torch.Tensor, b: torch.Tensor):
+        if _use_native(a, b):
+            return require_ext().bmm_nt(a.contiguous(), b.contiguous())
+        return a @ b.transpose(-1, -2)
+
+    @staticmethod
+    def setup_context(ctx, inputs, output):
+        ctx.save_for_backward(*inputs)
+
+    @staticmethod
+    def backward(ctx, gy: torch.Tensor):
+        a, b = ctx.saved_tensors
+        gy = gy.contiguous()
+        if _use_native(gy, a, b):
+            m = require_ext()
+            ga = m.bmm_nn(gy, b.contiguous())   # gy @ b
+            gb = m.bmm_tn(gy, a.contiguous())   # gy^T @ a
+        else:
+            ga = gy @ b
+            gb = gy.transpose(-1, -2) @ a
+        return ga, gb
+
+
+def colearn_matmul_nt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """Batched ``a @ b^T`` (both operands contraction-major)."""
+    if torch.is_autocast_enabled("cuda") and a.is_cuda:
+        dt = torch.get_autocast_dtype("cuda")
+        a, b = a.to(dt), b.to(dt)
+    return _MatmulNTFn.apply(a, b)
