@@ -272,16 +272,22 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b) {
   return gemm2(a, b, true, true, a.size(0), b.size(0), a.size(1));
 }
 
+// measured (BERT backward shapes): for 2-D flavors an explicit transpose
+// pass + the vector-staged NT kernel beats the scatter-staged K-strided
+// kernel by 15-30%; for small-batch bmm the native kernels win 2x.
 torch::Tensor gemm_nn(torch::Tensor a, torch::Tensor b) {
   TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(0),
               "gemm_nn shape mismatch");
-  return gemm2(a, b, true, false, a.size(0), b.size(1), a.size(1));
+  auto bt = b.transpose(0, 1).contiguous();
+  return gemm2(a, bt, true, true, a.size(0), b.size(1), a.size(1));
 }
 
 torch::Tensor gemm_tn(torch::Tensor a, torch::Tensor b) {
   TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(0) == b.size(0),
               "gemm_tn shape mismatch");
-  return gemm2(a, b, false, false, a.size(1), b.size(1), a.size(0));
+  auto at = a.transpose(0, 1).contiguous();
+  auto bt = b.transpose(0, 1).contiguous();
+  return gemm2(at, bt, true, true, a.size(1), b.size(1), a.size(0));
 }
 
 torch::Tensor bmm_nt(torch::Tensor a, torch::Tensor b) {
