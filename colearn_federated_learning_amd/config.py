@@ -82,6 +82,11 @@ class AggregationConfig:
     # all-reduce.
     topk_ratio: float = 0.0
     transport: str = "dense"  # dense | sparse
+    # >0: split the delta into bucket_bytes chunks all-reduced on this many
+    # extra communicators/streams (xGMI link-overlap experiment).  0 = one
+    # RCCL all-reduce — RCCL already drives multiple channels/links for a
+    # single large collective, so this is the default.
+    comm_streams: int = 0
 
 
 @dataclass

@@ -35,17 +35,18 @@ class Aggregator:
         self,
         cfg: AggregationConfig,
         device: torch.device,
-        num_comm_streams: int = 4,
     ) -> None:
         self.cfg = cfg
         self.device = device
         self.world = dist.get_world_size() if dist.is_initialized() else 1
         self._groups: List[object] = []
         self._streams: List[torch.cuda.Stream] = []
-        if self.world > 1 and device.type == "cuda":
-            # Several communicators let bucket all-reduces overlap; RCCL
-            # then drives multiple xGMI links concurrently.
-            for _ in range(num_comm_streams):
+        if self.world > 1 and device.type == "cuda" and cfg.comm_streams > 0:
+            # Optional: several communicators let bucket all-reduces overlap
+            # so RCCL can drive distinct xGMI links concurrently.  Off by
+            # default — a single large RCCL all-reduce already uses multiple
+            # channels, and this path can't be validated on 1-GPU boxes.
+            for _ in range(cfg.comm_streams):
                 self._groups.append(dist.new_group(backend=dist.get_backend()))
                 self._streams.append(torch.cuda.Stream(device=device))
 
