@@ -43,6 +43,32 @@ def main() -> None:
                 f"SELECT MIN(start), MAX(end) FROM rocpd_kernel_dispatch_{u}"
             ).fetchone()
             f.write(f"TOTAL {tot:.1f} ms kernels; span {(hi-lo)/1e6:.1f} ms\n\n")
+            # PMC events, aggregated per kernel per counter
+            npmc = cur.execute(
+                f"SELECT COUNT(*) FROM rocpd_pmc_event_{u}"
+            ).fetchone()[0]
+            if npmc:
+                cols = [r[1] for r in cur.execute(
+                    f"PRAGMA table_info(rocpd_pmc_event_{u})")]
+                f.write(f"PMC events: {npmc} (cols: {cols})\n")
+                try:
+                    q2 = (
+                        f"SELECT s.display_name, i.name,"
+                        f" SUM(p.value), COUNT(*)"
+                        f" FROM rocpd_pmc_event_{u} p"
+                        f" JOIN rocpd_info_pmc_{u} i ON p.pmc_id = i.id"
+                        f" JOIN rocpd_kernel_dispatch_{u} d"
+                        f"   ON p.event_id = d.event_id"
+                        f" JOIN rocpd_info_kernel_symbol_{u} s"
+                        f"   ON d.kernel_id = s.id"
+                        f" GROUP BY s.display_name, i.name"
+                        f" ORDER BY s.display_name, i.name"
+                    )
+                    for name, cname, val, n in cur.execute(q2):
+                        f.write(f"  {cname:34s} {val:16.0f} n={n:6d} "
+                                f"{name[:60]}\n")
+                except Exception as e:
+                    f.write(f"  pmc join failed: {e}\n")
 
 
 if __name__ == "__main__":
