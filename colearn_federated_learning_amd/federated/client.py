@@ -183,9 +183,29 @@ def _per_sample_clipped_backward(
     per-sample norm + clip-scale + reduction runs through the fused HIP
     kernel on gfx950 (ops.dp) when gradients live on the GPU.
     """
-    from torch.func import functional_call, grad, vmap
-
     from ..ops import dp as dp_ops
+
+    # fast path: hook-based per-sample grads through the native kernels
+    # (ops.grad_sample); vmap fallback for unsupported module sets.
+    try:
+        from ..ops.grad_sample import GradSampleEngine
+
+        eng = GradSampleEngine(model, loss_scale=float(x.shape[0]))
+        with eng:
+            with torch.autocast("cuda", torch.bfloat16, enabled=use_bf16):
+                loss_val = model.loss(x, y)
+            loss_val.backward()
+            per_sample = eng.grad_samples()
+            clipped = dp_ops.clip_and_reduce(per_sample, clip_norm)
+            pdict = dict(model.named_parameters())
+            for (k, p), g in zip(model.named_parameters(), clipped):
+                p.grad.copy_(g.to(p.dtype))
+            eng.clear()
+        return loss_val.detach()
+    except NotImplementedError:
+        pass
+
+    from torch.func import functional_call, grad, vmap
 
     params = {k: v.detach() for k, v in model.named_parameters()}
     buffers = {k: v.detach() for k, v in model.named_buffers()}

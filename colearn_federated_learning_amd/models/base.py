@@ -47,14 +47,31 @@ class ColearnModel(nn.Module):
         return loss, correct, y.numel()
 
 
+class CLinear(nn.Module):
+    """nn.Linear-compatible module routed through the colearn GEMM (so the
+    module CALL is visible to hooks — the per-sample grad engine needs it)."""
+
+    def __init__(self, in_features: int, out_features: int) -> None:
+        super().__init__()
+        self.weight = nn.Parameter(torch.empty(out_features, in_features))
+        self.bias = nn.Parameter(torch.empty(out_features))
+        # identical init to nn.Linear
+        nn.init.kaiming_uniform_(self.weight, a=5 ** 0.5)
+        bound = 1 / (in_features ** 0.5)
+        nn.init.uniform_(self.bias, -bound, bound)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        from ..ops.gemm import colearn_linear
+
+        return colearn_linear(x, self.weight, self.bias)
+
+
 class LogisticRegression(ColearnModel):
     """Multinomial logistic regression — the plumbing-config model."""
 
     def __init__(self, in_features: int, num_classes: int) -> None:
         super().__init__()
-        self.fc = nn.Linear(in_features, num_classes)
+        self.fc = CLinear(in_features, num_classes)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        from ..ops.gemm import colearn_linear
-
-        return colearn_linear(x.flatten(1), self.fc.weight, self.fc.bias)
+        return self.fc(x.flatten(1))

@@ -18,9 +18,8 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..ops.conv import colearn_conv2d
-from ..ops.gemm import colearn_linear
 from ..ops.norms import ColearnBatchNorm2d
-from .base import ColearnModel
+from .base import CLinear, ColearnModel
 
 
 class _Conv2d(nn.Module):
@@ -137,7 +136,7 @@ class ResNet(ColearnModel):
         self.layer2 = self._make_layer(block, 128, layers[1], 2, group_norm)
         self.layer3 = self._make_layer(block, 256, layers[2], 2, group_norm)
         self.layer4 = self._make_layer(block, 512, layers[3], 2, group_norm)
-        self.fc = nn.Linear(512 * block.expansion, num_classes)
+        self.fc = CLinear(512 * block.expansion, num_classes)
 
     def _make_layer(self, block, planes: int, n: int, stride: int,
                     group_norm: bool) -> nn.Sequential:
@@ -151,7 +150,7 @@ class ResNet(ColearnModel):
         x = self.pool(self.stem(x))
         x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
         x = F.adaptive_avg_pool2d(x, 1).flatten(1)
-        return colearn_linear(x, self.fc.weight, self.fc.bias)
+        return self.fc(x)
 
 
 def resnet18(**kw) -> ResNet:
