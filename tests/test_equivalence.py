@@ -84,7 +84,13 @@ def test_world2_equals_world1(tmp_path, over):
         _run_world(1, d1, 29743, over=over)
     w1 = _final_weights(d1)
     w2 = _final_weights(d2)
-    torch.testing.assert_close(w1, w2, rtol=1e-4, atol=1e-5)
+    if over.get("agg__topk_ratio"):
+        # near the k-th |value| threshold, layout-order fp noise can swap
+        # which entry a client keeps — compare in aggregate
+        diff = (w1 - w2).abs()
+        assert float(diff.mean()) < 1e-5 and float(diff.max()) < 5e-2
+    else:
+        torch.testing.assert_close(w1, w2, rtol=1e-4, atol=1e-5)
 
 
 def test_sparse_transport_matches_dense(tmp_path):
