@@ -254,8 +254,8 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
     float* __restrict__ gwf, ConvShape cs, int64_t NPQ, int64_t RSC,
     int nbm, int nbn, int splitk) {
   constexpr int TM = 128, TN = 128;
-  __shared__ short Gs[TM][BK + 2];
-  __shared__ short Xs[TN][BK + 2];
+  __shared__ short Gs[2][TM][BK + 2];
+  __shared__ short Xs[2][TN][BK + 2];
   const int g = blockIdx.z;
   const short* gyg = gy + static_cast<int64_t>(g) * NPQ * cs.K;
   const short* xg =
@@ -281,10 +281,10 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
   const int kk = threadIdx.x >> 3;         // 0..31
   const int e16 = (threadIdx.x & 7) * 16;  // 16-wide column block
 
-  for (int64_t k0 = k_begin; k0 < k_end; k0 += BK) {
+  short8v_t gv0, gv1, xv0, xv1;
+  auto load_chunk = [&](int64_t k0) {
     const int64_t npq = k0 + kk;
-    short8v_t gv0 = mfma_frag_zero(), gv1 = mfma_frag_zero();
-    short8v_t xv0 = mfma_frag_zero(), xv1 = mfma_frag_zero();
+    gv0 = gv1 = xv0 = xv1 = mfma_frag_zero();
     if (npq < k_end) {
       if (co0 + e16 < cs.K) {
         const short* p = gyg + npq * cs.K + co0 + e16;
@@ -305,27 +305,39 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
       conv_fwd_gather(xg, cs, npq, rc0 + e16, xv0);
       conv_fwd_gather(xg, cs, npq, rc0 + e16 + 8, xv1);
     }
-    __syncthreads();
+  };
+  auto write_chunk = [&](int buf) {
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      Gs[e16 + j][kk] = gv0[j];
-      Gs[e16 + 8 + j][kk] = gv1[j];
-      Xs[e16 + j][kk] = xv0[j];
-      Xs[e16 + 8 + j][kk] = xv1[j];
+      Gs[buf][e16 + j][kk] = gv0[j];
+      Gs[buf][e16 + 8 + j][kk] = gv1[j];
+      Xs[buf][e16 + j][kk] = xv0[j];
+      Xs[buf][e16 + 8 + j][kk] = xv1[j];
     }
-    __syncthreads();
+  };
+
+  // double-buffered pipeline: next chunk's global loads issue before the
+  // MFMA phase (HBM latency hides under compute — T14), LDS write lands
+  // in the other buffer, ONE barrier per K-step.
+  load_chunk(k_begin);
+  write_chunk(0);
+  __syncthreads();
+  int cur = 0;
+  for (int64_t k0 = k_begin; k0 < k_end; k0 += BK) {
+    const bool has_next = k0 + BK < k_end;
+    if (has_next) load_chunk(k0 + BK);
     const int lane = threadIdx.x % kWave;
     const int fr = lane & 15, kg = (lane >> 4) * 8;
     short8v_t af[4], bfr[4];
 #pragma unroll
     for (int m = 0; m < 4; ++m) {
-      const short* p = &Gs[wm + m * 16 + fr][kg];
+      const short* p = &Gs[cur][wm + m * 16 + fr][kg];
 #pragma unroll
       for (int j = 0; j < 8; ++j) af[m][j] = p[j];
     }
 #pragma unroll
     for (int n = 0; n < 4; ++n) {
-      const short* p = &Xs[wn + n * 16 + fr][kg];
+      const short* p = &Xs[cur][wn + n * 16 + fr][kg];
 #pragma unroll
       for (int j = 0; j < 8; ++j) bfr[n][j] = p[j];
     }
@@ -335,6 +347,9 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
       for (int n = 0; n < 4; ++n)
         acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af[m], bfr[n], acc[m][n], 0, 0, 0);
+    if (has_next) write_chunk(cur ^ 1);
+    __syncthreads();
+    cur ^= 1;
   }
   // deterministic slab epilogue
   const int lane = threadIdx.x % kWave;
