@@ -253,12 +253,8 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
     float* __restrict__ gwf, ConvShape cs, int64_t NPQ, int64_t RSC,
     int nbm, int nbn, int splitk) {
   constexpr int TM = 128, TN = 128;
-  // +8 pad = 16 B: keeps the 8-element fragment reads 16-B aligned (a +2
-  // pad forced scalar ds_read_u16) while the 80-B row stride spreads a
-  // 16-lane b128 read group over distinct banks (20*r mod 64 distinct for
-  // r in 0..15).
-  __shared__ short Gs[2][TM][BK + 8];
-  __shared__ short Xs[2][TN][BK + 8];
+  __shared__ short Gs[2][TM][BK + 2];
+  __shared__ short Xs[2][TN][BK + 2];
   const int g = blockIdx.z;
   const short* gyg = gy + static_cast<int64_t>(g) * NPQ * cs.K;
   const short* xg =
