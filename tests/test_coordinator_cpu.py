@@ -75,3 +75,25 @@ def test_fedprox_changes_update():
     s1.run_round(0)
     s2.run_round(0)
     assert not torch.allclose(s1.global_flat, s2.global_flat)
+
+
+def test_straggler_timeout_excludes_client():
+    # timeout of 0 seconds: every fit exceeds it => all results excluded,
+    # aggregation reweights to a no-op round
+    cfg = driver_config(1).override(rounds=1, num_clients=2,
+                                    straggler_timeout_s=0.0,
+                                    train__use_stacked=False)
+    s = FederatedServer(cfg)
+    g0 = s.global_flat.clone()
+    rec = s.run_round(0)
+    assert rec["client_losses"] == {}
+    torch.testing.assert_close(s.global_flat, g0)
+
+
+def test_target_accuracy_early_stop():
+    cfg = driver_config(1).override(rounds=50, num_clients=3,
+                                    num_evaluators=1, target_accuracy=0.5)
+    s = FederatedServer(cfg)
+    hist = s.train()
+    assert len(hist) < 50  # stopped well before the round budget
+    assert hist[-1]["eval_accuracy"] >= 0.5
