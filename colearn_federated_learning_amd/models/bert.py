@@ -64,7 +64,8 @@ class BertSelfAttention(nn.Module):
             for i in range(3)
         )
         scores = colearn_matmul_nt(q, k) / math.sqrt(self.dh)
-        probs = self.drop(F.softmax(scores.float(), dim=-1).to(v.dtype))
+        # torch's softmax accumulates in fp32 internally for bf16
+        probs = self.drop(F.softmax(scores, dim=-1))
         ctx = colearn_matmul(probs, v)
         ctx = ctx.view(b, self.h, s, self.dh).transpose(1, 2).reshape(b, s, d)
         return self.out(ctx)
