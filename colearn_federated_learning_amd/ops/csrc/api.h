@@ -65,5 +65,8 @@ void launch_conv_dgrad(const void* gy, const void* wt, void* gx,
 int conv_wgrad_splitk(ConvShape cs, int G);
 void launch_conv_wgrad(const void* gy, const void* x, float* gwf, void* gw,
                        ConvShape cs, int G, hipStream_t stream);
+void launch_conv_wgrad_variant(const void* gy, const void* x, float* gwf,
+                               void* gw, ConvShape cs, int G, int variant,
+                               hipStream_t stream);
 
 }  // namespace colearn

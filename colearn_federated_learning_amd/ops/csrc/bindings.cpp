@@ -10,6 +10,8 @@
 
 #include <torch/extension.h>
 
+#include <cstdlib>
+
 #include <ATen/hip/impl/HIPGuardImplMasqueradingAsCUDA.h>
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 
@@ -403,8 +405,15 @@ torch::Tensor conv2d_wgrad(torch::Tensor gy, torch::Tensor x, int64_t R,
   auto gwf = torch::empty({G * splitk, cs.K, rsc},
                           x.options().dtype(torch::kFloat32));
   auto gw_rsci = torch::empty({G, cs.K, cs.R, cs.S, cs.C}, x.options());
-  launch_conv_wgrad(gc.data_ptr(), xc.data_ptr(), gwf.data_ptr<float>(),
-                    gw_rsci.data_ptr(), cs, static_cast<int>(G), stream());
+  const char* env = getenv("COLEARN_WGRAD_VARIANT");
+  const int variant = env ? atoi(env) : 0;
+  if (variant)
+    launch_conv_wgrad_variant(gc.data_ptr(), xc.data_ptr(),
+                              gwf.data_ptr<float>(), gw_rsci.data_ptr(), cs,
+                              static_cast<int>(G), variant, stream());
+  else
+    launch_conv_wgrad(gc.data_ptr(), xc.data_ptr(), gwf.data_ptr<float>(),
+                      gw_rsci.data_ptr(), cs, static_cast<int>(G), stream());
   HIP_CHECK_LAST();
   auto out = gw_rsci.permute({0, 1, 4, 2, 3}).contiguous();
   return G == 1 ? out.squeeze(0) : out;

@@ -465,4 +465,219 @@ void launch_conv_wgrad(const void* gy, const void* x, float* gwf, void* gw,
       gwf, static_cast<bf16*>(gw), n_per_g, splitk);
 }
 
+// ==== wgrad layout variants (tuning bench only; prod kernel above) =======
+// V1: conv-tile layout (64-B rows, 16-B slot XOR swizzle) — scatter b16
+//     writes, proven conflict-free b128 fragment reads.
+// V2: natural [k][col] tile — vector b128 writes, strided b16 reads,
+//     +8-short row pad to spread read banks.
+__global__ __launch_bounds__(kBlockThreads) void conv_wgrad_v1_kernel(
+    const short* __restrict__ gy, const short* __restrict__ x,
+    float* __restrict__ gwf, ConvShape cs, int64_t NPQ, int64_t RSC,
+    int nbm, int nbn, int splitk) {
+  constexpr int TM = 128, TN = 128;
+  __shared__ short Gs[2][TM * BK];
+  __shared__ short Xs[2][TN * BK];
+  const int g = blockIdx.z;
+  const short* gyg = gy + static_cast<int64_t>(g) * NPQ * cs.K;
+  const short* xg = x + static_cast<int64_t>(g) * cs.N * cs.H * cs.W * cs.C;
+  float* slab = gwf + (static_cast<int64_t>(g) * splitk + blockIdx.y) *
+                          cs.K * RSC;
+  const int co0 = (blockIdx.x / nbn) * TM;
+  const int rc0 = (blockIdx.x % nbn) * TN;
+  const int64_t chunk = (NPQ + splitk - 1) / splitk;
+  const int64_t k_begin = static_cast<int64_t>(blockIdx.y) * chunk;
+  const int64_t k_end = min(k_begin + chunk, NPQ);
+  const int wid = threadIdx.x / kWave;
+  const int wm = (wid >> 1) * 64, wn = (wid & 1) * 64;
+  f32x4_t acc[4][4];
+#pragma unroll
+  for (int m = 0; m < 4; ++m)
+#pragma unroll
+    for (int n = 0; n < 4; ++n) acc[m][n] = {0.f, 0.f, 0.f, 0.f};
+  const int kk = threadIdx.x >> 3;
+  const int e16 = (threadIdx.x & 7) * 16;
+  short8v_t gv0, gv1, xv0, xv1;
+  auto load_chunk = [&](int64_t k0) {
+    const int64_t npq = k0 + kk;
+    gv0 = gv1 = xv0 = xv1 = mfma_frag_zero();
+    if (npq < k_end) {
+      if (co0 + e16 < cs.K) {
+        const short* p = gyg + npq * cs.K + co0 + e16;
+        if ((cs.K & 7) == 0 && co0 + e16 + 16 <= cs.K) {
+          gv0 = *reinterpret_cast<const short8v_t*>(p);
+          gv1 = *reinterpret_cast<const short8v_t*>(p + 8);
+        }
+      }
+      conv_fwd_gather(xg, cs, npq, rc0 + e16, xv0);
+      conv_fwd_gather(xg, cs, npq, rc0 + e16 + 8, xv1);
+    }
+  };
+  auto write_chunk = [&](int buf) {
+    char* G = reinterpret_cast<char*>(Gs[buf]);
+    char* X = reinterpret_cast<char*>(Xs[buf]);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      *reinterpret_cast<short*>(
+          G + lds_slot_byte(e16 + j, kk >> 3) + (kk & 7) * 2) = gv0[j];
+      *reinterpret_cast<short*>(
+          G + lds_slot_byte(e16 + 8 + j, kk >> 3) + (kk & 7) * 2) = gv1[j];
+      *reinterpret_cast<short*>(
+          X + lds_slot_byte(e16 + j, kk >> 3) + (kk & 7) * 2) = xv0[j];
+      *reinterpret_cast<short*>(
+          X + lds_slot_byte(e16 + 8 + j, kk >> 3) + (kk & 7) * 2) = xv1[j];
+    }
+  };
+  load_chunk(k_begin);
+  write_chunk(0);
+  __syncthreads();
+  int cur = 0;
+  for (int64_t k0 = k_begin; k0 < k_end; k0 += BK) {
+    const bool has_next = k0 + BK < k_end;
+    if (has_next) load_chunk(k0 + BK);
+    short8v_t af[4], bfr[4];
+#pragma unroll
+    for (int m = 0; m < 4; ++m)
+      af[m] = frag_from_lds(Gs[cur], wm + m * 16);
+#pragma unroll
+    for (int n = 0; n < 4; ++n)
+      bfr[n] = frag_from_lds(Xs[cur], wn + n * 16);
+#pragma unroll
+    for (int m = 0; m < 4; ++m)
+#pragma unroll
+      for (int n = 0; n < 4; ++n)
+        acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[m], bfr[n], acc[m][n], 0, 0, 0);
+    if (has_next) write_chunk(cur ^ 1);
+    __syncthreads();
+    cur ^= 1;
+  }
+  const int lane = threadIdx.x % kWave;
+  const int cif = lane & 15, rg = lane >> 4;
+#pragma unroll
+  for (int m = 0; m < 4; ++m)
+#pragma unroll
+    for (int n = 0; n < 4; ++n)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int co = co0 + wm + m * 16 + rg * 4 + j;
+        const int rc = rc0 + wn + n * 16 + cif;
+        if (co < cs.K && rc < RSC)
+          slab[static_cast<int64_t>(co) * RSC + rc] = acc[m][n][j];
+      }
+}
+
+__global__ __launch_bounds__(kBlockThreads) void conv_wgrad_v2_kernel(
+    const short* __restrict__ gy, const short* __restrict__ x,
+    float* __restrict__ gwf, ConvShape cs, int64_t NPQ, int64_t RSC,
+    int nbm, int nbn, int splitk) {
+  constexpr int TM = 128, TN = 128, PAD = 8;
+  __shared__ short Gs[2][BK][TM + PAD];
+  __shared__ short Xs[2][BK][TN + PAD];
+  const int g = blockIdx.z;
+  const short* gyg = gy + static_cast<int64_t>(g) * NPQ * cs.K;
+  const short* xg = x + static_cast<int64_t>(g) * cs.N * cs.H * cs.W * cs.C;
+  float* slab = gwf + (static_cast<int64_t>(g) * splitk + blockIdx.y) *
+                          cs.K * RSC;
+  const int co0 = (blockIdx.x / nbn) * TM;
+  const int rc0 = (blockIdx.x % nbn) * TN;
+  const int64_t chunk = (NPQ + splitk - 1) / splitk;
+  const int64_t k_begin = static_cast<int64_t>(blockIdx.y) * chunk;
+  const int64_t k_end = min(k_begin + chunk, NPQ);
+  const int wid = threadIdx.x / kWave;
+  const int wm = (wid >> 1) * 64, wn = (wid & 1) * 64;
+  f32x4_t acc[4][4];
+#pragma unroll
+  for (int m = 0; m < 4; ++m)
+#pragma unroll
+    for (int n = 0; n < 4; ++n) acc[m][n] = {0.f, 0.f, 0.f, 0.f};
+  const int kk = threadIdx.x >> 3;
+  const int e16 = (threadIdx.x & 7) * 16;
+  short8v_t gv0, gv1, xv0, xv1;
+  auto load_chunk = [&](int64_t k0) {
+    const int64_t npq = k0 + kk;
+    gv0 = gv1 = xv0 = xv1 = mfma_frag_zero();
+    if (npq < k_end) {
+      if (co0 + e16 < cs.K) {
+        const short* p = gyg + npq * cs.K + co0 + e16;
+        if ((cs.K & 7) == 0 && co0 + e16 + 16 <= cs.K) {
+          gv0 = *reinterpret_cast<const short8v_t*>(p);
+          gv1 = *reinterpret_cast<const short8v_t*>(p + 8);
+        }
+      }
+      conv_fwd_gather(xg, cs, npq, rc0 + e16, xv0);
+      conv_fwd_gather(xg, cs, npq, rc0 + e16 + 8, xv1);
+    }
+  };
+  auto write_chunk = [&](int buf) {
+    *reinterpret_cast<short8v_t*>(&Gs[buf][kk][e16]) = gv0;
+    *reinterpret_cast<short8v_t*>(&Gs[buf][kk][e16 + 8]) = gv1;
+    *reinterpret_cast<short8v_t*>(&Xs[buf][kk][e16]) = xv0;
+    *reinterpret_cast<short8v_t*>(&Xs[buf][kk][e16 + 8]) = xv1;
+  };
+  load_chunk(k_begin);
+  write_chunk(0);
+  __syncthreads();
+  int cur = 0;
+  const int lane = threadIdx.x % kWave;
+  const int fr = lane & 15, kg = (lane >> 4) * 8;
+  for (int64_t k0 = k_begin; k0 < k_end; k0 += BK) {
+    const bool has_next = k0 + BK < k_end;
+    if (has_next) load_chunk(k0 + BK);
+    short8v_t af[4], bfr[4];
+#pragma unroll
+    for (int m = 0; m < 4; ++m)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) af[m][j] = Gs[cur][kg + j][wm + m * 16 + fr];
+#pragma unroll
+    for (int n = 0; n < 4; ++n)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) bfr[n][j] = Xs[cur][kg + j][wn + n * 16 + fr];
+#pragma unroll
+    for (int m = 0; m < 4; ++m)
+#pragma unroll
+      for (int n = 0; n < 4; ++n)
+        acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[m], bfr[n], acc[m][n], 0, 0, 0);
+    if (has_next) write_chunk(cur ^ 1);
+    __syncthreads();
+    cur ^= 1;
+  }
+  const int cif = lane & 15, rg = lane >> 4;
+#pragma unroll
+  for (int m = 0; m < 4; ++m)
+#pragma unroll
+    for (int n = 0; n < 4; ++n)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int co = co0 + wm + m * 16 + rg * 4 + j;
+        const int rc = rc0 + wn + n * 16 + cif;
+        if (co < cs.K && rc < RSC)
+          slab[static_cast<int64_t>(co) * RSC + rc] = acc[m][n][j];
+      }
+}
+
+void launch_conv_wgrad_variant(const void* gy, const void* x, float* gwf,
+                               void* gw, ConvShape cs, int G, int variant,
+                               hipStream_t stream) {
+  const int64_t NPQ = static_cast<int64_t>(cs.N) * cs.P * cs.Q;
+  const int64_t RSC = static_cast<int64_t>(cs.R) * cs.S * cs.C;
+  const int nbm = (cs.K + 127) / 128;
+  const int nbn = static_cast<int>((RSC + 127) / 128);
+  const int splitk = conv_wgrad_splitk(cs, G);
+  dim3 grid(nbm * nbn, splitk, G);
+  if (variant == 1)
+   hipLaunchKernelGGL(( conv_wgrad_v1_kernel), dim3(grid), dim3(kBlockThreads), 0, stream, 
+        static_cast<const short*>(gy), static_cast<const short*>(x), gwf,
+        cs, NPQ, RSC, nbm, nbn, splitk);
+  else
+   hipLaunchKernelGGL(( conv_wgrad_v2_kernel), dim3(grid), dim3(kBlockThreads), 0, stream, 
+        static_cast<const short*>(gy), static_cast<const short*>(x), gwf,
+        cs, NPQ, RSC, nbm, nbn, splitk);
+  const int64_t n_per_g = static_cast<int64_t>(cs.K) * RSC;
+  int cgrid_x = elemwise_grid(n_per_g, 256, G > 0 ? 2048 / G : 2048);
+  if (cgrid_x < 1) cgrid_x = 1;
+ hipLaunchKernelGGL(( wgrad_combine_kernel), dim3(dim3(cgrid_x, G)), dim3(256), 0, stream, 
+      gwf, static_cast<bf16*>(gw), n_per_g, splitk);
+}
+
 }  // namespace colearn
