@@ -55,9 +55,9 @@ void launch_batchnorm_bwd(const void* gy, const void* y, const void* x,
                           bool training, bool relu, hipStream_t stream);
 void launch_gemm_nt(const void* A, const void* B, void* C, int64_t batch,
                     int64_t M, int64_t N, int64_t K, hipStream_t stream);
-void launch_gemm(const void* A, const void* B, void* C, int64_t batch,
-                 int64_t M, int64_t N, int64_t K, bool a_km, bool b_km,
-                 hipStream_t stream);
+void launch_gemm(const void* A, const void* B, void* C, const float* bias,
+                 int64_t batch, int64_t M, int64_t N, int64_t K, bool a_km,
+                 bool b_km, hipStream_t stream);
 void launch_conv_fwd(const void* x, const void* w, void* y, ConvShape cs,
                      int G, hipStream_t stream);
 void launch_conv_dgrad(const void* gy, const void* wt, void* gx,

@@ -254,7 +254,8 @@ batchnorm_bwd(torch::Tensor gy, torch::Tensor y, torch::Tensor x,
 // all four operand layouts hit one kernel with transposed LDS staging —
 // no host transpose passes.
 torch::Tensor gemm2(torch::Tensor a, torch::Tensor b, bool a_km, bool b_km,
-                    int64_t M, int64_t N, int64_t K) {
+                    int64_t M, int64_t N, int64_t K,
+                    const float* bias = nullptr) {
   check_bf16(a, "a");
   check_bf16(b, "b");
   const c10::hip::HIPGuardMasqueradingAsCUDA guard(a.device());
@@ -262,10 +263,24 @@ torch::Tensor gemm2(torch::Tensor a, torch::Tensor b, bool a_km, bool b_km,
   const int64_t batch = a.dim() == 3 ? a.size(0) : 1;
   auto c = batch > 1 ? torch::empty({batch, M, N}, a.options())
                      : torch::empty({M, N}, a.options());
-  launch_gemm(ac.data_ptr(), bc.data_ptr(), c.data_ptr(), batch, M, N, K,
-              a_km, b_km, stream());
+  launch_gemm(ac.data_ptr(), bc.data_ptr(), c.data_ptr(), bias, batch, M,
+              N, K, a_km, b_km, stream());
   HIP_CHECK_LAST();
   return c;
+}
+
+// linear forward: x @ w^T + bias with the bias fused into the epilogue
+torch::Tensor linear_nt(torch::Tensor x, torch::Tensor w,
+                        torch::Tensor bias) {
+  TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && x.size(1) == w.size(1),
+              "linear_nt shape mismatch");
+  torch::Tensor bf;
+  const float* bp = nullptr;
+  if (bias.numel() > 0) {
+    bf = bias.to(torch::kFloat32).contiguous();
+    bp = bf.data_ptr<float>();
+  }
+  return gemm2(x, w, true, true, x.size(0), w.size(0), x.size(1), bp);
 }
 
 torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b) {
@@ -435,6 +450,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("batchnorm_fwd", &batchnorm_fwd);
   m.def("batchnorm_bwd", &batchnorm_bwd);
   m.def("gemm_nt", &gemm_nt);
+  m.def("linear_nt", &linear_nt);
   m.def("gemm_nn", &gemm_nn);
   m.def("gemm_tn", &gemm_tn);
   m.def("bmm_nt", &bmm_nt);

@@ -31,8 +31,9 @@ DEV_INLINE int lds_elem_byte(int row, int k) {
 template <bool A_KM, bool B_KM>
 __global__ __launch_bounds__(kBlockThreads) void gemm_kernel(
     const short* __restrict__ A, const short* __restrict__ B,
-    bf16* __restrict__ C, int64_t M, int64_t N, int64_t K, int64_t sA,
-    int64_t sB, int64_t sC, int nbm, int nbn) {
+    bf16* __restrict__ C, const float* __restrict__ bias, int64_t M,
+    int64_t N, int64_t K, int64_t sA, int64_t sB, int64_t sC, int nbm,
+    int nbn) {
   __shared__ short lds[2][2][BM * BK];
   const int bid = xcd_swizzle(blockIdx.x, nbm * nbn);
   const int64_t bm = static_cast<int64_t>(bid / nbn) * BM;
@@ -165,13 +166,28 @@ __global__ __launch_bounds__(kBlockThreads) void gemm_kernel(
     __syncthreads();
     cur ^= 1;
   }
+  if (bias != nullptr) {
+    // fused bias epilogue: one add in registers instead of a whole-output
+    // elementwise pass afterwards
+    const int lane = threadIdx.x % kWave;
+    const int cif = lane & 15;
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      const int64_t col = bn + wn + n * 16 + cif;
+      const float bv = col < N ? bias[col] : 0.f;
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) acc[m][n][j] += bv;
+    }
+  }
   store_tile_bf16(reinterpret_cast<const float*>(acc), C, N, M, N, bm, bn,
                   wm, wn);
 }
 
-void launch_gemm(const void* A, const void* B, void* C, int64_t batch,
-                 int64_t M, int64_t N, int64_t K, bool a_km, bool b_km,
-                 hipStream_t stream) {
+void launch_gemm(const void* A, const void* B, void* C, const float* bias,
+                 int64_t batch, int64_t M, int64_t N, int64_t K, bool a_km,
+                 bool b_km, hipStream_t stream) {
   const int nbm = static_cast<int>((M + BM - 1) / BM);
   const int nbn = static_cast<int>((N + BN - 1) / BN);
   dim3 grid(nbm * nbn, 1, static_cast<unsigned>(batch));
@@ -181,22 +197,22 @@ void launch_gemm(const void* A, const void* B, void* C, int64_t batch,
   const int64_t sA = M * K, sB = N * K, sC = M * N;
   if (a_km && b_km)
    hipLaunchKernelGGL(( gemm_kernel<true, true>), dim3(grid), dim3(kBlockThreads), 0, stream, 
-        a, b, c, M, N, K, sA, sB, sC, nbm, nbn);
+        a, b, c, bias, M, N, K, sA, sB, sC, nbm, nbn);
   else if (a_km && !b_km)
    hipLaunchKernelGGL(( gemm_kernel<true, false>), dim3(grid), dim3(kBlockThreads), 0, stream, 
-        a, b, c, M, N, K, sA, sB, sC, nbm, nbn);
+        a, b, c, bias, M, N, K, sA, sB, sC, nbm, nbn);
   else if (!a_km && b_km)
    hipLaunchKernelGGL(( gemm_kernel<false, true>), dim3(grid), dim3(kBlockThreads), 0, stream, 
-        a, b, c, M, N, K, sA, sB, sC, nbm, nbn);
+        a, b, c, bias, M, N, K, sA, sB, sC, nbm, nbn);
   else
    hipLaunchKernelGGL(( gemm_kernel<false, false>), dim3(grid), dim3(kBlockThreads), 0, stream, 
-        a, b, c, M, N, K, sA, sB, sC, nbm, nbn);
+        a, b, c, bias, M, N, K, sA, sB, sC, nbm, nbn);
 }
 
 // legacy entry (conv path compatibility)
 void launch_gemm_nt(const void* A, const void* B, void* C, int64_t batch,
                     int64_t M, int64_t N, int64_t K, hipStream_t stream) {
-  launch_gemm(A, B, C, batch, M, N, K, true, true, stream);
+  launch_gemm(A, B, C, nullptr, batch, M, N, K, true, true, stream);
 }
 
 }  // namespace colearn

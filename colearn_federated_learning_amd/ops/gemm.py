@@ -31,10 +31,11 @@ class _LinearFn(torch.autograd.Function):
     def forward(x: torch.Tensor, w: torch.Tensor,
                 bias: Optional[torch.Tensor]):
         if _use_native(x, w):
-            out = require_ext().gemm_nt(x.contiguous(), w.contiguous())
-            if bias is not None:
-                out = out + bias
-            return out
+            return require_ext().linear_nt(
+                x.contiguous(), w.contiguous(),
+                bias if bias is not None
+                else torch.empty(0, device=x.device),
+            )
         return F.linear(x, w, bias)
 
     @staticmethod
