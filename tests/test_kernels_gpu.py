@@ -403,8 +403,10 @@ def test_stacked_gpu_matches_sequential():
     )
     s_stack.run_round(0)
     s_seq.run_round(0)
+    # the paths differ in benign fp details (e.g. fused fp32 bias vs eager
+    # bf16 bias): compare at bf16-rounding tolerance
     torch.testing.assert_close(s_stack.global_flat, s_seq.global_flat,
-                               rtol=2e-3, atol=2e-4)
+                               rtol=2e-3, atol=1e-3)
     torch.testing.assert_close(s_stack.global_buf, s_seq.global_buf,
                                rtol=1e-2, atol=1e-3)
 
