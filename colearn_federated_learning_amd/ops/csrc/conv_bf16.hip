@@ -22,6 +22,8 @@
 // C % 8 == 0 (every ResNet conv except the 3-channel stem).
 
 #include "api.h"
+
+#include <cstdlib>
 #include "tile_gemm.h"
 
 namespace colearn {
@@ -117,16 +119,17 @@ DEV_INLINE void conv_dgrad_gather(const short* __restrict__ gy,
 
 // ---- the shared implicit-GEMM NT kernel ----------------------------------
 // GATHER: 0 = conv fwd (gather x), 1 = dgrad stride1, 2 = dgrad strided
-// TN: output-channel tile width — 64 for the CO=64 layers (a 128-wide
-// tile would waste half its MFMAs there), 128 otherwise.
-template <int GATHER, int TN>
+// TN: output-channel tile width (64 for the CO=64 layers).
+// BKT: K-step depth; 64 halves the barrier count for long-K deep layers.
+template <int GATHER, int TN, int BKT>
 __global__ __launch_bounds__(kBlockThreads) void conv_nt_kernel(
     const short* __restrict__ Asrc, const short* __restrict__ Bw,
     bf16* __restrict__ C, ConvShape cs, int64_t Mc /* rows per client */,
     int64_t Nd, int64_t Kd, int64_t a_stride /* A elems per client */,
     int nbm, int nbn) {
-  __shared__ short lds_a[2][BM * BK];
-  __shared__ short lds_b[2][TN * BK];
+  constexpr int QV = BKT / 32;  // 16-element load pairs per thread-row
+  __shared__ short lds_a[2][BM * BKT];
+  __shared__ short lds_b[2][TN * BKT];
   const int bid = xcd_swizzle(blockIdx.x, nbm * nbn);
   const int64_t bm_local = static_cast<int64_t>(bid / nbn) * BM;
   const int64_t bn = static_cast<int64_t>(bid % nbn) * TN;
@@ -136,9 +139,8 @@ __global__ __launch_bounds__(kBlockThreads) void conv_nt_kernel(
   bf16* Cg = C + static_cast<int64_t>(g) * Mc * Nd;
 
   const int wid = threadIdx.x / kWave;
-  // TN=128: 2x2 waves of 64x64; TN=64: 4x1 waves of 32x64
-  constexpr int MF = TN == 128 ? 4 : 2;  // 16-row fragments per wave (M)
-  constexpr int NF = 4;                  // 16-col fragments per wave (N)
+  constexpr int MF = TN == 128 ? 4 : 2;
+  constexpr int NF = 4;
   const int wm = TN == 128 ? (wid >> 1) * 64 : wid * 32;
   const int wn = TN == 128 ? (wid & 1) * 64 : 0;
 
@@ -150,55 +152,61 @@ __global__ __launch_bounds__(kBlockThreads) void conv_nt_kernel(
 
   const int row = threadIdx.x >> 1;
   const int half = threadIdx.x & 1;
-  const int ntiles = static_cast<int>((Kd + BK - 1) / BK);
+  const int ntiles = static_cast<int>((Kd + BKT - 1) / BKT);
 
-  short8v_t a0, a1, b0, b1;
+  short8v_t a0[2 * QV], b0[2 * QV];
   auto load_tile = [&](int t) {
-    const int64_t k = static_cast<int64_t>(t) * BK + half * 16;
+    const int64_t kbase =
+        static_cast<int64_t>(t) * BKT + half * (BKT / 2);
     const int64_t am = bm_local + row;
-    if (am < Mc) {
-      if constexpr (GATHER == 0) {
-        conv_fwd_gather(A, cs, am, k, a0);
-        conv_fwd_gather(A, cs, am, k + 8, a1);
-      } else if constexpr (GATHER == 1) {
-        conv_dgrad_gather<true>(A, cs, am, k, a0);
-        conv_dgrad_gather<true>(A, cs, am, k + 8, a1);
-      } else {
-        conv_dgrad_gather<false>(A, cs, am, k, a0);
-        conv_dgrad_gather<false>(A, cs, am, k + 8, a1);
-      }
-    } else {
-      a0 = a1 = mfma_frag_zero();
-    }
-    b0 = b1 = mfma_frag_zero();
-    const int64_t bn_row = bn + row;
-    if (row < TN && bn_row < Nd) {
-      const short* p = Bg + bn_row * Kd + k;
-      if (k + 16 <= Kd) {
-        b0 = *reinterpret_cast<const short8v_t*>(p);
-        b1 = *reinterpret_cast<const short8v_t*>(p + 8);
-      } else {
 #pragma unroll
-        for (int j = 0; j < 16; ++j)
-          if (k + j < Kd) {
-            if (j < 8)
-              b0[j] = p[j];
-            else
-              b1[j - 8] = p[j];
-          }
+    for (int q = 0; q < QV; ++q) {
+      const int64_t k = kbase + q * 16;
+      if (am < Mc) {
+        if constexpr (GATHER == 0) {
+          conv_fwd_gather(A, cs, am, k, a0[2 * q]);
+          conv_fwd_gather(A, cs, am, k + 8, a0[2 * q + 1]);
+        } else if constexpr (GATHER == 1) {
+          conv_dgrad_gather<true>(A, cs, am, k, a0[2 * q]);
+          conv_dgrad_gather<true>(A, cs, am, k + 8, a0[2 * q + 1]);
+        } else {
+          conv_dgrad_gather<false>(A, cs, am, k, a0[2 * q]);
+          conv_dgrad_gather<false>(A, cs, am, k + 8, a0[2 * q + 1]);
+        }
+      } else {
+        a0[2 * q] = a0[2 * q + 1] = mfma_frag_zero();
+      }
+      b0[2 * q] = b0[2 * q + 1] = mfma_frag_zero();
+      const int64_t bn_row = bn + row;
+      if (row < TN && bn_row < Nd) {
+        const short* p = Bg + bn_row * Kd + k;
+        if (k + 16 <= Kd) {
+          b0[2 * q] = *reinterpret_cast<const short8v_t*>(p);
+          b0[2 * q + 1] = *reinterpret_cast<const short8v_t*>(p + 8);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 16; ++j)
+            if (k + j < Kd) {
+              if (j < 8)
+                b0[2 * q][j] = p[j];
+              else
+                b0[2 * q + 1][j - 8] = p[j];
+            }
+        }
       }
     }
   };
   auto write_tile = [&](int buf) {
-    *reinterpret_cast<short8v_t*>(reinterpret_cast<char*>(lds_a[buf]) +
-                                  lds_slot_byte(row, half * 2)) = a0;
-    *reinterpret_cast<short8v_t*>(reinterpret_cast<char*>(lds_a[buf]) +
-                                  lds_slot_byte(row, half * 2 + 1)) = a1;
-    if (row < TN) {
-      *reinterpret_cast<short8v_t*>(reinterpret_cast<char*>(lds_b[buf]) +
-                                    lds_slot_byte(row, half * 2)) = b0;
-      *reinterpret_cast<short8v_t*>(reinterpret_cast<char*>(lds_b[buf]) +
-                                    lds_slot_byte(row, half * 2 + 1)) = b1;
+    char* As = reinterpret_cast<char*>(lds_a[buf]);
+    char* Bs = reinterpret_cast<char*>(lds_b[buf]);
+    const int s0 = half * (2 * QV);
+#pragma unroll
+    for (int q = 0; q < 2 * QV; ++q) {
+      *reinterpret_cast<short8v_t*>(
+          As + lds_slot_byte_t<BKT>(row, s0 + q)) = a0[q];
+      if (row < TN)
+        *reinterpret_cast<short8v_t*>(
+            Bs + lds_slot_byte_t<BKT>(row, s0 + q)) = b0[q];
     }
   };
 
@@ -208,19 +216,22 @@ __global__ __launch_bounds__(kBlockThreads) void conv_nt_kernel(
   int cur = 0;
   for (int t = 0; t < ntiles; ++t) {
     if (t + 1 < ntiles) load_tile(t + 1);
-    short8v_t af[MF], bfr[NF];
 #pragma unroll
-    for (int m = 0; m < MF; ++m)
-      af[m] = frag_from_lds(lds_a[cur], wm + m * 16);
+    for (int kq = 0; kq < QV; ++kq) {
+      short8v_t af[MF], bfr[NF];
 #pragma unroll
-    for (int n = 0; n < NF; ++n)
-      bfr[n] = frag_from_lds(lds_b[cur], wn + n * 16);
-#pragma unroll
-    for (int m = 0; m < MF; ++m)
+      for (int m = 0; m < MF; ++m)
+        af[m] = frag_from_lds_t<BKT>(lds_a[cur], wm + m * 16, kq * 32);
 #pragma unroll
       for (int n = 0; n < NF; ++n)
-        acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            af[m], bfr[n], acc[m][n], 0, 0, 0);
+        bfr[n] = frag_from_lds_t<BKT>(lds_b[cur], wn + n * 16, kq * 32);
+#pragma unroll
+      for (int m = 0; m < MF; ++m)
+#pragma unroll
+        for (int n = 0; n < NF; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[m], bfr[n], acc[m][n], 0, 0, 0);
+    }
     if (t + 1 < ntiles) write_tile(cur ^ 1);
     __syncthreads();
     cur ^= 1;
@@ -240,6 +251,37 @@ __global__ __launch_bounds__(kBlockThreads) void conv_nt_kernel(
         if (r < Mc && c < Nd)
           Cg[r * Nd + c] = f2b(acc[m][n][j]);
       }
+}
+
+// pick the (TN, BKT) instantiation; BKT=64 when the K loop is long.
+template <int GATHER>
+static void conv_nt_dispatch(const short* A, const short* B, bf16* C,
+                             ConvShape cs, int64_t Mc, int64_t Nd,
+                             int64_t Kd, int64_t a_stride, int G,
+                             hipStream_t stream) {
+  const char* env = getenv("COLEARN_CONV_BK");
+  const int bkt = env ? atoi(env) : (Kd >= 4096 ? 64 : 32);
+  if (Nd <= 64) {
+    const int nbn = (static_cast<int>(Nd) + 63) / 64;
+    const int nbm = static_cast<int>((Mc + BM - 1) / BM);
+    dim3 grid(nbm * nbn, 1, G);
+    if (bkt == 64)
+      conv_nt_kernel<GATHER, 64, 64><<<grid, kBlockThreads, 0, stream>>>(
+          A, B, C, cs, Mc, Nd, Kd, a_stride, nbm, nbn);
+    else
+      conv_nt_kernel<GATHER, 64, 32><<<grid, kBlockThreads, 0, stream>>>(
+          A, B, C, cs, Mc, Nd, Kd, a_stride, nbm, nbn);
+  } else {
+    const int nbn = (static_cast<int>(Nd) + BN - 1) / BN;
+    const int nbm = static_cast<int>((Mc + BM - 1) / BM);
+    dim3 grid(nbm * nbn, 1, G);
+    if (bkt == 64)
+      conv_nt_kernel<GATHER, 128, 64><<<grid, kBlockThreads, 0, stream>>>(
+          A, B, C, cs, Mc, Nd, Kd, a_stride, nbm, nbn);
+    else
+      conv_nt_kernel<GATHER, 128, 32><<<grid, kBlockThreads, 0, stream>>>(
+          A, B, C, cs, Mc, Nd, Kd, a_stride, nbm, nbn);
+  }
 }
 
 // ==== wgrad ===============================================================
@@ -388,20 +430,9 @@ void launch_conv_fwd(const void* x, const void* w, void* y, ConvShape cs,
   const int64_t Mc = static_cast<int64_t>(cs.N) * cs.P * cs.Q;
   const int64_t Kd = static_cast<int64_t>(cs.R) * cs.S * cs.C;
   const int64_t a_stride = static_cast<int64_t>(cs.N) * cs.H * cs.W * cs.C;
-  const int nbm = static_cast<int>((Mc + BM - 1) / BM);
-  if (cs.K <= 64) {
-    const int nbn = (cs.K + 63) / 64;
-    conv_nt_kernel<0, 64><<<dim3(nbm * nbn, 1, G), kBlockThreads, 0,
-                            stream>>>(
-        static_cast<const short*>(x), static_cast<const short*>(w),
-        static_cast<bf16*>(y), cs, Mc, cs.K, Kd, a_stride, nbm, nbn);
-  } else {
-    const int nbn = (cs.K + BN - 1) / BN;
-    conv_nt_kernel<0, 128><<<dim3(nbm * nbn, 1, G), kBlockThreads, 0,
-                             stream>>>(
-        static_cast<const short*>(x), static_cast<const short*>(w),
-        static_cast<bf16*>(y), cs, Mc, cs.K, Kd, a_stride, nbm, nbn);
-  }
+  conv_nt_dispatch<0>(static_cast<const short*>(x),
+                      static_cast<const short*>(w), static_cast<bf16*>(y),
+                      cs, Mc, cs.K, Kd, a_stride, G, stream);
 }
 
 void launch_conv_dgrad(const void* gy, const void* wt, void* gx,
@@ -409,30 +440,16 @@ void launch_conv_dgrad(const void* gy, const void* wt, void* gx,
   const int64_t Mc = static_cast<int64_t>(cs.N) * cs.H * cs.W;
   const int64_t Kd = static_cast<int64_t>(cs.R) * cs.S * cs.K;
   const int64_t a_stride = static_cast<int64_t>(cs.N) * cs.P * cs.Q * cs.K;
-  const int nbm = static_cast<int>((Mc + BM - 1) / BM);
-  if (cs.C <= 64) {
-    const int nbn = (cs.C + 63) / 64;
-    dim3 grid(nbm * nbn, 1, G);
-    if (cs.stride == 1)
-      conv_nt_kernel<1, 64><<<grid, kBlockThreads, 0, stream>>>(
-          static_cast<const short*>(gy), static_cast<const short*>(wt),
-          static_cast<bf16*>(gx), cs, Mc, cs.C, Kd, a_stride, nbm, nbn);
-    else
-      conv_nt_kernel<2, 64><<<grid, kBlockThreads, 0, stream>>>(
-          static_cast<const short*>(gy), static_cast<const short*>(wt),
-          static_cast<bf16*>(gx), cs, Mc, cs.C, Kd, a_stride, nbm, nbn);
-  } else {
-    const int nbn = (cs.C + BN - 1) / BN;
-    dim3 grid(nbm * nbn, 1, G);
-    if (cs.stride == 1)
-      conv_nt_kernel<1, 128><<<grid, kBlockThreads, 0, stream>>>(
-          static_cast<const short*>(gy), static_cast<const short*>(wt),
-          static_cast<bf16*>(gx), cs, Mc, cs.C, Kd, a_stride, nbm, nbn);
-    else
-      conv_nt_kernel<2, 128><<<grid, kBlockThreads, 0, stream>>>(
-          static_cast<const short*>(gy), static_cast<const short*>(wt),
-          static_cast<bf16*>(gx), cs, Mc, cs.C, Kd, a_stride, nbm, nbn);
-  }
+  if (cs.stride == 1)
+    conv_nt_dispatch<1>(static_cast<const short*>(gy),
+                        static_cast<const short*>(wt),
+                        static_cast<bf16*>(gx), cs, Mc, cs.C, Kd, a_stride,
+                        G, stream);
+  else
+    conv_nt_dispatch<2>(static_cast<const short*>(gy),
+                        static_cast<const short*>(wt),
+                        static_cast<bf16*>(gx), cs, Mc, cs.C, Kd, a_stride,
+                        G, stream);
 }
 
 int conv_wgrad_splitk(ConvShape cs, int G) {

@@ -26,20 +26,38 @@ constexpr int BN = 128;
 constexpr int BK = 32;
 constexpr int kBlockThreads = 256;
 
-// LDS tile: [128 rows][32 k] bf16 = 8 KiB; 4 16-B slots per row.
-constexpr int kSlotsPerRow = BK * 2 / 16;  // 4
+// LDS tile: [128 rows][BKT k] bf16; 16-B slots per row with an XOR
+// swizzle that keeps every 16-lane ds_read_b128 group conflict-free:
+//   BKT=32: 64-B rows (16 banks) -> swz = (row>>2)&3 over 4 slots
+//   BKT=64: 128-B rows (32 banks) -> swz = (row>>1)&7 over 8 slots
+constexpr int kSlotsPerRow = BK * 2 / 16;  // 4 (BKT=32 default)
 
-DEV_INLINE int lds_slot_byte(int row, int slot) {
-  return row * (BK * 2) + ((slot ^ ((row >> 2) & 3)) * 16);
+template <int BKT = BK>
+DEV_INLINE int lds_slot_byte_t(int row, int slot) {
+  if constexpr (BKT == 64)
+    return row * (BKT * 2) + ((slot ^ ((row >> 1) & 7)) * 16);
+  else
+    return row * (BKT * 2) + ((slot ^ ((row >> 2) & 3)) * 16);
 }
 
-// read one A/B fragment (8 bf16) for this lane from a K-major LDS tile.
-DEV_INLINE short8v_t frag_from_lds(const short* tile, int row_base) {
+DEV_INLINE int lds_slot_byte(int row, int slot) {
+  return lds_slot_byte_t<BK>(row, slot);
+}
+
+// read one A/B fragment (8 bf16) for this lane from a K-major LDS tile;
+// k0 = K-offset of the fragment within the tile (multiples of 32).
+template <int BKT = BK>
+DEV_INLINE short8v_t frag_from_lds_t(const short* tile, int row_base,
+                                     int k0 = 0) {
   const int lane = threadIdx.x % kWave;
   const int row = row_base + (lane & 15);
-  const int slot = lane >> 4;  // k-group: k = slot*8 + j
+  const int slot = (k0 >> 3) + (lane >> 4);  // k-group: k = slot*8 + j
   return *reinterpret_cast<const short8v_t*>(
-      reinterpret_cast<const char*>(tile) + lds_slot_byte(row, slot));
+      reinterpret_cast<const char*>(tile) + lds_slot_byte_t<BKT>(row, slot));
+}
+
+DEV_INLINE short8v_t frag_from_lds(const short* tile, int row_base) {
+  return frag_from_lds_t<BK>(tile, row_base);
 }
 
 DEV_INLINE short8v_t mfma_frag_zero() {
