@@ -110,8 +110,8 @@ class FederatedClient:
                     x = x.to(dev, non_blocking=True)
                     y = y.to(dev, non_blocking=True)
                     if runner is not None and runner.matches(x):
-                        # hipGraph fast path: fwd+bwd+optimizer in ONE
-                        # graph replay (steprunner.py)
+                        # hipGraph fast path (for DP configs this replays
+                        # the whole hook+clip+optimizer pipeline)
                         loss = runner.step(x, y)
                     elif self.cfg.dp.enabled:
                         optimizer.zero_grad()

@@ -101,8 +101,20 @@ def make_step_runner(model, optimizer, flat, cfg, device,
                      sample_shape) -> Optional[GraphedStepRunner]:
     """Build a runner for the config's full-batch shape, or None when
     graphs don't apply (CPU, DP per-sample path, or disabled)."""
-    if device.type != "cuda" or cfg.dp.enabled or not cfg.train.use_graphs:
+    if device.type != "cuda" or not cfg.train.use_graphs:
         return None
+    if cfg.dp.enabled:
+        if cfg.data.dataset == "text_synth":
+            return None
+        bs = cfg.train.batch_size
+        x_shape = (bs,) + tuple(sample_shape)
+        try:
+            return DPGraphedStepRunner(
+                model, optimizer, flat, cfg.dp.clip_norm, x_shape, device,
+                cfg.train.dtype == "bfloat16",
+            )
+        except NotImplementedError:
+            return None  # no per-sample rule: eager vmap fallback
     if cfg.model.name == "bert_base":
         # hipGraph replays of the BERT step (sort/embedding kernels in the
         # graph) intermittently memory-fault on ROCm 7.2 once any other
@@ -124,3 +136,72 @@ def make_step_runner(model, optimizer, flat, cfg, device,
         model, optimizer, flat, x_shape, y_shape, x_dtype, y_dtype, device,
         use_bf16,
     )
+
+
+class DPGraphedStepRunner:
+    """hipGraph capture of the FULL DP-SGD step: forward, backward with
+    per-sample-gradient hooks (ops.grad_sample), fused clip+reduce
+    (ops.dp), clipped-grad copy-back, fused optimizer.  The hook Python
+    runs once at capture; its kernels replay.  No RNG-consuming ops are in
+    the DP models (GroupNorm, no dropout), and no sort/embedding kernels
+    (the ROCm graph-replay hazard), so capture is safe here.
+    """
+
+    def __init__(self, model, optimizer, flat, clip_norm: float,
+                 batch_shape, device: torch.device, use_bf16: bool,
+                 warmup_iters: int = 2) -> None:
+        from ..ops import dp as dp_ops
+        from ..ops.grad_sample import GradSampleEngine
+
+        self.device = device
+        self.optimizer = optimizer
+        self.enabled = device.type == "cuda"
+        if not self.enabled:
+            return
+        bs = batch_shape[0]
+        self.static_x = torch.zeros(batch_shape, dtype=torch.float32,
+                                    device=device)
+        self.static_y = torch.zeros((bs,), dtype=torch.long, device=device)
+        eng = GradSampleEngine(model, loss_scale=float(bs))
+
+        def one_step():
+            flat.flat_g.zero_()
+            with eng:
+                with torch.autocast("cuda", torch.bfloat16,
+                                    enabled=use_bf16):
+                    loss = model.loss(self.static_x, self.static_y)
+                loss.backward()
+                per_sample = eng.grad_samples()
+                clipped = dp_ops.clip_and_reduce(per_sample, clip_norm)
+                for (_, p), g in zip(model.named_parameters(), clipped):
+                    p.grad.copy_(g.to(p.dtype))
+                eng.clear()
+            optimizer.device_step()
+            return loss
+
+        model.train()
+        self.static_x.normal_()
+        self.static_y.random_(0, 2)
+        s = torch.cuda.Stream(device=device)
+        s.wait_stream(torch.cuda.current_stream(device))
+        with torch.cuda.stream(s):
+            for _ in range(warmup_iters):
+                optimizer.host_prepare()
+                one_step()
+        torch.cuda.current_stream(device).wait_stream(s)
+        self.graph = torch.cuda.CUDAGraph()
+        optimizer.host_prepare()
+        with torch.cuda.graph(self.graph):
+            self.static_loss = one_step()
+        optimizer.t = 0
+        torch.cuda.synchronize(device)
+
+    def matches(self, x: torch.Tensor) -> bool:
+        return self.enabled and tuple(x.shape) == tuple(self.static_x.shape)
+
+    def step(self, x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
+        self.static_x.copy_(x, non_blocking=True)
+        self.static_y.copy_(y, non_blocking=True)
+        self.optimizer.host_prepare()
+        self.graph.replay()
+        return self.static_loss

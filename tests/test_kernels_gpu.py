@@ -449,3 +449,27 @@ def test_stacked_graphed_matches_eager():
     s_e.run_round(0)
     torch.testing.assert_close(s_g.global_flat, s_e.global_flat,
                                rtol=1e-6, atol=1e-7)
+
+
+def test_dp_graphed_matches_eager():
+    """Graphed DP step (hooks+clip+optimizer in one replay) == eager DP."""
+    from colearn_federated_learning_amd.config import driver_config
+    from colearn_federated_learning_amd.federated.coordinator import (
+        FederatedServer,
+    )
+
+    base = dict(rounds=1, num_clients=2, data__samples_per_client=64,
+                train__batch_size=32, train__lr=0.01,
+                dp__noise_multiplier=0.0)
+    dev = torch.device("cuda", 0)
+    s_g = FederatedServer(driver_config(5).override(**base), device=dev)
+    assert s_g.step_runner is not None
+    s_e = FederatedServer(
+        driver_config(5).override(train__use_graphs=False, **base),
+        device=dev,
+    )
+    assert s_e.step_runner is None
+    s_g.run_round(0)
+    s_e.run_round(0)
+    torch.testing.assert_close(s_g.global_flat, s_e.global_flat,
+                               rtol=1e-5, atol=1e-6)
