@@ -190,13 +190,15 @@ def _per_sample_clipped_backward(
     try:
         from ..ops.grad_sample import GradSampleEngine
 
-        eng = GradSampleEngine(model, loss_scale=float(x.shape[0]))
+        eng = GradSampleEngine(model)
         with eng:
             with torch.autocast("cuda", torch.bfloat16, enabled=use_bf16):
                 loss_val = model.loss(x, y)
             loss_val.backward()
             per_sample = eng.grad_samples()
-            clipped = dp_ops.clip_and_reduce(per_sample, clip_norm)
+            clipped = dp_ops.clip_and_reduce(
+                per_sample, clip_norm, pre_scale=float(x.shape[0])
+            )
             pdict = dict(model.named_parameters())
             for (k, p), g in zip(model.named_parameters(), clipped):
                 p.grad.copy_(g.to(p.dtype))

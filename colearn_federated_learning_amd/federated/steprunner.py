@@ -162,7 +162,7 @@ class DPGraphedStepRunner:
         self.static_x = torch.zeros(batch_shape, dtype=torch.float32,
                                     device=device)
         self.static_y = torch.zeros((bs,), dtype=torch.long, device=device)
-        eng = GradSampleEngine(model, loss_scale=float(bs))
+        eng = GradSampleEngine(model)
 
         def one_step():
             flat.flat_g.zero_()
@@ -172,7 +172,8 @@ class DPGraphedStepRunner:
                     loss = model.loss(self.static_x, self.static_y)
                 loss.backward()
                 per_sample = eng.grad_samples()
-                clipped = dp_ops.clip_and_reduce(per_sample, clip_norm)
+                clipped = dp_ops.clip_and_reduce(per_sample, clip_norm,
+                                                 pre_scale=float(bs))
                 for (_, p), g in zip(model.named_parameters(), clipped):
                     p.grad.copy_(g.to(p.dtype))
                 eng.clear()

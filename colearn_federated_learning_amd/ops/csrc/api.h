@@ -25,13 +25,14 @@ void launch_fused_adam(float* w, const float* g, float* m, float* v,
                        int64_t ref_period, float lr, float b1, float b2,
                        float eps, float wd, float mu, float bc1, float bc2,
                        bool has_prox, hipStream_t stream);
-void launch_sq_norm_per_row(const float* g, float* sq, int64_t rows,
-                            int64_t cols, hipStream_t stream);
+void launch_sq_norm_per_row(const float* g, float* sq, float* partials,
+                            int64_t rows, int64_t cols, int chunks,
+                            hipStream_t stream);
 void launch_scale_rows_mean(const float* g, const float* scale, float* out,
                             int64_t rows, int64_t cols, hipStream_t stream);
 void launch_topk(const float* v, int64_t n, int64_t k, int* idx, float* val,
-                 unsigned* workspace, int* cand, int cand_cap,
-                 hipStream_t stream);
+                 unsigned* workspace /* >= 1036 u32 */, int* cand,
+                 int cand_cap, hipStream_t stream);
 void launch_scatter_add_pairs(float* out, const int* idx, const float* val,
                               int64_t k, hipStream_t stream);
 void launch_layernorm_fwd(const void* x, const float* w, const float* b,
@@ -63,6 +64,8 @@ void launch_conv_fwd(const void* x, const void* w, void* y, ConvShape cs,
 void launch_conv_dgrad(const void* gy, const void* wt, void* gx,
                        ConvShape cs, int G, hipStream_t stream);
 int conv_wgrad_splitk(ConvShape cs, int G);
+void launch_conv_wgrad_nosum(const void* gy, const void* x, float* gwf,
+                             ConvShape cs, int G, hipStream_t stream);
 void launch_conv_wgrad(const void* gy, const void* x, float* gwf, void* gw,
                        ConvShape cs, int G, hipStream_t stream);
 void launch_conv_wgrad_variant(const void* gy, const void* x, float* gwf,

@@ -92,8 +92,14 @@ void sq_norm_per_row(torch::Tensor g, torch::Tensor sq) {
   check_f32(sq, "sq");
   const c10::hip::HIPGuardMasqueradingAsCUDA guard(g.device());
   auto gc = g.contiguous();
+  const int64_t rows = gc.size(0), cols = gc.size(1);
+  int chunks = static_cast<int>(
+      std::min<int64_t>(std::max<int64_t>(2048 / rows, 1), 64));
+  auto partials = torch::empty({rows * chunks},
+                               g.options().dtype(torch::kFloat32));
   launch_sq_norm_per_row(gc.data_ptr<float>(), sq.data_ptr<float>(),
-                         gc.size(0), gc.size(1), stream());
+                         partials.data_ptr<float>(), rows, cols, chunks,
+                         stream());
   HIP_CHECK_LAST();
 }
 
@@ -116,7 +122,7 @@ std::tuple<torch::Tensor, torch::Tensor> topk_compress(torch::Tensor flat,
   k = std::min<int64_t>(k, fc.numel());
   auto idx = torch::empty({k}, fc.options().dtype(torch::kInt32));
   auto val = torch::empty({k}, fc.options());
-  auto ws = torch::zeros({520}, fc.options().dtype(torch::kInt32));
+  auto ws = torch::zeros({1036}, fc.options().dtype(torch::kInt32));
   const int cand_cap = static_cast<int>(k) + 4096;
   auto cand = torch::empty({cand_cap}, fc.options().dtype(torch::kInt32));
   launch_topk(fc.data_ptr<float>(), fc.numel(), k, idx.data_ptr<int>(),
@@ -396,6 +402,39 @@ torch::Tensor conv2d_dgrad(torch::Tensor gy, torch::Tensor w, int64_t H,
   return gx;
 }
 
+torch::Tensor conv2d_wgrad_fp32(torch::Tensor gy, torch::Tensor x,
+                                int64_t R, int64_t S, int64_t stride,
+                                int64_t pad, int64_t G) {
+  // per-sample gradient path: returns the fp32 split-K sum directly
+  // (splitk == 1 there), skipping the bf16 cast + re-float round trip
+  check_bf16(gy, "gy");
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(gy.device());
+  ConvShape cs;
+  cs.N = static_cast<int>(x.size(0) / G);
+  cs.C = static_cast<int>(x.size(1));
+  cs.H = static_cast<int>(x.size(2));
+  cs.W = static_cast<int>(x.size(3));
+  cs.K = static_cast<int>(gy.size(1));
+  cs.P = static_cast<int>(gy.size(2));
+  cs.Q = static_cast<int>(gy.size(3));
+  cs.R = static_cast<int>(R);
+  cs.S = static_cast<int>(S);
+  cs.stride = static_cast<int>(stride);
+  cs.pad = static_cast<int>(pad);
+  auto gc = gy.contiguous(torch::MemoryFormat::ChannelsLast);
+  auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
+  const int64_t rsc = static_cast<int64_t>(cs.R) * cs.S * cs.C;
+  const int splitk = conv_wgrad_splitk(cs, static_cast<int>(G));
+  TORCH_CHECK(splitk == 1, "fp32 wgrad path expects splitk==1");
+  auto gwf = torch::empty({G, cs.K, cs.R, cs.S, cs.C},
+                          x.options().dtype(torch::kFloat32));
+  launch_conv_wgrad_nosum(gc.data_ptr(), xc.data_ptr(),
+                          gwf.data_ptr<float>(), cs, static_cast<int>(G),
+                          stream());
+  HIP_CHECK_LAST();
+  return gwf.permute({0, 1, 4, 2, 3}).contiguous();
+}
+
 torch::Tensor conv2d_wgrad(torch::Tensor gy, torch::Tensor x, int64_t R,
                            int64_t S, int64_t stride, int64_t pad,
                            int64_t G) {
@@ -420,6 +459,7 @@ torch::Tensor conv2d_wgrad(torch::Tensor gy, torch::Tensor x, int64_t R,
   auto gwf = torch::empty({G * splitk, cs.K, rsc},
                           x.options().dtype(torch::kFloat32));
   auto gw_rsci = torch::empty({G, cs.K, cs.R, cs.S, cs.C}, x.options());
+  (void)0;
   const char* env = getenv("COLEARN_WGRAD_VARIANT");
   const int variant = env ? atoi(env) : 0;
   if (variant)
@@ -459,4 +499,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_fwd", &conv2d_fwd);
   m.def("conv2d_dgrad", &conv2d_dgrad);
   m.def("conv2d_wgrad", &conv2d_wgrad);
+  m.def("conv2d_wgrad_fp32", &conv2d_wgrad_fp32);
 }

@@ -463,6 +463,17 @@ int conv_wgrad_splitk(ConvShape cs, int G) {
   return static_cast<int>(want < cap ? want : cap);
 }
 
+void launch_conv_wgrad_nosum(const void* gy, const void* x, float* gwf,
+                             ConvShape cs, int G, hipStream_t stream) {
+  const int64_t NPQ = static_cast<int64_t>(cs.N) * cs.P * cs.Q;
+  const int64_t RSC = static_cast<int64_t>(cs.R) * cs.S * cs.C;
+  const int nbm = (cs.K + 127) / 128;
+  const int nbn = static_cast<int>((RSC + 127) / 128);
+  conv_wgrad_kernel<<<dim3(nbm * nbn, 1, G), kBlockThreads, 0, stream>>>(
+      static_cast<const short*>(gy), static_cast<const short*>(x), gwf, cs,
+      NPQ, RSC, nbm, nbn, 1);
+}
+
 void launch_conv_wgrad(const void* gy, const void* x, float* gwf, void* gw,
                        ConvShape cs, int G, hipStream_t stream) {
   const int64_t NPQ = static_cast<int64_t>(cs.N) * cs.P * cs.Q;

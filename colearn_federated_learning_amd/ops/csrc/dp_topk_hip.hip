@@ -16,9 +16,12 @@
 namespace colearn {
 
 // ---- per-sample squared norms: sq[b] += sum_d g[b,d]^2 --------------------
-__global__ void sq_norm_per_row_kernel(const float* __restrict__ g,
-                                       float* __restrict__ sq, int64_t rows,
-                                       int64_t cols) {
+// deterministic AND parallel: per-(row, chunk) partials in fixed slots,
+// combined per row in chunk order.
+__global__ void sq_norm_partial_kernel(const float* __restrict__ g,
+                                       float* __restrict__ partials,
+                                       int64_t rows, int64_t cols,
+                                       int chunks) {
   __shared__ float scratch[4];
   const int row = blockIdx.x;
   if (row >= rows) return;
@@ -35,7 +38,20 @@ __global__ void sq_norm_per_row_kernel(const float* __restrict__ g,
     acc += v * v;
   }
   acc = block_allreduce_sum<256>(acc, scratch);
-  if (threadIdx.x == 0) atomicAdd(sq + row, acc);
+  if (threadIdx.x == 0)
+    partials[static_cast<int64_t>(row) * chunks + blockIdx.y] = acc;
+}
+
+__global__ void sq_norm_combine_kernel(const float* __restrict__ partials,
+                                       float* __restrict__ sq, int64_t rows,
+                                       int chunks) {
+  const int64_t row = static_cast<int64_t>(blockIdx.x) * blockDim.x +
+                      threadIdx.x;
+  if (row >= rows) return;
+  float acc = 0.f;
+  for (int c = 0; c < chunks; ++c)
+    acc += partials[row * chunks + c];
+  sq[row] += acc;
 }
 
 // ---- out[d] = (1/rows) * sum_b scale[b] * g[b,d] --------------------------
@@ -132,17 +148,16 @@ __global__ void topk_hist2_kernel(const float* __restrict__ v, int64_t n,
     atomicAdd(hist2 + threadIdx.x, lh[threadIdx.x]);
 }
 
-// threshold = (bin1<<24)|(bin2<<16); counts strictly above threshold plus a
-// ticket budget for ties at the threshold make the output exactly k.
-// out2: [0]=threshold, [1]=need_eq budget at threshold
+// select round 2: refine within the chosen top-8-bit bin.
+// sel2: [0]=24-bit prefix (bin1<<24 | bin2<<16), [1]=entries still needed
 __global__ void topk_select2_kernel(const unsigned* __restrict__ sel1,
                                     const unsigned* __restrict__ hist2,
-                                    unsigned* __restrict__ out2) {
+                                    unsigned* __restrict__ sel2) {
   __shared__ unsigned h[256];
   h[threadIdx.x] = hist2[threadIdx.x];
   __syncthreads();
   if (threadIdx.x == 0) {
-    const unsigned rem = sel1[2];  // entries of k still needed inside bin1
+    const unsigned rem = sel1[2];
     unsigned above = 0;
     int bin = 0;
     for (int b = 255; b >= 0; --b) {
@@ -152,11 +167,55 @@ __global__ void topk_select2_kernel(const unsigned* __restrict__ sel1,
       }
       above += h[b];
     }
-    out2[0] = (sel1[0] << 24) | (static_cast<unsigned>(bin) << 16);
-    out2[1] = rem - above;  // tickets for keys exactly at the threshold
-    // counters used by the compaction pass
-    out2[2] = 0;  // ticket counter (== threshold keys taken)
-    out2[3] = 0;  // output cursor
+    sel2[0] = (sel1[0] << 24) | (static_cast<unsigned>(bin) << 16);
+    sel2[1] = rem - above;
+  }
+}
+
+// round 3: histogram of bits 15..8 within the 16-bit prefix (shrinks the
+// candidate bucket ~256x — with k ~ 1e5 the 16-bit bucket alone held
+// thousands of entries and the O(c^2) tie ranking exploded).
+__global__ void topk_hist3_kernel(const float* __restrict__ v, int64_t n,
+                                  const unsigned* __restrict__ sel2,
+                                  unsigned* __restrict__ hist3) {
+  __shared__ unsigned lh[256];
+  if (threadIdx.x < 256) lh[threadIdx.x] = 0;
+  __syncthreads();
+  const unsigned prefix = sel2[0];
+  int64_t i = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+  for (; i < n; i += grid_stride()) {
+    const unsigned key = absbits(v[i]);
+    if ((key & 0xFFFF0000u) == prefix)
+      atomicAdd(&lh[(key >> 8) & 0xFFu], 1u);
+  }
+  __syncthreads();
+  if (threadIdx.x < 256 && lh[threadIdx.x])
+    atomicAdd(hist3 + threadIdx.x, lh[threadIdx.x]);
+}
+
+// out2: [0]=24-bit threshold, [1]=need_eq at threshold, [2]=ticket,
+// [3]=output cursor
+__global__ void topk_select3_kernel(const unsigned* __restrict__ sel2,
+                                    const unsigned* __restrict__ hist3,
+                                    unsigned* __restrict__ out2) {
+  __shared__ unsigned h[256];
+  h[threadIdx.x] = hist3[threadIdx.x];
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    const unsigned rem = sel2[1];
+    unsigned above = 0;
+    int bin = 0;
+    for (int b = 255; b >= 0; --b) {
+      if (above + h[b] >= rem) {
+        bin = b;
+        break;
+      }
+      above += h[b];
+    }
+    out2[0] = sel2[0] | (static_cast<unsigned>(bin) << 8);
+    out2[1] = rem - above;
+    out2[2] = 0;
+    out2[3] = 0;
   }
 }
 
@@ -170,7 +229,7 @@ __global__ void topk_compact_kernel(const float* __restrict__ v, int64_t n,
   const unsigned thresh = ctl[0];
   int64_t i = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
   for (; i < n; i += grid_stride()) {
-    const unsigned key = absbits(v[i]) & 0xFFFF0000u;
+    const unsigned key = absbits(v[i]) & 0xFFFFFF00u;
     if (key < thresh) continue;
     if (key == thresh) {
       const unsigned c = atomicAdd(&ctl[2], 1u);
@@ -186,7 +245,7 @@ __global__ void topk_compact_kernel(const float* __restrict__ v, int64_t n,
 }
 
 // exact tie resolution: among the candidates (all sharing the threshold's
-// 16-bit key), rank by (|v| desc, index asc) and keep the need_eq best.
+// 24-bit key), rank by (|v| desc, index asc) and keep the need_eq best.
 // One block; O(c^2/threads) comparisons, c is the candidate count (small:
 // one 16-bit bucket of the distribution).
 __global__ void topk_ties_kernel(const float* __restrict__ v,
@@ -226,12 +285,14 @@ __global__ void scatter_add_pairs_kernel(float* __restrict__ out,
 }
 
 // ---- launchers ------------------------------------------------------------
-void launch_sq_norm_per_row(const float* g, float* sq, int64_t rows,
-                            int64_t cols, hipStream_t stream) {
-  // one block per row: the per-row sum is a single deterministic
-  // block-reduction (atomicAdd across chunk-blocks would not be).
-  dim3 grid(static_cast<unsigned>(rows), 1);
- hipLaunchKernelGGL(( sq_norm_per_row_kernel), dim3(grid), dim3(256), 0, stream, g, sq, rows, cols);
+void launch_sq_norm_per_row(const float* g, float* sq, float* partials,
+                            int64_t rows, int64_t cols, int chunks,
+                            hipStream_t stream) {
+  dim3 grid(static_cast<unsigned>(rows), chunks);
+ hipLaunchKernelGGL(( sq_norm_partial_kernel), dim3(grid), dim3(256), 0, stream, g, partials, rows, cols,
+                                                   chunks);
+ hipLaunchKernelGGL(( sq_norm_combine_kernel), dim3((rows + 255) / 256), dim3(256), 0, stream, 
+      partials, sq, rows, chunks);
 }
 
 void launch_scale_rows_mean(const float* g, const float* scale, float* out,
@@ -241,17 +302,21 @@ void launch_scale_rows_mean(const float* g, const float* scale, float* out,
 }
 
 void launch_topk(const float* v, int64_t n, int64_t k, int* idx, float* val,
-                 unsigned* workspace /* >= 520 unsigned, zeroed */,
+                 unsigned* workspace /* >= 1036 unsigned, zeroed */,
                  int* cand, int cand_cap, hipStream_t stream) {
   unsigned* hist1 = workspace;        // 256
   unsigned* hist2 = workspace + 256;  // 256
-  unsigned* sel1 = workspace + 512;   // 4
-  unsigned* ctl = workspace + 516;    // 4
+  unsigned* hist3 = workspace + 512;  // 256
+  unsigned* sel1 = workspace + 768;   // 4
+  unsigned* sel2 = workspace + 772;   // 4
+  unsigned* ctl = workspace + 776;    // 4
   const int grid = elemwise_grid(n);
  hipLaunchKernelGGL(( topk_hist1_kernel), dim3(grid), dim3(256), 0, stream, v, n, hist1);
  hipLaunchKernelGGL(( topk_select_kernel), dim3(1), dim3(256), 0, stream, hist1, k, sel1);
  hipLaunchKernelGGL(( topk_hist2_kernel), dim3(grid), dim3(256), 0, stream, v, n, sel1, hist2);
- hipLaunchKernelGGL(( topk_select2_kernel), dim3(1), dim3(256), 0, stream, sel1, hist2, ctl);
+ hipLaunchKernelGGL(( topk_select2_kernel), dim3(1), dim3(256), 0, stream, sel1, hist2, sel2);
+ hipLaunchKernelGGL(( topk_hist3_kernel), dim3(grid), dim3(256), 0, stream, v, n, sel2, hist3);
+ hipLaunchKernelGGL(( topk_select3_kernel), dim3(1), dim3(256), 0, stream, sel2, hist3, ctl);
  hipLaunchKernelGGL(( topk_compact_kernel), dim3(grid), dim3(256), 0, stream, v, n, ctl, idx, val, k,
                                                 cand, cand_cap);
  hipLaunchKernelGGL(( topk_ties_kernel), dim3(1), dim3(256), 0, stream, v, ctl, cand, cand_cap, idx, val,

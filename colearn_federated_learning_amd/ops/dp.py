@@ -33,12 +33,19 @@ def per_sample_norms(per_sample: Sequence[torch.Tensor]) -> torch.Tensor:
 
 
 def clip_and_reduce(
-    per_sample: Sequence[torch.Tensor], clip_norm: float
+    per_sample: Sequence[torch.Tensor], clip_norm: float,
+    pre_scale: float = 1.0,
 ) -> List[torch.Tensor]:
-    """Clip each sample's total grad to clip_norm; return per-param means."""
+    """Clip each sample's total grad to clip_norm; return per-param means.
+
+    ``pre_scale``: the per-sample grads are treated as pre_scale * g (the
+    hook engine passes unscaled grads of a MEAN loss with pre_scale = B)
+    — the factor folds into the clip scale so no extra sweep is needed.
+    """
     b = per_sample[0].shape[0]
-    norms = per_sample_norms(per_sample)
-    scale = (clip_norm / norms.clamp_min(1e-12)).clamp_max(1.0)  # [B]
+    norms = per_sample_norms(per_sample) * pre_scale
+    scale = ((clip_norm / norms.clamp_min(1e-12)).clamp_max(1.0)
+             * pre_scale)  # [B]
     out = []
     if per_sample[0].is_cuda and native_enabled():
         m = require_ext()

@@ -17,8 +17,9 @@ A forward hook records each module's input; a full-backward hook computes
 fills the flat grad buffer — DP overwrites it with the clipped mean
 (ops.dp.clip_and_reduce, the fused norm/clip kernels).
 
-Per-sample grads are w.r.t. the PER-SAMPLE loss: hooks scale by B when
-the loss was a batch mean (``loss_scale=B``).
+The rules return grads w.r.t. the MEAN loss, unscaled: the batch factor
+folds into ops.dp.clip_and_reduce(pre_scale=B) so no extra sweep ever
+touches the [B, *param] tensors.
 """
 
 from __future__ import annotations
@@ -32,61 +33,63 @@ import torch.nn.functional as F
 from . import native_enabled, require_ext
 
 
-def _conv_grad_sample(module, x: torch.Tensor, gy: torch.Tensor,
-                      scale: float) -> Dict[str, torch.Tensor]:
+def _conv_grad_sample(module, x: torch.Tensor,
+                      gy: torch.Tensor) -> Dict[str, torch.Tensor]:
+    """NOTE: rules return UNSCALED per-sample grads; the loss-mean factor
+    folds into the clip pass (ops.dp.clip_and_reduce pre_scale) so no
+    extra elementwise sweep touches the [B, *param] tensors."""
     b = x.shape[0]
     if x.is_cuda and gy.dtype == torch.bfloat16 and x.dtype != torch.bfloat16:
         x = x.bfloat16()  # stem input arrives fp32 under autocast
     if (x.is_cuda and native_enabled() and x.dtype == torch.bfloat16
             and gy.dtype == torch.bfloat16):
-        gw = require_ext().conv2d_wgrad(
+        gw = require_ext().conv2d_wgrad_fp32(
             gy.contiguous(), x.contiguous(), module.weight.shape[-2],
             module.weight.shape[-1], module.stride, module.padding, b,
         )
-    else:
-        gws = [
-            torch.nn.grad.conv2d_weight(
-                x[i : i + 1].float(), module.weight.shape, gy[i : i + 1].float(),
-                module.stride, module.padding,
-            )
-            for i in range(b)
-        ]
-        gw = torch.stack(gws)
-    return {"weight": gw.float() * scale}
+        return {"weight": gw}
+    gws = [
+        torch.nn.grad.conv2d_weight(
+            x[i : i + 1].float(), module.weight.shape, gy[i : i + 1].float(),
+            module.stride, module.padding,
+        )
+        for i in range(b)
+    ]
+    return {"weight": torch.stack(gws).float()}
 
 
-def _linear_grad_sample(module, x: torch.Tensor, gy: torch.Tensor,
-                        scale: float) -> Dict[str, torch.Tensor]:
+def _linear_grad_sample(module, x: torch.Tensor,
+                        gy: torch.Tensor) -> Dict[str, torch.Tensor]:
     x2 = x.reshape(x.shape[0], -1, x.shape[-1]).float()
     g2 = gy.reshape(gy.shape[0], -1, gy.shape[-1]).float()
-    out = {"weight": torch.einsum("bsi,bsj->bij", g2, x2) * scale}
+    out = {"weight": torch.einsum("bsi,bsj->bij", g2, x2)}
     if getattr(module, "bias", None) is not None:
-        out["bias"] = g2.sum(dim=1) * scale
+        out["bias"] = g2.sum(dim=1)
     return out
 
 
-def _groupnorm_grad_sample(module, x: torch.Tensor, gy: torch.Tensor,
-                           scale: float) -> Dict[str, torch.Tensor]:
+def _groupnorm_grad_sample(module, x: torch.Tensor,
+                           gy: torch.Tensor) -> Dict[str, torch.Tensor]:
     xf, gf = x.float(), gy.float()
     xhat = F.group_norm(xf, module.num_groups, eps=module.eps)
     dims = tuple(range(2, x.dim()))
     return {
-        "weight": (gf * xhat).sum(dim=dims) * scale,
-        "bias": gf.sum(dim=dims) * scale,
+        "weight": (gf * xhat).sum(dim=dims),
+        "bias": gf.sum(dim=dims),
     }
 
 
 class GradSampleEngine:
-    """Attach to a model; after ``loss.backward()`` every supported
-    parameter has ``.grad_sample`` ([B, *shape], fp32)."""
+    """Attach to a model; after ``loss.backward()`` on a MEAN loss, every
+    supported parameter has ``.grad_sample`` ([B, *shape], fp32, unscaled
+    — multiply by B for per-sample-loss gradients)."""
 
-    def __init__(self, model: nn.Module, loss_scale: float = 1.0) -> None:
+    def __init__(self, model: nn.Module) -> None:
         from ..models.base import CLinear
         from ..models.bert import _Linear as BertLinear
         from ..models.resnet import _Conv2d
 
         self.model = model
-        self.loss_scale = loss_scale
         self._handles = []
         self._rules = []
         for mod in model.modules():
@@ -132,11 +135,9 @@ class GradSampleEngine:
         mod._gs_input = inputs[0].detach()
 
     def _make_bwd(self, rule):
-        scale = self.loss_scale
-
         def hook(mod, _grad_input, grad_output):
             gy = grad_output[0].detach()
-            gs = rule(mod, mod._gs_input, gy, scale)
+            gs = rule(mod, mod._gs_input, gy)
             for name, g in gs.items():
                 getattr(mod, name).grad_sample = g
 

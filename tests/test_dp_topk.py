@@ -102,12 +102,15 @@ def test_grad_sample_engine_matches_vmap():
     x = torch.randn(4, 3, 32, 32)
     y = torch.randint(0, 10, (4,))
 
-    eng = GradSampleEngine(m, loss_scale=float(x.shape[0]))
+    eng = GradSampleEngine(m)
     with eng:
         loss = m.loss(x, y)
         loss.backward()
+        # engine grads are w.r.t. the MEAN loss: scale by B to compare
+        # against per-sample-loss vmap grads
         hook_gs = {
-            n: p.grad_sample.clone() for n, p in m.named_parameters()
+            n: p.grad_sample.clone() * x.shape[0]
+            for n, p in m.named_parameters()
         }
         eng.clear()
     m.zero_grad(set_to_none=False)
@@ -142,11 +145,11 @@ def test_grad_sample_engine_logreg():
     m = LogisticRegression(20, 5)
     x = torch.randn(6, 20)
     y = torch.randint(0, 5, (6,))
-    eng = GradSampleEngine(m, loss_scale=6.0)
+    eng = GradSampleEngine(m)
     with eng:
         m.loss(x, y).backward()
-        gw = m.fc.weight.grad_sample.clone()
-        gb = m.fc.bias.grad_sample.clone()
+        gw = m.fc.weight.grad_sample.clone() * 6.0
+        gb = m.fc.bias.grad_sample.clone() * 6.0
         eng.clear()
 
     params = {k: v.detach() for k, v in m.named_parameters()}
