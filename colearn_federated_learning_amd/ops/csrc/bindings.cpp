@@ -423,9 +423,8 @@ torch::Tensor conv2d_wgrad_fp32(torch::Tensor gy, torch::Tensor x,
   cs.pad = static_cast<int>(pad);
   auto gc = gy.contiguous(torch::MemoryFormat::ChannelsLast);
   auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
-  const int64_t rsc = static_cast<int64_t>(cs.R) * cs.S * cs.C;
-  const int splitk = conv_wgrad_splitk(cs, static_cast<int>(G));
-  TORCH_CHECK(splitk == 1, "fp32 wgrad path expects splitk==1");
+  // always splitk == 1 here: per-sample launches already provide
+  // G = batch blocks of z-parallelism
   auto gwf = torch::empty({G, cs.K, cs.R, cs.S, cs.C},
                           x.options().dtype(torch::kFloat32));
   launch_conv_wgrad_nosum(gc.data_ptr(), xc.data_ptr(),
