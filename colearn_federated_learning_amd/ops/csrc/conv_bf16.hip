@@ -29,15 +29,14 @@
 namespace colearn {
 
 // ---- forward A-gather: row m = (n,p,q), k = (r,s,ci) ---------------------
-DEV_INLINE void conv_fwd_gather(const short* __restrict__ x,
-                                const ConvShape cs, int64_t m, int64_t k,
-                                short8v_t& out) {
+// The (n,p,q) decomposition costs two 64-bit divisions; hot callers hoist
+// it out of their K-loops via conv_fwd_gather_at.
+DEV_INLINE void conv_fwd_gather_at(const short* __restrict__ x,
+                                   const ConvShape cs, int n, int p, int q,
+                                   int64_t k, short8v_t& out) {
   out = mfma_frag_zero();
   const int64_t K_dim = static_cast<int64_t>(cs.R) * cs.S * cs.C;
   if (k >= K_dim) return;
-  const int q = static_cast<int>(m % cs.Q);
-  const int p = static_cast<int>((m / cs.Q) % cs.P);
-  const int n = static_cast<int>(m / (static_cast<int64_t>(cs.P) * cs.Q));
   if ((cs.C & 7) == 0) {
     const int ci = static_cast<int>(k % cs.C);
     const int s = static_cast<int>((k / cs.C) % cs.S);
@@ -65,19 +64,27 @@ DEV_INLINE void conv_fwd_gather(const short* __restrict__ x,
   }
 }
 
+// decompose helper + wrapper keeping the old signature (wgrad's npq
+// varies per K-step, so it keeps decomposing)
+DEV_INLINE void conv_fwd_gather(const short* __restrict__ x,
+                                const ConvShape cs, int64_t m, int64_t k,
+                                short8v_t& out) {
+  const int q = static_cast<int>(m % cs.Q);
+  const int p = static_cast<int>((m / cs.Q) % cs.P);
+  const int n = static_cast<int>(m / (static_cast<int64_t>(cs.P) * cs.Q));
+  conv_fwd_gather_at(x, cs, n, p, q, k, out);
+}
+
 // ---- dgrad A-gather: row m = (n,h,w), k = (r,s,co) -----------------------
 // STRIDE1: compile-time fast path without the divisibility checks (most
 // ResNet convs are stride 1).
 template <bool STRIDE1>
-DEV_INLINE void conv_dgrad_gather(const short* __restrict__ gy,
-                                  const ConvShape cs, int64_t m, int64_t k,
-                                  short8v_t& out) {
+DEV_INLINE void conv_dgrad_gather_at(const short* __restrict__ gy,
+                                     const ConvShape cs, int n, int h,
+                                     int w, int64_t k, short8v_t& out) {
   out = mfma_frag_zero();
   const int64_t K_dim = static_cast<int64_t>(cs.R) * cs.S * cs.K;
   if (k >= K_dim) return;
-  const int w = static_cast<int>(m % cs.W);
-  const int h = static_cast<int>((m / cs.W) % cs.H);
-  const int n = static_cast<int>(m / (static_cast<int64_t>(cs.H) * cs.W));
   if ((cs.K & 7) == 0) {
     const int co = static_cast<int>(k % cs.K);
     const int s = static_cast<int>((k / cs.K) % cs.S);
@@ -117,6 +124,16 @@ DEV_INLINE void conv_dgrad_gather(const short* __restrict__ gy,
   }
 }
 
+template <bool STRIDE1>
+DEV_INLINE void conv_dgrad_gather(const short* __restrict__ gy,
+                                  const ConvShape cs, int64_t m, int64_t k,
+                                  short8v_t& out) {
+  const int w = static_cast<int>(m % cs.W);
+  const int h = static_cast<int>((m / cs.W) % cs.H);
+  const int n = static_cast<int>(m / (static_cast<int64_t>(cs.H) * cs.W));
+  conv_dgrad_gather_at<STRIDE1>(gy, cs, n, h, w, k, out);
+}
+
 // ---- the shared implicit-GEMM NT kernel ----------------------------------
 // GATHER: 0 = conv fwd (gather x), 1 = dgrad stride1, 2 = dgrad strided
 // TN: output-channel tile width (64 for the CO=64 layers).
@@ -154,24 +171,36 @@ __global__ __launch_bounds__(kBlockThreads) void conv_nt_kernel(
   const int half = threadIdx.x & 1;
   const int ntiles = static_cast<int>((Kd + BKT - 1) / BKT);
 
+  // this thread's output pixel is K-loop-invariant: decompose ONCE
+  const int64_t am = bm_local + row;
+  int dn, dp_, dq;  // (n, p, q) for fwd / (n, h, w) for dgrad
+  {
+    const int inner = GATHER == 0 ? cs.Q : cs.W;
+    const int mid = GATHER == 0 ? cs.P : cs.H;
+    dq = static_cast<int>(am % inner);
+    dp_ = static_cast<int>((am / inner) % mid);
+    dn = static_cast<int>(am / (static_cast<int64_t>(mid) * inner));
+  }
+
   short8v_t a0[2 * QV], b0[2 * QV];
   auto load_tile = [&](int t) {
     const int64_t kbase =
         static_cast<int64_t>(t) * BKT + half * (BKT / 2);
-    const int64_t am = bm_local + row;
 #pragma unroll
     for (int q = 0; q < QV; ++q) {
       const int64_t k = kbase + q * 16;
       if (am < Mc) {
         if constexpr (GATHER == 0) {
-          conv_fwd_gather(A, cs, am, k, a0[2 * q]);
-          conv_fwd_gather(A, cs, am, k + 8, a0[2 * q + 1]);
+          conv_fwd_gather_at(A, cs, dn, dp_, dq, k, a0[2 * q]);
+          conv_fwd_gather_at(A, cs, dn, dp_, dq, k + 8, a0[2 * q + 1]);
         } else if constexpr (GATHER == 1) {
-          conv_dgrad_gather<true>(A, cs, am, k, a0[2 * q]);
-          conv_dgrad_gather<true>(A, cs, am, k + 8, a0[2 * q + 1]);
+          conv_dgrad_gather_at<true>(A, cs, dn, dp_, dq, k, a0[2 * q]);
+          conv_dgrad_gather_at<true>(A, cs, dn, dp_, dq, k + 8,
+                                     a0[2 * q + 1]);
         } else {
-          conv_dgrad_gather<false>(A, cs, am, k, a0[2 * q]);
-          conv_dgrad_gather<false>(A, cs, am, k + 8, a0[2 * q + 1]);
+          conv_dgrad_gather_at<false>(A, cs, dn, dp_, dq, k, a0[2 * q]);
+          conv_dgrad_gather_at<false>(A, cs, dn, dp_, dq, k + 8,
+                                      a0[2 * q + 1]);
         }
       } else {
         a0[2 * q] = a0[2 * q + 1] = mfma_frag_zero();
