@@ -346,6 +346,7 @@ ConvShape make_shape(const torch::Tensor& x, const torch::Tensor& w,
   cs.pad = static_cast<int>(pad);
   cs.P = (cs.H + 2 * cs.pad - cs.R) / cs.stride + 1;
   cs.Q = (cs.W + 2 * cs.pad - cs.S) / cs.stride + 1;
+  conv_shape_fill_magic(cs);
   return cs;
 }
 
@@ -388,6 +389,7 @@ torch::Tensor conv2d_dgrad(torch::Tensor gy, torch::Tensor w, int64_t H,
   cs.W = static_cast<int>(W);
   cs.stride = static_cast<int>(stride);
   cs.pad = static_cast<int>(pad);
+  conv_shape_fill_magic(cs);
   auto gc = gy.contiguous(torch::MemoryFormat::ChannelsLast);
   // wT K-major: [(G)][CI][R][S][CO]
   auto wt = (w.dim() == 5 ? w.permute({0, 2, 3, 4, 1})
@@ -421,6 +423,7 @@ torch::Tensor conv2d_wgrad_fp32(torch::Tensor gy, torch::Tensor x,
   cs.S = static_cast<int>(S);
   cs.stride = static_cast<int>(stride);
   cs.pad = static_cast<int>(pad);
+  conv_shape_fill_magic(cs);
   auto gc = gy.contiguous(torch::MemoryFormat::ChannelsLast);
   auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
   // always splitk == 1 here: per-sample launches already provide
@@ -451,6 +454,7 @@ torch::Tensor conv2d_wgrad(torch::Tensor gy, torch::Tensor x, int64_t R,
   cs.S = static_cast<int>(S);
   cs.stride = static_cast<int>(stride);
   cs.pad = static_cast<int>(pad);
+  conv_shape_fill_magic(cs);
   auto gc = gy.contiguous(torch::MemoryFormat::ChannelsLast);
   auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
   const int64_t rsc = static_cast<int64_t>(cs.R) * cs.S * cs.C;

@@ -28,6 +28,10 @@
 
 namespace colearn {
 
+DEV_INLINE unsigned u32div(unsigned n, U32Div d) {
+  return __umulhi(n, d.m) >> d.sh;
+}
+
 // ---- forward A-gather: row m = (n,p,q), k = (r,s,ci) ---------------------
 // The (n,p,q) decomposition costs two 64-bit divisions; hot callers hoist
 // it out of their K-loops via conv_fwd_gather_at.
@@ -38,9 +42,12 @@ DEV_INLINE void conv_fwd_gather_at(const short* __restrict__ x,
   const int64_t K_dim = static_cast<int64_t>(cs.R) * cs.S * cs.C;
   if (k >= K_dim) return;
   if ((cs.C & 7) == 0) {
-    const int ci = static_cast<int>(k % cs.C);
-    const int s = static_cast<int>((k / cs.C) % cs.S);
-    const int r = static_cast<int>(k / (static_cast<int64_t>(cs.C) * cs.S));
+    const unsigned ki = static_cast<unsigned>(k);
+    const unsigned kc = u32div(ki, cs.dC);       // k / C
+    const int ci = static_cast<int>(ki - kc * cs.C);
+    const unsigned r_ = u32div(ki, cs.dSC);      // k / (S*C)
+    const int s = static_cast<int>(kc - r_ * cs.S);
+    const int r = static_cast<int>(r_);
     const int ih = p * cs.stride - cs.pad + r;
     const int iw = q * cs.stride - cs.pad + s;
     if (ih < 0 || ih >= cs.H || iw < 0 || iw >= cs.W) return;
@@ -69,10 +76,12 @@ DEV_INLINE void conv_fwd_gather_at(const short* __restrict__ x,
 DEV_INLINE void conv_fwd_gather(const short* __restrict__ x,
                                 const ConvShape cs, int64_t m, int64_t k,
                                 short8v_t& out) {
-  const int q = static_cast<int>(m % cs.Q);
-  const int p = static_cast<int>((m / cs.Q) % cs.P);
-  const int n = static_cast<int>(m / (static_cast<int64_t>(cs.P) * cs.Q));
-  conv_fwd_gather_at(x, cs, n, p, q, k, out);
+  const unsigned mi = static_cast<unsigned>(m);
+  const unsigned mq = u32div(mi, cs.dQ);   // m / Q
+  const int q = static_cast<int>(mi - mq * cs.Q);
+  const unsigned n_ = u32div(mi, cs.dPQ);  // m / (P*Q)
+  const int p = static_cast<int>(mq - n_ * cs.P);
+  conv_fwd_gather_at(x, cs, static_cast<int>(n_), p, q, k, out);
 }
 
 // ---- dgrad A-gather: row m = (n,h,w), k = (r,s,co) -----------------------
@@ -86,9 +95,12 @@ DEV_INLINE void conv_dgrad_gather_at(const short* __restrict__ gy,
   const int64_t K_dim = static_cast<int64_t>(cs.R) * cs.S * cs.K;
   if (k >= K_dim) return;
   if ((cs.K & 7) == 0) {
-    const int co = static_cast<int>(k % cs.K);
-    const int s = static_cast<int>((k / cs.K) % cs.S);
-    const int r = static_cast<int>(k / (static_cast<int64_t>(cs.K) * cs.S));
+    const unsigned ki = static_cast<unsigned>(k);
+    const unsigned kk_ = u32div(ki, cs.dK);      // k / K
+    const int co = static_cast<int>(ki - kk_ * cs.K);
+    const unsigned r_ = u32div(ki, cs.dSK);      // k / (S*K)
+    const int s = static_cast<int>(kk_ - r_ * cs.S);
+    const int r = static_cast<int>(r_);
     const int ph = h + cs.pad - r;
     const int pw = w + cs.pad - s;
     if (ph < 0 || pw < 0) return;
