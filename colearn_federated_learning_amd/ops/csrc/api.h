@@ -22,8 +22,12 @@ inline U32Div make_u32div(unsigned d) {
   int sh = 0;
   while ((1u << sh) < d) ++sh;
   r.sh = sh;
-  r.m = static_cast<unsigned>(
-      ((static_cast<unsigned long long>(1) << (32 + sh)) + d - 1) / d);
+  if ((d & (d - 1)) == 0) {
+    r.m = 0;  // power of two: ceil(2^(32+sh)/d) == 2^32 overflows; shift
+  } else {
+    r.m = static_cast<unsigned>(
+        ((static_cast<unsigned long long>(1) << (32 + sh)) + d - 1) / d);
+  }
   return r;
 }
 

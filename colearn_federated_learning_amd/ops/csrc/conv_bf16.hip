@@ -29,7 +29,7 @@
 namespace colearn {
 
 DEV_INLINE unsigned u32div(unsigned n, U32Div d) {
-  return __umulhi(n, d.m) >> d.sh;
+  return d.m == 0 ? (n >> d.sh) : (__umulhi(n, d.m) >> d.sh);
 }
 
 // ---- forward A-gather: row m = (n,p,q), k = (r,s,ci) ---------------------
