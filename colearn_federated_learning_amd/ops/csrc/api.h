@@ -18,16 +18,22 @@ struct U32Div {
 };
 
 inline U32Div make_u32div(unsigned d) {
+  // round-up variant with sh = floor(log2 d): m = ceil(2^(32+sh)/d) then
+  // fits in 32 bits for non-powers-of-two (2^sh < d); powers of two use a
+  // plain shift (m = 0 sentinel).  Exact for n < 2^31.
   U32Div r;
-  int sh = 0;
-  while ((1u << sh) < d) ++sh;
-  r.sh = sh;
   if ((d & (d - 1)) == 0) {
-    r.m = 0;  // power of two: ceil(2^(32+sh)/d) == 2^32 overflows; shift
-  } else {
-    r.m = static_cast<unsigned>(
-        ((static_cast<unsigned long long>(1) << (32 + sh)) + d - 1) / d);
+    int sh = 0;
+    while ((1u << sh) < d) ++sh;
+    r.sh = sh;
+    r.m = 0;
+    return r;
   }
+  int sh = 0;
+  while ((2u << sh) < d) ++sh;  // sh = floor(log2 d)
+  r.sh = sh;
+  r.m = static_cast<unsigned>(
+      ((static_cast<unsigned long long>(1) << (32 + sh)) + d - 1) / d);
   return r;
 }
 
