@@ -278,8 +278,11 @@ def test_conv_bwd(n, c, hw, co, ks, stride, pad):
 
 # ---- end-to-end ------------------------------------------------------------
 def test_resnet18_step_native_vs_cpu():
-    """One bf16 training step on GPU (native kernels) vs fp32 CPU oracle:
-    losses must agree to bf16 tolerance."""
+    """A few bf16 training steps on GPU (native kernels) vs the fp32 CPU
+    oracle.  NOTE: a single random-init loss comparison is too weak (a
+    globally wrong conv still gives ~uniform logits and loss ~ln 10);
+    training a few steps amplifies any kernel error into divergent
+    losses, so the TRAINED loss must also match."""
     from colearn_federated_learning_amd.config import DataConfig, ModelConfig
     from colearn_federated_learning_amd.models import build_model
 
@@ -290,15 +293,23 @@ def test_resnet18_step_native_vs_cpu():
     mgpu = mgpu.to(DEV)
     x = torch.randn(8, 3, 32, 32)
     y = torch.randint(0, 10, (8,))
-    loss_cpu = mcpu.loss(x, y)
-    with torch.autocast("cuda", torch.bfloat16):
-        loss_gpu = mgpu.loss(x.to(DEV), y.to(DEV))
-    assert abs(loss_cpu.item() - loss_gpu.item()) < 0.1
-    loss_gpu.backward()
-    gnorm = sum(
-        float(p.grad.norm()) for p in mgpu.parameters() if p.grad is not None
-    )
-    assert gnorm > 0 and torch.isfinite(torch.tensor(gnorm))
+    opt_c = torch.optim.SGD(mcpu.parameters(), lr=0.05)
+    opt_g = torch.optim.SGD(mgpu.parameters(), lr=0.05)
+    for step in range(5):
+        opt_c.zero_grad()
+        loss_cpu = mcpu.loss(x, y)
+        loss_cpu.backward()
+        opt_c.step()
+        opt_g.zero_grad()
+        with torch.autocast("cuda", torch.bfloat16):
+            loss_gpu = mgpu.loss(x.to(DEV), y.to(DEV))
+        loss_gpu.backward()
+        opt_g.step()
+    # after 5 steps the CPU loss has dropped well below ln(10); the GPU
+    # path must track it (bf16 tolerance, amplified by training)
+    assert loss_cpu.item() < 1.8
+    assert abs(loss_cpu.item() - loss_gpu.item()) < 0.35, (
+        loss_cpu.item(), loss_gpu.item())
 
 
 def test_native_extension_required_on_gpu(monkeypatch):
