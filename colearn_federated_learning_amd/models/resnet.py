@@ -23,12 +23,17 @@ from .base import CLinear, ColearnModel
 
 
 class _Conv2d(nn.Module):
+    """Conv with K-MAJOR weight storage [CO, KH, KW, CI] — the HIP conv
+    kernels' native operand layout (zero weight permutes in the hot path).
+    Init draws in the standard [CO, CI, KH, KW] layout first so the
+    distribution (and RNG stream consumption) matches nn.Conv2d."""
+
     def __init__(self, cin: int, cout: int, k: int, stride: int = 1,
                  padding: int = 0) -> None:
         super().__init__()
-        self.weight = nn.Parameter(torch.empty(cout, cin, k, k))
-        nn.init.kaiming_normal_(self.weight, mode="fan_out",
-                                nonlinearity="relu")
+        w = torch.empty(cout, cin, k, k)
+        nn.init.kaiming_normal_(w, mode="fan_out", nonlinearity="relu")
+        self.weight = nn.Parameter(w.permute(0, 2, 3, 1).contiguous())
         self.stride, self.padding = stride, padding
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:

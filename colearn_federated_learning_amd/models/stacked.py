@@ -35,17 +35,19 @@ def _native(x: torch.Tensor) -> bool:
 
 
 class _SConv2dFn(torch.autograd.Function):
+    # stacked weight is K-MAJOR: [G, CO, KH, KW, CI] (see ops.conv._Conv2dFn)
     @staticmethod
     def forward(x, w, stride: int, padding: int, G: int):
         if _native(x):
-            return require_ext().conv2d_fwd(x, w, stride, padding)
+            return require_ext().conv2d_fwd_km(x, w, stride, padding)
         # CPU oracle: grouped conv with the client dim folded into channels
         Gb, C, H, W = x.shape
         B = Gb // G
-        co, ci, kh, kw = w.shape[1:]
+        wstd = w.permute(0, 1, 4, 2, 3)  # [G, CO, CI, KH, KW]
+        co, ci, kh, kw = wstd.shape[1:]
         xr = (x.view(G, B, C, H, W).transpose(0, 1)
               .reshape(B, G * C, H, W))
-        yr = F.conv2d(xr, w.reshape(G * co, ci, kh, kw), None, stride,
+        yr = F.conv2d(xr, wstd.reshape(G * co, ci, kh, kw), None, stride,
                       padding, groups=G)
         P, Q = yr.shape[-2:]
         return (yr.view(B, G, co, P, Q).transpose(0, 1)
@@ -63,25 +65,26 @@ class _SConv2dFn(torch.autograd.Function):
         s, p, G = ctx.stride, ctx.padding, ctx.G
         if _native(gy):
             m = require_ext()
-            gx = m.conv2d_dgrad(gy, w, x.shape[2], x.shape[3], s, p)
-            gw = m.conv2d_wgrad(gy, x, w.shape[-2], w.shape[-1], s, p, G)
+            gx = m.conv2d_dgrad_km(gy, w, x.shape[2], x.shape[3], s, p)
+            gw = m.conv2d_wgrad_km(gy, x, w.shape[2], w.shape[3], s, p, G)
         else:
             Gb, C, H, W = x.shape
             B = Gb // G
-            co, ci, kh, kw = w.shape[1:]
+            wstd = w.permute(0, 1, 4, 2, 3)
+            co, ci, kh, kw = wstd.shape[1:]
             xr = (x.view(G, B, C, H, W).transpose(0, 1)
                   .reshape(B, G * C, H, W))
             gyr = (gy.view(G, B, co, gy.shape[2], gy.shape[3])
                    .transpose(0, 1)
                    .reshape(B, G * co, gy.shape[2], gy.shape[3]))
-            wg = w.reshape(G * co, ci, kh, kw)
+            wg = wstd.reshape(G * co, ci, kh, kw)
             gxr = torch.nn.grad.conv2d_input(xr.shape, wg, gyr, s, p,
                                              groups=G)
             gwr = torch.nn.grad.conv2d_weight(xr, wg.shape, gyr, s, p,
                                               groups=G)
             gx = (gxr.view(B, G, C, H, W).transpose(0, 1)
                   .reshape(G * B, C, H, W))
-            gw = gwr.view(G, co, ci, kh, kw)
+            gw = gwr.view(G, co, ci, kh, kw).permute(0, 1, 3, 4, 2)
         return gx, gw, None, None, None
 
 
@@ -89,9 +92,9 @@ class SConv2d(nn.Module):
     def __init__(self, G: int, cin: int, cout: int, k: int,
                  stride: int = 1, padding: int = 0) -> None:
         super().__init__()
-        self.weight = nn.Parameter(torch.empty(G, cout, cin, k, k))
-        nn.init.kaiming_normal_(self.weight, mode="fan_out",
-                                nonlinearity="relu")
+        w = torch.empty(G, cout, cin, k, k)
+        nn.init.kaiming_normal_(w, mode="fan_out", nonlinearity="relu")
+        self.weight = nn.Parameter(w.permute(0, 1, 3, 4, 2).contiguous())
         self.G, self.stride, self.padding = G, stride, padding
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:

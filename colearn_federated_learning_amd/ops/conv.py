@@ -21,14 +21,19 @@ def _use_native(*ts: torch.Tensor) -> bool:
 
 
 class _Conv2dFn(torch.autograd.Function):
+    """Weights are K-MAJOR: [CO, KH, KW, CI].  This is the conv kernels'
+    native operand layout, so the model path runs with ZERO per-call weight
+    permute passes (fwd: none; dgrad: one for the transposed view; wgrad:
+    none — the fp32 slab output IS [CO, R, S, CI])."""
+
     generate_vmap_rule = True
 
     @staticmethod
     def forward(x, w, stride: int, padding: int):
         if _use_native(x, w):
             # layout normalization (channels-last) happens in the binding
-            return require_ext().conv2d_fwd(x, w, stride, padding)
-        return F.conv2d(x, w, None, stride, padding)
+            return require_ext().conv2d_fwd_km(x, w, stride, padding)
+        return F.conv2d(x, w.permute(0, 3, 1, 2), None, stride, padding)
 
     @staticmethod
     def setup_context(ctx, inputs, output):
@@ -42,17 +47,20 @@ class _Conv2dFn(torch.autograd.Function):
         s, p = ctx.stride, ctx.padding
         if _use_native(gy, w, x):
             m = require_ext()
-            gx = m.conv2d_dgrad(gy, w, x.shape[2], x.shape[3], s, p)
-            gw = m.conv2d_wgrad(gy, x, w.shape[2], w.shape[3], s, p, 1)
+            gx = m.conv2d_dgrad_km(gy, w, x.shape[2], x.shape[3], s, p)
+            gw = m.conv2d_wgrad_km(gy, x, w.shape[1], w.shape[2], s, p, 1)
         else:
-            gx = torch.nn.grad.conv2d_input(x.shape, w, gy, s, p)
-            gw = torch.nn.grad.conv2d_weight(x, w.shape, gy, s, p)
+            ws = w.permute(0, 3, 1, 2)
+            gx = torch.nn.grad.conv2d_input(x.shape, ws, gy, s, p)
+            gw = torch.nn.grad.conv2d_weight(x, ws.shape, gy, s, p)
+            gw = gw.permute(0, 2, 3, 1)
         return gx, gw, None, None
 
 
 def colearn_conv2d(
     x: torch.Tensor, w: torch.Tensor, stride: int = 1, padding: int = 0
 ) -> torch.Tensor:
+    """w is K-major [CO, KH, KW, CI] (see _Conv2dFn)."""
     if torch.is_autocast_enabled("cuda") and x.is_cuda:
         dt = torch.get_autocast_dtype("cuda")
         x, w = x.to(dt), w.to(dt)

@@ -350,6 +350,26 @@ ConvShape make_shape(const torch::Tensor& x, const torch::Tensor& w,
   return cs;
 }
 
+// shape helper for K-major weights ([*, CO, R, S, CI])
+ConvShape make_shape_km(const torch::Tensor& x, const torch::Tensor& w,
+                        int64_t stride, int64_t pad, int64_t G) {
+  ConvShape cs;
+  cs.N = static_cast<int>(x.size(0) / G);
+  cs.C = static_cast<int>(x.size(1));
+  cs.H = static_cast<int>(x.size(2));
+  cs.W = static_cast<int>(x.size(3));
+  const int wd = w.dim();
+  cs.K = static_cast<int>(w.size(wd - 4));
+  cs.R = static_cast<int>(w.size(wd - 3));
+  cs.S = static_cast<int>(w.size(wd - 2));
+  cs.stride = static_cast<int>(stride);
+  cs.pad = static_cast<int>(pad);
+  cs.P = (cs.H + 2 * cs.pad - cs.R) / cs.stride + 1;
+  cs.Q = (cs.W + 2 * cs.pad - cs.S) / cs.stride + 1;
+  conv_shape_fill_magic(cs);
+  return cs;
+}
+
 torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
                          int64_t pad) {
   check_bf16(x, "x");
@@ -369,6 +389,92 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
                   static_cast<int>(G), stream());
   HIP_CHECK_LAST();
   return y;
+}
+
+// K-major weight API (the model path: weights live as [*, CO, R, S, CI],
+// so no permute pass per call)
+torch::Tensor conv2d_fwd_km(torch::Tensor x, torch::Tensor w,
+                            int64_t stride, int64_t pad) {
+  check_bf16(x, "x");
+  check_bf16(w, "w");
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
+  const int64_t G = w.dim() == 5 ? w.size(0) : 1;
+  ConvShape cs = make_shape_km(x, w, stride, pad, G);
+  auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
+  auto wp = w.contiguous();
+  auto y = torch::empty({x.size(0), cs.K, cs.P, cs.Q},
+                        x.options().memory_format(
+                            torch::MemoryFormat::ChannelsLast));
+  launch_conv_fwd(xc.data_ptr(), wp.data_ptr(), y.data_ptr(), cs,
+                  static_cast<int>(G), stream());
+  HIP_CHECK_LAST();
+  return y;
+}
+
+torch::Tensor conv2d_dgrad_km(torch::Tensor gy, torch::Tensor w, int64_t H,
+                              int64_t W, int64_t stride, int64_t pad) {
+  check_bf16(gy, "gy");
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(gy.device());
+  const int64_t G = w.dim() == 5 ? w.size(0) : 1;
+  const int wd = w.dim();
+  ConvShape cs;
+  cs.N = static_cast<int>(gy.size(0) / G);
+  cs.K = static_cast<int>(gy.size(1));
+  cs.P = static_cast<int>(gy.size(2));
+  cs.Q = static_cast<int>(gy.size(3));
+  cs.C = static_cast<int>(w.size(wd - 1));
+  cs.R = static_cast<int>(w.size(wd - 3));
+  cs.S = static_cast<int>(w.size(wd - 2));
+  cs.H = static_cast<int>(H);
+  cs.W = static_cast<int>(W);
+  cs.stride = static_cast<int>(stride);
+  cs.pad = static_cast<int>(pad);
+  conv_shape_fill_magic(cs);
+  auto gc = gy.contiguous(torch::MemoryFormat::ChannelsLast);
+  // wT K-major for dgrad: [(G)][CI][R][S][CO] (one permute; fwd has none)
+  auto wt = (w.dim() == 5 ? w.permute({0, 4, 2, 3, 1})
+                          : w.permute({3, 1, 2, 0}))
+                .contiguous();
+  auto gx = torch::empty({gy.size(0), cs.C, cs.H, cs.W},
+                         gy.options().memory_format(
+                             torch::MemoryFormat::ChannelsLast));
+  launch_conv_dgrad(gc.data_ptr(), wt.data_ptr(), gx.data_ptr(), cs,
+                    static_cast<int>(G), stream());
+  HIP_CHECK_LAST();
+  return gx;
+}
+
+// K-major wgrad: returns [(G,) CO, R, S, CI] — the kernel's native
+// output layout, no permute
+torch::Tensor conv2d_wgrad_km(torch::Tensor gy, torch::Tensor x, int64_t R,
+                              int64_t S, int64_t stride, int64_t pad,
+                              int64_t G) {
+  check_bf16(gy, "gy");
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(gy.device());
+  ConvShape cs;
+  cs.N = static_cast<int>(x.size(0) / G);
+  cs.C = static_cast<int>(x.size(1));
+  cs.H = static_cast<int>(x.size(2));
+  cs.W = static_cast<int>(x.size(3));
+  cs.K = static_cast<int>(gy.size(1));
+  cs.P = static_cast<int>(gy.size(2));
+  cs.Q = static_cast<int>(gy.size(3));
+  cs.R = static_cast<int>(R);
+  cs.S = static_cast<int>(S);
+  cs.stride = static_cast<int>(stride);
+  cs.pad = static_cast<int>(pad);
+  conv_shape_fill_magic(cs);
+  auto gc = gy.contiguous(torch::MemoryFormat::ChannelsLast);
+  auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
+  const int64_t rsc = static_cast<int64_t>(cs.R) * cs.S * cs.C;
+  const int splitk = conv_wgrad_splitk(cs, static_cast<int>(G));
+  auto gwf = torch::empty({G * splitk, cs.K, rsc},
+                          x.options().dtype(torch::kFloat32));
+  auto gw = torch::empty({G, cs.K, cs.R, cs.S, cs.C}, x.options());
+  launch_conv_wgrad(gc.data_ptr(), xc.data_ptr(), gwf.data_ptr<float>(),
+                    gw.data_ptr(), cs, static_cast<int>(G), stream());
+  HIP_CHECK_LAST();
+  return G == 1 ? gw.squeeze(0) : gw;
 }
 
 torch::Tensor conv2d_dgrad(torch::Tensor gy, torch::Tensor w, int64_t H,
@@ -437,6 +543,37 @@ torch::Tensor conv2d_wgrad_fp32(torch::Tensor gy, torch::Tensor x,
   return gwf.permute({0, 1, 4, 2, 3}).contiguous();
 }
 
+// K-major per-sample wgrad: [B, CO, R, S, CI] fp32 — the kernel's native
+// output, zero permute passes (matches K-major module weights)
+torch::Tensor conv2d_wgrad_fp32_km(torch::Tensor gy, torch::Tensor x,
+                                   int64_t R, int64_t S, int64_t stride,
+                                   int64_t pad, int64_t G) {
+  check_bf16(gy, "gy");
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(gy.device());
+  ConvShape cs;
+  cs.N = static_cast<int>(x.size(0) / G);
+  cs.C = static_cast<int>(x.size(1));
+  cs.H = static_cast<int>(x.size(2));
+  cs.W = static_cast<int>(x.size(3));
+  cs.K = static_cast<int>(gy.size(1));
+  cs.P = static_cast<int>(gy.size(2));
+  cs.Q = static_cast<int>(gy.size(3));
+  cs.R = static_cast<int>(R);
+  cs.S = static_cast<int>(S);
+  cs.stride = static_cast<int>(stride);
+  cs.pad = static_cast<int>(pad);
+  conv_shape_fill_magic(cs);
+  auto gc = gy.contiguous(torch::MemoryFormat::ChannelsLast);
+  auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
+  auto gwf = torch::empty({G, cs.K, cs.R, cs.S, cs.C},
+                          x.options().dtype(torch::kFloat32));
+  launch_conv_wgrad_nosum(gc.data_ptr(), xc.data_ptr(),
+                          gwf.data_ptr<float>(), cs, static_cast<int>(G),
+                          stream());
+  HIP_CHECK_LAST();
+  return gwf;
+}
+
 torch::Tensor conv2d_wgrad(torch::Tensor gy, torch::Tensor x, int64_t R,
                            int64_t S, int64_t stride, int64_t pad,
                            int64_t G) {
@@ -500,7 +637,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bmm_nn", &bmm_nn);
   m.def("bmm_tn", &bmm_tn);
   m.def("conv2d_fwd", &conv2d_fwd);
+  m.def("conv2d_fwd_km", &conv2d_fwd_km);
+  m.def("conv2d_dgrad_km", &conv2d_dgrad_km);
+  m.def("conv2d_wgrad_km", &conv2d_wgrad_km);
   m.def("conv2d_dgrad", &conv2d_dgrad);
   m.def("conv2d_wgrad", &conv2d_wgrad);
   m.def("conv2d_wgrad_fp32", &conv2d_wgrad_fp32);
+  m.def("conv2d_wgrad_fp32_km", &conv2d_wgrad_fp32_km);
 }

@@ -41,18 +41,21 @@ def _conv_grad_sample(module, x: torch.Tensor,
     b = x.shape[0]
     if x.is_cuda and gy.dtype == torch.bfloat16 and x.dtype != torch.bfloat16:
         x = x.bfloat16()  # stem input arrives fp32 under autocast
+    # module.weight is K-major [CO, KH, KW, CI]
+    kh, kw = module.weight.shape[1], module.weight.shape[2]
     if (x.is_cuda and native_enabled() and x.dtype == torch.bfloat16
             and gy.dtype == torch.bfloat16):
-        gw = require_ext().conv2d_wgrad_fp32(
-            gy.contiguous(), x.contiguous(), module.weight.shape[-2],
-            module.weight.shape[-1], module.stride, module.padding, b,
+        gw = require_ext().conv2d_wgrad_fp32_km(
+            gy.contiguous(), x.contiguous(), kh, kw,
+            module.stride, module.padding, b,
         )
         return {"weight": gw}
+    std_shape = (module.weight.shape[0], module.weight.shape[3], kh, kw)
     gws = [
         torch.nn.grad.conv2d_weight(
-            x[i : i + 1].float(), module.weight.shape, gy[i : i + 1].float(),
+            x[i : i + 1].float(), std_shape, gy[i : i + 1].float(),
             module.stride, module.padding,
-        )
+        ).permute(0, 2, 3, 1)
         for i in range(b)
     ]
     return {"weight": torch.stack(gws).float()}

@@ -31,7 +31,7 @@ def test_stacked_matches_sequential_resnet():
     rec_b = s_seq.run_round(0)
     assert abs(rec_a["mean_client_loss"] - rec_b["mean_client_loss"]) < 1e-2
     torch.testing.assert_close(s_stack.global_flat, s_seq.global_flat,
-                               rtol=1e-3, atol=1e-4)
+                               rtol=1e-3, atol=3e-4)
     torch.testing.assert_close(s_stack.global_buf, s_seq.global_buf,
                                rtol=1e-2, atol=1e-3)
 
@@ -69,7 +69,7 @@ def test_stacked_fedprox():
     s_stack.run_round(0)
     s_seq.run_round(0)
     torch.testing.assert_close(s_stack.global_flat, s_seq.global_flat,
-                               rtol=1e-3, atol=1e-4)
+                               rtol=1e-3, atol=3e-4)
 
 
 def test_stacked_topk_per_client():
