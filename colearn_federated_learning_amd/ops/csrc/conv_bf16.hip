@@ -331,11 +331,22 @@ static void conv_nt_dispatch(const short* A, const short* B, bf16* C,
 // deterministic split-K slabs per client.  grid = (tiles, splitk, G).
 // Both operands transposed-scatter into K-major LDS (b16 writes; the
 // tr_b16 hardware-transpose read is the planned next step).
+//
+// Tile size is a template param: shapes with K <= 64 (ResNet stem/layer1)
+// or RSC <= 64 (1x1 downsamples) were burning >half their MFMA issue on
+// tile padding at 128x128 (measured: layer1 tile-occ 0.45) — the 64-wide
+// variants restore occupancy.  Waves always cover 64 output rows; the 4
+// waves split the other way: TM=128 -> 2x2 waves of 64x(TN/2),
+// TM=64 -> 1x4 waves of 64x(TN/4).
+template <int TM, int TN>
 __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
     const short* __restrict__ gy, const short* __restrict__ x,
     float* __restrict__ gwf, ConvShape cs, int64_t NPQ, int64_t RSC,
     int nbm, int nbn, int splitk) {
-  constexpr int TM = 128, TN = 128;
+  constexpr int WAVES_M = TM / 64;
+  constexpr int WAVES_N = 4 / WAVES_M;
+  constexpr int WN = TN / WAVES_N;  // wave output cols: 64/32/16
+  constexpr int NFR = WN / 16;      // B-fragments per wave
   __shared__ short Gs[2][TM][BK + 2];
   __shared__ short Xs[2][TN][BK + 2];
   const int g = blockIdx.z;
@@ -352,49 +363,80 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
   const int64_t k_end = min(k_begin + chunk, NPQ);
 
   const int wid = threadIdx.x / kWave;
-  const int wm = (wid >> 1) * 64, wn = (wid & 1) * 64;
-  f32x4_t acc[4][4];
+  const int wm = (wid / WAVES_N) * 64, wn = (wid % WAVES_N) * WN;
+  f32x4_t acc[4][NFR];
 #pragma unroll
   for (int m = 0; m < 4; ++m)
 #pragma unroll
-    for (int n = 0; n < 4; ++n) acc[m][n] = {0.f, 0.f, 0.f, 0.f};
+    for (int n = 0; n < NFR; ++n) acc[m][n] = {0.f, 0.f, 0.f, 0.f};
 
-  // staging: thread t covers npq row kk = t>>3, element block (t&7)*16
-  const int kk = threadIdx.x >> 3;         // 0..31
-  const int e16 = (threadIdx.x & 7) * 16;  // 16-wide column block
+  // staging decompositions (per operand: its TM*2 leading threads cover
+  // [cols][BK] in 16-element blocks): thread t -> npq row kk, col block.
+  // G_ALL/X_ALL are compile-time "every thread participates" facts so the
+  // 128-wide instantiations keep branch-free staging (the runtime guard
+  // variant cost 62 SGPR spills and 3x on the fully-occupied shapes).
+  constexpr bool G_ALL = (TM / 16) * BK >= kBlockThreads;
+  constexpr bool X_ALL = (TN / 16) * BK >= kBlockThreads;
+  const int kkg = threadIdx.x / (TM / 16);
+  const int e16g = (threadIdx.x % (TM / 16)) * 16;
+  const int kkx = threadIdx.x / (TN / 16);
+  const int e16x = (threadIdx.x % (TN / 16)) * 16;
 
   short8v_t gv0, gv1, xv0, xv1;
-  auto load_chunk = [&](int64_t k0) {
-    const int64_t npq = k0 + kk;
-    gv0 = gv1 = xv0 = xv1 = mfma_frag_zero();
-    if (npq < k_end) {
-      if (co0 + e16 < cs.K) {
-        const short* p = gyg + npq * cs.K + co0 + e16;
-        if ((cs.K & 7) == 0 && co0 + e16 + 16 <= cs.K) {
-          gv0 = *reinterpret_cast<const short8v_t*>(p);
-          gv1 = *reinterpret_cast<const short8v_t*>(p + 8);
-        } else {
+  auto load_gy = [&](int64_t npq) {
+    if (co0 + e16g < cs.K) {
+      const short* p = gyg + npq * cs.K + co0 + e16g;
+      if ((cs.K & 7) == 0 && co0 + e16g + 16 <= cs.K) {
+        gv0 = *reinterpret_cast<const short8v_t*>(p);
+        gv1 = *reinterpret_cast<const short8v_t*>(p + 8);
+      } else {
 #pragma unroll
-          for (int j = 0; j < 16; ++j)
-            if (co0 + e16 + j < cs.K) {
-              if (j < 8)
-                gv0[j] = p[j];
-              else
-                gv1[j - 8] = p[j];
-            }
-        }
+        for (int j = 0; j < 16; ++j)
+          if (co0 + e16g + j < cs.K) {
+            if (j < 8)
+              gv0[j] = p[j];
+            else
+              gv1[j - 8] = p[j];
+          }
       }
-      conv_fwd_gather(xg, cs, npq, rc0 + e16, xv0);
-      conv_fwd_gather(xg, cs, npq, rc0 + e16 + 8, xv1);
+    }
+  };
+  auto load_chunk = [&](int64_t k0) {
+    gv0 = gv1 = xv0 = xv1 = mfma_frag_zero();
+    if constexpr (TM == TN && G_ALL) {
+      // equal tiles, every thread stages both operands from the same npq
+      // row: single guard, merged branch (matches the pre-template
+      // codegen — the split-guard form spilled 53 SGPRs and ran 3x slower
+      // on the fully-occupied shapes)
+      const int64_t npq = k0 + kkg;
+      if (npq < k_end) {
+        load_gy(npq);
+        conv_fwd_gather(xg, cs, npq, rc0 + e16x, xv0);
+        conv_fwd_gather(xg, cs, npq, rc0 + e16x + 8, xv1);
+      }
+    } else {
+      if ((G_ALL || kkg < BK) && k0 + kkg < k_end) load_gy(k0 + kkg);
+      if ((X_ALL || kkx < BK) && k0 + kkx < k_end) {
+        const int64_t npq = k0 + kkx;
+        conv_fwd_gather(xg, cs, npq, rc0 + e16x, xv0);
+        conv_fwd_gather(xg, cs, npq, rc0 + e16x + 8, xv1);
+      }
     }
   };
   auto write_chunk = [&](int buf) {
+    if (G_ALL || kkg < BK) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      Gs[buf][e16 + j][kk] = gv0[j];
-      Gs[buf][e16 + 8 + j][kk] = gv1[j];
-      Xs[buf][e16 + j][kk] = xv0[j];
-      Xs[buf][e16 + 8 + j][kk] = xv1[j];
+      for (int j = 0; j < 8; ++j) {
+        Gs[buf][e16g + j][kkg] = gv0[j];
+        Gs[buf][e16g + 8 + j][kkg] = gv1[j];
+      }
+    }
+    if (X_ALL || kkx < BK) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        Xs[buf][e16x + j][kkx] = xv0[j];
+        Xs[buf][e16x + 8 + j][kkx] = xv1[j];
+      }
     }
   };
 
@@ -410,7 +452,7 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
     if (has_next) load_chunk(k0 + BK);
     const int lane = threadIdx.x % kWave;
     const int fr = lane & 15, kg = (lane >> 4) * 8;
-    short8v_t af[4], bfr[4];
+    short8v_t af[4], bfr[NFR];
 #pragma unroll
     for (int m = 0; m < 4; ++m) {
       const short* p = &Gs[cur][wm + m * 16 + fr][kg];
@@ -418,7 +460,7 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
       for (int j = 0; j < 8; ++j) af[m][j] = p[j];
     }
 #pragma unroll
-    for (int n = 0; n < 4; ++n) {
+    for (int n = 0; n < NFR; ++n) {
       const short* p = &Xs[cur][wn + n * 16 + fr][kg];
 #pragma unroll
       for (int j = 0; j < 8; ++j) bfr[n][j] = p[j];
@@ -426,7 +468,7 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
 #pragma unroll
     for (int m = 0; m < 4; ++m)
 #pragma unroll
-      for (int n = 0; n < 4; ++n)
+      for (int n = 0; n < NFR; ++n)
         acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af[m], bfr[n], acc[m][n], 0, 0, 0);
     if (has_next) write_chunk(cur ^ 1);
@@ -439,7 +481,7 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
 #pragma unroll
   for (int m = 0; m < 4; ++m)
 #pragma unroll
-    for (int n = 0; n < 4; ++n)
+    for (int n = 0; n < NFR; ++n)
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
         const int co = co0 + wm + m * 16 + rg * 4 + j;
@@ -493,39 +535,64 @@ void launch_conv_dgrad(const void* gy, const void* wt, void* gx,
                         G, stream);
 }
 
+// tile selection: small K / small RSC shapes use 64-wide tiles so the
+// padding waste stays bounded (see kernel comment)
+static void wgrad_tiles(ConvShape cs, int64_t RSC, int& tm, int& tn,
+                        int& nbm, int& nbn) {
+  tm = cs.K <= 64 ? 64 : 128;
+  tn = RSC <= 64 ? 64 : 128;
+  nbm = (cs.K + tm - 1) / tm;
+  nbn = static_cast<int>((RSC + tn - 1) / tn);
+}
+
 int conv_wgrad_splitk(ConvShape cs, int G) {
   const int64_t NPQ = static_cast<int64_t>(cs.N) * cs.P * cs.Q;
   const int64_t RSC = static_cast<int64_t>(cs.R) * cs.S * cs.C;
-  const int nbm = (cs.K + 127) / 128;
-  const int nbn = static_cast<int>((RSC + 127) / 128);
+  int tm, tn, nbm, nbn;
+  wgrad_tiles(cs, RSC, tm, tn, nbm, nbn);
   int64_t want = 512 / (static_cast<int64_t>(nbm) * nbn * G);
   if (want < 1) want = 1;
   int64_t cap = (NPQ + BK - 1) / BK;
   return static_cast<int>(want < cap ? want : cap);
 }
 
+static void wgrad_dispatch(const short* gy, const short* x, float* gwf,
+                           ConvShape cs, int64_t NPQ, int64_t RSC, int G,
+                           int splitk, hipStream_t stream) {
+  int tm, tn, nbm, nbn;
+  wgrad_tiles(cs, RSC, tm, tn, nbm, nbn);
+  dim3 grid(nbm * nbn, splitk, G);
+  if (tm == 64 && tn == 64)
+    conv_wgrad_kernel<64, 64><<<grid, kBlockThreads, 0, stream>>>(
+        gy, x, gwf, cs, NPQ, RSC, nbm, nbn, splitk);
+  else if (tm == 64)
+    conv_wgrad_kernel<64, 128><<<grid, kBlockThreads, 0, stream>>>(
+        gy, x, gwf, cs, NPQ, RSC, nbm, nbn, splitk);
+  else if (tn == 64)
+    conv_wgrad_kernel<128, 64><<<grid, kBlockThreads, 0, stream>>>(
+        gy, x, gwf, cs, NPQ, RSC, nbm, nbn, splitk);
+  else
+    conv_wgrad_kernel<128, 128><<<grid, kBlockThreads, 0, stream>>>(
+        gy, x, gwf, cs, NPQ, RSC, nbm, nbn, splitk);
+}
+
 void launch_conv_wgrad_nosum(const void* gy, const void* x, float* gwf,
                              ConvShape cs, int G, hipStream_t stream) {
   const int64_t NPQ = static_cast<int64_t>(cs.N) * cs.P * cs.Q;
   const int64_t RSC = static_cast<int64_t>(cs.R) * cs.S * cs.C;
-  const int nbm = (cs.K + 127) / 128;
-  const int nbn = static_cast<int>((RSC + 127) / 128);
-  conv_wgrad_kernel<<<dim3(nbm * nbn, 1, G), kBlockThreads, 0, stream>>>(
-      static_cast<const short*>(gy), static_cast<const short*>(x), gwf, cs,
-      NPQ, RSC, nbm, nbn, 1);
+  wgrad_dispatch(static_cast<const short*>(gy),
+                 static_cast<const short*>(x), gwf, cs, NPQ, RSC, G, 1,
+                 stream);
 }
 
 void launch_conv_wgrad(const void* gy, const void* x, float* gwf, void* gw,
                        ConvShape cs, int G, hipStream_t stream) {
   const int64_t NPQ = static_cast<int64_t>(cs.N) * cs.P * cs.Q;
   const int64_t RSC = static_cast<int64_t>(cs.R) * cs.S * cs.C;
-  const int nbm = (cs.K + 127) / 128;
-  const int nbn = static_cast<int>((RSC + 127) / 128);
   const int splitk = conv_wgrad_splitk(cs, G);
-  conv_wgrad_kernel<<<dim3(nbm * nbn, splitk, G), kBlockThreads, 0,
-                      stream>>>(static_cast<const short*>(gy),
-                                static_cast<const short*>(x), gwf, cs, NPQ,
-                                RSC, nbm, nbn, splitk);
+  wgrad_dispatch(static_cast<const short*>(gy),
+                 static_cast<const short*>(x), gwf, cs, NPQ, RSC, G,
+                 splitk, stream);
   const int64_t n_per_g = static_cast<int64_t>(cs.K) * RSC;
   int cgrid_x = elemwise_grid(n_per_g, 256, G > 0 ? 2048 / G : 2048);
   if (cgrid_x < 1) cgrid_x = 1;
