@@ -60,6 +60,8 @@ class GraphedStepRunner:
                         else model.loss(self.static_x, self.static_y))
             (loss.sum() if loss.dim() else loss).backward()
             optimizer.device_step()
+            if hasattr(flat, "sync_shadow"):
+                flat.sync_shadow()  # refresh bf16 shadow inside the graph
             return loss
 
         model.train()

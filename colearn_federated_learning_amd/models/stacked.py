@@ -88,6 +88,33 @@ class _SConv2dFn(torch.autograd.Function):
         return gx, gw, None, None, None
 
 
+class _SConv2dShadowFn(torch.autograd.Function):
+    """Shadow-weight conv: compute reads the persistent bf16 shadow view
+    (one fused flat cast per step instead of a per-layer autocast cast),
+    and the weight gradient is emitted in fp32 straight into the autograd
+    slot of the fp32 master parameter (no bf16 grad round trip)."""
+
+    @staticmethod
+    def forward(x, w, shadow, stride: int, padding: int, G: int):
+        return require_ext().conv2d_fwd_km(x, shadow, stride, padding)
+
+    @staticmethod
+    def setup_context(ctx, inputs, output):
+        x, w, shadow, stride, padding, G = inputs
+        ctx.save_for_backward(x, shadow)
+        ctx.stride, ctx.padding, ctx.G = stride, padding, G
+
+    @staticmethod
+    def backward(ctx, gy):
+        x, shadow = ctx.saved_tensors
+        s, p, G = ctx.stride, ctx.padding, ctx.G
+        m = require_ext()
+        gx = m.conv2d_dgrad_km(gy, shadow, x.shape[2], x.shape[3], s, p)
+        gw = m.conv2d_wgrad_km32(gy, x, shadow.shape[2], shadow.shape[3],
+                                 s, p, G)
+        return gx, gw, None, None, None, None
+
+
 class SConv2d(nn.Module):
     def __init__(self, G: int, cin: int, cout: int, k: int,
                  stride: int = 1, padding: int = 0) -> None:
@@ -95,9 +122,15 @@ class SConv2d(nn.Module):
         w = torch.empty(G, cout, cin, k, k)
         nn.init.kaiming_normal_(w, mode="fan_out", nonlinearity="relu")
         self.weight = nn.Parameter(w.permute(0, 1, 3, 4, 2).contiguous())
+        self.wshadow = None  # bf16 flat-buffer view (StackedFlatParams)
         self.G, self.stride, self.padding = G, stride, padding
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if self.wshadow is not None and x.is_cuda and native_enabled():
+            if x.dtype != torch.bfloat16:
+                x = x.bfloat16()
+            return _SConv2dShadowFn.apply(x, self.weight, self.wshadow,
+                                          self.stride, self.padding, self.G)
         w = self.weight
         if torch.is_autocast_enabled("cuda") and x.is_cuda:
             dt = torch.get_autocast_dtype("cuda")

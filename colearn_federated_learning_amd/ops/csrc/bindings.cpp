@@ -509,6 +509,44 @@ torch::Tensor conv2d_wgrad_km(torch::Tensor gy, torch::Tensor x, int64_t R,
   return G == 1 ? gw.squeeze(0) : gw;
 }
 
+// fp32-grad variant for the shadow-weight path: grads are produced in the
+// flat fp32 layout directly (no bf16 round trip on the weight gradient)
+torch::Tensor conv2d_wgrad_km32(torch::Tensor gy, torch::Tensor x,
+                                int64_t R, int64_t S, int64_t stride,
+                                int64_t pad, int64_t G) {
+  check_bf16(gy, "gy");
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(gy.device());
+  const int64_t ci = x.size(1);
+  if (ci < 8) x = pad_x_c8(x.contiguous(torch::MemoryFormat::ChannelsLast));
+  ConvShape cs;
+  cs.N = static_cast<int>(x.size(0) / G);
+  cs.C = static_cast<int>(x.size(1));
+  cs.H = static_cast<int>(x.size(2));
+  cs.W = static_cast<int>(x.size(3));
+  cs.K = static_cast<int>(gy.size(1));
+  cs.P = static_cast<int>(gy.size(2));
+  cs.Q = static_cast<int>(gy.size(3));
+  cs.R = static_cast<int>(R);
+  cs.S = static_cast<int>(S);
+  cs.stride = static_cast<int>(stride);
+  cs.pad = static_cast<int>(pad);
+  conv_shape_fill_magic(cs);
+  auto gc = gy.contiguous(torch::MemoryFormat::ChannelsLast);
+  auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
+  const int64_t rsc = static_cast<int64_t>(cs.R) * cs.S * cs.C;
+  const int splitk = conv_wgrad_splitk(cs, static_cast<int>(G));
+  auto gwf = torch::empty({G * splitk, cs.K, rsc},
+                          x.options().dtype(torch::kFloat32));
+  auto gw = torch::empty({G, cs.K, cs.R, cs.S, cs.C},
+                         x.options().dtype(torch::kFloat32));
+  launch_conv_wgrad_out(gc.data_ptr(), xc.data_ptr(), gwf.data_ptr<float>(),
+                        gw.data_ptr(), cs, static_cast<int>(G), true,
+                        stream());
+  HIP_CHECK_LAST();
+  if (ci < cs.C) gw = gw.narrow(-1, 0, ci).contiguous();
+  return G == 1 ? gw.squeeze(0) : gw;
+}
+
 torch::Tensor conv2d_dgrad(torch::Tensor gy, torch::Tensor w, int64_t H,
                            int64_t W, int64_t stride, int64_t pad) {
   check_bf16(gy, "gy");
@@ -675,6 +713,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_fwd_km", &conv2d_fwd_km);
   m.def("conv2d_dgrad_km", &conv2d_dgrad_km);
   m.def("conv2d_wgrad_km", &conv2d_wgrad_km);
+  m.def("conv2d_wgrad_km32", &conv2d_wgrad_km32);
   m.def("conv2d_dgrad", &conv2d_dgrad);
   m.def("conv2d_wgrad", &conv2d_wgrad);
   m.def("conv2d_wgrad_fp32", &conv2d_wgrad_fp32);

@@ -492,19 +492,25 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
       }
 }
 
-// sum the split-K slabs of each client in slice order and emit bf16.
+// sum the split-K slabs of each client in slice order; OUT is bf16 for
+// the autocast grad path or float for the shadow-weight path (grads land
+// directly in the fp32 flat grad layout, no bf16 round trip).
+template <typename OUT>
 __global__ void wgrad_combine_kernel(const float* __restrict__ gwf,
-                                     bf16* __restrict__ gw, int64_t n_per_g,
+                                     OUT* __restrict__ gw, int64_t n_per_g,
                                      int splitk) {
   const int g = blockIdx.y;
   const float* base = gwf + static_cast<int64_t>(g) * splitk * n_per_g;
-  bf16* out = gw + static_cast<int64_t>(g) * n_per_g;
+  OUT* out = gw + static_cast<int64_t>(g) * n_per_g;
   int64_t i = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
   const int64_t stride = static_cast<int64_t>(blockDim.x) * gridDim.x;
   for (; i < n_per_g; i += stride) {
     float acc = 0.f;
     for (int sk = 0; sk < splitk; ++sk) acc += base[sk * n_per_g + i];
-    out[i] = f2b(acc);
+    if constexpr (sizeof(OUT) == 2)
+      out[i] = f2b(acc);
+    else
+      out[i] = acc;
   }
 }
 
@@ -588,6 +594,12 @@ void launch_conv_wgrad_nosum(const void* gy, const void* x, float* gwf,
 
 void launch_conv_wgrad(const void* gy, const void* x, float* gwf, void* gw,
                        ConvShape cs, int G, hipStream_t stream) {
+  launch_conv_wgrad_out(gy, x, gwf, gw, cs, G, false, stream);
+}
+
+void launch_conv_wgrad_out(const void* gy, const void* x, float* gwf,
+                           void* gw, ConvShape cs, int G, bool f32_out,
+                           hipStream_t stream) {
   const int64_t NPQ = static_cast<int64_t>(cs.N) * cs.P * cs.Q;
   const int64_t RSC = static_cast<int64_t>(cs.R) * cs.S * cs.C;
   const int splitk = conv_wgrad_splitk(cs, G);
@@ -597,8 +609,12 @@ void launch_conv_wgrad(const void* gy, const void* x, float* gwf, void* gw,
   const int64_t n_per_g = static_cast<int64_t>(cs.K) * RSC;
   int cgrid_x = elemwise_grid(n_per_g, 256, G > 0 ? 2048 / G : 2048);
   if (cgrid_x < 1) cgrid_x = 1;
- hipLaunchKernelGGL(( wgrad_combine_kernel), dim3(dim3(cgrid_x, G)), dim3(256), 0, stream, 
-      gwf, static_cast<bf16*>(gw), n_per_g, splitk);
+  if (f32_out)
+   hipLaunchKernelGGL(( wgrad_combine_kernel), dim3(dim3(cgrid_x, G)), dim3(256), 0, stream, 
+        gwf, static_cast<float*>(gw), n_per_g, splitk);
+  else
+   hipLaunchKernelGGL(( wgrad_combine_kernel), dim3(dim3(cgrid_x, G)), dim3(256), 0, stream, 
+        gwf, static_cast<bf16*>(gw), n_per_g, splitk);
 }
 
 // ==== wgrad layout variants (tuning bench only; prod kernel above) =======
