@@ -11,6 +11,8 @@
 // LDS round-trip for the row stats.
 
 #include "api.h"
+
+#include <cstdlib>
 #include "common.h"
 
 namespace colearn {
@@ -542,7 +544,9 @@ void launch_batchnorm_fwd(const void* x_, const void* res_, const float* w,
   const bool fixed = (256 * 8) % C == 0;
   const size_t lds =
       (fixed ? 256 * 16 : 2 * C) * sizeof(float);
-  const int grid = elemwise_grid(R * C / 8, 256, 512);
+  const char* env = getenv("COLEARN_BN_GX_GRID");
+  const int gx_cap = env ? atoi(env) : 512;
+  const int grid = elemwise_grid(R * C / 8, 256, gx_cap);
   const int rgrid = norms_reduce_grid(R * C / 8);
   if (training) {
     // sums layout: [G][rgrid][2C] partials + [G][2C] combined at the end
@@ -594,7 +598,9 @@ void launch_batchnorm_bwd(const void* gy_, const void* y_, const void* x_,
   const bool fixed = (256 * 8) % C == 0;
   const size_t lds =
       (fixed ? 256 * 16 : 2 * C) * sizeof(float);
-  const int grid = elemwise_grid(R * C / 8, 256, 512);
+  const char* env = getenv("COLEARN_BN_GX_GRID");
+  const int gx_cap = env ? atoi(env) : 512;
+  const int grid = elemwise_grid(R * C / 8, 256, gx_cap);
   const int rgrid = norms_reduce_grid(R * C / 8);
   float* partials = acc;
   float* combined = acc + static_cast<int64_t>(G) * rgrid * 2 * C;
