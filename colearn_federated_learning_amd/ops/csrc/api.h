@@ -93,10 +93,10 @@ void launch_batchnorm_fwd(const void* x, const void* res, const float* w,
                           float* invstd, float* sums, int64_t R, int C,
                           int G, float momentum, float eps, bool training,
                           bool relu, hipStream_t stream);
-void launch_batchnorm_bwd(const void* gy, const void* y, const void* x,
+void launch_batchnorm_bwd(const void* gy_, const void* y_, const void* x_,
                           const float* w, const float* mean,
-                          const float* invstd, void* gx, void* gym,
-                          float* acc, int64_t R, int C, int G,
+                          const float* invstd, void* gx_, void* gym_,
+                          float* acc, float* gwgb, int64_t R, int C, int G,
                           bool training, bool relu, hipStream_t stream);
 void launch_gemm_nt(const void* A, const void* B, void* C, int64_t batch,
                     int64_t M, int64_t N, int64_t K, hipStream_t stream);

@@ -239,21 +239,20 @@ batchnorm_bwd(torch::Tensor gy, torch::Tensor y, torch::Tensor x,
   torch::Tensor gym;  // masked gy == the residual grad
   if (relu) gym = torch::empty_like(gc);
   const int rgrid = norms_reduce_grid(R * C / 8);
-  auto acc = torch::empty({G * static_cast<int64_t>(rgrid + 1) * 2 * C},
+  auto acc = torch::empty({G * static_cast<int64_t>(rgrid) * 2 * C},
                           x.options().dtype(torch::kFloat32));
+  // plane layout [2][G*C]: row 0 = gb, row 1 = gw — contiguous views,
+  // no clone kernels (see partial_combine_planes_kernel)
+  auto gwgb = torch::empty({2, G * C},
+                           x.options().dtype(torch::kFloat32));
   launch_batchnorm_bwd(gc.data_ptr(), relu ? yc.data_ptr() : nullptr,
                        xc.data_ptr(), w.data_ptr<float>(),
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
                        gx.data_ptr(), relu ? gym.data_ptr() : nullptr,
-                       acc.data_ptr<float>(), R, C, static_cast<int>(G),
-                       training, relu, stream());
+                       acc.data_ptr<float>(), gwgb.data_ptr<float>(), R, C,
+                       static_cast<int>(G), training, relu, stream());
   HIP_CHECK_LAST();
-  const int64_t off = static_cast<int64_t>(G) * rgrid * 2 * C;
-  // combined region is [G][2C]: gb = even C-halves, gw = odd C-halves
-  auto comb = acc.narrow(0, off, G * 2 * C).view({G, 2, C});
-  auto gb = comb.select(1, 0).reshape({G * C}).clone();
-  auto gw = comb.select(1, 1).reshape({G * C}).clone();
-  return {gx, gw, gb, relu ? gym : gc};
+  return {gx, gwgb.select(0, 1), gwgb.select(0, 0), relu ? gym : gc};
 }
 
 // ---- GEMM family ----------------------------------------------------------

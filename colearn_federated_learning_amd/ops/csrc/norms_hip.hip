@@ -231,6 +231,31 @@ __global__ void partial_combine_kernel(const float* __restrict__ partials,
   if (lane == 0) out[i] = acc;
 }
 
+// BN-bwd variant: writes the combined sums into PLANE layout
+// [2][G][C] (plane 0 = sum(gym) = gb, plane 1 = sum(gym*xhat) = gw) so
+// the binding can return gb/gw as contiguous views — this removes two
+// clone kernels per BN backward (40 extra ~4.5us launches per ResNet-18
+// step; small-kernel latency was ~20% of step time).
+__global__ void partial_combine_planes_kernel(
+    const float* __restrict__ partials, float* __restrict__ out, int C,
+    int nblk) {
+  const int width = 2 * C;
+  partials += static_cast<int64_t>(blockIdx.y) * nblk * width;
+  const int lane = threadIdx.x % kWave;
+  const int i = blockIdx.x * (blockDim.x / kWave) + threadIdx.x / kWave;
+  if (i >= width) return;
+  float acc = 0.f;
+  for (int b = lane; b < nblk; b += kWave)
+    acc += partials[static_cast<int64_t>(b) * width + i];
+  acc = wave_allreduce_sum(acc);
+  if (lane == 0) {
+    const int plane = i < C ? 0 : 1;
+    const int c = i < C ? i : i - C;
+    out[(static_cast<int64_t>(plane) * gridDim.y + blockIdx.y) * C + c] =
+        acc;
+  }
+}
+
 // Pass 2: finalize stats + update running stats over all G*C channel
 // instances.  The combined sums are laid out [G][2C]: sum at g*2C + c,
 // sumsq at g*2C + C + c.
@@ -430,7 +455,7 @@ __global__ void bn_bwd_gx_kernel(const bf16* __restrict__ gym,
                                  const float* __restrict__ w,
                                  const float* __restrict__ acc /*[2*C]*/,
                                  bf16* __restrict__ gx, int64_t R, int C,
-                                 bool training) {
+                                 int64_t plane, bool training) {
   const int64_t n = R * C;
   const int64_t goff = static_cast<int64_t>(blockIdx.z) * n;
   gym += goff;
@@ -439,7 +464,7 @@ __global__ void bn_bwd_gx_kernel(const bf16* __restrict__ gym,
   mean += blockIdx.z * C;
   invstd += blockIdx.z * C;
   w += blockIdx.z * C;
-  acc += static_cast<int64_t>(blockIdx.z) * 2 * C;
+  acc += static_cast<int64_t>(blockIdx.z) * C;  // plane layout [2][G][C]
   const float invn = 1.f / R;
   int64_t i = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) * 8;
   const int64_t stride = grid_stride() * 8;
@@ -454,7 +479,7 @@ __global__ void bn_bwd_gx_kernel(const bf16* __restrict__ gym,
       const float wi = w[c] * invstd[c];
       a[j] = wi;
       if (training) {
-        const float gwn = acc[C + c] * invn * invstd[c];
+        const float gwn = acc[plane + c] * invn * invstd[c];
         bco[j] = -wi * (acc[c] * invn - gwn * mean[c]);
         d[j] = -wi * gwn;
       } else {
@@ -491,7 +516,7 @@ __global__ void bn_bwd_gx_kernel(const bf16* __restrict__ gym,
       if (training) {
         const float xhat = (bits2f(v.x[j]) - mean[c]) * invstd[c];
         r = w[c] * invstd[c] *
-            (gf - acc[c] * invn - xhat * acc[C + c] * invn);
+            (gf - acc[c] * invn - xhat * acc[plane + c] * invn);
       } else {
         r = w[c] * invstd[c] * gf;
       }
@@ -589,7 +614,8 @@ void launch_batchnorm_fwd(const void* x_, const void* res_, const float* w,
 void launch_batchnorm_bwd(const void* gy_, const void* y_, const void* x_,
                           const float* w, const float* mean,
                           const float* invstd, void* gx_, void* gym_,
-                          float* acc, int64_t R, int C, int G,
+                          float* acc, float* gwgb /* [2][G*C] planes */,
+                          int64_t R, int C, int G,
                           bool training, bool relu, hipStream_t stream) {
   const bf16* gy = static_cast<const bf16*>(gy_);
   const bf16* y = static_cast<const bf16*>(y_);
@@ -604,7 +630,6 @@ void launch_batchnorm_bwd(const void* gy_, const void* y_, const void* x_,
   const int grid = elemwise_grid(R * C / 8, 256, gx_cap);
   const int rgrid = norms_reduce_grid(R * C / 8);
   float* partials = acc;
-  float* combined = acc + static_cast<int64_t>(G) * rgrid * 2 * C;
   acc = partials;
   dim3 rg(rgrid, 1, G);
   if (relu) {
@@ -622,11 +647,12 @@ void launch_batchnorm_bwd(const void* gy_, const void* y_, const void* x_,
      hipLaunchKernelGGL(( bn_bwd_reduce_kernel<false, false>), dim3(rg), dim3(256), lds, stream, 
           gy, y, x, mean, invstd, R, C, acc, nullptr);
   }
- hipLaunchKernelGGL(( partial_combine_kernel), dim3(dim3((2 * C + 3) / 4, G)), dim3(256), 0, stream, 
-      partials, combined, 2 * C, rgrid);
+ hipLaunchKernelGGL(( partial_combine_planes_kernel), dim3(dim3((2 * C + 3) / 4, G)), dim3(256), 0,
+                                  stream, partials, gwgb, C, rgrid);
   const bf16* gym_in = relu ? gym : gy;
  hipLaunchKernelGGL(( bn_bwd_gx_kernel), dim3(dim3(grid, 1, G)), dim3(256), 0, stream, 
-      gym_in, x, mean, invstd, w, combined, gx, R, C, training);
+      gym_in, x, mean, invstd, w, gwgb, gx, R, C,
+      static_cast<int64_t>(G) * C, training);
 }
 
 }  // namespace colearn

@@ -69,6 +69,22 @@ def main() -> None:
                                 f"{name[:60]}\n")
                 except Exception as e:
                     f.write(f"  pmc join failed: {e}\n")
+            # optional: dump dispatch order (argv[3] = count, argv[4] = skip)
+            if len(sys.argv) > 3:
+                cnt = int(sys.argv[3])
+                skip = int(sys.argv[4]) if len(sys.argv) > 4 else 0
+                q3 = (
+                    f"SELECT s.display_name, d.grid_size_x, d.grid_size_y,"
+                    f" d.grid_size_z, (d.end-d.start)/1e3"
+                    f" FROM rocpd_kernel_dispatch_{u} d"
+                    f" JOIN rocpd_info_kernel_symbol_{u} s"
+                    f"   ON d.kernel_id=s.id"
+                    f" ORDER BY d.start LIMIT {cnt} OFFSET {skip}"
+                )
+                f.write("== dispatch order ==\n")
+                for name, gx, gy, gz, us in cur.execute(q3):
+                    f.write(f"{us:8.1f}us ({gx:>9},{gy:>3},{gz:>3}) "
+                            f"{name[:80]}\n")
 
 
 if __name__ == "__main__":
