@@ -73,18 +73,23 @@ class StackedFlatParams:
         the optimizer) replaces the per-layer autocast cast kernels that
         were ~5% of flagship step time.
         """
-        from ..models.stacked import SConv2d
+        from ..models.stacked import SBatchNorm2d, SConv2d
 
         self.flat_bf16 = torch.empty_like(self.flat_w,
                                           dtype=torch.bfloat16)
         by_name = dict(zip(self.names,
                            zip(self._offsets, self.t_numels)))
         for mod_name, mod in model.named_modules():
-            if not isinstance(mod, SConv2d):
-                continue
-            off, n = by_name[f"{mod_name}.weight" if mod_name else "weight"]
-            mod.wshadow = (self.flat_bf16[off : off + self.G * n]
-                           .view(mod.weight.shape))
+            if isinstance(mod, SConv2d):
+                off, n = by_name[f"{mod_name}.weight" if mod_name
+                                 else "weight"]
+                mod.wshadow = (self.flat_bf16[off : off + self.G * n]
+                               .view(mod.weight.shape))
+                # direct wgrad target: the param's flat-grad view
+                mod.wgslot = mod.weight.grad
+            elif isinstance(mod, SBatchNorm2d):
+                mod.gslot_w = mod.weight.grad.view(-1)
+                mod.gslot_b = mod.bias.grad.view(-1)
         self.sync_shadow()
 
     @torch.no_grad()

@@ -95,13 +95,14 @@ class _SConv2dShadowFn(torch.autograd.Function):
     slot of the fp32 master parameter (no bf16 grad round trip)."""
 
     @staticmethod
-    def forward(x, w, shadow, stride: int, padding: int, G: int):
+    def forward(x, w, shadow, gslot, stride: int, padding: int, G: int):
         return require_ext().conv2d_fwd_km(x, shadow, stride, padding)
 
     @staticmethod
     def setup_context(ctx, inputs, output):
-        x, w, shadow, stride, padding, G = inputs
+        x, w, shadow, gslot, stride, padding, G = inputs
         ctx.save_for_backward(x, shadow)
+        ctx.gslot = gslot
         ctx.stride, ctx.padding, ctx.G = stride, padding, G
 
     @staticmethod
@@ -110,9 +111,17 @@ class _SConv2dShadowFn(torch.autograd.Function):
         s, p, G = ctx.stride, ctx.padding, ctx.G
         m = require_ext()
         gx = m.conv2d_dgrad_km(gy, shadow, x.shape[2], x.shape[3], s, p)
-        gw = m.conv2d_wgrad_km32(gy, x, shadow.shape[2], shadow.shape[3],
-                                 s, p, G)
-        return gx, gw, None, None, None, None
+        if ctx.gslot is not None:
+            # write the fp32 grad straight into the flat-grad view; no
+            # AccumulateGrad add kernel runs for this parameter
+            m.conv2d_wgrad_km32(gy, x, shadow.shape[2], shadow.shape[3],
+                                s, p, G, ctx.gslot)
+            gw = None
+        else:
+            e = torch.empty(0, device=x.device, dtype=torch.float32)
+            gw = m.conv2d_wgrad_km32(gy, x, shadow.shape[2],
+                                     shadow.shape[3], s, p, G, e)
+        return gx, gw, None, None, None, None, None
 
 
 class SConv2d(nn.Module):
@@ -123,6 +132,7 @@ class SConv2d(nn.Module):
         nn.init.kaiming_normal_(w, mode="fan_out", nonlinearity="relu")
         self.weight = nn.Parameter(w.permute(0, 1, 3, 4, 2).contiguous())
         self.wshadow = None  # bf16 flat-buffer view (StackedFlatParams)
+        self.wgslot = None   # fp32 flat-grad view (direct wgrad target)
         self.G, self.stride, self.padding = G, stride, padding
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
@@ -130,7 +140,8 @@ class SConv2d(nn.Module):
             if x.dtype != torch.bfloat16:
                 x = x.bfloat16()
             return _SConv2dShadowFn.apply(x, self.weight, self.wshadow,
-                                          self.stride, self.padding, self.G)
+                                          self.wgslot, self.stride,
+                                          self.padding, self.G)
         w = self.weight
         if torch.is_autocast_enabled("cuda") and x.is_cuda:
             dt = torch.get_autocast_dtype("cuda")
@@ -152,6 +163,8 @@ class SBatchNorm2d(nn.Module):
         self.register_buffer("running_var", torch.ones(G, num_features))
         self.register_buffer("num_batches_tracked",
                              torch.tensor(0, dtype=torch.long))
+        self.gslot_w = None  # fp32 flat-grad views (direct grad targets)
+        self.gslot_b = None
 
     def forward(self, x: torch.Tensor, residual=None) -> torch.Tensor:
         if self.training:
@@ -162,7 +175,7 @@ class SBatchNorm2d(nn.Module):
                 x, self.weight.view(-1), self.bias.view(-1),
                 self.running_mean.view(-1), self.running_var.view(-1),
                 residual, self.training, self.momentum, self.eps,
-                self.relu, self.G,
+                self.relu, self.G, self.gslot_w, self.gslot_b,
             )
             return y
         # CPU oracle: per-client F.batch_norm

@@ -96,7 +96,8 @@ void launch_batchnorm_fwd(const void* x, const void* res, const float* w,
 void launch_batchnorm_bwd(const void* gy_, const void* y_, const void* x_,
                           const float* w, const float* mean,
                           const float* invstd, void* gx_, void* gym_,
-                          float* acc, float* gwgb, int64_t R, int C, int G,
+                          float* acc, float* gwgb_b, float* gwgb_w,
+                          int64_t R, int C, int G,
                           bool training, bool relu, hipStream_t stream);
 void launch_gemm_nt(const void* A, const void* B, void* C, int64_t batch,
                     int64_t M, int64_t N, int64_t K, hipStream_t stream);

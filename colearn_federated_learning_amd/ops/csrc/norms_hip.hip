@@ -237,8 +237,8 @@ __global__ void partial_combine_kernel(const float* __restrict__ partials,
 // clone kernels per BN backward (40 extra ~4.5us launches per ResNet-18
 // step; small-kernel latency was ~20% of step time).
 __global__ void partial_combine_planes_kernel(
-    const float* __restrict__ partials, float* __restrict__ out, int C,
-    int nblk) {
+    const float* __restrict__ partials, float* __restrict__ out_b,
+    float* __restrict__ out_w, int C, int nblk) {
   const int width = 2 * C;
   partials += static_cast<int64_t>(blockIdx.y) * nblk * width;
   const int lane = threadIdx.x % kWave;
@@ -249,10 +249,9 @@ __global__ void partial_combine_planes_kernel(
     acc += partials[static_cast<int64_t>(b) * width + i];
   acc = wave_allreduce_sum(acc);
   if (lane == 0) {
-    const int plane = i < C ? 0 : 1;
+    float* out = i < C ? out_b : out_w;
     const int c = i < C ? i : i - C;
-    out[(static_cast<int64_t>(plane) * gridDim.y + blockIdx.y) * C + c] =
-        acc;
+    out[static_cast<int64_t>(blockIdx.y) * C + c] = acc;
   }
 }
 
@@ -453,9 +452,10 @@ __global__ void bn_bwd_gx_kernel(const bf16* __restrict__ gym,
                                  const float* __restrict__ mean,
                                  const float* __restrict__ invstd,
                                  const float* __restrict__ w,
-                                 const float* __restrict__ acc /*[2*C]*/,
+                                 const float* __restrict__ acc_b,
+                                 const float* __restrict__ acc_w,
                                  bf16* __restrict__ gx, int64_t R, int C,
-                                 int64_t plane, bool training) {
+                                 bool training) {
   const int64_t n = R * C;
   const int64_t goff = static_cast<int64_t>(blockIdx.z) * n;
   gym += goff;
@@ -464,7 +464,8 @@ __global__ void bn_bwd_gx_kernel(const bf16* __restrict__ gym,
   mean += blockIdx.z * C;
   invstd += blockIdx.z * C;
   w += blockIdx.z * C;
-  acc += static_cast<int64_t>(blockIdx.z) * C;  // plane layout [2][G][C]
+  acc_b += static_cast<int64_t>(blockIdx.z) * C;
+  acc_w += static_cast<int64_t>(blockIdx.z) * C;
   const float invn = 1.f / R;
   int64_t i = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) * 8;
   const int64_t stride = grid_stride() * 8;
@@ -479,8 +480,8 @@ __global__ void bn_bwd_gx_kernel(const bf16* __restrict__ gym,
       const float wi = w[c] * invstd[c];
       a[j] = wi;
       if (training) {
-        const float gwn = acc[plane + c] * invn * invstd[c];
-        bco[j] = -wi * (acc[c] * invn - gwn * mean[c]);
+        const float gwn = acc_w[c] * invn * invstd[c];
+        bco[j] = -wi * (acc_b[c] * invn - gwn * mean[c]);
         d[j] = -wi * gwn;
       } else {
         bco[j] = 0.f;
@@ -516,7 +517,7 @@ __global__ void bn_bwd_gx_kernel(const bf16* __restrict__ gym,
       if (training) {
         const float xhat = (bits2f(v.x[j]) - mean[c]) * invstd[c];
         r = w[c] * invstd[c] *
-            (gf - acc[c] * invn - xhat * acc[plane + c] * invn);
+            (gf - acc_b[c] * invn - xhat * acc_w[c] * invn);
       } else {
         r = w[c] * invstd[c] * gf;
       }
@@ -614,7 +615,7 @@ void launch_batchnorm_fwd(const void* x_, const void* res_, const float* w,
 void launch_batchnorm_bwd(const void* gy_, const void* y_, const void* x_,
                           const float* w, const float* mean,
                           const float* invstd, void* gx_, void* gym_,
-                          float* acc, float* gwgb /* [2][G*C] planes */,
+                          float* acc, float* gwgb_b, float* gwgb_w,
                           int64_t R, int C, int G,
                           bool training, bool relu, hipStream_t stream) {
   const bf16* gy = static_cast<const bf16*>(gy_);
@@ -648,11 +649,11 @@ void launch_batchnorm_bwd(const void* gy_, const void* y_, const void* x_,
           gy, y, x, mean, invstd, R, C, acc, nullptr);
   }
  hipLaunchKernelGGL(( partial_combine_planes_kernel), dim3(dim3((2 * C + 3) / 4, G)), dim3(256), 0,
-                                  stream, partials, gwgb, C, rgrid);
+                                  stream, partials, gwgb_b, gwgb_w, C,
+                                            rgrid);
   const bf16* gym_in = relu ? gym : gy;
  hipLaunchKernelGGL(( bn_bwd_gx_kernel), dim3(dim3(grid, 1, G)), dim3(256), 0, stream, 
-      gym_in, x, mean, invstd, w, gwgb, gx, R, C,
-      static_cast<int64_t>(G) * C, training);
+      gym_in, x, mean, invstd, w, gwgb_b, gwgb_w, gx, R, C, training);
 }
 
 }  // namespace colearn

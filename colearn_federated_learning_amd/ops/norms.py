@@ -31,7 +31,10 @@ class _BatchNormFn(torch.autograd.Function):
     @staticmethod
     def forward(x, weight, bias, running_mean, running_var, residual,
                 training: bool, momentum: float, eps: float, relu: bool,
-                groups: int = 1):
+                groups: int = 1, gslot_w=None, gslot_b=None):
+        # gslot_w/gslot_b: flat-grad views; when set, the backward writes
+        # the param grads straight into them and returns None for w/b, so
+        # no AccumulateGrad add kernels run (stacked engine fast path).
         if _native(x):
             m = require_ext()
             y, save_mean, save_invstd = m.batchnorm_fwd(
@@ -77,16 +80,23 @@ class _BatchNormFn(torch.autograd.Function):
         ctx.training = inputs[6]
         ctx.relu = inputs[9]
         ctx.groups = inputs[10] if len(inputs) > 10 else 1
+        ctx.gslot_w = inputs[11] if len(inputs) > 11 else None
+        ctx.gslot_b = inputs[12] if len(inputs) > 12 else None
+        ctx.n_inputs = len(inputs)
         ctx.has_res = inputs[5] is not None
         ctx.mark_non_differentiable(output[1], output[2])
 
     @staticmethod
     def backward(ctx, gy, _gmean, _ginvstd):
         x, weight, mean, invstd, y = ctx.saved_tensors
+        direct = ctx.gslot_w is not None and _native(x)
         if _native(x):
+            e = torch.empty(0, device=x.device, dtype=torch.float32)
             gx, gw, gb, gres = require_ext().batchnorm_bwd(
                 gy, y, x, weight, mean, invstd, ctx.training, ctx.relu,
                 ctx.groups,
+                ctx.gslot_w if direct else e,
+                ctx.gslot_b if direct else e,
             )
         else:
             gyf = gy.float()
@@ -106,9 +116,10 @@ class _BatchNormFn(torch.autograd.Function):
             else:
                 gx = (weight * invstd)[None, :, None, None] * gyf
             gx = gx.to(x.dtype)
-        return (gx, gw, gb, None, None,
-                gres if ctx.has_res else None, None, None, None, None,
-                None)
+        out = [gx, None if direct else gw, None if direct else gb,
+               None, None, gres if ctx.has_res else None,
+               None, None, None, None, None, None, None]
+        return tuple(out[: ctx.n_inputs])
 
 
 class ColearnBatchNorm2d(nn.Module):
