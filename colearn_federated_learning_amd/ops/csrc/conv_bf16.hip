@@ -150,17 +150,24 @@ DEV_INLINE void conv_dgrad_gather(const short* __restrict__ gy,
 // GATHER: 0 = conv fwd (gather x), 1 = dgrad stride1, 2 = dgrad strided
 // TN: output-channel tile width (64 for the CO=64 layers).
 // BKT: K-step depth; 64 halves the barrier count for long-K deep layers.
-template <int GATHER, int TN, int BKT>
+// TM: M-tile height.  256 is used for the small-N shapes (N <= 64: the
+// K=64 layers' fwd and the C=64 layers' dgrad): a 128x64 tile spends as
+// many gather instructions per block as MFMA issue; doubling M amortizes
+// nothing on the A side (A rows are per-thread) but doubles MFMA per
+// B-stage and per barrier, and halves the block count the B operand is
+// re-read by.
+template <int GATHER, int TM, int TN, int BKT>
 __global__ __launch_bounds__(kBlockThreads) void conv_nt_kernel(
     const short* __restrict__ Asrc, const short* __restrict__ Bw,
     bf16* __restrict__ C, ConvShape cs, int64_t Mc /* rows per client */,
     int64_t Nd, int64_t Kd, int64_t a_stride /* A elems per client */,
     int nbm, int nbn) {
   constexpr int QV = BKT / 32;  // 16-element load pairs per thread-row
-  __shared__ short lds_a[2][BM * BKT];
+  constexpr int RPT = TM / 128;  // A staging rows per thread
+  __shared__ short lds_a[2][TM * BKT];
   __shared__ short lds_b[2][TN * BKT];
   const int bid = xcd_swizzle(blockIdx.x, nbm * nbn);
-  const int64_t bm_local = static_cast<int64_t>(bid / nbn) * BM;
+  const int64_t bm_local = static_cast<int64_t>(bid / nbn) * TM;
   const int64_t bn = static_cast<int64_t>(bid % nbn) * TN;
   const int g = blockIdx.z;  // stacked client index
   const short* A = Asrc + static_cast<int64_t>(g) * a_stride;
@@ -168,10 +175,11 @@ __global__ __launch_bounds__(kBlockThreads) void conv_nt_kernel(
   bf16* Cg = C + static_cast<int64_t>(g) * Mc * Nd;
 
   const int wid = threadIdx.x / kWave;
-  constexpr int MF = TN == 128 ? 4 : 2;
+  constexpr int MF = TM == 256 ? 4 : (TN == 128 ? 4 : 2);
   constexpr int NF = 4;
-  const int wm = TN == 128 ? (wid >> 1) * 64 : wid * 32;
-  const int wn = TN == 128 ? (wid & 1) * 64 : 0;
+  const int wm = TM == 256 ? wid * 64
+                           : (TN == 128 ? (wid >> 1) * 64 : wid * 32);
+  const int wn = (TM != 256 && TN == 128) ? (wid & 1) * 64 : 0;
 
   f32x4_t acc[MF][NF];
 #pragma unroll
@@ -183,39 +191,47 @@ __global__ __launch_bounds__(kBlockThreads) void conv_nt_kernel(
   const int half = threadIdx.x & 1;
   const int ntiles = static_cast<int>((Kd + BKT - 1) / BKT);
 
-  // this thread's output pixel is K-loop-invariant: decompose ONCE
-  const int64_t am = bm_local + row;
-  int dn, dp_, dq;  // (n, p, q) for fwd / (n, h, w) for dgrad
-  {
+  // this thread's output pixels are K-loop-invariant: decompose ONCE
+  int dn[RPT], dp_[RPT], dq[RPT];
+#pragma unroll
+  for (int rr = 0; rr < RPT; ++rr) {
+    const int64_t am = bm_local + row + rr * 128;
     const int inner = GATHER == 0 ? cs.Q : cs.W;
     const int mid = GATHER == 0 ? cs.P : cs.H;
-    dq = static_cast<int>(am % inner);
-    dp_ = static_cast<int>((am / inner) % mid);
-    dn = static_cast<int>(am / (static_cast<int64_t>(mid) * inner));
+    dq[rr] = static_cast<int>(am % inner);
+    dp_[rr] = static_cast<int>((am / inner) % mid);
+    dn[rr] = static_cast<int>(am / (static_cast<int64_t>(mid) * inner));
   }
 
-  short8v_t a0[2 * QV], b0[2 * QV];
+  short8v_t a0[RPT][2 * QV], b0[2 * QV];
   auto load_tile = [&](int t) {
     const int64_t kbase =
         static_cast<int64_t>(t) * BKT + half * (BKT / 2);
 #pragma unroll
     for (int q = 0; q < QV; ++q) {
       const int64_t k = kbase + q * 16;
-      if (am < Mc) {
-        if constexpr (GATHER == 0) {
-          conv_fwd_gather_at(A, cs, dn, dp_, dq, k, a0[2 * q]);
-          conv_fwd_gather_at(A, cs, dn, dp_, dq, k + 8, a0[2 * q + 1]);
-        } else if constexpr (GATHER == 1) {
-          conv_dgrad_gather_at<true>(A, cs, dn, dp_, dq, k, a0[2 * q]);
-          conv_dgrad_gather_at<true>(A, cs, dn, dp_, dq, k + 8,
-                                     a0[2 * q + 1]);
+#pragma unroll
+      for (int rr = 0; rr < RPT; ++rr) {
+        if (bm_local + row + rr * 128 < Mc) {
+          if constexpr (GATHER == 0) {
+            conv_fwd_gather_at(A, cs, dn[rr], dp_[rr], dq[rr], k,
+                               a0[rr][2 * q]);
+            conv_fwd_gather_at(A, cs, dn[rr], dp_[rr], dq[rr], k + 8,
+                               a0[rr][2 * q + 1]);
+          } else if constexpr (GATHER == 1) {
+            conv_dgrad_gather_at<true>(A, cs, dn[rr], dp_[rr], dq[rr], k,
+                                       a0[rr][2 * q]);
+            conv_dgrad_gather_at<true>(A, cs, dn[rr], dp_[rr], dq[rr],
+                                       k + 8, a0[rr][2 * q + 1]);
+          } else {
+            conv_dgrad_gather_at<false>(A, cs, dn[rr], dp_[rr], dq[rr], k,
+                                        a0[rr][2 * q]);
+            conv_dgrad_gather_at<false>(A, cs, dn[rr], dp_[rr], dq[rr],
+                                        k + 8, a0[rr][2 * q + 1]);
+          }
         } else {
-          conv_dgrad_gather_at<false>(A, cs, dn, dp_, dq, k, a0[2 * q]);
-          conv_dgrad_gather_at<false>(A, cs, dn, dp_, dq, k + 8,
-                                      a0[2 * q + 1]);
+          a0[rr][2 * q] = a0[rr][2 * q + 1] = mfma_frag_zero();
         }
-      } else {
-        a0[2 * q] = a0[2 * q + 1] = mfma_frag_zero();
       }
       b0[2 * q] = b0[2 * q + 1] = mfma_frag_zero();
       const int64_t bn_row = bn + row;
@@ -243,8 +259,10 @@ __global__ __launch_bounds__(kBlockThreads) void conv_nt_kernel(
     const int s0 = half * (2 * QV);
 #pragma unroll
     for (int q = 0; q < 2 * QV; ++q) {
-      *reinterpret_cast<short8v_t*>(
-          As + lds_slot_byte_t<BKT>(row, s0 + q)) = a0[q];
+#pragma unroll
+      for (int rr = 0; rr < RPT; ++rr)
+        *reinterpret_cast<short8v_t*>(
+            As + lds_slot_byte_t<BKT>(row + rr * 128, s0 + q)) = a0[rr][q];
       if (row < TN)
         *reinterpret_cast<short8v_t*>(
             Bs + lds_slot_byte_t<BKT>(row, s0 + q)) = b0[q];
@@ -305,25 +323,44 @@ static void conv_nt_dispatch(const short* A, const short* B, bf16* C,
   // (fewer barriers per MFMA), 32 wins for shallow-K stem shapes
   const int bkt = env ? atoi(env) : (Kd >= 1024 ? 64 : 32);
   if (Nd <= 64) {
+    // small-N shapes: a 256x64 tile variant exists (see kernel comment)
+    // but measured ~2% SLOWER end-to-end than 128x64 (lower occupancy
+    // outweighs the amortized B-staging) — off unless COLEARN_CONV_TM256
+    const char* envtm = getenv("COLEARN_CONV_TM256");
     const int nbn = (static_cast<int>(Nd) + 63) / 64;
-    const int nbm = static_cast<int>((Mc + BM - 1) / BM);
+    const int tm =
+        (envtm && atoi(envtm) && Mc >= 256 * 256) ? 256 : 128;
+    const int nbm = static_cast<int>((Mc + tm - 1) / tm);
     dim3 grid(nbm * nbn, 1, G);
-    if (bkt == 64)
-      conv_nt_kernel<GATHER, 64, 64><<<grid, kBlockThreads, 0, stream>>>(
-          A, B, C, cs, Mc, Nd, Kd, a_stride, nbm, nbn);
+    if (tm == 256) {
+      if (bkt == 64)
+        conv_nt_kernel<GATHER, 256, 64, 64>
+            <<<grid, kBlockThreads, 0, stream>>>(A, B, C, cs, Mc, Nd, Kd,
+                                                 a_stride, nbm, nbn);
+      else
+        conv_nt_kernel<GATHER, 256, 64, 32>
+            <<<grid, kBlockThreads, 0, stream>>>(A, B, C, cs, Mc, Nd, Kd,
+                                                 a_stride, nbm, nbn);
+    } else if (bkt == 64)
+      conv_nt_kernel<GATHER, 128, 64, 64>
+          <<<grid, kBlockThreads, 0, stream>>>(A, B, C, cs, Mc, Nd, Kd,
+                                               a_stride, nbm, nbn);
     else
-      conv_nt_kernel<GATHER, 64, 32><<<grid, kBlockThreads, 0, stream>>>(
-          A, B, C, cs, Mc, Nd, Kd, a_stride, nbm, nbn);
+      conv_nt_kernel<GATHER, 128, 64, 32>
+          <<<grid, kBlockThreads, 0, stream>>>(A, B, C, cs, Mc, Nd, Kd,
+                                               a_stride, nbm, nbn);
   } else {
     const int nbn = (static_cast<int>(Nd) + BN - 1) / BN;
     const int nbm = static_cast<int>((Mc + BM - 1) / BM);
     dim3 grid(nbm * nbn, 1, G);
     if (bkt == 64)
-      conv_nt_kernel<GATHER, 128, 64><<<grid, kBlockThreads, 0, stream>>>(
-          A, B, C, cs, Mc, Nd, Kd, a_stride, nbm, nbn);
+      conv_nt_kernel<GATHER, 128, 128, 64>
+          <<<grid, kBlockThreads, 0, stream>>>(A, B, C, cs, Mc, Nd, Kd,
+                                               a_stride, nbm, nbn);
     else
-      conv_nt_kernel<GATHER, 128, 32><<<grid, kBlockThreads, 0, stream>>>(
-          A, B, C, cs, Mc, Nd, Kd, a_stride, nbm, nbn);
+      conv_nt_kernel<GATHER, 128, 128, 32>
+          <<<grid, kBlockThreads, 0, stream>>>(A, B, C, cs, Mc, Nd, Kd,
+                                               a_stride, nbm, nbn);
   }
 }
 
