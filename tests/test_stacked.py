@@ -82,3 +82,21 @@ def test_stacked_topk_per_client():
     # paths may keep different entries, so compare loosely
     diff = (s_stack.global_flat - s_seq.global_flat).abs()
     assert float(diff.mean()) < 1e-4
+
+
+def test_stacked_single_client_matches_sequential():
+    """G=1 stacks too (keeps shadow weights + direct grad writes on the
+    8-GPU layout where each rank holds one client)."""
+    over = dict(rounds=1, num_clients=1, samples_per_client=64,
+                train__batch_size=32, train__max_local_steps=2,
+                train__lr=0.01, num_evaluators=0)
+    s_stack = FederatedServer(driver_config(2).override(**over))
+    assert s_stack.stacked is not None
+    s_seq = FederatedServer(
+        driver_config(2).override(train__use_stacked=False, **over))
+    assert s_seq.stacked is None
+    rec_a = s_stack.run_round(0)
+    rec_b = s_seq.run_round(0)
+    assert abs(rec_a["mean_client_loss"] - rec_b["mean_client_loss"]) < 1e-2
+    torch.testing.assert_close(s_stack.global_flat, s_seq.global_flat,
+                               rtol=1e-3, atol=3e-4)
