@@ -280,8 +280,10 @@ torch::Tensor gemm2(torch::Tensor a, torch::Tensor b, bool a_km, bool b_km,
   const c10::hip::HIPGuardMasqueradingAsCUDA guard(a.device());
   auto ac = a.contiguous(), bc = b.contiguous();
   const int64_t batch = a.dim() == 3 ? a.size(0) : 1;
-  auto c = batch > 1 ? torch::empty({batch, M, N}, a.options())
-                     : torch::empty({M, N}, a.options());
+  // keep the caller's dimensionality: a 3-D batch-1 bmm must return 3-D
+  // (autograd feeds the output shape back into the backward bmm calls)
+  auto c = a.dim() == 3 ? torch::empty({batch, M, N}, a.options())
+                        : torch::empty({M, N}, a.options());
   launch_gemm(ac.data_ptr(), bc.data_ptr(), c.data_ptr(), bias, batch, M,
               N, K, a_km, b_km, stream());
   HIP_CHECK_LAST();

@@ -228,6 +228,18 @@ def test_bmm():
     torch.testing.assert_close(out2.float().cpu(), ref2, rtol=2e-2, atol=0.1)
 
 
+def test_bmm_batch1_stays_3d():
+    # regression: 3-D batch-1 bmm must return 3-D (autograd feeds the
+    # output shape back into the backward bmm calls — G=1 stacked fc)
+    torch.manual_seed(9)
+    a = torch.randn(1, 64, 32).bfloat16().to(DEV)
+    b = torch.randn(1, 48, 32).bfloat16().to(DEV)
+    out = _ext().bmm_nt(a, b)
+    assert out.shape == (1, 64, 48)
+    ref = torch.matmul(a.float().cpu(), b.float().cpu().transpose(1, 2))
+    torch.testing.assert_close(out.float().cpu(), ref, rtol=2e-2, atol=0.1)
+
+
 # ---- conv family -----------------------------------------------------------
 @pytest.mark.parametrize(
     "n,c,hw,co,ks,stride,pad",
