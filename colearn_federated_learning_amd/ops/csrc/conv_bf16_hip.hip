@@ -321,8 +321,12 @@ static void conv_nt_dispatch(const short* A, const short* B, bf16* C,
                              hipStream_t stream) {
   const char* env = getenv("COLEARN_CONV_BK");
   // A/B-measured on the flagship bench: BKT=64 wins from Kd ~1k up
-  // (fewer barriers per MFMA), 32 wins for shallow-K stem shapes
-  const int bkt = env ? atoi(env) : (Kd >= 1024 ? 64 : 32);
+  // (fewer barriers per MFMA), 32 wins for shallow-K stem shapes.
+  // Strided dgrad (GATHER=2) keeps BKT=32: half its gathers early-exit
+  // on the stride check, so deeper staging only adds latency per barrier
+  // (measured: stride-2 dgrad kernels ran 84->197 us with BKT=64).
+  const int bkt =
+      env ? atoi(env) : ((GATHER != 2 && Kd >= 1024) ? 64 : 32);
   if (Nd <= 64) {
     // small-N shapes: a 256x64 tile variant exists (see kernel comment)
     // but measured ~2% SLOWER end-to-end than 128x64 (lower occupancy
