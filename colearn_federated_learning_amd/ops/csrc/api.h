@@ -111,6 +111,7 @@ void launch_conv_dgrad(const void* gy, const void* wt, void* gx,
 int conv_wgrad_splitk(ConvShape cs, int G);
 void launch_conv_wgrad_nosum(const void* gy, const void* x, float* gwf,
                              ConvShape cs, int G, hipStream_t stream);
+void launch_tr_probe(void* out, int scheme, hipStream_t stream);
 void launch_conv_wgrad_out(const void* gy, const void* x, float* gwf,
                            void* gw, ConvShape cs, int G, bool f32_out,
                            hipStream_t stream);

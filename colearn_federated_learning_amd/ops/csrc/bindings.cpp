@@ -719,6 +719,14 @@ torch::Tensor conv2d_wgrad(torch::Tensor gy, torch::Tensor x, int64_t R,
   return G == 1 ? out.squeeze(0) : out;
 }
 
+torch::Tensor tr_probe(int64_t scheme) {
+  auto out = torch::empty({64, 4},
+                          torch::device(torch::kCUDA).dtype(torch::kInt16));
+  launch_tr_probe(out.data_ptr(), static_cast<int>(scheme), stream());
+  HIP_CHECK_LAST();
+  return out;
+}
+
 }  // namespace
 }  // namespace colearn
 
@@ -746,6 +754,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_dgrad_km", &conv2d_dgrad_km);
   m.def("conv2d_wgrad_km", &conv2d_wgrad_km);
   m.def("conv2d_wgrad_km32", &conv2d_wgrad_km32);
+  m.def("tr_probe", &tr_probe);
   m.def("conv2d_dgrad", &conv2d_dgrad);
   m.def("conv2d_wgrad", &conv2d_wgrad);
   m.def("conv2d_wgrad_fp32", &conv2d_wgrad_fp32);

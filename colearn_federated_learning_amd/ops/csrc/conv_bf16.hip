@@ -874,4 +874,39 @@ void launch_conv_wgrad_variant(const void* gy, const void* x, float* gwf,
       gwf, static_cast<bf16*>(gw), n_per_g, splitk);
 }
 
+// ---- ds_read_b64_tr_b16 mapping probe (tools/gpu_debug) -----------------
+// One wave; LDS filled with its own linear index; each lane reads 8 B at
+// addr = lane*8 (+scheme variations); dump shows which LDS element lands
+// in (lane, j) so the wgrad staging image can be designed for the
+// hardware transpose read.
+// STATUS: the instruction assembles (dest VGPR pair + per-lane address)
+// but returned zeros for all schemes on hardware — the operand/addressing
+// contract needs more probing before wgrad staging can use it.  Parked as
+// the documented next wgrad step (SURVEY.md deferred items).
+typedef __attribute__((ext_vector_type(4))) short short4v_probe;
+__global__ void tr_probe_kernel(short* __restrict__ out, int scheme) {
+  __shared__ short lds[1024];
+  for (int i = threadIdx.x; i < 1024; i += 64)
+    lds[i] = static_cast<short>(i);
+  __syncthreads();
+  const int l = threadIdx.x;
+  unsigned addr;
+  if (scheme == 0)
+    addr = l * 8;  // plain per-lane contiguous
+  else if (scheme == 1)
+    addr = ((l & 15) * 4 + (l >> 4) * 256) * 2;  // 4-col groups per lane
+  else
+    addr = ((l >> 4) * 4 + (l & 15) * 32) * 2;   // k-group-major
+  short4v_probe v;
+  asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
+               : "=v"(v)
+               : "v"(addr));
+#pragma unroll
+  for (int j = 0; j < 4; ++j) out[l * 4 + j] = v[j];
+}
+
+void launch_tr_probe(void* out, int scheme, hipStream_t stream) {
+  tr_probe_kernel<<<1, 64, 0, stream>>>(static_cast<short*>(out), scheme);
+}
+
 }  // namespace colearn
