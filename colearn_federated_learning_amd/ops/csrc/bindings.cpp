@@ -704,7 +704,6 @@ torch::Tensor conv2d_wgrad(torch::Tensor gy, torch::Tensor x, int64_t R,
   auto gwf = torch::empty({G * splitk, cs.K, rsc},
                           x.options().dtype(torch::kFloat32));
   auto gw_rsci = torch::empty({G, cs.K, cs.R, cs.S, cs.C}, x.options());
-  (void)0;
   const char* env = getenv("COLEARN_WGRAD_VARIANT");
   const int variant = env ? atoi(env) : 0;
   if (variant)

@@ -880,6 +880,10 @@ void launch_conv_wgrad_variant(const void* gy, const void* x, float* gwf,
 // addr = lane*8 (+scheme variations); dump shows which LDS element lands
 // in (lane, j) so the wgrad staging image can be designed for the
 // hardware transpose read.
+// STATUS: the instruction assembles (dest VGPR pair + per-lane address)
+// but returned zeros for all schemes on hardware — the operand/addressing
+// contract needs more probing before wgrad staging can use it.  Parked as
+// the documented next wgrad step (SURVEY.md deferred items).
 typedef __attribute__((ext_vector_type(4))) short short4v_probe;
 __global__ void tr_probe_kernel(short* __restrict__ out, int scheme) {
   __shared__ short lds[1024];

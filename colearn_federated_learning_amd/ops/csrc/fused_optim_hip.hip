@@ -89,7 +89,6 @@ __global__ void fused_adam_kernel(float* __restrict__ w,
   const int64_t stride = grid_stride() * 4;
   const float step = lr / bc1;
   const float inv_bc2 = 1.f / bc2;
-  (void)0;
   for (; i + 3 < n; i += stride) {
     float4 wv = *reinterpret_cast<float4*>(w + i);
     const float4 gv = *reinterpret_cast<const float4*>(g + i);
