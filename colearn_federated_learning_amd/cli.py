@@ -68,6 +68,12 @@ def _launch(rank: int, args) -> None:
 
     cfg = _build_cfg(args)
     history = run(cfg)
+    import torch.distributed as dist
+
+    if dist.is_initialized():
+        # exit with the group destroyed: a live gloo group at interpreter
+        # shutdown can SIGABRT in its background thread
+        dist.destroy_process_group()
     if rank == 0 and history:
         last = history[-1]
         print(f"session done: round {last['round']} "

@@ -44,10 +44,15 @@ def init_distributed(backend: str = "auto") -> Tuple[int, int, torch.device]:
             torch.cuda.set_device(device)
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29711")
-        dist.init_process_group(
-            backend=backend, rank=rank, world_size=world,
-            timeout=datetime.timedelta(seconds=300),
-        )
+        # COLEARN_INIT_FILE: file-store rendezvous (no TCP listener) —
+        # used by the multi-process test harness, where back-to-back
+        # TCPStore setups occasionally raced on teardown
+        init_file = os.environ.get("COLEARN_INIT_FILE")
+        kwargs = dict(backend=backend, rank=rank, world_size=world,
+                      timeout=datetime.timedelta(seconds=300))
+        if init_file:
+            kwargs["init_method"] = f"file://{init_file}"
+        dist.init_process_group(**kwargs)
     elif device.type == "cuda":
         torch.cuda.set_device(device)
     return rank, world, device

@@ -32,13 +32,33 @@ def _worker(rank: int, world: int, tmpdir: str, port: int, over_json: str):
     os.environ["LOCAL_RANK"] = str(rank)
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
+    # file-store rendezvous: no TCP listener to race on (see utils.dist)
+    os.environ["COLEARN_INIT_FILE"] = os.path.join(tmpdir, ".pg_init")
     from colearn_federated_learning_amd import run
 
     cfg = _make_cfg(tmpdir, **json.loads(over_json))
     run(cfg)
+    # explicit teardown: letting the interpreter exit with a live gloo
+    # group occasionally SIGABRTs in its background thread (flaky CI)
+    import torch.distributed as dist
+
+    if dist.is_initialized():
+        dist.destroy_process_group()
 
 
-def _run_world(world: int, tmpdir: str, port: int, over: dict = None):
+def _free_port() -> int:
+    # fixed ports collide with TIME_WAIT sockets when the suite runs
+    # back-to-back (observed transient failures) — ask the kernel instead
+    import socket
+
+    with socket.socket() as sock:
+        sock.bind(("127.0.0.1", 0))
+        return sock.getsockname()[1]
+
+
+def _run_world(world: int, tmpdir: str, port: int = None, over: dict = None):
+    if port is None:
+        port = _free_port()
     ctx = mp.get_context("spawn")
     procs = [
         ctx.Process(target=_worker,
@@ -75,13 +95,13 @@ def test_world2_equals_world1(tmp_path, over):
     d2 = str(tmp_path / "w2")
     os.makedirs(d1), os.makedirs(d2)
     # single process (in-proc, no dist)
-    _run_world(1, d1, 29741)
-    _run_world(2, d2, 29742, over=None if not over else over)
+    _run_world(1, d1)
+    _run_world(2, d2, over=None if not over else over)
     if over:
         # layouts must agree WITH the feature on: rerun world1 with it
         d1 = str(tmp_path / "w1b")
         os.makedirs(d1)
-        _run_world(1, d1, 29743, over=over)
+        _run_world(1, d1, over=over)
     w1 = _final_weights(d1)
     w2 = _final_weights(d2)
     if over.get("agg__topk_ratio"):
@@ -101,7 +121,7 @@ def test_sparse_transport_matches_dense(tmp_path):
     os.makedirs(dd), os.makedirs(ds)
     over_d = {"agg__topk_ratio": 0.05, "agg__transport": "dense"}
     over_s = {"agg__topk_ratio": 0.05, "agg__transport": "sparse"}
-    _run_world(2, dd, 29744, over=over_d)
-    _run_world(2, ds, 29745, over=over_s)
+    _run_world(2, dd, over=over_d)
+    _run_world(2, ds, over=over_s)
     torch.testing.assert_close(_final_weights(dd), _final_weights(ds),
                                rtol=1e-4, atol=1e-5)

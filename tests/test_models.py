@@ -117,3 +117,29 @@ def test_bert_masked_head_matches_full_logits():
         logits.float().view(-1, 300), labels.reshape(-1), ignore_index=-100
     )
     torch.testing.assert_close(loss_masked, loss_full, rtol=1e-5, atol=1e-6)
+
+
+def test_kmajor_conv_cpu_fallback_matches_torch():
+    """_Conv2dFn stores weights K-major [CO, KH, KW, CI]; the CPU path
+    permutes for F.conv2d and permutes the weight grad back.  Guard the
+    permute logic (fwd + both grads) against the plain torch oracle."""
+    import torch.nn.functional as F
+
+    from colearn_federated_learning_amd.ops.conv import colearn_conv2d
+
+    torch.manual_seed(0)
+    x = torch.randn(2, 8, 9, 9, requires_grad=True)
+    w_std = torch.randn(16, 8, 3, 3)
+    w_km = w_std.permute(0, 2, 3, 1).contiguous().requires_grad_()
+    x2 = x.detach().clone().requires_grad_()
+    w2 = w_std.clone().requires_grad_()
+
+    y = colearn_conv2d(x, w_km, stride=2, padding=1)
+    ref = F.conv2d(x2, w2, None, 2, 1)
+    torch.testing.assert_close(y, ref)
+
+    gy = torch.randn_like(y)
+    y.backward(gy)
+    ref.backward(gy)
+    torch.testing.assert_close(x.grad, x2.grad)
+    torch.testing.assert_close(w_km.grad, w2.grad.permute(0, 2, 3, 1))
