@@ -144,13 +144,22 @@ class FederatedConfig:
         return cls(**kw)
 
     def override(self, **kv: Any) -> "FederatedConfig":
-        """Return a copy with dotted-key overrides, e.g. train__lr=0.1."""
+        """Return a copy with dotted-key overrides, e.g. train__lr=0.1.
+
+        Unknown keys raise: a typo'd override (``samples_per_client=`` for
+        ``data__samples_per_client=``) silently configuring nothing cost a
+        64x-too-long test before this check existed.
+        """
         d = self.to_dict()
         for key, val in kv.items():
             parts = key.split("__")
             node = d
             for p in parts[:-1]:
+                if p not in node:
+                    raise KeyError(f"unknown config group {p!r} in {key!r}")
                 node = node[p]
+            if parts[-1] not in node:
+                raise KeyError(f"unknown config key {key!r}")
             node[parts[-1]] = val
         return FederatedConfig.from_dict(d)
 

@@ -53,3 +53,29 @@ def test_save_load_direct(tmp_path):
     ck = load_checkpoint(str(tmp_path))
     assert ck["round"] == 7
     torch.testing.assert_close(ck["model_state_dict"]["w"], state["w"])
+
+
+def test_resume_matches_uninterrupted_resnet_stacked(tmp_path):
+    """Resume equivalence for the flagship model family: BN running stats
+    and the stacked engine's round-start reload must all restore."""
+    over = dict(num_clients=2, data__samples_per_client=8,
+                train__batch_size=8, train__lr=0.01, num_evaluators=0,
+                model__image_size=16)  # quarter the conv work: CPU grouped
+                # conv is slow and this test builds 4 engines
+    cfg_full = driver_config(2).override(rounds=2, **over)
+    s_full = FederatedServer(cfg_full)
+    assert s_full.stacked is not None
+    s_full.train()
+
+    ck = str(tmp_path / "ck")
+    cfg_a = driver_config(2).override(rounds=1, checkpoint_dir=ck,
+                                      checkpoint_every=1, **over)
+    FederatedServer(cfg_a).train()
+    cfg_b = driver_config(2).override(rounds=2, checkpoint_dir=ck,
+                                      checkpoint_every=1, resume=True,
+                                      **over)
+    s_b = FederatedServer(cfg_b)
+    assert s_b.start_round == 1
+    s_b.train()
+    torch.testing.assert_close(s_b.global_flat, s_full.global_flat,
+                               rtol=1e-5, atol=1e-6)

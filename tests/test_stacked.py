@@ -87,7 +87,7 @@ def test_stacked_topk_per_client():
 def test_stacked_single_client_matches_sequential():
     """G=1 stacks too (keeps shadow weights + direct grad writes on the
     8-GPU layout where each rank holds one client)."""
-    over = dict(rounds=1, num_clients=1, samples_per_client=64,
+    over = dict(rounds=1, num_clients=1, data__samples_per_client=64,
                 train__batch_size=32, train__max_local_steps=2,
                 train__lr=0.01, num_evaluators=0)
     s_stack = FederatedServer(driver_config(2).override(**over))
