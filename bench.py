@@ -37,6 +37,9 @@ def main() -> None:
     ap.add_argument("--samples-per-client", type=int, default=2048)
     ap.add_argument("--set", action="append", default=[],
                     help="config overrides, dotted key=value")
+    ap.add_argument("--target-accuracy", type=float, default=None,
+                    help="also run a rounds-to-target measurement after "
+                         "the throughput steps (eval per round, early stop)")
     args = ap.parse_args()
 
     from colearn_federated_learning_amd.config import driver_config
@@ -79,6 +82,19 @@ def main() -> None:
     barrier_sync()
     elapsed = time.perf_counter() - t0
 
+    # optional companion metric from BASELINE.json: rounds-to-target.
+    # Runs the engine's eval + early-stop loop on a fresh server (the
+    # throughput run above has no evaluators in its timed rounds).
+    rounds_to_target = None
+    if args.target_accuracy is not None:
+        cfg_t = cfg.override(rounds=200, num_evaluators=1,
+                             target_accuracy=args.target_accuracy)
+        server_t = FederatedServer(cfg_t, rank=rank, world=world,
+                                   device=device)
+        hist = server_t.train()
+        if hist and hist[-1].get("eval_accuracy", 0) >= args.target_accuracy:
+            rounds_to_target = len(hist)
+
     # MAX elapsed over ranks (the job is as slow as its slowest rank)
     if dist.is_initialized():
         t = torch.tensor([elapsed], dtype=torch.float64)
@@ -107,6 +123,8 @@ def main() -> None:
             "higher_is_better": True,
             "scaling": "strong",
             "vs_baseline": None,
+            "rounds_to_target": rounds_to_target,
+            "target_accuracy": args.target_accuracy,
             "dtype": "bf16" if on_gpu else "float32",
             "data": "synthetic",
             "config": {
