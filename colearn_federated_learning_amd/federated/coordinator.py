@@ -212,6 +212,11 @@ class FederatedServer:
 
         with self.timer.phase("train"):
             if self.stacked is not None:
+                # the stacked engine trains ALL local clients in one
+                # launch; unsampled/dropped clients' deltas are simply not
+                # folded (their compute is wasted only when
+                # clients_per_round < num_clients, which the stacked
+                # configs don't use)
                 sampled = set(plan.trainers)
                 deltas, bufs, losses, n_per_client = \
                     self.stacked.train_round(

@@ -55,3 +55,30 @@ def test_stable_hash_range(parts):
     h = stable_hash(*parts)
     assert 0 <= h < 2**31
     assert h == stable_hash(*parts)
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.floats(1e-5, 10.0), st.integers(1, 64), st.integers(1, 1024))
+def test_config_override_yaml_roundtrip(lr, bs, spc):
+    from colearn_federated_learning_amd.config import (
+        FederatedConfig,
+        driver_config,
+    )
+
+    cfg = driver_config(2).override(
+        train__lr=lr, train__batch_size=bs, data__samples_per_client=spc)
+    assert cfg.train.lr == lr and cfg.train.batch_size == bs
+    assert cfg.data.samples_per_client == spc
+    back = FederatedConfig.from_dict(cfg.to_dict())
+    assert back.to_dict() == cfg.to_dict()
+
+
+def test_config_override_unknown_key_raises():
+    import pytest
+
+    from colearn_federated_learning_amd.config import driver_config
+
+    with pytest.raises(KeyError):
+        driver_config(2).override(samples_per_client=8)  # missing data__
+    with pytest.raises(KeyError):
+        driver_config(2).override(train__learning_rate=0.1)
