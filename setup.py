@@ -4,13 +4,15 @@
 
 The .so lands next to the Python wrappers
 (colearn_federated_learning_amd/ops/_colearn_hip*.so) so repo snapshots
-carry it to GPU boxes.  Sources are native HIP written for CDNA4 — the
-build does not hipify anything.
+carry it to GPU boxes.  Sources are native HIP written for CDNA4; torch's
+build machinery mechanically copies them to ``*_hip.hip`` (its hipify
+pass), but there is no CUDA in them to translate — the copies are
+byte-identical modulo the rename and are gitignored.
 """
 
 import os
 
-from setuptools import setup
+from setuptools import find_packages, setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
@@ -37,7 +39,7 @@ ext = CUDAExtension(
 setup(
     name="colearn_federated_learning_amd",
     version="0.1.0",
-    packages=["colearn_federated_learning_amd"],
+    packages=find_packages(include=["colearn_federated_learning_amd*"]),
     ext_modules=[ext],
     cmdclass={"build_ext": BuildExtension.with_options(no_python_abi_suffix=False)},
 )
