@@ -160,3 +160,21 @@ def test_grad_sample_engine_logreg():
     vg = vmap(grad(sample_loss), in_dims=(None, 0, 0))(params, x, y)
     torch.testing.assert_close(gw, vg["fc.weight"], rtol=1e-5, atol=1e-6)
     torch.testing.assert_close(gb, vg["fc.bias"], rtol=1e-5, atol=1e-6)
+
+
+def test_accountant_monotonicity():
+    from colearn_federated_learning_amd.ops.dp import PrivacyAccountant
+
+    a = PrivacyAccountant(noise_multiplier=1.0, sample_rate=0.1)
+    a.step(10)
+    e10 = a.epsilon()
+    a.step(90)
+    e100 = a.epsilon()
+    assert e100 > e10 > 0  # epsilon grows with composition
+
+    quiet = PrivacyAccountant(noise_multiplier=4.0, sample_rate=0.1)
+    quiet.step(100)
+    assert quiet.epsilon() < e100  # more noise => stronger guarantee
+
+    off = PrivacyAccountant(noise_multiplier=0.0, sample_rate=0.1)
+    assert off.epsilon() == float("inf")
