@@ -199,8 +199,13 @@ def _per_sample_clipped_backward(
             clipped = dp_ops.clip_and_reduce(
                 per_sample, clip_norm, pre_scale=float(x.shape[0])
             )
-            pdict = dict(model.named_parameters())
-            for (k, p), g in zip(model.named_parameters(), clipped):
+            # grad_samples() covers only requires_grad params — zip against
+            # the same filtered sequence so a frozen param can't misalign
+            # the copy-back.
+            trainable = [p for _, p in model.named_parameters()
+                         if p.requires_grad]
+            assert len(trainable) == len(clipped)
+            for p, g in zip(trainable, clipped):
                 p.grad.copy_(g.to(p.dtype))
             eng.clear()
         return loss_val.detach()

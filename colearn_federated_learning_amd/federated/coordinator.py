@@ -203,6 +203,15 @@ class FederatedServer:
                 k = max(int(delta.numel() * topk_ratio), 1)
                 idx, val = topk_compress(delta.contiguous(), k)
                 delta = topk_decompress(idx, val, delta.numel())
+            if cfg.dp.enabled:
+                # Client-level sensitivity: the noise is sigma*C on the
+                # SUMMED deltas, so each client's contribution must be
+                # bounded by C.  Per-sample clipping inside local steps does
+                # NOT bound the S-step delta (it can reach ~S*lr*C), so
+                # clip the whole delta here (McMahan et al. 2018 DP-FedAvg).
+                nrm = float(delta.norm())
+                if nrm > cfg.dp.clip_norm:
+                    delta = delta * (cfg.dp.clip_norm / nrm)
             self.delta_acc.add_(delta, alpha=w)
             if bufs is not None and bufs.numel():
                 self.buf_acc.add_(bufs.to(self.buf_acc.device), alpha=w)
@@ -342,6 +351,7 @@ class FederatedServer:
                          {"state": "start", "config": self.cfg.to_dict()},
                          retain=True)
         history = []
+        r = self.start_round - 1  # resume-at-end: loop may not run at all
         for r in range(self.start_round, self.cfg.rounds):
             rec = self.run_round(r)
             history.append(rec)
@@ -349,7 +359,7 @@ class FederatedServer:
                     and rec.get("eval_accuracy", 0.0)
                     >= self.cfg.target_accuracy):
                 break
-        if self.rank == 0 and self.cfg.checkpoint_dir:
+        if self.rank == 0 and self.cfg.checkpoint_dir and r >= 0:
             self._checkpoint(r)
         self.bus.publish(Topics.SESSION, {"state": "stop"}, retain=True)
         self.metrics.close()

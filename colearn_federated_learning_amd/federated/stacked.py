@@ -223,41 +223,49 @@ class StackedEngine:
         use_bf16 = tc.dtype == "bfloat16" and self.device.type == "cuda"
 
         gens = [c._gen(round_idx, "batches") for c in self.clients]
-        iters = [
-            c.shard.batches(tc.batch_size, generator=g, drop_last=False)
-            for c, g in zip(self.clients, gens)
-        ]
         loss_acc = torch.zeros(self.G, device=self.device)
         total_n = 0
         steps = 0
-        while True:
-            batches = []
-            for it in iters:
-                batches.append(next(it, None))
-            if any(b is None for b in batches):
-                break  # equal shard sizes => all exhaust together
-            if (tc.max_local_steps is not None
-                    and steps >= tc.max_local_steps):
+        budget_hit = False
+        # one persistent generator per client across epochs, fresh batch
+        # iterators each epoch — mirrors the sequential client.fit loop
+        # (local_epochs was silently ignored here before).
+        for _epoch in range(tc.local_epochs):
+            if budget_hit:
                 break
-            xs = torch.stack([b[0] for b in batches])  # [G, B, ...]
-            ys = torch.stack([b[1] for b in batches])
-            x = xs.reshape(-1, *xs.shape[2:]).to(self.device,
-                                                 non_blocking=True)
-            y = ys.reshape(-1).to(self.device, non_blocking=True)
-            if self.runner is not None and self.runner.matches(x):
-                loss_vec = self.runner.step(x, y)
-            else:
-                self.optimizer.zero_grad()
-                with torch.autocast("cuda", torch.bfloat16,
-                                    enabled=use_bf16):
-                    loss_vec = self.model.loss_terms(x, y)
-                loss_vec.sum().backward()
-                self.optimizer.step()
-                self.sflat.sync_shadow()
-            bs = batches[0][0].shape[0]
-            loss_acc += loss_vec.detach() * bs
-            total_n += bs
-            steps += 1
+            iters = [
+                c.shard.batches(tc.batch_size, generator=g, drop_last=False)
+                for c, g in zip(self.clients, gens)
+            ]
+            while True:
+                batches = []
+                for it in iters:
+                    batches.append(next(it, None))
+                if any(b is None for b in batches):
+                    break  # equal shard sizes => all exhaust together
+                if (tc.max_local_steps is not None
+                        and steps >= tc.max_local_steps):
+                    budget_hit = True
+                    break
+                xs = torch.stack([b[0] for b in batches])  # [G, B, ...]
+                ys = torch.stack([b[1] for b in batches])
+                x = xs.reshape(-1, *xs.shape[2:]).to(self.device,
+                                                     non_blocking=True)
+                y = ys.reshape(-1).to(self.device, non_blocking=True)
+                if self.runner is not None and self.runner.matches(x):
+                    loss_vec = self.runner.step(x, y)
+                else:
+                    self.optimizer.zero_grad()
+                    with torch.autocast("cuda", torch.bfloat16,
+                                        enabled=use_bf16):
+                        loss_vec = self.model.loss_terms(x, y)
+                    loss_vec.sum().backward()
+                    self.optimizer.step()
+                    self.sflat.sync_shadow()
+                bs = batches[0][0].shape[0]
+                loss_acc += loss_vec.detach() * bs
+                total_n += bs
+                steps += 1
         losses = {
             cid: float(v) / max(total_n, 1)
             for cid, v in zip(self.client_ids, loss_acc)

@@ -176,7 +176,10 @@ class DPGraphedStepRunner:
                 per_sample = eng.grad_samples()
                 clipped = dp_ops.clip_and_reduce(per_sample, clip_norm,
                                                  pre_scale=float(bs))
-                for (_, p), g in zip(model.named_parameters(), clipped):
+                trainable = [p for _, p in model.named_parameters()
+                             if p.requires_grad]
+                assert len(trainable) == len(clipped)
+                for p, g in zip(trainable, clipped):
                     p.grad.copy_(g.to(p.dtype))
                 eng.clear()
             optimizer.device_step()
