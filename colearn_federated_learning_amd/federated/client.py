@@ -29,6 +29,7 @@ from .bus import MessageBus, Topics
 class FitResult:
     n_samples: int
     loss: float
+    timed_out: bool = False  # straggler cut off mid-fit (excluded upstream)
     # new weights live in the shared work model; coordinator extracts deltas
 
 
@@ -81,6 +82,7 @@ class FederatedClient:
         optimizer,
         round_idx: int,
         runner=None,
+        deadline: Optional[float] = None,
     ) -> FitResult:
         """Local training for ``local_epochs`` over this client's shard.
 
@@ -98,14 +100,25 @@ class FederatedClient:
         # GPU sync per training step and serialize the whole pipeline.
         loss_acc = torch.zeros((), device=dev)
         total_n, steps = 0, 0
+        timed_out = False
+        import time as _time
         devices = [dev] if dev.type == "cuda" else []
         with torch.random.fork_rng(devices=devices):
             torch.manual_seed(
                 stable_hash(self.cfg.seed, "train", round_idx, self.client_id)
             )
             for _ in range(tc.local_epochs):
+                if timed_out:
+                    break
                 for x, y in self.shard.batches(tc.batch_size, generator=gen):
                     if tc.max_local_steps is not None and steps >= tc.max_local_steps:
+                        break
+                    if (deadline is not None
+                            and _time.perf_counter() > deadline):
+                        # straggler bound: abort mid-fit so the ROUND time
+                        # is bounded by the timeout, not by this client's
+                        # full fit
+                        timed_out = True
                         break
                     x = x.to(dev, non_blocking=True)
                     y = y.to(dev, non_blocking=True)
@@ -131,7 +144,7 @@ class FederatedClient:
                     total_n += bs
                     steps += 1
         avg = float(loss_acc) / max(total_n, 1)
-        res = FitResult(n_samples=total_n, loss=avg)
+        res = FitResult(n_samples=total_n, loss=avg, timed_out=timed_out)
         if self.bus is not None:
             self.bus.publish(
                 Topics.RESULT.format(client_id=self.client_id),

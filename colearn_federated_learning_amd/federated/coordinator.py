@@ -117,13 +117,13 @@ class FederatedServer:
         self.flatbuf = FlatBuffers(self.model)
         self.global_buf = self.flatbuf.gather()
         self.buf_acc = torch.zeros_like(self.global_buf)
-        # stacked-client engine: all local clients train simultaneously
-        # with per-client weights (federated/stacked.py) when applicable
-        from .stacked import StackedEngine, stackable
+        # stacked-client engine: all active local clients train
+        # simultaneously with per-client weights (federated/stacked.py)
+        from .stacked import StackedEnginePool, stackable
 
         local_clients = [self.clients[c] for c in sorted(self.clients)]
         self.stacked = (
-            StackedEngine(cfg, local_clients, self.device)
+            StackedEnginePool(cfg, local_clients, self.device)
             if stackable(cfg, local_clients) else None
         )
         self.step_runner = None
@@ -221,22 +221,26 @@ class FederatedServer:
 
         with self.timer.phase("train"):
             if self.stacked is not None:
-                # the stacked engine trains ALL local clients in one
-                # launch; unsampled/dropped clients' deltas are simply not
-                # folded (their compute is wasted only when
-                # clients_per_round < num_clients, which the stacked
-                # configs don't use)
+                # train ONLY this round's active clients: the engine pool
+                # provides a stacked engine sized to the active count, so
+                # clients_per_round < num_clients wastes no compute
                 sampled = set(plan.trainers)
-                deltas, bufs, losses, n_per_client = \
-                    self.stacked.train_round(
-                        self.global_flat, self.global_buf, round_idx
-                    )
-                for i, cid in enumerate(self.stacked.client_ids):
-                    if cid not in sampled or self._dropped(round_idx, cid):
-                        continue
-                    fold_client(cid, deltas[i],
-                                bufs[i] if bufs.numel() else None,
-                                losses[cid], n_per_client)
+                active = [
+                    c for c in self.stacked.all_clients
+                    if c.client_id in sampled
+                    and not self._dropped(round_idx, c.client_id)
+                ]
+                if active:
+                    engine = self.stacked.get(len(active))
+                    deltas, bufs, losses, n_per_client = \
+                        engine.train_round(
+                            active, self.global_flat, self.global_buf,
+                            round_idx,
+                        )
+                    for i, c in enumerate(active):
+                        fold_client(c.client_id, deltas[i],
+                                    bufs[i] if bufs.numel() else None,
+                                    losses[c.client_id], n_per_client)
             else:
                 for cid in plan.trainers:
                     client = self.clients.get(cid)
@@ -244,16 +248,21 @@ class FederatedServer:
                         continue
                     if self._dropped(round_idx, cid):
                         continue
-                    t_fit = time.perf_counter()
+                    deadline = (
+                        time.perf_counter() + cfg.straggler_timeout_s
+                        if cfg.straggler_timeout_s is not None else None
+                    )
                     self.flat.load_flat(self.global_flat)
                     self.flatbuf.load(self.global_buf)
                     self.optimizer.reset()
                     res = client.fit(self.model, self.optimizer, round_idx,
-                                     runner=self.step_runner)
-                    if (cfg.straggler_timeout_s is not None
-                            and time.perf_counter() - t_fit
-                            > cfg.straggler_timeout_s):
-                        continue  # straggler excluded; reweighted below
+                                     runner=self.step_runner,
+                                     deadline=deadline)
+                    if res.timed_out:
+                        # straggler actually CUT OFF mid-fit (round time is
+                        # bounded); its partial result is excluded and the
+                        # aggregation reweights over the rest
+                        continue
                     fold_client(cid, self.flat.flat_w - self.global_flat,
                                 self.flatbuf.gather(), res.loss,
                                 res.n_samples)

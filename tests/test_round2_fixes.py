@@ -56,6 +56,53 @@ def test_resume_at_end_no_crash(tmp_path):
     assert hist == []
 
 
+def test_straggler_cut_off_mid_fit():
+    """the timeout must bound fit time, not just
+    exclude a client retroactively after it fully trained."""
+    import time
+
+    cfg = driver_config(1).override(
+        rounds=1, num_clients=2, data__samples_per_client=4096,
+        train__batch_size=4, train__local_epochs=50,
+        straggler_timeout_s=0.2, train__use_stacked=False,
+        num_evaluators=0,
+    )
+    s = FederatedServer(cfg)
+    g0 = s.global_flat.clone()
+    t0 = time.perf_counter()
+    rec = s.run_round(0)
+    elapsed = time.perf_counter() - t0
+    # 50 epochs x 1024 tiny batches per client would take many seconds;
+    # the deadline must abort each fit within ~the timeout
+    assert elapsed < 10 * cfg.straggler_timeout_s + 2.0
+    assert rec["client_losses"] == {}          # both cut off => excluded
+    torch.testing.assert_close(s.global_flat, g0)  # no-op aggregation
+
+
+def test_stacked_subsampled_matches_sequential():
+    """clients_per_round < num_clients must train only
+    the sampled clients (engine pool) and match the sequential path."""
+    over = dict(
+        rounds=2, num_clients=4, clients_per_round=2,
+        data__samples_per_client=32, train__batch_size=8,
+        train__dtype="float32", train__lr=0.05, num_evaluators=0,
+    )
+    cfg = driver_config(1).override(**over)
+    s_stack = FederatedServer(cfg)
+    assert s_stack.stacked is not None
+    s_seq = FederatedServer(cfg.override(train__use_stacked=False))
+    for r in range(2):
+        ra = s_stack.run_round(r)
+        rb = s_seq.run_round(r)
+        assert set(ra["client_losses"]) == set(rb["client_losses"])
+        assert len(ra["client_losses"]) == 2
+    torch.testing.assert_close(s_stack.global_flat, s_seq.global_flat,
+                               rtol=1e-4, atol=1e-5)
+    # the pool built a right-sized engine (2 slots), not the full one
+    assert set(s_stack.stacked._engines) >= {2}
+    assert s_stack.stacked._engines[2].G == 2
+
+
 def test_dp_client_delta_clipped():
     """client-level sensitivity requires the per-client
     DELTA bounded by clip_norm, not just per-sample grads within steps."""
