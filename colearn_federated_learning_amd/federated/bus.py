@@ -44,6 +44,9 @@ class MessageBus:
     _retained: Dict[str, Any] = field(default_factory=dict)
     _lock: threading.RLock = field(default_factory=threading.RLock)
     _log: List[Tuple[str, Any]] = field(default_factory=list)
+    # locally-originated messages pending cross-rank relay (BusRelay);
+    # relayed messages are republished with relayed=True and never re-enter
+    _outbox: List[Tuple[str, Any, bool]] = field(default_factory=list)
 
     @staticmethod
     def _match(pattern: str, topic: str) -> bool:
@@ -59,11 +62,14 @@ class MessageBus:
             return False
         return all(p == "+" or p == t for p, t in zip(pp, tp))
 
-    def publish(self, topic: str, payload: Any, retain: bool = False) -> None:
+    def publish(self, topic: str, payload: Any, retain: bool = False,
+                relayed: bool = False) -> None:
         with self._lock:
             if retain:
                 self._retained[topic] = payload
             self._log.append((topic, payload))
+            if not relayed:
+                self._outbox.append((topic, payload, retain))
             cbs = [s.callback for s in self._subs if self._match(s.pattern, topic)]
         for cb in cbs:
             cb(topic, payload)
@@ -88,3 +94,46 @@ class MessageBus:
         with self._lock:
             return {t: p for t, p in self._retained.items()
                     if self._match(pattern, t)}
+
+
+class BusRelay:
+    """Cross-rank control plane: relays locally-originated bus traffic over
+    the torch.distributed process group at round boundaries.
+
+    CoLearn's MQTT broker makes every participant's availability / result /
+    metrics messages visible to the coordinator regardless of which device
+    hosts the client.  With clients sharded across
+    RCCL ranks, each rank's in-process bus only sees its own clients;
+    ``sync()`` is a collective ``all_gather_object`` of each rank's pending
+    outbox, republished everywhere with ``relayed=True`` — so rank 0's
+    registry, logs and retained-message snapshots reflect the whole node,
+    exactly as a broker would.  Payloads are tiny control dicts (no model
+    bytes), so the object collective's pickle cost is irrelevant.
+
+    ``sync()`` must be called at the same points on every rank (it is a
+    collective); the coordinator calls it after client announcement and at
+    the end of every round.
+    """
+
+    def __init__(self, bus: MessageBus) -> None:
+        self.bus = bus
+        self._synced = 0  # outbox prefix already relayed
+
+    def sync(self) -> None:
+        import torch.distributed as dist
+
+        if not dist.is_initialized() or dist.get_world_size() == 1:
+            with self.bus._lock:
+                self._synced = len(self.bus._outbox)
+            return
+        with self.bus._lock:
+            pending = self.bus._outbox[self._synced:]
+            self._synced = len(self.bus._outbox)
+        gathered: List[Any] = [None] * dist.get_world_size()
+        dist.all_gather_object(gathered, pending)
+        rank = dist.get_rank()
+        for r, msgs in enumerate(gathered):
+            if r == rank or not msgs:
+                continue
+            for topic, payload, retain in msgs:
+                self.bus.publish(topic, payload, retain=retain, relayed=True)

@@ -57,9 +57,12 @@ class FederatedClient:
 
     # -- control plane -------------------------------------------------
     def announce(self) -> None:
+        # retained per-client availability subtopic (MQTT idiom) so the
+        # coordinator's registry snapshot sees EVERY client, not just the
+        # last announcer on a shared topic
         if self.bus is not None:
             self.bus.publish(
-                Topics.AVAILABILITY,
+                f"{Topics.AVAILABILITY}/{self.client_id}",
                 {"client_id": self.client_id, "n_samples": len(self.shard),
                  "device": str(self.device)},
                 retain=True,

@@ -89,8 +89,20 @@ class FederatedServer:
             c = FederatedClient(cid, cfg, shard, self.device, bus=self.bus)
             c.announce()
             self.clients[cid] = c
-        # global registry is deterministic: every configured client exists.
-        self.registry: List[int] = list(range(cfg.num_clients))
+        # relay availability announcements across ranks (collective): every
+        # rank's bus — in particular rank 0's, the coordinator's — now
+        # holds the retained availability messages of ALL clients, exactly
+        # as CoLearn's MQTT broker would.
+        from .bus import BusRelay
+
+        self.bus_relay = BusRelay(self.bus)
+        self.bus_relay.sync()
+        # registry built FROM the control plane (deterministic: every
+        # configured client announces at construction).
+        self.registry: List[int] = sorted(
+            p["client_id"]
+            for p in self.bus.collect(Topics.AVAILABILITY + "/#").values()
+        ) or list(range(cfg.num_clients))
         self.start_round = 0
         self.accountant = (
             PrivacyAccountant(
@@ -308,6 +320,11 @@ class FederatedServer:
                 if ev[2] > 0:
                     eval_metrics = {"eval_loss": ev[0] / ev[2],
                                     "eval_accuracy": ev[1] / ev[2]}
+
+        # --- control-plane relay (collective): results/metrics published
+        # by other ranks' clients this round land on every bus, so rank
+        # 0's log mirrors the whole node like CoLearn's broker would
+        self.bus_relay.sync()
 
         # --- bookkeeping -------------------------------------------------
         n_total = self.aggregator.allreduce_scalar(float(n_samples_round))

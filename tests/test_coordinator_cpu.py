@@ -33,8 +33,10 @@ def test_control_plane_messages():
     server = FederatedServer(cfg, bus=bus)
     server.train()
     topics = [t for t, _ in events]
-    # availability announced by both clients
-    assert topics.count(Topics.AVAILABILITY) == 2
+    # availability announced by both clients (per-client subtopics)
+    avail = [t for t in topics
+             if t.startswith(Topics.AVAILABILITY + "/")]
+    assert len(avail) == 2
     # roles assigned
     assert any(t.startswith("colearn/role/") for t in topics)
     # round + session lifecycle
