@@ -285,11 +285,9 @@ class FederatedServer:
                 self.delta_acc, local_weight
             )
             if cfg.dp.enabled and cfg.dp.noise_multiplier > 0:
-                g = torch.Generator().manual_seed(
-                    stable_hash(cfg.seed, "dp", round_idx)
-                )
                 add_dp_noise(self.delta_acc, cfg.dp.clip_norm,
-                             cfg.dp.noise_multiplier, generator=g)
+                             cfg.dp.noise_multiplier,
+                             key=stable_hash(cfg.seed, "dp", round_idx))
                 if self.accountant is not None:
                     self.accountant.step()
             self.aggregator.allreduce_tensor(self.buf_acc)

@@ -77,6 +77,9 @@ void launch_scale_rows_mean(const float* g, const float* scale, float* out,
 void launch_topk(const float* v, int64_t n, int64_t k, int* idx, float* val,
                  unsigned* workspace /* >= 1036 u32 */, int* cand,
                  int cand_cap, hipStream_t stream);
+void launch_philox_gaussian_add(float* out, int64_t n,
+                                unsigned long long seed, float std,
+                                hipStream_t stream);
 void launch_scatter_add_pairs(float* out, const int* idx, const float* val,
                               int64_t k, hipStream_t stream);
 void launch_layernorm_fwd(const void* x, const float* w, const float* b,

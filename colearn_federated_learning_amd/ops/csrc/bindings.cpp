@@ -114,6 +114,16 @@ void scale_rows_mean(torch::Tensor g, torch::Tensor scale,
   HIP_CHECK_LAST();
 }
 
+void philox_gaussian_add(torch::Tensor flat, int64_t seed, double std) {
+  check_f32(flat, "flat");
+  TORCH_CHECK(flat.is_contiguous(), "flat must be contiguous");
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(flat.device());
+  launch_philox_gaussian_add(flat.data_ptr<float>(), flat.numel(),
+                             static_cast<unsigned long long>(seed),
+                             static_cast<float>(std), stream());
+  HIP_CHECK_LAST();
+}
+
 std::tuple<torch::Tensor, torch::Tensor> topk_compress(torch::Tensor flat,
                                                        int64_t k) {
   check_f32(flat, "flat");
@@ -735,6 +745,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adam", &fused_adam);
   m.def("sq_norm_per_row", &sq_norm_per_row);
   m.def("scale_rows_mean", &scale_rows_mean);
+  m.def("philox_gaussian_add", &philox_gaussian_add);
   m.def("topk_compress", &topk_compress);
   m.def("scatter_add_pairs", &scatter_add_pairs);
   m.def("layernorm_fwd", &layernorm_fwd);

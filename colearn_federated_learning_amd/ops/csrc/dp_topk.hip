@@ -283,6 +283,68 @@ __global__ void scatter_add_pairs_kernel(float* __restrict__ out,
   for (; i < k; i += grid_stride()) atomicAdd(out + idx[i], val[i]);
 }
 
+// ---- DP Gaussian noise: device-side philox4x32-10 + Box-Muller ------------
+// The Gaussian mechanism's noise is generated ON DEVICE, keyed only on
+// (seed, element index): independent of grid shape and rank layout, so
+// every rank adds the SAME noise vector to the aggregated delta (the
+// simulation-equivalence contract) with no host randn + H2D round trip.
+DEV_INLINE unsigned mulhilo32(unsigned a, unsigned b, unsigned* hi) {
+  const unsigned long long p = static_cast<unsigned long long>(a) * b;
+  *hi = static_cast<unsigned>(p >> 32);
+  return static_cast<unsigned>(p);
+}
+
+__global__ void philox_gaussian_add_kernel(float* __restrict__ out,
+                                           int64_t n,
+                                           unsigned long long seed,
+                                           float std) {
+  const unsigned key0 = static_cast<unsigned>(seed & 0xffffffffull);
+  const unsigned key1 = static_cast<unsigned>(seed >> 32);
+  const int64_t nq = (n + 3) >> 2;  // one philox block = 4 outputs
+  int64_t q = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+  for (; q < nq; q += grid_stride()) {
+    unsigned c0 = static_cast<unsigned>(q & 0xffffffff);
+    unsigned c1 = static_cast<unsigned>(static_cast<unsigned long long>(q)
+                                        >> 32);
+    unsigned c2 = 0, c3 = 0;
+    unsigned k0 = key0, k1 = key1;
+#pragma unroll
+    for (int r = 0; r < 10; ++r) {
+      unsigned hi0, hi1;
+      const unsigned lo0 = mulhilo32(0xD2511F53u, c0, &hi0);
+      const unsigned lo1 = mulhilo32(0xCD9E8D57u, c2, &hi1);
+      const unsigned n0 = hi1 ^ c1 ^ k0;
+      const unsigned n2 = hi0 ^ c3 ^ k1;
+      c0 = n0;
+      c1 = lo1;
+      c2 = n2;
+      c3 = lo0;
+      k0 += 0x9E3779B9u;
+      k1 += 0xBB67AE85u;
+    }
+    // Box-Muller; +0.5 keeps u in (0,1) so log() never sees 0
+    constexpr float kInv32 = 2.3283064365386963e-10f;  // 2^-32
+    constexpr float kTwoPi = 6.28318530717958647692f;
+    const float r1 = sqrtf(-2.f * logf((static_cast<float>(c0) + 0.5f)
+                                       * kInv32));
+    const float t1 = kTwoPi * (static_cast<float>(c1) * kInv32);
+    const float r2 = sqrtf(-2.f * logf((static_cast<float>(c2) + 0.5f)
+                                       * kInv32));
+    const float t2 = kTwoPi * (static_cast<float>(c3) * kInv32);
+    float z[4];
+    sincosf(t1, &z[0], &z[1]);
+    sincosf(t2, &z[2], &z[3]);
+    z[0] *= r1;
+    z[1] *= r1;
+    z[2] *= r2;
+    z[3] *= r2;
+    const int64_t base = q << 2;
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      if (base + j < n) out[base + j] += std * z[j];
+  }
+}
+
 // ---- launchers ------------------------------------------------------------
 void launch_sq_norm_per_row(const float* g, float* sq, float* partials,
                             int64_t rows, int64_t cols, int chunks,
@@ -326,6 +388,13 @@ void launch_scatter_add_pairs(float* out, const int* idx, const float* val,
                               int64_t k, hipStream_t stream) {
   scatter_add_pairs_kernel<<<elemwise_grid(k), 256, 0, stream>>>(out, idx,
                                                                  val, k);
+}
+
+void launch_philox_gaussian_add(float* out, int64_t n,
+                                unsigned long long seed, float std,
+                                hipStream_t stream) {
+  philox_gaussian_add_kernel<<<elemwise_grid((n + 3) / 4), 256, 0, stream>>>(
+      out, n, seed, std);
 }
 
 }  // namespace colearn
