@@ -34,7 +34,9 @@ def main() -> None:
     ap.add_argument("--config", type=int, default=2,
                     help="BASELINE.json driver config index (default: "
                          "flagship ResNet-18 FedAvg)")
-    ap.add_argument("--samples-per-client", type=int, default=2048)
+    ap.add_argument("--samples-per-client", type=int, default=None,
+                    help="override shard size (default: the driver "
+                         "config's own value)")
     ap.add_argument("--set", action="append", default=[],
                     help="config overrides, dotted key=value")
     ap.add_argument("--target-accuracy", type=float, default=None,
@@ -54,12 +56,14 @@ def main() -> None:
     on_gpu = torch.cuda.is_available()
     cfg = cfg.override(
         rounds=args.warmup + args.steps,
-        data__samples_per_client=args.samples_per_client,
         train__dtype="bfloat16" if on_gpu else "float32",
         num_evaluators=0,
         checkpoint_dir=None,
         metrics_path=None,
     )
+    if args.samples_per_client is not None:
+        cfg = cfg.override(
+            data__samples_per_client=args.samples_per_client)
     if args.set:
         cfg = cfg.override(**parse_overrides(args.set))
     rank, world, device = init_distributed(cfg.agg.backend)
