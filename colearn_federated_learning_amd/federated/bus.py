@@ -118,6 +118,21 @@ class BusRelay:
     def __init__(self, bus: MessageBus) -> None:
         self.bus = bus
         self._synced = 0  # outbox prefix already relayed
+        self._group = None
+
+    def _gloo_group(self):
+        """Dedicated gloo subgroup for the control plane: object
+        collectives stay on host (no GPU round trip through RCCL for
+        pickled dicts), and control messages can't interleave with the
+        data plane's in-flight RCCL collectives."""
+        import torch.distributed as dist
+
+        if self._group is None:
+            if dist.get_backend() == "gloo":
+                self._group = dist.group.WORLD
+            else:
+                self._group = dist.new_group(backend="gloo")
+        return self._group
 
     def sync(self) -> None:
         import torch.distributed as dist
@@ -130,7 +145,7 @@ class BusRelay:
             pending = self.bus._outbox[self._synced:]
             self._synced = len(self.bus._outbox)
         gathered: List[Any] = [None] * dist.get_world_size()
-        dist.all_gather_object(gathered, pending)
+        dist.all_gather_object(gathered, pending, group=self._gloo_group())
         rank = dist.get_rank()
         for r, msgs in enumerate(gathered):
             if r == rank or not msgs:

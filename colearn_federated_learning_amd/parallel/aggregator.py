@@ -74,10 +74,26 @@ class Aggregator:
         return flat_delta, total_weight
 
     # ------------------------------------------------------------------
+    @staticmethod
+    def _staged(t: torch.Tensor) -> bool:
+        """gloo can't reduce CUDA tensors: stage through host.  This lets
+        the whole multi-process GPU engine run over gloo transport — the
+        1-GPU-box validation path for multi-rank training (RCCL refuses
+        two ranks on one device: 'Duplicate GPU detected')."""
+        return t.is_cuda and dist.get_backend() == "gloo"
+
+    def _allreduce_inplace(self, t: torch.Tensor) -> None:
+        if self._staged(t):
+            c = t.cpu()
+            dist.all_reduce(c, op=dist.ReduceOp.SUM)
+            t.copy_(c)
+        else:
+            dist.all_reduce(t, op=dist.ReduceOp.SUM)
+
     def _dense_allreduce(self, flat: torch.Tensor) -> None:
         bucket_elems = max(self.cfg.bucket_bytes // flat.element_size(), 1)
         if not self._streams or flat.numel() <= bucket_elems:
-            dist.all_reduce(flat, op=dist.ReduceOp.SUM)
+            self._allreduce_inplace(flat)
             return
         works = []
         n = flat.numel()
@@ -121,10 +137,15 @@ class Aggregator:
         idx[: nz.numel()] = nz.to(torch.int32)
         val[: nz.numel()] = flat[nz].float()
         if dist.get_backend() == "gloo":
+            if idx.is_cuda:  # host-staged (see _staged)
+                idx, val = idx.cpu(), val.cpu()
             idx_list = [torch.empty_like(idx) for _ in range(self.world)]
             val_list = [torch.empty_like(val) for _ in range(self.world)]
             dist.all_gather(idx_list, idx)
             dist.all_gather(val_list, val)
+            if flat.is_cuda:
+                idx_list = [i.to(flat.device) for i in idx_list]
+                val_list = [v.to(flat.device) for v in val_list]
         else:
             idx_all = torch.empty(self.world * kmax, dtype=idx.dtype,
                                   device=idx.device)
@@ -141,11 +162,16 @@ class Aggregator:
     # ------------------------------------------------------------------
     def broadcast_flat(self, flat: torch.Tensor, src: int = 0) -> None:
         if self.world > 1:
-            dist.broadcast(flat, src=src)
+            if self._staged(flat):
+                c = flat.cpu()
+                dist.broadcast(c, src=src)
+                flat.copy_(c)
+            else:
+                dist.broadcast(flat, src=src)
 
     def allreduce_tensor(self, t: torch.Tensor) -> None:
         if self.world > 1 and t.numel():
-            dist.all_reduce(t, op=dist.ReduceOp.SUM)
+            self._allreduce_inplace(t)
 
     def allreduce_scalar(self, value: float, op=None) -> float:
         if self.world == 1:
