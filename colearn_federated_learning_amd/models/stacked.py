@@ -167,9 +167,8 @@ class SBatchNorm2d(nn.Module):
         self.gslot_b = None
 
     def forward(self, x: torch.Tensor, residual=None) -> torch.Tensor:
-        if self.training:
-            with torch.no_grad():
-                self.num_batches_tracked += 1
+        # no num_batches_tracked increment: unused with fixed momentum and
+        # it cost one launch per BN layer per step (see ops/norms.py)
         if _native(x):
             y, _, _ = _BatchNormFn.apply(
                 x, self.weight.view(-1), self.bias.view(-1),
@@ -338,14 +337,25 @@ class StackedResNet(ColearnModel):
             blocks.append(block(G, self.inplanes, planes, 1, group_norm))
         return nn.Sequential(*blocks)
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def features(self, x: torch.Tensor) -> torch.Tensor:
         x = self.pool(self.stem(x))
-        x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
-        x = F.adaptive_avg_pool2d(x, 1).flatten(1)
+        return self.layer4(self.layer3(self.layer2(self.layer1(x))))
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        x = F.adaptive_avg_pool2d(self.features(x), 1).flatten(1)
         return self.fc(x)  # [G*B, classes]
 
     def loss_terms(self, x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
-        logits = self(x)
+        from ..ops.head import fused_head_applicable, fused_head_loss
+
+        feat = self.features(x)
+        if fused_head_applicable(feat, self.fc.weight,
+                                 self.fc.weight.shape[1]):
+            # fused avgpool+FC+CE (4 launches fwd+bwd, fp32 head math,
+            # direct grad writes) — csrc/head.hip
+            return fused_head_loss(feat, self.fc.weight, self.fc.bias,
+                                   y, self.G)
+        logits = self.fc(F.adaptive_avg_pool2d(feat, 1).flatten(1))
         per_sample = F.cross_entropy(logits.float(), y, reduction="none")
         return per_sample.view(self.G, -1).mean(dim=1)  # [G]
 

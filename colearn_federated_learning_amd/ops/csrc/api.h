@@ -77,6 +77,15 @@ void launch_scale_rows_mean(const float* g, const float* scale, float* out,
 void launch_topk(const float* v, int64_t n, int64_t k, int* idx, float* val,
                  unsigned* workspace /* >= 1036 u32 */, int* cand,
                  int cand_cap, hipStream_t stream);
+void launch_head_fwd(const void* x, const float* w, const float* b,
+                     const int64_t* y, float* pooled, float* probs,
+                     float* loss_n, float* loss_vec, int N, int C, int HW,
+                     int K, int B, int G, hipStream_t stream);
+void launch_head_bwd(const float* probs, const float* pooled,
+                     const float* w, const int64_t* y, const float* gloss,
+                     float* dlogits, void* dx, float* gw, float* gb,
+                     int N, int C, int HW, int K, int B, int G,
+                     hipStream_t stream);
 void launch_philox_gaussian_add(float* out, int64_t n,
                                 unsigned long long seed, float std,
                                 hipStream_t stream);

@@ -114,6 +114,65 @@ void scale_rows_mean(torch::Tensor g, torch::Tensor scale,
   HIP_CHECK_LAST();
 }
 
+// ---- fused classifier head (avgpool + per-client FC + softmax CE) --------
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> head_fwd(
+    torch::Tensor x, torch::Tensor w, torch::Tensor b, torch::Tensor y,
+    int64_t G) {
+  check_bf16(x, "x");
+  check_f32(w, "w");
+  check_f32(b, "b");
+  TORCH_CHECK(x.dim() == 4 &&
+                  x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "x must be channels-last NCHW");
+  TORCH_CHECK(y.scalar_type() == torch::kInt64, "y must be int64");
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
+  const int N = x.size(0), C = x.size(1);
+  const int HW = x.size(2) * x.size(3);
+  const int K = w.size(1);
+  TORCH_CHECK(K <= 32, "head supports <=32 classes");
+  TORCH_CHECK(N % G == 0, "N must be divisible by G");
+  const int B = N / static_cast<int>(G);
+  auto f32 = x.options().dtype(torch::kFloat32);
+  auto pooled = torch::empty({N, C}, f32);
+  auto probs = torch::empty({N, K}, f32);
+  auto loss_n = torch::empty({N}, f32);
+  auto loss_vec = torch::empty({G}, f32);
+  launch_head_fwd(x.data_ptr(), w.data_ptr<float>(), b.data_ptr<float>(),
+                  y.data_ptr<int64_t>(), pooled.data_ptr<float>(),
+                  probs.data_ptr<float>(), loss_n.data_ptr<float>(),
+                  loss_vec.data_ptr<float>(), N, C, HW,
+                  K, B, static_cast<int>(G), stream());
+  HIP_CHECK_LAST();
+  return {loss_vec, pooled, probs};
+}
+
+torch::Tensor head_bwd(torch::Tensor probs, torch::Tensor pooled,
+                       torch::Tensor w, torch::Tensor y,
+                       torch::Tensor gloss, torch::Tensor gw,
+                       torch::Tensor gb, int64_t H, int64_t W) {
+  check_f32(probs, "probs");
+  check_f32(pooled, "pooled");
+  check_f32(w, "w");
+  check_f32(gloss, "gloss");
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(probs.device());
+  const int N = pooled.size(0), C = pooled.size(1);
+  const int K = w.size(1), G = w.size(0);
+  const int HW = static_cast<int>(H * W);
+  const int B = N / G;
+  auto dlogits = torch::empty({N, K},
+                              probs.options().dtype(torch::kFloat32));
+  auto dx = torch::empty({N, C, H, W},
+                         probs.options().dtype(torch::kBFloat16),
+                         at::MemoryFormat::ChannelsLast);
+  launch_head_bwd(probs.data_ptr<float>(), pooled.data_ptr<float>(),
+                  w.data_ptr<float>(), y.data_ptr<int64_t>(),
+                  gloss.data_ptr<float>(), dlogits.data_ptr<float>(),
+                  dx.data_ptr(), gw.data_ptr<float>(),
+                  gb.data_ptr<float>(), N, C, HW, K, B, G, stream());
+  HIP_CHECK_LAST();
+  return dx;
+}
+
 void philox_gaussian_add(torch::Tensor flat, int64_t seed, double std) {
   check_f32(flat, "flat");
   TORCH_CHECK(flat.is_contiguous(), "flat must be contiguous");
@@ -746,6 +805,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sq_norm_per_row", &sq_norm_per_row);
   m.def("scale_rows_mean", &scale_rows_mean);
   m.def("philox_gaussian_add", &philox_gaussian_add);
+  m.def("head_fwd", &head_fwd);
+  m.def("head_bwd", &head_bwd);
   m.def("topk_compress", &topk_compress);
   m.def("scatter_add_pairs", &scatter_add_pairs);
   m.def("layernorm_fwd", &layernorm_fwd);

@@ -140,9 +140,12 @@ class ColearnBatchNorm2d(nn.Module):
                              torch.tensor(0, dtype=torch.long))
 
     def forward(self, x: torch.Tensor, residual=None) -> torch.Tensor:
-        if self.training:
-            with torch.no_grad():
-                self.num_batches_tracked += 1
+        # num_batches_tracked is kept for state-dict compatibility but NOT
+        # incremented: torch only consults it when momentum is None (ours
+        # is fixed), and the on-device `+= 1` cost one extra launch per BN
+        # layer per captured step (20/step on ResNet-18 — measured in
+        # profiles/final_resnet18_kernel_stats.txt as the 1980-launch
+        # CUDAFunctorOnSelf_add<long> row).
         y, _, _ = _BatchNormFn.apply(
             x, self.weight, self.bias, self.running_mean, self.running_var,
             residual, self.training, self.momentum, self.eps, self.relu, 1,

@@ -29,6 +29,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "norms.hip"),
         os.path.join(CSRC, "gemm_bf16.hip"),
         os.path.join(CSRC, "conv_bf16.hip"),
+        os.path.join(CSRC, "head.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

@@ -536,3 +536,68 @@ def test_add_dp_noise_device_path_used():
     ref = torch.zeros(10_000, device=DEV)
     m.philox_gaussian_add(ref, 99, 3.0)
     torch.testing.assert_close(flat, ref)
+
+
+# ---- fused classifier head (avgpool + FC + CE) -----------------------------
+def test_fused_head_matches_torch_oracle():
+    import torch.nn.functional as F
+
+    from colearn_federated_learning_amd.ops.head import fused_head_loss
+
+    _ext()
+    torch.manual_seed(7)
+    G, B, C, H, W, K = 3, 16, 64, 4, 4, 10
+    x = (torch.randn(G * B, C, H, W, device=DEV)
+         .to(torch.bfloat16)
+         .contiguous(memory_format=torch.channels_last))
+    w = torch.randn(G, K, C, device=DEV) * 0.1
+    b = torch.randn(G, K, device=DEV) * 0.1
+    y = torch.randint(0, K, (G * B,), device=DEV)
+
+    xf = x.clone().requires_grad_(True)
+    wf, bf = w.clone(), b.clone()
+    wf.grad = torch.full_like(wf, 7.0)  # overwritten, not accumulated
+    bf.grad = torch.full_like(bf, 7.0)
+    loss_vec = fused_head_loss(xf, wf, bf, y, G)
+    loss_vec.sum().backward()
+
+    xo = x.float().clone().requires_grad_(True)
+    wo = w.clone().requires_grad_(True)
+    bo = b.clone().requires_grad_(True)
+    pooled = F.adaptive_avg_pool2d(xo, 1).flatten(1).view(G, B, C)
+    logits = torch.einsum("gbc,gkc->gbk", pooled, wo) + bo[:, None, :]
+    ls = F.cross_entropy(logits.reshape(-1, K), y,
+                         reduction="none").view(G, B).mean(dim=1)
+    ls.sum().backward()
+
+    torch.testing.assert_close(loss_vec, ls, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(wf.grad, wo.grad, rtol=1e-4, atol=1e-6)
+    torch.testing.assert_close(bf.grad, bo.grad, rtol=1e-4, atol=1e-6)
+    torch.testing.assert_close(xf.grad.float(), xo.grad,
+                               rtol=1e-2, atol=1e-4)  # bf16 output grad
+
+
+def test_fused_head_deterministic():
+    from colearn_federated_learning_amd.ops.head import fused_head_loss
+
+    _ext()
+    torch.manual_seed(8)
+    G, B, C, H, W, K = 8, 64, 512, 4, 4, 10
+    x = (torch.randn(G * B, C, H, W, device=DEV)
+         .to(torch.bfloat16)
+         .contiguous(memory_format=torch.channels_last))
+    w = torch.randn(G, K, C, device=DEV) * 0.05
+    b = torch.zeros(G, K, device=DEV)
+    y = torch.randint(0, K, (G * B,), device=DEV)
+    outs = []
+    for _ in range(2):
+        xf = x.clone().requires_grad_(True)
+        wf, bf = w.clone(), b.clone()
+        wf.grad = torch.zeros_like(wf)
+        bf.grad = torch.zeros_like(bf)
+        lv = fused_head_loss(xf, wf, bf, y, G)
+        lv.sum().backward()
+        outs.append((lv.detach().clone(), xf.grad.clone(),
+                     wf.grad.clone(), bf.grad.clone()))
+    for a, bb in zip(outs[0], outs[1]):
+        assert torch.equal(a, bb)
