@@ -1,0 +1,47 @@
+"""Attribute the remaining torch-native kernels in the stacked step.
+
+Runs ONE eager (non-graphed) stacked flagship step under torch.profiler
+with shapes, printing ops that launch FillFunctor / elementwise /
+copy kernels — the residual small-kernel tail after the head fusion.
+"""
+import os
+import sys
+
+sys.path.insert(0, os.getcwd())
+
+import torch
+
+from colearn_federated_learning_amd import FederatedServer, driver_config
+
+cfg = driver_config(2).override(
+    rounds=1, data__samples_per_client=256, train__use_graphs=False,
+    num_evaluators=0,
+)
+s = FederatedServer(cfg, device=torch.device("cuda", 0))
+s.run_round(0)  # warm
+from torch.profiler import ProfilerActivity, profile
+
+eng = s.stacked.full
+x = torch.randn(8 * 64, 3, 32, 32, device="cuda")
+y = torch.randint(0, 10, (8 * 64,), device="cuda")
+torch.cuda.synchronize()
+with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA],
+             record_shapes=True) as prof:
+    eng.optimizer.zero_grad()
+    with torch.autocast("cuda", torch.bfloat16):
+        lv = eng.model.loss_terms(x, y)
+    lv.sum().backward()
+    eng.optimizer.step()
+    eng.sflat.sync_shadow()
+    torch.cuda.synchronize()
+print(prof.key_averages(group_by_input_shape=True)
+      .table(sort_by="self_cuda_time_total", row_limit=40,
+             max_src_column_width=60))
+# kernel-level counts for the tail kernels
+ka = prof.key_averages()
+print("\n---- per-op CUDA kernel counts (fills/copies/elementwise) ----")
+for ev in sorted(ka, key=lambda e: -e.self_device_time_total)[:60]:
+    n = ev.key
+    if any(t in n for t in ("fill", "Fill", "copy", "add", "elementwise",
+                            "zero", "div", "mul", "mean", "sum")):
+        print(f"{ev.self_device_time_total/1000.0:9.3f} ms  n={ev.count:5d}  {n[:100]}")
