@@ -66,12 +66,16 @@ class StackedFlatParams:
         self.flat_bf16 = None  # allocated by attach_shadows
 
     def attach_shadows(self, model: torch.nn.Module) -> None:
-        """Give every stacked conv a persistent bf16 view of its weights.
+        """Give every stacked conv a persistent bf16 view of its weights,
+        plus a persistent PRE-TRANSPOSED dgrad-layout view.
 
-        The views alias one flat bf16 buffer refreshed by sync_shadow() —
-        ONE fused cast per train step (inside the captured graph, after
-        the optimizer) replaces the per-layer autocast cast kernels that
-        were ~5% of flagship step time.
+        sync_shadow() is then ONE fused kernel per train step (inside the
+        captured graph, after the optimizer) that casts the whole flat
+        fp32 buffer to bf16 AND regenerates every conv's [G][CI][R][S][CO]
+        dgrad weight in tiled-transpose passes — replacing both the
+        per-layer autocast casts (~5% of step time, round 1) and the
+        ~20 per-call permute+contiguous copies the dgrad path used to
+        issue each step (~10% of step time — gpurun_out/eager_attrib.txt).
         """
         from ..models.stacked import SBatchNorm2d, SConv2d
 
@@ -79,6 +83,9 @@ class StackedFlatParams:
                                           dtype=torch.bfloat16)
         by_name = dict(zip(self.names,
                            zip(self._offsets, self.t_numels)))
+        segA, segB = [], []
+        t_off, blk = 0, 0
+        conv_t = []  # (mod, t_off, shape)
         for mod_name, mod in model.named_modules():
             if isinstance(mod, SConv2d):
                 off, n = by_name[f"{mod_name}.weight" if mod_name
@@ -87,14 +94,48 @@ class StackedFlatParams:
                                .view(mod.weight.shape))
                 # direct wgrad target: the param's flat-grad view
                 mod.wgslot = mod.weight.grad
+                G, CO, KH, KW, CI = mod.weight.shape
+                if CO % 32 == 0 and CI % 32 == 0:
+                    RS = KH * KW
+                    tco, tci = CO // 32, CI // 32
+                    nblk = G * RS * tco * tci
+                    segA.append([off, t_off, blk, blk + nblk])
+                    segB.append([CO, CI, RS, tco, tci, G])
+                    conv_t.append((mod, t_off, (G, CI, KH, KW, CO)))
+                    t_off += G * n
+                    blk += nblk
+                # else (stem CI=3): dgrad is skipped anyway — the input
+                # batch never needs a gradient (needs_input_grad gate)
             elif isinstance(mod, SBatchNorm2d):
                 mod.gslot_w = mod.weight.grad.view(-1)
                 mod.gslot_b = mod.bias.grad.view(-1)
+        dev = self.device
+        self.flat_bf16_t = torch.empty(max(t_off, 1),
+                                       dtype=torch.bfloat16, device=dev)
+        self._segA = torch.tensor(segA, dtype=torch.int64,
+                                  device=dev).reshape(-1, 4)
+        self._segB = torch.tensor(segB, dtype=torch.int32,
+                                  device=dev).reshape(-1, 6)
+        self._transpose_blocks = blk
+        for mod, toff, shape in conv_t:
+            n = 1
+            for d in shape:
+                n *= d
+            mod.wshadow_t = self.flat_bf16_t[toff : toff + n].view(shape)
         self.sync_shadow()
 
     @torch.no_grad()
     def sync_shadow(self) -> None:
-        if self.flat_bf16 is not None:
+        if self.flat_bf16 is None:
+            return
+        from ..ops import native_enabled, require_ext
+
+        if self.flat_w.is_cuda and native_enabled():
+            require_ext().shadow_sync(
+                self.flat_w, self.flat_bf16, self.flat_bf16_t,
+                self._segA, self._segB, self._transpose_blocks,
+            )
+        else:
             self.flat_bf16.copy_(self.flat_w)
 
     # -- global <-> stacked ------------------------------------------------

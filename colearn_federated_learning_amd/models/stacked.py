@@ -95,22 +95,34 @@ class _SConv2dShadowFn(torch.autograd.Function):
     slot of the fp32 master parameter (no bf16 grad round trip)."""
 
     @staticmethod
-    def forward(x, w, shadow, gslot, stride: int, padding: int, G: int):
+    def forward(x, w, shadow, shadow_t, gslot, stride: int, padding: int,
+                G: int):
         return require_ext().conv2d_fwd_km(x, shadow, stride, padding)
 
     @staticmethod
     def setup_context(ctx, inputs, output):
-        x, w, shadow, gslot, stride, padding, G = inputs
-        ctx.save_for_backward(x, shadow)
+        x, w, shadow, shadow_t, gslot, stride, padding, G = inputs
+        ctx.save_for_backward(x, shadow, shadow_t)
         ctx.gslot = gslot
         ctx.stride, ctx.padding, ctx.G = stride, padding, G
 
     @staticmethod
     def backward(ctx, gy):
-        x, shadow = ctx.saved_tensors
+        x, shadow, shadow_t = ctx.saved_tensors
         s, p, G = ctx.stride, ctx.padding, ctx.G
         m = require_ext()
-        gx = m.conv2d_dgrad_km(gy, shadow, x.shape[2], x.shape[3], s, p)
+        if not ctx.needs_input_grad[0]:
+            # stem conv: the input batch never needs a gradient — skip
+            # the whole dgrad kernel
+            gx = None
+        elif shadow_t is not None:
+            # pre-transposed dgrad weight (refreshed once per step by
+            # shadow_sync) — no per-call permute+contiguous copy
+            gx = m.conv2d_dgrad_km_pre(gy, shadow_t, x.shape[2],
+                                       x.shape[3], s, p)
+        else:
+            gx = m.conv2d_dgrad_km(gy, shadow, x.shape[2], x.shape[3],
+                                   s, p)
         if ctx.gslot is not None:
             # write the fp32 grad straight into the flat-grad view; no
             # AccumulateGrad add kernel runs for this parameter
@@ -121,7 +133,7 @@ class _SConv2dShadowFn(torch.autograd.Function):
             e = torch.empty(0, device=x.device, dtype=torch.float32)
             gw = m.conv2d_wgrad_km32(gy, x, shadow.shape[2],
                                      shadow.shape[3], s, p, G, e)
-        return gx, gw, None, None, None, None, None
+        return gx, gw, None, None, None, None, None, None
 
 
 class SConv2d(nn.Module):
@@ -131,8 +143,9 @@ class SConv2d(nn.Module):
         w = torch.empty(G, cout, cin, k, k)
         nn.init.kaiming_normal_(w, mode="fan_out", nonlinearity="relu")
         self.weight = nn.Parameter(w.permute(0, 1, 3, 4, 2).contiguous())
-        self.wshadow = None  # bf16 flat-buffer view (StackedFlatParams)
-        self.wgslot = None   # fp32 flat-grad view (direct wgrad target)
+        self.wshadow = None    # bf16 flat-buffer view (StackedFlatParams)
+        self.wshadow_t = None  # pre-transposed dgrad layout [G,CI,R,S,CO]
+        self.wgslot = None     # fp32 flat-grad view (direct wgrad target)
         self.G, self.stride, self.padding = G, stride, padding
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
@@ -140,8 +153,9 @@ class SConv2d(nn.Module):
             if x.dtype != torch.bfloat16:
                 x = x.bfloat16()
             return _SConv2dShadowFn.apply(x, self.weight, self.wshadow,
-                                          self.wgslot, self.stride,
-                                          self.padding, self.G)
+                                          self.wshadow_t, self.wgslot,
+                                          self.stride, self.padding,
+                                          self.G)
         w = self.weight
         if torch.is_autocast_enabled("cuda") and x.is_cuda:
             dt = torch.get_autocast_dtype("cuda")

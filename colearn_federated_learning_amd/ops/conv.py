@@ -45,13 +45,16 @@ class _Conv2dFn(torch.autograd.Function):
     def backward(ctx, gy):
         x, w = ctx.saved_tensors
         s, p = ctx.stride, ctx.padding
+        need_gx = ctx.needs_input_grad[0]  # stem: input needs no grad
         if _use_native(gy, w, x):
             m = require_ext()
-            gx = m.conv2d_dgrad_km(gy, w, x.shape[2], x.shape[3], s, p)
+            gx = (m.conv2d_dgrad_km(gy, w, x.shape[2], x.shape[3], s, p)
+                  if need_gx else None)
             gw = m.conv2d_wgrad_km(gy, x, w.shape[1], w.shape[2], s, p, 1)
         else:
             ws = w.permute(0, 3, 1, 2)
-            gx = torch.nn.grad.conv2d_input(x.shape, ws, gy, s, p)
+            gx = (torch.nn.grad.conv2d_input(x.shape, ws, gy, s, p)
+                  if need_gx else None)
             gw = torch.nn.grad.conv2d_weight(x, ws.shape, gy, s, p)
             gw = gw.permute(0, 2, 3, 1)
         return gx, gw, None, None

@@ -77,6 +77,10 @@ void launch_scale_rows_mean(const float* g, const float* scale, float* out,
 void launch_topk(const float* v, int64_t n, int64_t k, int* idx, float* val,
                  unsigned* workspace /* >= 1036 u32 */, int* cand,
                  int cand_cap, hipStream_t stream);
+void launch_shadow_sync(const float* w, void* shadow, int64_t n,
+                        const int64_t* segA, const int* segB, int nseg,
+                        void* shadow_t, int64_t transpose_blocks,
+                        hipStream_t stream);
 void launch_head_fwd(const void* x, const float* w, const float* b,
                      const int64_t* y, float* pooled, float* probs,
                      float* loss_n, float* loss_vec, int N, int C, int HW,

@@ -114,6 +114,55 @@ void scale_rows_mean(torch::Tensor g, torch::Tensor scale,
   HIP_CHECK_LAST();
 }
 
+// ---- fused shadow sync (flat cast + per-conv dgrad-layout transposes) ----
+void shadow_sync(torch::Tensor flat_w, torch::Tensor shadow,
+                 torch::Tensor shadow_t, torch::Tensor segA,
+                 torch::Tensor segB, int64_t transpose_blocks) {
+  check_f32(flat_w, "flat_w");
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(flat_w.device());
+  const int nseg = segA.numel() ? static_cast<int>(segA.size(0)) : 0;
+  launch_shadow_sync(
+      flat_w.data_ptr<float>(), shadow.data_ptr(), flat_w.numel(),
+      nseg ? segA.data_ptr<int64_t>() : nullptr,
+      nseg ? segB.data_ptr<int>() : nullptr, nseg,
+      nseg ? shadow_t.data_ptr() : nullptr, transpose_blocks, stream());
+  HIP_CHECK_LAST();
+}
+
+// dgrad with a PRE-TRANSPOSED weight ([G][CI][R][S][CO] bf16, produced by
+// shadow_sync once per step) — no per-call permute+contiguous copy.
+torch::Tensor conv2d_dgrad_km_pre(torch::Tensor gy, torch::Tensor wt,
+                                  int64_t H, int64_t W, int64_t stride,
+                                  int64_t pad) {
+  check_bf16(gy, "gy");
+  check_bf16(wt, "wt");
+  TORCH_CHECK(wt.is_contiguous(), "wt must be contiguous");
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(gy.device());
+  const int64_t G = wt.dim() == 5 ? wt.size(0) : 1;
+  const int wd = wt.dim();
+  ConvShape cs;
+  cs.N = static_cast<int>(gy.size(0) / G);
+  cs.K = static_cast<int>(gy.size(1));
+  cs.P = static_cast<int>(gy.size(2));
+  cs.Q = static_cast<int>(gy.size(3));
+  cs.C = static_cast<int>(wt.size(wd - 4));  // [*, CI, R, S, CO]
+  cs.R = static_cast<int>(wt.size(wd - 3));
+  cs.S = static_cast<int>(wt.size(wd - 2));
+  cs.H = static_cast<int>(H);
+  cs.W = static_cast<int>(W);
+  cs.stride = static_cast<int>(stride);
+  cs.pad = static_cast<int>(pad);
+  conv_shape_fill_magic(cs);
+  auto gc = gy.contiguous(torch::MemoryFormat::ChannelsLast);
+  auto gx = torch::empty({gy.size(0), cs.C, cs.H, cs.W},
+                         gy.options().memory_format(
+                             torch::MemoryFormat::ChannelsLast));
+  launch_conv_dgrad(gc.data_ptr(), wt.data_ptr(), gx.data_ptr(), cs,
+                    static_cast<int>(G), stream());
+  HIP_CHECK_LAST();
+  return gx;
+}
+
 // ---- fused classifier head (avgpool + per-client FC + softmax CE) --------
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> head_fwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor b, torch::Tensor y,
@@ -807,6 +856,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("philox_gaussian_add", &philox_gaussian_add);
   m.def("head_fwd", &head_fwd);
   m.def("head_bwd", &head_bwd);
+  m.def("shadow_sync", &shadow_sync);
+  m.def("conv2d_dgrad_km_pre", &conv2d_dgrad_km_pre);
   m.def("topk_compress", &topk_compress);
   m.def("scatter_add_pairs", &scatter_add_pairs);
   m.def("layernorm_fwd", &layernorm_fwd);
