@@ -98,7 +98,7 @@ class StackedFlatParams:
                 if CO % 32 == 0 and CI % 32 == 0:
                     RS = KH * KW
                     tco, tci = CO // 32, CI // 32
-                    nblk = G * RS * tco * tci
+                    nblk = G * tco * tci  # RS looped inside the block
                     segA.append([off, t_off, blk, blk + nblk])
                     segB.append([CO, CI, RS, tco, tci, G])
                     conv_t.append((mod, t_off, (G, CI, KH, KW, CO)))

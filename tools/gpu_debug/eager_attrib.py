@@ -19,6 +19,21 @@ cfg = driver_config(2).override(
 )
 s = FederatedServer(cfg, device=torch.device("cuda", 0))
 s.run_round(0)  # warm
+
+# trace who calls Tensor.zero_ / torch.zeros during the step
+import traceback
+from collections import Counter
+
+_zero_callers = Counter()
+_orig_zero = torch.Tensor.zero_
+
+
+def _traced_zero(self):
+    _zero_callers["".join(traceback.format_stack()[-4:-1])] += 1
+    return _orig_zero(self)
+
+
+torch.Tensor.zero_ = _traced_zero
 from torch.profiler import ProfilerActivity, profile
 
 eng = s.stacked.full
@@ -39,6 +54,11 @@ print(prof.key_averages(group_by_input_shape=True)
              max_src_column_width=60))
 # kernel-level counts for the tail kernels
 ka = prof.key_averages()
+torch.Tensor.zero_ = _orig_zero
+print("\n---- zero_ call sites ----")
+for stack, n in _zero_callers.most_common(10):
+    print(f"x{n}:\n{stack}")
+
 print("\n---- per-op CUDA kernel counts (fills/copies/elementwise) ----")
 for ev in sorted(ka, key=lambda e: -e.self_device_time_total)[:60]:
     n = ev.key
