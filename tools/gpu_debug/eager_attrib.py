@@ -59,6 +59,37 @@ print("\n---- zero_ call sites ----")
 for stack, n in _zero_callers.most_common(10):
     print(f"x{n}:\n{stack}")
 
+# dispatch-level: catch C++-originated fills/zeros with shapes
+from torch.utils._python_dispatch import TorchDispatchMode
+
+
+class _FillSpy(TorchDispatchMode):
+    def __init__(self):
+        super().__init__()
+        self.hits = Counter()
+
+    def __torch_dispatch__(self, func, types, args=(), kwargs=None):
+        name = str(func)
+        if "zero_" in name or "fill" in name or "copy_" in name:
+            shape = tuple(args[0].shape) if args and hasattr(
+                args[0], "shape") else None
+            self.hits[(name, shape)] += 1
+        return func(*args, **(kwargs or {}))
+
+
+spy = _FillSpy()
+with spy:
+    eng.optimizer.zero_grad()
+    with torch.autocast("cuda", torch.bfloat16):
+        lv = eng.model.loss_terms(x, y)
+    lv.sum().backward()
+    eng.optimizer.step()
+    eng.sflat.sync_shadow()
+    torch.cuda.synchronize()
+print("\n---- dispatch-level fills/zeros/copies ----")
+for (name, shape), n in spy.hits.most_common(25):
+    print(f"x{n:4d}  {name}  {shape}")
+
 print("\n---- per-op CUDA kernel counts (fills/copies/elementwise) ----")
 for ev in sorted(ka, key=lambda e: -e.self_device_time_total)[:60]:
     n = ev.key
