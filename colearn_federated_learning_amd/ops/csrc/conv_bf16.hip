@@ -390,8 +390,12 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
   constexpr int WAVES_N = 4 / WAVES_M;
   constexpr int WN = TN / WAVES_N;  // wave output cols: 64/32/16
   constexpr int NFR = WN / 16;      // B-fragments per wave
-  __shared__ short Gs[2][TM][BK + 2];
-  __shared__ short Xs[2][TN][BK + 2];
+  // XOR-swizzled 16-B slot layout (same scheme as conv_nt): fragment
+  // reads are single conflict-free ds_read_b128s instead of 8 scalar
+  // u16 loads — the wgrad loop was issue-bound at 0.76 MFMA-busy ratio
+  // on exactly those reads (profiles/pmc_mfma_evidence.md)
+  __shared__ short Gs[2][TM * BK];
+  __shared__ short Xs[2][TN * BK];
   const int g = blockIdx.z;
   const short* gyg = gy + static_cast<int64_t>(g) * NPQ * cs.K;
   const short* xg =
@@ -468,17 +472,27 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
   };
   auto write_chunk = [&](int buf) {
     if (G_ALL || kkg < BK) {
+      const int slot_g = kkg >> 3, sub_g = (kkg & 7) * 2;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        Gs[buf][e16g + j][kkg] = gv0[j];
-        Gs[buf][e16g + 8 + j][kkg] = gv1[j];
+        *reinterpret_cast<short*>(
+            reinterpret_cast<char*>(Gs[buf]) +
+            lds_slot_byte_t<BK>(e16g + j, slot_g) + sub_g) = gv0[j];
+        *reinterpret_cast<short*>(
+            reinterpret_cast<char*>(Gs[buf]) +
+            lds_slot_byte_t<BK>(e16g + 8 + j, slot_g) + sub_g) = gv1[j];
       }
     }
     if (X_ALL || kkx < BK) {
+      const int slot_x = kkx >> 3, sub_x = (kkx & 7) * 2;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        Xs[buf][e16x + j][kkx] = xv0[j];
-        Xs[buf][e16x + 8 + j][kkx] = xv1[j];
+        *reinterpret_cast<short*>(
+            reinterpret_cast<char*>(Xs[buf]) +
+            lds_slot_byte_t<BK>(e16x + j, slot_x) + sub_x) = xv0[j];
+        *reinterpret_cast<short*>(
+            reinterpret_cast<char*>(Xs[buf]) +
+            lds_slot_byte_t<BK>(e16x + 8 + j, slot_x) + sub_x) = xv1[j];
       }
     }
   };
@@ -494,20 +508,18 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
     const bool has_next = k0 + BK < k_end;
     if (has_next) load_chunk(k0 + BK);
     const int lane = threadIdx.x % kWave;
-    const int fr = lane & 15, kg = (lane >> 4) * 8;
+    const int fr = lane & 15, slot = lane >> 4;
     short8v_t af[4], bfr[NFR];
 #pragma unroll
-    for (int m = 0; m < 4; ++m) {
-      const short* p = &Gs[cur][wm + m * 16 + fr][kg];
+    for (int m = 0; m < 4; ++m)
+      af[m] = *reinterpret_cast<const short8v_t*>(
+          reinterpret_cast<const char*>(Gs[cur]) +
+          lds_slot_byte_t<BK>(wm + m * 16 + fr, slot));
 #pragma unroll
-      for (int j = 0; j < 8; ++j) af[m][j] = p[j];
-    }
-#pragma unroll
-    for (int n = 0; n < NFR; ++n) {
-      const short* p = &Xs[cur][wn + n * 16 + fr][kg];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) bfr[n][j] = p[j];
-    }
+    for (int n = 0; n < NFR; ++n)
+      bfr[n] = *reinterpret_cast<const short8v_t*>(
+          reinterpret_cast<const char*>(Xs[cur]) +
+          lds_slot_byte_t<BK>(wn + n * 16 + fr, slot));
 #pragma unroll
     for (int m = 0; m < 4; ++m)
 #pragma unroll
