@@ -390,12 +390,21 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
   constexpr int WAVES_N = 4 / WAVES_M;
   constexpr int WN = TN / WAVES_N;  // wave output cols: 64/32/16
   constexpr int NFR = WN / 16;      // B-fragments per wave
-  // XOR-swizzled 16-B slot layout (same scheme as conv_nt): fragment
-  // reads are single conflict-free ds_read_b128s instead of 8 scalar
-  // u16 loads — the wgrad loop was issue-bound at 0.76 MFMA-busy ratio
-  // on exactly those reads (profiles/pmc_mfma_evidence.md)
+  // XOR-swizzled 16-B slot layout: fragment reads are single
+  // conflict-free ds_read_b128s instead of 8 scalar u16 loads — the
+  // wgrad loop was issue-bound at 0.76 MFMA-busy ratio on exactly those
+  // reads (profiles/pmc_mfma_evidence.md).  The swizzle
+  // adds a (row>>4) term on top of conv_nt's (row>>2): the staging
+  // writes scatter rows 16 apart (the transpose), which land on
+  // identical banks under the plain scheme (64-B rows: 16*16 ≡ 0 mod 32
+  // banks) — the extra term spreads them while fragment reads (16
+  // CONSECUTIVE rows, so row>>4 constant) stay conflict-free.
   __shared__ short Gs[2][TM * BK];
   __shared__ short Xs[2][TN * BK];
+  auto wg_slot = [](int row, int slot) {
+    return row * (BK * 2) +
+           ((slot ^ ((row >> 2) & 3) ^ ((row >> 4) & 3)) * 16);
+  };
   const int g = blockIdx.z;
   const short* gyg = gy + static_cast<int64_t>(g) * NPQ * cs.K;
   const short* xg =
@@ -477,10 +486,10 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
       for (int j = 0; j < 8; ++j) {
         *reinterpret_cast<short*>(
             reinterpret_cast<char*>(Gs[buf]) +
-            lds_slot_byte_t<BK>(e16g + j, slot_g) + sub_g) = gv0[j];
+            wg_slot(e16g + j, slot_g) + sub_g) = gv0[j];
         *reinterpret_cast<short*>(
             reinterpret_cast<char*>(Gs[buf]) +
-            lds_slot_byte_t<BK>(e16g + 8 + j, slot_g) + sub_g) = gv1[j];
+            wg_slot(e16g + 8 + j, slot_g) + sub_g) = gv1[j];
       }
     }
     if (X_ALL || kkx < BK) {
@@ -489,10 +498,10 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
       for (int j = 0; j < 8; ++j) {
         *reinterpret_cast<short*>(
             reinterpret_cast<char*>(Xs[buf]) +
-            lds_slot_byte_t<BK>(e16x + j, slot_x) + sub_x) = xv0[j];
+            wg_slot(e16x + j, slot_x) + sub_x) = xv0[j];
         *reinterpret_cast<short*>(
             reinterpret_cast<char*>(Xs[buf]) +
-            lds_slot_byte_t<BK>(e16x + 8 + j, slot_x) + sub_x) = xv1[j];
+            wg_slot(e16x + 8 + j, slot_x) + sub_x) = xv1[j];
       }
     }
   };
@@ -514,12 +523,12 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
     for (int m = 0; m < 4; ++m)
       af[m] = *reinterpret_cast<const short8v_t*>(
           reinterpret_cast<const char*>(Gs[cur]) +
-          lds_slot_byte_t<BK>(wm + m * 16 + fr, slot));
+          wg_slot(wm + m * 16 + fr, slot));
 #pragma unroll
     for (int n = 0; n < NFR; ++n)
       bfr[n] = *reinterpret_cast<const short8v_t*>(
           reinterpret_cast<const char*>(Xs[cur]) +
-          lds_slot_byte_t<BK>(wn + n * 16 + fr, slot));
+          wg_slot(wn + n * 16 + fr, slot));
 #pragma unroll
     for (int m = 0; m < 4; ++m)
 #pragma unroll
