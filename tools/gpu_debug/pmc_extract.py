@@ -31,7 +31,7 @@ def main():
             # dispatches through rocpd_event / kernel dispatch event_id
             try:
                 q = f"""
-                SELECT s.display_name, ip.name, SUM(pe.value),
+                SELECT ks.display_name, ip.name, SUM(pe.value),
                        COUNT(*)
                 FROM rocpd_pmc_event_{u} pe
                 JOIN rocpd_info_pmc_{u} ip ON pe.pmc_id = ip.id
@@ -39,7 +39,6 @@ def main():
                      ON pe.event_id = d.event_id
                 JOIN rocpd_info_kernel_symbol_{u} ks
                      ON d.kernel_id = ks.id
-                JOIN rocpd_string_{u} s ON ks.display_name_id = s.id
                 GROUP BY 1, 2 ORDER BY 1, 2"""
                 for name, cn, v, n in cur.execute(q):
                     out.write(f"  {cn:28s} {v:18.0f} n={n:6d} "
