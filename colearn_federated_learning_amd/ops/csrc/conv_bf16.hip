@@ -479,11 +479,17 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
       }
     }
   };
+  // scatter writes rotate their row order by the lane's 16-row group
+  // (j -> (jj + row>>4) & 7): combined with the row>>4 swizzle term this
+  // makes each write cycle's 8 concurrent row-groups hit 8 DISTINCT
+  // (parity, slot) bank groups — verified injective by enumeration.
   auto write_chunk = [&](int buf) {
     if (G_ALL || kkg < BK) {
       const int slot_g = kkg >> 3, sub_g = (kkg & 7) * 2;
+      const int m_g = e16g >> 4;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
+      for (int jj = 0; jj < 8; ++jj) {
+        const int j = (jj + m_g) & 7;
         *reinterpret_cast<short*>(
             reinterpret_cast<char*>(Gs[buf]) +
             wg_slot(e16g + j, slot_g) + sub_g) = gv0[j];
@@ -494,8 +500,10 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
     }
     if (X_ALL || kkx < BK) {
       const int slot_x = kkx >> 3, sub_x = (kkx & 7) * 2;
+      const int m_x = e16x >> 4;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
+      for (int jj = 0; jj < 8; ++jj) {
+        const int j = (jj + m_x) & 7;
         *reinterpret_cast<short*>(
             reinterpret_cast<char*>(Xs[buf]) +
             wg_slot(e16x + j, slot_x) + sub_x) = xv0[j];
