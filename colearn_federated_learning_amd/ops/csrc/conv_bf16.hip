@@ -401,8 +401,14 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
   // CONSECUTIVE rows, so row>>4 constant) stay conflict-free.
   __shared__ short Gs[2][TM * BK];
   __shared__ short Xs[2][TN * BK];
+  // physical-row remap (row ^ bit6->bit0) + slot XOR swizzle: together
+  // they put all 3 bits of a write lane's 16-row group index into the
+  // bank bits, so a staging write cycle's 8 concurrent row-groups hit 8
+  // distinct bank groups; fragment reads (16 consecutive logical rows)
+  // stay conflict-free — the remap only permutes parity within a group.
   auto wg_slot = [](int row, int slot) {
-    return row * (BK * 2) +
+    const int pr = row ^ ((row >> 6) & 1);
+    return pr * (BK * 2) +
            ((slot ^ ((row >> 2) & 3) ^ ((row >> 4) & 3)) * 16);
   };
   const int g = blockIdx.z;
@@ -479,17 +485,11 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
       }
     }
   };
-  // scatter writes rotate their row order by the lane's 16-row group
-  // (j -> (jj + row>>4) & 7): combined with the row>>4 swizzle term this
-  // makes each write cycle's 8 concurrent row-groups hit 8 DISTINCT
-  // (parity, slot) bank groups — verified injective by enumeration.
   auto write_chunk = [&](int buf) {
     if (G_ALL || kkg < BK) {
       const int slot_g = kkg >> 3, sub_g = (kkg & 7) * 2;
-      const int m_g = e16g >> 4;
 #pragma unroll
-      for (int jj = 0; jj < 8; ++jj) {
-        const int j = (jj + m_g) & 7;
+      for (int j = 0; j < 8; ++j) {
         *reinterpret_cast<short*>(
             reinterpret_cast<char*>(Gs[buf]) +
             wg_slot(e16g + j, slot_g) + sub_g) = gv0[j];
@@ -500,10 +500,8 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
     }
     if (X_ALL || kkx < BK) {
       const int slot_x = kkx >> 3, sub_x = (kkx & 7) * 2;
-      const int m_x = e16x >> 4;
 #pragma unroll
-      for (int jj = 0; jj < 8; ++jj) {
-        const int j = (jj + m_x) & 7;
+      for (int j = 0; j < 8; ++j) {
         *reinterpret_cast<short*>(
             reinterpret_cast<char*>(Xs[buf]) +
             wg_slot(e16x + j, slot_x) + sub_x) = xv0[j];
