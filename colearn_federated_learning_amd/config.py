@@ -38,6 +38,7 @@ class ModelConfig:
     num_classes: int = 10
     in_channels: int = 3
     image_size: int = 32
+    dropout: float = 0.1  # bert_base only (0 => deterministic paths)
 
 
 @dataclass

@@ -233,7 +233,12 @@ class StackedEngine:
             cfg.train, flat_ref=self.flat_ref,
         )
         self.runner = None
-        if device.type == "cuda" and cfg.train.use_graphs:
+        if cfg.model.name == "bert_base":
+            # hipGraph replays with sort/embedding kernels intermittently
+            # fault on ROCm 7.2 (see steprunner.py), and the stacked BERT
+            # kernels are large enough that graphs buy ~nothing — eager.
+            pass
+        elif device.type == "cuda" and cfg.train.use_graphs:
             from .steprunner import GraphedStepRunner
 
             bs = cfg.train.batch_size
@@ -300,7 +305,8 @@ class StackedEngine:
                 ys = torch.stack([b[1] for b in batches])
                 x = xs.reshape(-1, *xs.shape[2:]).to(self.device,
                                                      non_blocking=True)
-                y = ys.reshape(-1).to(self.device, non_blocking=True)
+                y = ys.reshape(-1, *ys.shape[2:]).to(self.device,
+                                                     non_blocking=True)
                 if self.runner is not None and self.runner.matches(x):
                     loss_vec = self.runner.step(x, y)
                 else:
@@ -374,7 +380,8 @@ def stackable(cfg: FederatedConfig, clients: List[FederatedClient]) -> bool:
     shard size (lock-step batches) and the model family is implemented."""
     if not cfg.train.use_stacked:
         return False
-    if cfg.model.name not in ("logreg", "resnet18", "resnet50"):
+    if cfg.model.name not in ("logreg", "resnet18", "resnet50",
+                              "bert_base"):
         return False
     if cfg.dp.enabled:  # per-sample path stays sequential
         return False

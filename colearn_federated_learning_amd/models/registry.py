@@ -32,5 +32,6 @@ def build_model(model_cfg: ModelConfig, data_cfg: DataConfig,
         return BertForMaskedLM(BertConfig(
             vocab_size=data_cfg.vocab_size,
             max_position_embeddings=max(data_cfg.seq_len, 512),
+            dropout=model_cfg.dropout,
         ))
     raise ValueError(f"unknown model {name!r}")

@@ -1,0 +1,199 @@
+"""Stacked-client BERT-base MLM: G clients as batched MFMA GEMMs.
+
+VERDICT r1 next #4: BERT was the least-native path — 8 clients trained
+SEQUENTIALLY, 222 GEMM launches per client step, 65% of step time in
+small torch kernels (gpurun_out/bert_prof.txt).  Stacking trains all G
+clients at once with per-client weights:
+
+- every projection is ONE batched GEMM over G clients
+  (SLinear: [G, M, in] @ [G, out, in]^T via the bmm MFMA kernels),
+- LayerNorm runs the grouped native kernel (weight [G, H], client-major
+  rows — ops/norms.py groups path),
+- attention is the same batched score/context GEMMs with batch G*B*heads,
+- embeddings gather from per-client tables [G, V, H],
+- the MLM head keeps the fixed-candidate-budget trick (25% of tokens,
+  stable argsort per client) so shapes stay static per step.
+
+Parameter attribute ORDER mirrors models/bert.py exactly, so per-client
+delta extraction stays a segment-wise reshape (federated/stacked.py).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Tuple
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ..ops.gemm import colearn_matmul, colearn_matmul_nt
+from ..ops.norms import _LayerNormFn
+from .base import ColearnModel
+from .bert import BertConfig
+from .stacked import SLinear
+
+
+class SLayerNorm(nn.Module):
+    """Per-client LayerNorm: weight/bias [G, H]; input rows client-major."""
+
+    def __init__(self, G: int, hidden: int, eps: float = 1e-12) -> None:
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(G, hidden))
+        self.bias = nn.Parameter(torch.zeros(G, hidden))
+        self.G, self.eps = G, eps
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        y, _, _ = _LayerNormFn.apply(x, self.weight, self.bias, self.eps,
+                                     self.G)
+        return y
+
+
+class SEmbedding(nn.Module):
+    """Per-client embedding tables [G, n, H]."""
+
+    def __init__(self, G: int, n: int, hidden: int) -> None:
+        super().__init__()
+        self.weight = nn.Parameter(torch.empty(G, n, hidden))
+        nn.init.normal_(self.weight, std=0.02)
+        self.G = G
+
+    def forward(self, ids: torch.Tensor) -> torch.Tensor:
+        # ids [G*B, s] client-major -> [G*B, s, H]
+        G = self.G
+        ids3 = ids.reshape(G, -1)
+        g_idx = torch.arange(G, device=ids.device).unsqueeze(1)
+        out = self.weight[g_idx, ids3]  # [G, B*s, H]
+        return out.reshape(*ids.shape, self.weight.shape[-1])
+
+
+class SBertSelfAttention(nn.Module):
+    def __init__(self, G: int, cfg: BertConfig) -> None:
+        super().__init__()
+        self.h = cfg.num_heads
+        self.dh = cfg.hidden_size // cfg.num_heads
+        self.qkv = SLinear(G, cfg.hidden_size, 3 * cfg.hidden_size)
+        self.out = SLinear(G, cfg.hidden_size, cfg.hidden_size)
+        self.drop = nn.Dropout(cfg.dropout)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        b, s, d = x.shape  # b = G*B
+        qkv = self.qkv(x.reshape(b * s, d)).view(b, s, 3, self.h, self.dh)
+        q, k, v = (
+            qkv[:, :, i].transpose(1, 2).reshape(b * self.h, s, self.dh)
+            for i in range(3)
+        )
+        scores = colearn_matmul_nt(q, k) / math.sqrt(self.dh)
+        probs = self.drop(F.softmax(scores, dim=-1))
+        ctx = colearn_matmul(probs, v)
+        ctx = (ctx.view(b, self.h, s, self.dh).transpose(1, 2)
+               .reshape(b * s, d))
+        return self.out(ctx).view(b, s, d)
+
+
+class SBertLayer(nn.Module):
+    def __init__(self, G: int, cfg: BertConfig) -> None:
+        super().__init__()
+        self.attn = SBertSelfAttention(G, cfg)
+        self.ln1 = SLayerNorm(G, cfg.hidden_size, cfg.layer_norm_eps)
+        self.fc1 = SLinear(G, cfg.hidden_size, cfg.intermediate_size)
+        self.fc2 = SLinear(G, cfg.intermediate_size, cfg.hidden_size)
+        self.ln2 = SLayerNorm(G, cfg.hidden_size, cfg.layer_norm_eps)
+        self.drop = nn.Dropout(cfg.dropout)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        b, s, d = x.shape
+        x = self.ln1(x + self.drop(self.attn(x)))
+        ff = self.fc2(F.gelu(self.fc1(x.reshape(b * s, d)))).view(b, s, d)
+        x = self.ln2(x + self.drop(ff))
+        return x
+
+
+class StackedBert(ColearnModel):
+    def __init__(self, G: int, cfg: BertConfig) -> None:
+        super().__init__()
+        self.G = G
+        self.cfg = cfg
+        self.tok_emb = SEmbedding(G, cfg.vocab_size, cfg.hidden_size)
+        self.pos_emb = SEmbedding(G, cfg.max_position_embeddings,
+                                  cfg.hidden_size)
+        self.emb_ln = SLayerNorm(G, cfg.hidden_size, cfg.layer_norm_eps)
+        self.drop = nn.Dropout(cfg.dropout)
+        self.layers = nn.ModuleList(
+            SBertLayer(G, cfg) for _ in range(cfg.num_layers)
+        )
+        self.head_fc = SLinear(G, cfg.hidden_size, cfg.hidden_size)
+        self.head_ln = SLayerNorm(G, cfg.hidden_size, cfg.layer_norm_eps)
+        self.head_bias = nn.Parameter(torch.zeros(G, cfg.vocab_size))
+
+    # -- encoder -------------------------------------------------------
+    def encode(self, ids: torch.Tensor) -> torch.Tensor:
+        n, s = ids.shape  # n = G*B
+        H = self.cfg.hidden_size
+        # per-client position slice: pos table is [G, maxpos, H]
+        pos_vec = self.pos_emb.weight[:, :s]  # [G, s, H]
+        x = (self.tok_emb(ids).view(self.G, -1, s, H)
+             + pos_vec[:, None]).view(n, s, H)
+        if torch.is_autocast_enabled("cuda") and x.is_cuda:
+            x = x.to(torch.get_autocast_dtype("cuda"))
+        x = self.drop(self.emb_ln(x))
+        for layer in self.layers:
+            x = layer(x)
+        return x
+
+    def forward(self, ids: torch.Tensor) -> torch.Tensor:
+        n, s = ids.shape
+        h = self.encode(ids)
+        d = h.shape[-1]
+        h = self.head_ln(F.gelu(
+            self.head_fc(h.reshape(n * s, d))))
+        h3 = h.view(self.G, -1, d)
+        w = self.tok_emb.weight
+        if h3.dtype != w.dtype:
+            w = w.to(h3.dtype)
+        logits = colearn_matmul_nt(h3, w) + self.head_bias[:, None, :]
+        return logits.view(n, s, -1)
+
+    # -- MLM loss ------------------------------------------------------
+    def loss_terms(self, x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
+        """Per-client mean MLM loss [G]; fixed candidate budget (25% of
+        tokens per client, stable argsort — static shapes, graph-safe;
+        see models/bert.py loss for rationale)."""
+        G = self.G
+        hidden = self.encode(x)          # [G*B, s, H]
+        d = hidden.shape[-1]
+        h3 = hidden.reshape(G, -1, d)    # [G, B*s, H]
+        y2 = y.reshape(G, -1)
+        k = max((y2.shape[1] * 25 + 99) // 100, 8)
+        order = y2.ne(-100).to(torch.int8).argsort(
+            dim=1, descending=True, stable=True)[:, :k]
+        hk = torch.gather(h3, 1, order[..., None].expand(-1, -1, d))
+        yk = torch.gather(y2, 1, order)  # [G, k]
+        hk = self.head_ln(F.gelu(self.head_fc(hk.reshape(G * k, d))))
+        w = self.tok_emb.weight
+        if hk.dtype != w.dtype:
+            w = w.to(hk.dtype)
+        logits = (colearn_matmul_nt(hk.view(G, k, d), w)
+                  + self.head_bias[:, None, :])     # [G, k, V]
+        ce = F.cross_entropy(
+            logits.float().reshape(G * k, -1), yk.reshape(-1),
+            ignore_index=-100, reduction="none",
+        ).view(G, k)
+        valid = yk.ne(-100)
+        return (ce * valid).sum(dim=1) / valid.sum(dim=1).clamp_min(1)
+
+    def loss(self, x, y, params=None, buffers=None):
+        return self.loss_terms(x, y).sum()
+
+    @torch.no_grad()
+    def loss_and_correct(
+        self, x: torch.Tensor, y: torch.Tensor
+    ) -> Tuple[torch.Tensor, int, int]:
+        logits = self(x)
+        loss = F.cross_entropy(
+            logits.float().reshape(-1, logits.shape[-1]), y.reshape(-1),
+            ignore_index=-100,
+        )
+        mask = y != -100
+        correct = int((logits.argmax(-1)[mask] == y[mask]).sum())
+        return loss, correct, int(mask.sum())

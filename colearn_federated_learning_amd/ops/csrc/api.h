@@ -97,12 +97,12 @@ void launch_scatter_add_pairs(float* out, const int* idx, const float* val,
                               int64_t k, hipStream_t stream);
 void launch_layernorm_fwd(const void* x, const float* w, const float* b,
                           void* y, float* mean, float* invstd, int64_t rows,
-                          int H, float eps, hipStream_t stream);
+                          int H, float eps, int G, hipStream_t stream);
 int norms_reduce_grid(int64_t n8);
 void launch_layernorm_bwd(const void* gy, const void* x, const float* w,
                           const float* mean, const float* invstd, void* gx,
                           float* gw_gb, float* partials, int64_t rows, int H,
-                          hipStream_t stream);
+                          int G, hipStream_t stream);
 void launch_batchnorm_fwd(const void* x, const void* res, const float* w,
                           const float* b, float* running_mean,
                           float* running_var, void* y, float* mean,

@@ -262,12 +262,15 @@ void scatter_add_pairs(torch::Tensor out, torch::Tensor idx,
 
 // ---- LayerNorm ------------------------------------------------------------
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> layernorm_fwd(
-    torch::Tensor x, torch::Tensor w, torch::Tensor b, double eps) {
+    torch::Tensor x, torch::Tensor w, torch::Tensor b, double eps,
+    int64_t G) {
   check_bf16(x, "x");
   const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   auto xc = x.contiguous();
   const int H = static_cast<int>(x.size(-1));
   const int64_t rows = x.numel() / H;
+  TORCH_CHECK(w.numel() == G * H && rows % G == 0,
+              "grouped LN: weight [G*H], rows % G == 0");
   auto y = torch::empty_like(xc);
   auto mean = torch::empty({rows}, x.options().dtype(torch::kFloat32));
   auto invstd = torch::empty_like(mean);
@@ -275,7 +278,7 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> layernorm_fwd(
       xc.data_ptr(), w.data_ptr<float>(),
       b.data_ptr<float>(), y.data_ptr(),
       mean.data_ptr<float>(), invstd.data_ptr<float>(), rows, H,
-      static_cast<float>(eps), stream());
+      static_cast<float>(eps), static_cast<int>(G), stream());
   HIP_CHECK_LAST();
   auto stat_shape = x.sizes().vec();
   stat_shape.pop_back();
@@ -284,7 +287,7 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> layernorm_fwd(
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> layernorm_bwd(
     torch::Tensor gy, torch::Tensor x, torch::Tensor w, torch::Tensor mean,
-    torch::Tensor invstd) {
+    torch::Tensor invstd, int64_t G) {
   check_bf16(gy, "gy");
   const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   auto gc = gy.contiguous(), xc = x.contiguous();
@@ -292,17 +295,18 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> layernorm_bwd(
   const int H = static_cast<int>(x.size(-1));
   const int64_t rows = x.numel() / H;
   auto gx = torch::empty_like(xc);
-  auto gw_gb = torch::empty({2 * H}, x.options().dtype(torch::kFloat32));
-  auto partials = torch::empty({64 * 2 * H},
+  auto gw_gb = torch::empty({G, 2 * H},
+                            x.options().dtype(torch::kFloat32));
+  auto partials = torch::empty({G * 64 * 2 * H},
                                x.options().dtype(torch::kFloat32));
   launch_layernorm_bwd(
       gc.data_ptr(), xc.data_ptr(), w.data_ptr<float>(),
       mc.data_ptr<float>(), ic.data_ptr<float>(), gx.data_ptr(),
       gw_gb.data_ptr<float>(), partials.data_ptr<float>(), rows, H,
-      stream());
+      static_cast<int>(G), stream());
   HIP_CHECK_LAST();
-  return {gx, gw_gb.narrow(0, 0, H).clone(),
-          gw_gb.narrow(0, H, H).clone()};
+  return {gx, gw_gb.narrow(1, 0, H).clone(),
+          gw_gb.narrow(1, H, H).clone()};
 }
 
 // ---- BatchNorm (channels-last) -------------------------------------------

@@ -26,12 +26,12 @@ int norms_reduce_grid(int64_t n8) { return elemwise_grid(n8, 256, 256); }
 // ==== LayerNorm ============================================================
 // x: [rows, H] bf16, w/b: [H] fp32, y: bf16, mean/invstd: [rows] fp32
 __global__ void layernorm_fwd_kernel(const bf16* __restrict__ x,
-                                     const float* __restrict__ w,
-                                     const float* __restrict__ b,
+                                     const float* __restrict__ wg,
+                                     const float* __restrict__ bg,
                                      bf16* __restrict__ y,
                                      float* __restrict__ mean,
                                      float* __restrict__ invstd, int64_t rows,
-                                     int H, float eps) {
+                                     int H, float eps, int64_t rpg) {
   const int wid = threadIdx.x / kWave;
   const int lane = threadIdx.x % kWave;
   const int waves = blockDim.x / kWave;
@@ -39,6 +39,9 @@ __global__ void layernorm_fwd_kernel(const bf16* __restrict__ x,
   for (int64_t row = static_cast<int64_t>(blockIdx.x) * waves + wid;
        row < rows; row += static_cast<int64_t>(gridDim.x) * waves) {
     const short* xr = reinterpret_cast<const short*>(x + row * H);
+    // grouped (stacked-client) weights: rows are client-major
+    const float* w = wg + (row / rpg) * H;
+    const float* b = bg + (row / rpg) * H;
     float s = 0.f, ss = 0.f;
     for (int h = lane * 8; h + 7 < H; h += kWave * 8) {
       const short8 v = *reinterpret_cast<const short8*>(xr + h);
@@ -84,9 +87,9 @@ __global__ void layernorm_fwd_kernel(const bf16* __restrict__ x,
 // per-row part of the backward: gx.  (gw/gb accumulate in a column kernel.)
 __global__ void layernorm_bwd_gx_kernel(
     const bf16* __restrict__ gy, const bf16* __restrict__ x,
-    const float* __restrict__ w, const float* __restrict__ mean,
+    const float* __restrict__ wg, const float* __restrict__ mean,
     const float* __restrict__ invstd, bf16* __restrict__ gx, int64_t rows,
-    int H) {
+    int H, int64_t rpg) {
   const int wid = threadIdx.x / kWave;
   const int lane = threadIdx.x % kWave;
   const int waves = blockDim.x / kWave;
@@ -95,6 +98,7 @@ __global__ void layernorm_bwd_gx_kernel(
        row < rows; row += static_cast<int64_t>(gridDim.x) * waves) {
     const short* gr = reinterpret_cast<const short*>(gy + row * H);
     const short* xr = reinterpret_cast<const short*>(x + row * H);
+    const float* w = wg + (row / rpg) * H;
     const float mu = mean[row], istd = invstd[row];
     float s1 = 0.f, s2 = 0.f;  // sum(wg), sum(wg * xhat)
     for (int h = lane; h < H; h += kWave) {
@@ -119,11 +123,16 @@ __global__ void layernorm_bwd_gx_kernel(
 __global__ void layernorm_bwd_gwb_kernel(
     const bf16* __restrict__ gy, const bf16* __restrict__ x,
     const float* __restrict__ mean, const float* __restrict__ invstd,
-    float* __restrict__ partials /*[gridDim.y][2H]*/, int64_t rows, int H) {
+    float* __restrict__ partials /*[G][gridDim.y][2H]*/, int64_t rpg,
+    int H) {
   const int h = blockIdx.x * blockDim.x + threadIdx.x;
   if (h >= H) return;
+  // blockIdx.z = group (stacked client); rows client-major
+  const int64_t row0 = static_cast<int64_t>(blockIdx.z) * rpg;
+  const int64_t rows = row0 + rpg;
+  partials += static_cast<int64_t>(blockIdx.z) * gridDim.y * 2 * H;
   float aw = 0.f, ab = 0.f;
-  for (int64_t row = blockIdx.y; row < rows; row += gridDim.y) {
+  for (int64_t row = row0 + blockIdx.y; row < rows; row += gridDim.y) {
     const float g = bits2f(reinterpret_cast<const short*>(gy)[row * H + h]);
     const float xv = bits2f(reinterpret_cast<const short*>(x)[row * H + h]);
     aw += g * (xv - mean[row]) * invstd[row];
@@ -529,19 +538,19 @@ __global__ void bn_bwd_gx_kernel(const bf16* __restrict__ gym,
 // ---- launchers ------------------------------------------------------------
 void launch_layernorm_fwd(const void* x, const float* w, const float* b,
                           void* y, float* mean, float* invstd, int64_t rows,
-                          int H, float eps, hipStream_t stream) {
+                          int H, float eps, int G, hipStream_t stream) {
   const int waves = 4;  // 256-thread block, one wave per row
   int64_t g = (rows + waves - 1) / waves;
   const int grid = static_cast<int>(g < 2048 ? g : 2048);
   layernorm_fwd_kernel<<<grid, waves * kWave, 0, stream>>>(
       static_cast<const bf16*>(x), w, b, static_cast<bf16*>(y), mean,
-      invstd, rows, H, eps);
+      invstd, rows, H, eps, rows / G);
 }
 
 void launch_layernorm_bwd(const void* gy_, const void* x_, const float* w,
                           const float* mean, const float* invstd, void* gx_,
-                          float* gw_gb /*[2H] out*/, float* partials,
-                          int64_t rows, int H, hipStream_t stream) {
+                          float* gw_gb /*[G*2H] out*/, float* partials,
+                          int64_t rows, int H, int G, hipStream_t stream) {
   const bf16* gy = static_cast<const bf16*>(gy_);
   const bf16* x = static_cast<const bf16*>(x_);
   bf16* gx = static_cast<bf16*>(gx_);
@@ -549,12 +558,12 @@ void launch_layernorm_bwd(const void* gy_, const void* x_, const float* w,
   int64_t g = (rows + waves - 1) / waves;
   const int grid = static_cast<int>(g < 2048 ? g : 2048);
   layernorm_bwd_gx_kernel<<<grid, waves * kWave, 0, stream>>>(
-      gy, x, w, mean, invstd, gx, rows, H);
+      gy, x, w, mean, invstd, gx, rows, H, rows / G);
   const int nchunk = 64;
-  dim3 g2((H + 255) / 256, nchunk);
+  dim3 g2((H + 255) / 256, nchunk, G);
   layernorm_bwd_gwb_kernel<<<g2, 256, 0, stream>>>(gy, x, mean, invstd,
-                                                   partials, rows, H);
-  partial_combine_kernel<<<(2 * H + 3) / 4, 256, 0, stream>>>(
+                                                   partials, rows / G, H);
+  partial_combine_kernel<<<dim3((2 * H + 3) / 4, G), 256, 0, stream>>>(
       partials, gw_gb, 2 * H, nchunk);
 }
 

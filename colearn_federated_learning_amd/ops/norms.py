@@ -158,15 +158,34 @@ class ColearnBatchNorm2d(nn.Module):
 
 
 class _LayerNormFn(torch.autograd.Function):
+    """LayerNorm over the last dim; ``groups`` > 1 = stacked-client LN
+    (weight/bias [G, H], client-major rows — the BERT stacking path)."""
+
     generate_vmap_rule = True
 
     @staticmethod
-    def forward(x, weight, bias, eps: float):
+    def forward(x, weight, bias, eps: float, groups: int = 1):
         if _native(x):
             y, mean, invstd = require_ext().layernorm_fwd(
-                x.contiguous(), weight, bias, eps
+                x.contiguous(), weight.view(-1), bias.view(-1), eps,
+                groups,
             )
             return y, mean, invstd
+        if groups > 1:
+            # CPU oracle: per-client loop
+            G = groups
+            rows = x.numel() // x.shape[-1]
+            x3 = x.reshape(G, rows // G, x.shape[-1])
+            w2, b2 = weight.view(G, -1), bias.view(G, -1)
+            xf = x3.float()
+            mean = xf.mean(dim=-1)
+            var = xf.var(dim=-1, unbiased=False)
+            invstd = (var + eps).rsqrt()
+            y = ((xf - mean[..., None]) * invstd[..., None]
+                 * w2[:, None, :] + b2[:, None, :])
+            return (y.to(x.dtype).reshape(x.shape),
+                    mean.reshape(x.shape[:-1]),
+                    invstd.reshape(x.shape[:-1]))
         xf = x.float()
         mean = xf.mean(dim=-1)
         var = xf.var(dim=-1, unbiased=False)
@@ -180,17 +199,41 @@ class _LayerNormFn(torch.autograd.Function):
         x, weight = inputs[0], inputs[1]
         _, mean, invstd = output
         ctx.save_for_backward(x, weight, mean, invstd)
+        ctx.groups = inputs[4] if len(inputs) > 4 else 1
+        ctx.n_inputs = len(inputs)
         ctx.mark_non_differentiable(output[1], output[2])
         ctx.set_materialize_grads(False)  # see _BatchNormFn
 
     @staticmethod
     def backward(ctx, gy, _gmean, _ginvstd):
         x, weight, mean, invstd = ctx.saved_tensors
+        G = ctx.groups
         gy = gy.contiguous()
         if _native(x):
             gx, gw, gb = require_ext().layernorm_bwd(
-                gy, x.contiguous(), weight, mean, invstd
+                gy, x.contiguous(), weight.view(-1), mean, invstd, G
             )
+            if G > 1:
+                gw = gw.view(weight.shape)
+                gb = gb.view(weight.shape)
+            else:
+                gw, gb = gw.view(-1), gb.view(-1)
+        elif G > 1:
+            h = x.shape[-1]
+            rows = x.numel() // h
+            x3 = x.reshape(G, rows // G, h).float()
+            gy3 = gy.reshape(G, rows // G, h).float()
+            m3 = mean.reshape(G, rows // G)
+            i3 = invstd.reshape(G, rows // G)
+            w2 = weight.view(G, h)
+            xhat = (x3 - m3[..., None]) * i3[..., None]
+            gw = (gy3 * xhat).sum(dim=1).view(weight.shape)
+            gb = gy3.sum(dim=1).view(weight.shape)
+            wg = gy3 * w2[:, None, :]
+            gx = (i3[..., None] * (
+                wg - wg.mean(dim=-1, keepdim=True)
+                - xhat * (wg * xhat).mean(dim=-1, keepdim=True)
+            )).to(x.dtype).reshape(x.shape)
         else:
             xf, gyf = x.float(), gy.float()
             xhat = (xf - mean[..., None]) * invstd[..., None]
@@ -203,7 +246,8 @@ class _LayerNormFn(torch.autograd.Function):
                 - xhat * (wg * xhat).mean(dim=-1, keepdim=True)
             )
             gx = gx.to(x.dtype)
-        return gx, gw, gb, None
+        out = [gx, gw, gb, None, None]
+        return tuple(out[: ctx.n_inputs])
 
 
 class ColearnLayerNorm(nn.Module):

@@ -100,3 +100,42 @@ def test_stacked_single_client_matches_sequential():
     assert abs(rec_a["mean_client_loss"] - rec_b["mean_client_loss"]) < 1e-2
     torch.testing.assert_close(s_stack.global_flat, s_seq.global_flat,
                                rtol=1e-3, atol=3e-4)
+
+
+def test_stacked_bert_matches_sequential():
+    """Stacked BERT (batched per-client GEMMs + grouped LN + per-client
+    embeddings) == sequential per-client training, fp32, dropout 0."""
+    cfg = driver_config(4).override(
+        rounds=1, num_clients=2, clients_per_round=None,
+        data__samples_per_client=8, data__seq_len=16,
+        data__vocab_size=256, train__batch_size=4,
+        train__dtype="float32", train__max_local_steps=1,
+        model__dropout=0.0, num_evaluators=0,
+    )
+    s_stack = FederatedServer(cfg)
+    assert s_stack.stacked is not None
+    s_seq = FederatedServer(cfg.override(train__use_stacked=False))
+    assert s_seq.stacked is None
+    rec_a = s_stack.run_round(0)
+    rec_b = s_seq.run_round(0)
+    assert abs(rec_a["mean_client_loss"] - rec_b["mean_client_loss"]) < 1e-3
+    torch.testing.assert_close(s_stack.global_flat, s_seq.global_flat,
+                               rtol=1e-3, atol=5e-5)
+
+
+def test_stacked_bert_param_order_matches_template():
+    from colearn_federated_learning_amd.models.registry import build_model
+    from colearn_federated_learning_amd.models.stacked import (
+        build_stacked_model,
+    )
+
+    cfg = driver_config(4).override(data__vocab_size=64, data__seq_len=16)
+    tmpl = build_model(cfg.model, cfg.data)
+    stk = build_stacked_model(cfg.model, cfg.data, G=3)
+    tnames = [n for n, _ in tmpl.named_parameters()]
+    snames = [n for n, _ in stk.named_parameters()]
+    assert len(tnames) == len(snames)
+    for tn, sn, (_, tp), (_, sp) in zip(
+        tnames, snames, tmpl.named_parameters(), stk.named_parameters()
+    ):
+        assert sp.shape == (3,) + tuple(tp.shape), (tn, sn)
