@@ -629,3 +629,55 @@ def test_shadow_sync_transposes():
                 assert torch.equal(mod.wshadow_t, ref_t)
                 checked += 1
     assert checked >= 15  # every non-stem conv has a transposed shadow
+
+
+# ---- grouped LayerNorm (stacked BERT) --------------------------------------
+def test_grouped_layernorm_matches_oracle():
+    m = _ext()
+    torch.manual_seed(11)
+    G, R, H = 4, 96, 768  # R rows per client
+    x = torch.randn(G * R, H, device=DEV).bfloat16()
+    w = torch.randn(G, H, device=DEV) * 0.2 + 1.0
+    b = torch.randn(G, H, device=DEV) * 0.1
+    y, mean, invstd = m.layernorm_fwd(x, w.view(-1), b.view(-1), 1e-12, G)
+    xf = x.float().view(G, R, H)
+    mu = xf.mean(-1)
+    istd = (xf.var(-1, unbiased=False) + 1e-12).rsqrt()
+    ref = ((xf - mu[..., None]) * istd[..., None] * w[:, None]
+           + b[:, None]).reshape(G * R, H)
+    torch.testing.assert_close(y.float(), ref, rtol=1e-2, atol=1e-2)
+    gy = torch.randn_like(x)
+    gx, gw, gb = m.layernorm_bwd(gy, x, w.view(-1), mean, invstd, G)
+    xo = x.float().view(G, R, H).requires_grad_(True)
+    wo = w.clone().requires_grad_(True)
+    bo = b.clone().requires_grad_(True)
+    muo = xo.mean(-1, keepdim=True)
+    io = (xo.var(-1, unbiased=False, keepdim=True) + 1e-12).rsqrt()
+    yo = (xo - muo) * io * wo[:, None] + bo[:, None]
+    yo.backward(gy.float().view(G, R, H))
+    torch.testing.assert_close(gw.view(G, H), wo.grad, rtol=1e-2,
+                               atol=3e-2)
+    torch.testing.assert_close(gb.view(G, H), bo.grad, rtol=1e-2,
+                               atol=3e-2)
+    torch.testing.assert_close(gx.float().view(G, R, H), xo.grad,
+                               rtol=2e-2, atol=2e-2)
+
+
+# ---- BERT end-to-end trained loss (stacked, native kernels) ----------------
+def test_bert_stacked_step_trains():
+    """Config-4 style stacked BERT on GPU: loss must drop over local
+    steps and per-round aggregation must run end-to-end (mirrors
+    test_resnet18_step_native_vs_cpu for the BERT family)."""
+    from colearn_federated_learning_amd import FederatedServer, driver_config
+
+    _ext()
+    cfg = driver_config(4).override(
+        rounds=2, num_clients=4, data__samples_per_client=64,
+        train__batch_size=8, num_evaluators=0,
+    )
+    s = FederatedServer(cfg, device=torch.device(DEV, 0))
+    assert s.stacked is not None  # stacked BERT engine active
+    r0 = s.run_round(0)
+    r1 = s.run_round(1)
+    assert r1["mean_client_loss"] < r0["mean_client_loss"]
+    assert r0["n_samples"] == 4 * 64
