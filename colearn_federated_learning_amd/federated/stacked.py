@@ -77,7 +77,7 @@ class StackedFlatParams:
         ~20 per-call permute+contiguous copies the dgrad path used to
         issue each step (~10% of step time — gpurun_out/eager_attrib.txt).
         """
-        from ..models.stacked import SBatchNorm2d, SConv2d
+        from ..models.stacked import SBatchNorm2d, SConv2d, SLinear
 
         self.flat_bf16 = torch.empty_like(self.flat_w,
                                           dtype=torch.bfloat16)
@@ -109,6 +109,11 @@ class StackedFlatParams:
             elif isinstance(mod, SBatchNorm2d):
                 mod.gslot_w = mod.weight.grad.view(-1)
                 mod.gslot_b = mod.bias.grad.view(-1)
+            elif isinstance(mod, SLinear):
+                off, n = by_name[f"{mod_name}.weight" if mod_name
+                                 else "weight"]
+                mod.wshadow = (self.flat_bf16[off : off + self.G * n]
+                               .view(mod.weight.shape))
         dev = self.device
         self.flat_bf16_t = torch.empty(max(t_off, 1),
                                        dtype=torch.bfloat16, device=dev)

@@ -175,8 +175,15 @@ class StackedBert(ColearnModel):
             w = w.to(hk.dtype)
         logits = (colearn_matmul_nt(hk.view(G, k, d), w)
                   + self.head_bias[:, None, :])     # [G, k, V]
+        if logits.is_cuda and logits.dtype == torch.bfloat16:
+            # CE straight on bf16 logits: torch's log_softmax accumulates
+            # in fp32 internally; the explicit .float() materialized a
+            # [G*k, vocab] fp32 tensor (~1 GB/step at config 4)
+            flat_logits = logits.reshape(G * k, -1)
+        else:
+            flat_logits = logits.float().reshape(G * k, -1)
         ce = F.cross_entropy(
-            logits.float().reshape(G * k, -1), yk.reshape(-1),
+            flat_logits, yk.reshape(-1),
             ignore_index=-100, reduction="none",
         ).view(G, k)
         valid = yk.ne(-100)

@@ -119,7 +119,8 @@ void launch_gemm_nt(const void* A, const void* B, void* C, int64_t batch,
                     int64_t M, int64_t N, int64_t K, hipStream_t stream);
 void launch_gemm(const void* A, const void* B, void* C, const float* bias,
                  int64_t batch, int64_t M, int64_t N, int64_t K, bool a_km,
-                 bool b_km, hipStream_t stream);
+                 bool b_km, hipStream_t stream, bool f32_out = false,
+                 int64_t sBias = 0);
 void launch_conv_fwd(const void* x, const void* w, void* y, ConvShape cs,
                      int G, hipStream_t stream);
 void launch_conv_dgrad(const void* gy, const void* wt, void* gx,

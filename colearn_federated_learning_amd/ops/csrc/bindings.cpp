@@ -465,6 +465,47 @@ torch::Tensor bmm_tn(torch::Tensor a, torch::Tensor b) {
   return gemm2(a, b, false, false, a.size(2), b.size(2), a.size(1));
 }
 
+// a @ b^T + bias, per-batch bias rows [batch, N] fused into the epilogue
+// (stacked-client linear forward)
+torch::Tensor bmm_nt_bias(torch::Tensor a, torch::Tensor b,
+                          torch::Tensor bias) {
+  TORCH_CHECK(a.dim() == 3 && b.dim() == 3, "bmm expects 3-D");
+  check_bf16(a, "a");
+  check_bf16(b, "b");
+  check_f32(bias, "bias");
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(a.device());
+  auto ac = a.contiguous(), bc = b.contiguous(),
+       bs = bias.contiguous();
+  const int64_t batch = a.size(0), M = a.size(1), N = b.size(1),
+                K = a.size(2);
+  TORCH_CHECK(bs.numel() == batch * N, "bias must be [batch, N]");
+  auto c = torch::empty({batch, M, N}, a.options());
+  launch_gemm(ac.data_ptr(), bc.data_ptr(), c.data_ptr(),
+              bs.data_ptr<float>(), batch, M, N, K, true, true, stream(),
+              false, N);
+  HIP_CHECK_LAST();
+  return c;
+}
+
+// a^T @ b per batch with FP32 output written straight into ``out`` (the
+// flat-grad slot): the stacked-linear weight gradient with no bf16
+// round trip and no AccumulateGrad add.
+void bmm_tn_f32(torch::Tensor a, torch::Tensor b, torch::Tensor out) {
+  TORCH_CHECK(a.dim() == 3 && b.dim() == 3, "bmm expects 3-D");
+  check_bf16(a, "a");
+  check_bf16(b, "b");
+  check_f32(out, "out");
+  TORCH_CHECK(out.is_contiguous(), "out must be contiguous");
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(a.device());
+  auto ac = a.contiguous(), bc = b.contiguous();
+  const int64_t batch = a.size(0), M = a.size(2), N = b.size(2),
+                K = a.size(1);
+  TORCH_CHECK(out.numel() == batch * M * N, "out shape mismatch");
+  launch_gemm(ac.data_ptr(), bc.data_ptr(), out.data_ptr(), nullptr,
+              batch, M, N, K, false, false, stream(), true, 0);
+  HIP_CHECK_LAST();
+}
+
 // ---- conv family ----------------------------------------------------------
 // weight 4-D [CO,CI,R,S] = single model; 5-D [G,CO,CI,R,S] = stacked
 // clients (x batch is then G*B, client-major).
@@ -875,6 +916,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bmm_nt", &bmm_nt);
   m.def("bmm_nn", &bmm_nn);
   m.def("bmm_tn", &bmm_tn);
+  m.def("bmm_nt_bias", &bmm_nt_bias);
+  m.def("bmm_tn_f32", &bmm_tn_f32);
   m.def("conv2d_fwd", &conv2d_fwd);
   m.def("conv2d_fwd_km", &conv2d_fwd_km);
   m.def("conv2d_dgrad_km", &conv2d_dgrad_km);

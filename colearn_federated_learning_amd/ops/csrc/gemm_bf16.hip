@@ -27,12 +27,12 @@ DEV_INLINE int lds_elem_byte(int row, int k) {
 
 // A_KM: A is [M,K] (k contiguous); else A is [K,M] (m contiguous).
 // B_KM: B is [N,K] (k contiguous); else B is [K,N] (n contiguous).
-template <bool A_KM, bool B_KM>
+template <bool A_KM, bool B_KM, typename OUT = bf16>
 __global__ __launch_bounds__(kBlockThreads) void gemm_kernel(
     const short* __restrict__ A, const short* __restrict__ B,
-    bf16* __restrict__ C, const float* __restrict__ bias, int64_t M,
+    OUT* __restrict__ C, const float* __restrict__ bias, int64_t M,
     int64_t N, int64_t K, int64_t sA, int64_t sB, int64_t sC, int nbm,
-    int nbn) {
+    int nbn, int64_t sBias = 0) {
   __shared__ short lds[2][2][BM * BK];
   const int bid = xcd_swizzle(blockIdx.x, nbm * nbn);
   const int64_t bm = static_cast<int64_t>(bid / nbn) * BM;
@@ -167,7 +167,8 @@ __global__ __launch_bounds__(kBlockThreads) void gemm_kernel(
   }
   if (bias != nullptr) {
     // fused bias epilogue: one add in registers instead of a whole-output
-    // elementwise pass afterwards
+    // elementwise pass afterwards (sBias: per-batch bias rows, stacked)
+    bias += static_cast<int64_t>(blockIdx.z) * sBias;
     const int lane = threadIdx.x % kWave;
     const int cif = lane & 15;
 #pragma unroll
@@ -180,38 +181,56 @@ __global__ __launch_bounds__(kBlockThreads) void gemm_kernel(
         for (int j = 0; j < 4; ++j) acc[m][n][j] += bv;
     }
   }
-  store_tile_bf16(reinterpret_cast<const float*>(acc), C, N, M, N, bm, bn,
+  store_tile<OUT>(reinterpret_cast<const float*>(acc), C, N, M, N, bm, bn,
                   wm, wn);
 }
 
 void launch_gemm(const void* A, const void* B, void* C, const float* bias,
                  int64_t batch, int64_t M, int64_t N, int64_t K, bool a_km,
-                 bool b_km, hipStream_t stream) {
+                 bool b_km, hipStream_t stream, bool f32_out,
+                 int64_t sBias) {
   const int nbm = static_cast<int>((M + BM - 1) / BM);
   const int nbn = static_cast<int>((N + BN - 1) / BN);
   dim3 grid(nbm * nbn, 1, static_cast<unsigned>(batch));
   const short* a = static_cast<const short*>(A);
   const short* b = static_cast<const short*>(B);
-  bf16* c = static_cast<bf16*>(C);
   const int64_t sA = M * K, sB = N * K, sC = M * N;
+  if (f32_out) {
+    float* c = static_cast<float*>(C);
+    if (a_km && b_km)
+      gemm_kernel<true, true, float><<<grid, kBlockThreads, 0, stream>>>(
+          a, b, c, bias, M, N, K, sA, sB, sC, nbm, nbn, sBias);
+    else if (a_km && !b_km)
+      gemm_kernel<true, false, float><<<grid, kBlockThreads, 0, stream>>>(
+          a, b, c, bias, M, N, K, sA, sB, sC, nbm, nbn, sBias);
+    else if (!a_km && b_km)
+      gemm_kernel<false, true, float><<<grid, kBlockThreads, 0, stream>>>(
+          a, b, c, bias, M, N, K, sA, sB, sC, nbm, nbn, sBias);
+    else
+      gemm_kernel<false, false, float><<<grid, kBlockThreads, 0, stream>>>(
+          a, b, c, bias, M, N, K, sA, sB, sC, nbm, nbn, sBias);
+    return;
+  }
+  bf16* c = static_cast<bf16*>(C);
   if (a_km && b_km)
     gemm_kernel<true, true><<<grid, kBlockThreads, 0, stream>>>(
-        a, b, c, bias, M, N, K, sA, sB, sC, nbm, nbn);
+        a, b, c, bias, M, N, K, sA, sB, sC, nbm, nbn, sBias);
   else if (a_km && !b_km)
     gemm_kernel<true, false><<<grid, kBlockThreads, 0, stream>>>(
-        a, b, c, bias, M, N, K, sA, sB, sC, nbm, nbn);
+        a, b, c, bias, M, N, K, sA, sB, sC, nbm, nbn, sBias);
   else if (!a_km && b_km)
     gemm_kernel<false, true><<<grid, kBlockThreads, 0, stream>>>(
-        a, b, c, bias, M, N, K, sA, sB, sC, nbm, nbn);
+        a, b, c, bias, M, N, K, sA, sB, sC, nbm, nbn, sBias);
   else
     gemm_kernel<false, false><<<grid, kBlockThreads, 0, stream>>>(
-        a, b, c, bias, M, N, K, sA, sB, sC, nbm, nbn);
+        a, b, c, bias, M, N, K, sA, sB, sC, nbm, nbn, sBias);
 }
 
 // legacy entry (conv path compatibility)
 void launch_gemm_nt(const void* A, const void* B, void* C, int64_t batch,
                     int64_t M, int64_t N, int64_t K, hipStream_t stream) {
-  launch_gemm(A, B, C, nullptr, batch, M, N, K, true, true, stream);
+  launch_gemm(A, B, C, nullptr, batch, M, N, K, true, true, stream, false,
+              0);
 }
 
 }  // namespace colearn

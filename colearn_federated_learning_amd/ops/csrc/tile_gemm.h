@@ -65,13 +65,15 @@ DEV_INLINE short8v_t mfma_frag_zero() {
   return z;
 }
 
-// Epilogue: write acc[4][4] (each f32x4_t) of one wave into C (bf16,
-// row-major, ldc) with bounds checks.  Wave origin (wm, wn) in the block
-// tile; block origin (bm, bn) in the matrix.
-DEV_INLINE void store_tile_bf16(float const* acc_flat /* [4][4][4] */,
-                                bf16* __restrict__ c, int64_t ldc, int64_t M,
-                                int64_t N, int64_t bm, int64_t bn, int wm,
-                                int wn) {
+// Epilogue: write acc[4][4] (each f32x4_t) of one wave into C (bf16 or
+// fp32 — OUT=float is the direct-to-flat-grad path), row-major, ldc,
+// with bounds checks.  Wave origin (wm, wn) in the block tile; block
+// origin (bm, bn) in the matrix.
+template <typename OUT>
+DEV_INLINE void store_tile(float const* acc_flat /* [4][4][4] */,
+                           OUT* __restrict__ c, int64_t ldc, int64_t M,
+                           int64_t N, int64_t bm, int64_t bn, int wm,
+                           int wn) {
   const int lane = threadIdx.x % kWave;
   const int col_in_frag = lane & 15;
   const int row_grp = lane >> 4;  // rows row_grp*4 + j
@@ -83,11 +85,22 @@ DEV_INLINE void store_tile_bf16(float const* acc_flat /* [4][4][4] */,
       for (int j = 0; j < 4; ++j) {
         const int64_t row = bm + wm + m * 16 + row_grp * 4 + j;
         const int64_t col = bn + wn + n * 16 + col_in_frag;
-        if (row < M && col < N)
-          c[row * ldc + col] = f2b(acc_flat[(m * 4 + n) * 4 + j]);
+        if (row < M && col < N) {
+          const float v = acc_flat[(m * 4 + n) * 4 + j];
+          if constexpr (sizeof(OUT) == 2)
+            c[row * ldc + col] = f2b(v);
+          else
+            c[row * ldc + col] = v;
+        }
       }
     }
   }
+}
+
+DEV_INLINE void store_tile_bf16(float const* acc_flat, bf16* c, int64_t ldc,
+                                int64_t M, int64_t N, int64_t bm,
+                                int64_t bn, int wm, int wn) {
+  store_tile<bf16>(acc_flat, c, ldc, M, N, bm, bn, wm, wn);
 }
 
 // XCD-aware bijective block swizzle (guide T1): contiguous chunks per XCD.
