@@ -681,3 +681,21 @@ def test_bert_stacked_step_trains():
     r1 = s.run_round(1)
     assert r1["mean_client_loss"] < r0["mean_client_loss"]
     assert r0["n_samples"] == 4 * 64
+
+
+def test_bmm_nt_bias_and_tn_f32():
+    m = _ext()
+    torch.manual_seed(13)
+    G, M, N, K = 3, 192, 96, 128
+    a = (torch.randn(G, M, K, device=DEV) * 0.3).bfloat16()
+    b = (torch.randn(G, N, K, device=DEV) * 0.3).bfloat16()
+    bias = torch.randn(G, N, device=DEV)
+    y = m.bmm_nt_bias(a, b, bias)
+    ref = (a.float() @ b.float().transpose(1, 2)) + bias[:, None, :]
+    torch.testing.assert_close(y.float(), ref, rtol=2e-2, atol=2e-2)
+    # tn wgrad with fp32 slab output
+    gy = (torch.randn(G, M, N, device=DEV) * 0.3).bfloat16()
+    out = torch.full((G, N, K), 7.0, device=DEV)  # overwritten
+    m.bmm_tn_f32(gy, a, out)
+    ref_gw = gy.float().transpose(1, 2) @ a.float()
+    torch.testing.assert_close(out, ref_gw, rtol=2e-2, atol=2e-2)
