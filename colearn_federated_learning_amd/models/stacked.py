@@ -226,11 +226,16 @@ class _SLinearShadowFn(torch.autograd.Function):
     @staticmethod
     def setup_context(ctx, inputs, output):
         x3, w, b, wsh, gslot_w, gslot_b = inputs
-        ctx.save_for_backward(x3, wsh, gslot_w, gslot_b)
+        ctx.save_for_backward(x3, wsh)
+        # grad slots stash on ctx directly: they are written in place by
+        # every layer's backward, so the saved-tensor version check would
+        # (correctly for normal tensors, wrongly here) reject them
+        ctx.gslot_w, ctx.gslot_b = gslot_w, gslot_b
 
     @staticmethod
     def backward(ctx, gy):
-        x3, wsh, gslot_w, gslot_b = ctx.saved_tensors
+        x3, wsh = ctx.saved_tensors
+        gslot_w, gslot_b = ctx.gslot_w, ctx.gslot_b
         m = require_ext()
         gy = gy.contiguous()
         gx = m.bmm_nn(gy, wsh)
