@@ -19,10 +19,31 @@
 
 namespace colearn {
 
-// byte offset of element (row, k) in a swizzled K-major LDS tile
+// GEMM LDS addressing: conv_nt's XOR slot swizzle PLUS a (row>>4) term
+// and a physical-row bit6->bit0 remap.  The K-strided operands
+// scatter-transpose 16-row groups into the tile (same pattern as conv
+// wgrad staging); under the plain scheme those rows collapse onto
+// identical banks (16*16 = 0 mod 32) — the extra terms spread them
+// while fragment reads (16 CONSECUTIVE rows) stay conflict-free.
+// Used consistently for all tile writes and reads in this file.
+DEV_INLINE int gemm_slot_byte(int row, int slot) {
+  const int pr = row ^ ((row >> 6) & 1);
+  return pr * (BK * 2) +
+         ((slot ^ ((row >> 2) & 3) ^ ((row >> 4) & 3)) * 16);
+}
+
+// byte offset of element (row, k) in the swizzled K-major LDS tile
 DEV_INLINE int lds_elem_byte(int row, int k) {
-  return row * (BK * 2) + (((k >> 3) ^ ((row >> 2) & 3)) * 16) +
-         (k & 7) * 2;
+  return gemm_slot_byte(row, k >> 3) + (k & 7) * 2;
+}
+
+// fragment read (8 bf16) for this lane; k0 = K-offset (multiple of 32)
+DEV_INLINE short8v_t gemm_frag_from_lds(const short* tile, int row_base) {
+  const int lane = threadIdx.x % kWave;
+  const int row = row_base + (lane & 15);
+  const int slot = lane >> 4;
+  return *reinterpret_cast<const short8v_t*>(
+      reinterpret_cast<const char*>(tile) + gemm_slot_byte(row, slot));
 }
 
 // A_KM: A is [M,K] (k contiguous); else A is [K,M] (m contiguous).
@@ -117,9 +138,10 @@ __global__ __launch_bounds__(kBlockThreads) void gemm_kernel(
     char* As = reinterpret_cast<char*>(lds[buf][0]);
     char* Bs = reinterpret_cast<char*>(lds[buf][1]);
     if constexpr (A_KM) {
-      *reinterpret_cast<short8v_t*>(As + lds_slot_byte(row, half * 2)) = ra0;
+      *reinterpret_cast<short8v_t*>(As + gemm_slot_byte(row, half * 2)) =
+          ra0;
       *reinterpret_cast<short8v_t*>(
-          As + lds_slot_byte(row, half * 2 + 1)) = ra1;
+          As + gemm_slot_byte(row, half * 2 + 1)) = ra1;
     } else {
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -129,9 +151,10 @@ __global__ __launch_bounds__(kBlockThreads) void gemm_kernel(
       }
     }
     if constexpr (B_KM) {
-      *reinterpret_cast<short8v_t*>(Bs + lds_slot_byte(row, half * 2)) = rb0;
+      *reinterpret_cast<short8v_t*>(Bs + gemm_slot_byte(row, half * 2)) =
+          rb0;
       *reinterpret_cast<short8v_t*>(
-          Bs + lds_slot_byte(row, half * 2 + 1)) = rb1;
+          Bs + gemm_slot_byte(row, half * 2 + 1)) = rb1;
     } else {
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -151,10 +174,10 @@ __global__ __launch_bounds__(kBlockThreads) void gemm_kernel(
     short8v_t af[4], bfr[4];
 #pragma unroll
     for (int m = 0; m < 4; ++m)
-      af[m] = frag_from_lds(lds[cur][0], wm + m * 16);
+      af[m] = gemm_frag_from_lds(lds[cur][0], wm + m * 16);
 #pragma unroll
     for (int n = 0; n < 4; ++n)
-      bfr[n] = frag_from_lds(lds[cur][1], wn + n * 16);
+      bfr[n] = gemm_frag_from_lds(lds[cur][1], wn + n * 16);
 #pragma unroll
     for (int m = 0; m < 4; ++m)
 #pragma unroll
