@@ -75,10 +75,26 @@ class SBertSelfAttention(nn.Module):
         self.qkv = SLinear(G, cfg.hidden_size, 3 * cfg.hidden_size)
         self.out = SLinear(G, cfg.hidden_size, cfg.hidden_size)
         self.drop = nn.Dropout(cfg.dropout)
+        # fused-attention dropout stream: deterministic run-to-run
+        # (keyed on the module's creation-time torch seed + call count)
+        self._seed_base = int(torch.initial_seed()) & 0x7FFFFFFF
+        self._call = 0
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        from ..ops.attention import (
+            fused_attention, fused_attention_applicable,
+        )
+
         b, s, d = x.shape  # b = G*B
         qkv = self.qkv(x.reshape(b * s, d)).view(b, s, 3, self.h, self.dh)
+        if fused_attention_applicable(qkv):
+            p = float(self.drop.p) if self.training else 0.0
+            self._call += 1
+            seed = (self._seed_base * 2654435761 + self._call) & (
+                (1 << 63) - 1)
+            ctx = fused_attention(qkv, self.h,
+                                  1.0 / math.sqrt(self.dh), p, seed)
+            return self.out(ctx.reshape(b * s, d)).view(b, s, d)
         q, k, v = (
             qkv[:, :, i].transpose(1, 2).reshape(b * self.h, s, self.dh)
             for i in range(3)

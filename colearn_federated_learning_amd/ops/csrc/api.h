@@ -90,6 +90,13 @@ void launch_head_bwd(const float* probs, const float* pooled,
                      float* dlogits, void* dx, float* gw, float* gb,
                      int N, int C, int HW, int K, int B, int G,
                      hipStream_t stream);
+void launch_attn_fwd(const void* qkv, void* ctx, void* Psave, int N,
+                     int S, int H, float scale, float p,
+                     unsigned long long seed, hipStream_t stream);
+void launch_attn_bwd(const void* dctx, const void* qkv, const void* Psave,
+                     void* dS, void* dqkv, int N, int S, int H,
+                     float scale, float p, unsigned long long seed,
+                     hipStream_t stream);
 void launch_philox_gaussian_add(float* out, int64_t n,
                                 unsigned long long seed, float std,
                                 hipStream_t stream);

@@ -222,6 +222,47 @@ torch::Tensor head_bwd(torch::Tensor probs, torch::Tensor pooled,
   return dx;
 }
 
+// ---- fused attention (seq<=128, dh=64) -----------------------------------
+std::tuple<torch::Tensor, torch::Tensor> attn_fwd(
+    torch::Tensor qkv, int64_t H, double scale, double p, int64_t seed) {
+  check_bf16(qkv, "qkv");
+  TORCH_CHECK(qkv.dim() == 5 && qkv.size(2) == 3 && qkv.size(4) == 64,
+              "qkv must be [N, S, 3, H, 64]");
+  const int N = qkv.size(0), S = qkv.size(1);
+  TORCH_CHECK(S % 32 == 0 && S <= 128, "seq must be <=128, mult of 32");
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(qkv.device());
+  auto qc = qkv.contiguous();
+  auto ctx = torch::empty({N, S, H * 64}, qkv.options());
+  auto P = torch::empty({static_cast<int64_t>(N) * H, S, S},
+                        qkv.options());
+  launch_attn_fwd(qc.data_ptr(), ctx.data_ptr(), P.data_ptr(), N, S,
+                  static_cast<int>(H), static_cast<float>(scale),
+                  static_cast<float>(p),
+                  static_cast<unsigned long long>(seed), stream());
+  HIP_CHECK_LAST();
+  return {ctx, P};
+}
+
+torch::Tensor attn_bwd(torch::Tensor dctx, torch::Tensor qkv,
+                       torch::Tensor P, int64_t H, double scale, double p,
+                       int64_t seed) {
+  check_bf16(dctx, "dctx");
+  check_bf16(qkv, "qkv");
+  const int N = qkv.size(0), S = qkv.size(1);
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(qkv.device());
+  auto dc = dctx.contiguous();
+  auto qc = qkv.contiguous();
+  auto dS = torch::empty_like(P);
+  auto dqkv = torch::empty_like(qc);
+  launch_attn_bwd(dc.data_ptr(), qc.data_ptr(), P.data_ptr(),
+                  dS.data_ptr(), dqkv.data_ptr(), N, S,
+                  static_cast<int>(H), static_cast<float>(scale),
+                  static_cast<float>(p),
+                  static_cast<unsigned long long>(seed), stream());
+  HIP_CHECK_LAST();
+  return dqkv;
+}
+
 void philox_gaussian_add(torch::Tensor flat, int64_t seed, double std) {
   check_f32(flat, "flat");
   TORCH_CHECK(flat.is_contiguous(), "flat must be contiguous");
@@ -899,6 +940,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sq_norm_per_row", &sq_norm_per_row);
   m.def("scale_rows_mean", &scale_rows_mean);
   m.def("philox_gaussian_add", &philox_gaussian_add);
+  m.def("attn_fwd", &attn_fwd);
+  m.def("attn_bwd", &attn_bwd);
   m.def("head_fwd", &head_fwd);
   m.def("head_bwd", &head_bwd);
   m.def("shadow_sync", &shadow_sync);

@@ -31,6 +31,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "conv_bf16.hip"),
         os.path.join(CSRC, "head.hip"),
         os.path.join(CSRC, "shadow.hip"),
+        os.path.join(CSRC, "attn.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

@@ -699,3 +699,55 @@ def test_bmm_nt_bias_and_tn_f32():
     m.bmm_tn_f32(gy, a, out)
     ref_gw = gy.float().transpose(1, 2) @ a.float()
     torch.testing.assert_close(out, ref_gw, rtol=2e-2, atol=2e-2)
+
+
+# ---- fused attention (seq<=128, dh=64) -------------------------------------
+def test_fused_attention_matches_torch_p0():
+    import math
+
+    import torch.nn.functional as F
+
+    from colearn_federated_learning_amd.ops.attention import fused_attention
+
+    m = _ext()
+    torch.manual_seed(17)
+    N, S, H, dh = 6, 128, 4, 64
+    qkv = (torch.randn(N, S, 3, H, dh, device=DEV) * 0.5).bfloat16()
+    scale = 1.0 / math.sqrt(dh)
+
+    qkv_f = qkv.clone().requires_grad_(True)
+    ctx = fused_attention(qkv_f, H, scale, 0.0, 1234)
+    gout = (torch.randn(N, S, H * dh, device=DEV) * 0.3).bfloat16()
+    ctx.backward(gout)
+
+    qkv_o = qkv.float().clone().requires_grad_(True)
+    q, k, v = (qkv_o[:, :, i].transpose(1, 2) for i in range(3))
+    scores = (q @ k.transpose(-1, -2)) * scale
+    probs = F.softmax(scores, dim=-1)
+    ref = ((probs @ v).transpose(1, 2).reshape(N, S, H * dh))
+    ref.backward(gout.float())
+
+    torch.testing.assert_close(ctx.float(), ref, rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(qkv_f.grad.float(), qkv_o.grad,
+                               rtol=5e-2, atol=3e-2)
+
+
+def test_fused_attention_dropout_stats_and_determinism():
+    import math
+
+    from colearn_federated_learning_amd.ops.attention import fused_attention
+
+    _ext()
+    torch.manual_seed(18)
+    N, S, H, dh = 4, 128, 4, 64
+    qkv = (torch.randn(N, S, 3, H, dh, device=DEV) * 0.5).bfloat16()
+    scale = 1.0 / math.sqrt(dh)
+    a = fused_attention(qkv, H, scale, 0.1, 42)
+    b = fused_attention(qkv, H, scale, 0.1, 42)
+    assert torch.equal(a, b)  # keyed dropout: bitwise deterministic
+    c = fused_attention(qkv, H, scale, 0.1, 43)
+    assert not torch.allclose(a, c)
+    # dropout is mean-preserving: E[ctx_p] ~ ctx_0
+    d0 = fused_attention(qkv, H, scale, 0.0, 0)
+    assert (a.float() - d0.float()).abs().mean() < 0.15
+    assert abs(a.float().mean() - d0.float().mean()) < 2e-2
