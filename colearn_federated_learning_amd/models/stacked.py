@@ -240,7 +240,7 @@ class _SLinearShadowFn(torch.autograd.Function):
         gy = gy.contiguous()
         gx = m.bmm_nn(gy, wsh)
         m.bmm_tn_f32(gy, x3, gslot_w)
-        torch.sum(gy, dim=1, dtype=torch.float32, out=gslot_b)
+        m.colsum_f32(gy, gslot_b)  # deterministic fixed-order bias grads
         return gx, None, None, None, None, None
 
 

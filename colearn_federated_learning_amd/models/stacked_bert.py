@@ -27,6 +27,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops.elemwise import colearn_gelu
 from ..ops.gemm import colearn_matmul, colearn_matmul_nt
 from ..ops.norms import _LayerNormFn
 from .base import ColearnModel
@@ -120,7 +121,7 @@ class SBertLayer(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         b, s, d = x.shape
         x = self.ln1(x + self.drop(self.attn(x)))
-        ff = self.fc2(F.gelu(self.fc1(x.reshape(b * s, d)))).view(b, s, d)
+        ff = self.fc2(colearn_gelu(self.fc1(x.reshape(b * s, d)))).view(b, s, d)
         x = self.ln2(x + self.drop(ff))
         return x
 
@@ -161,7 +162,7 @@ class StackedBert(ColearnModel):
         n, s = ids.shape
         h = self.encode(ids)
         d = h.shape[-1]
-        h = self.head_ln(F.gelu(
+        h = self.head_ln(colearn_gelu(
             self.head_fc(h.reshape(n * s, d))))
         h3 = h.view(self.G, -1, d)
         w = self.tok_emb.weight
@@ -185,7 +186,7 @@ class StackedBert(ColearnModel):
             dim=1, descending=True, stable=True)[:, :k]
         hk = torch.gather(h3, 1, order[..., None].expand(-1, -1, d))
         yk = torch.gather(y2, 1, order)  # [G, k]
-        hk = self.head_ln(F.gelu(self.head_fc(hk.reshape(G * k, d))))
+        hk = self.head_ln(colearn_gelu(self.head_fc(hk.reshape(G * k, d))))
         w = self.tok_emb.weight
         if hk.dtype != w.dtype:
             w = w.to(hk.dtype)

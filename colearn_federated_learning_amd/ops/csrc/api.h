@@ -90,6 +90,12 @@ void launch_head_bwd(const float* probs, const float* pooled,
                      float* dlogits, void* dx, float* gw, float* gb,
                      int N, int C, int HW, int K, int B, int G,
                      hipStream_t stream);
+void launch_gelu_fwd(const void* x, void* y, int64_t n,
+                     hipStream_t stream);
+void launch_gelu_bwd(const void* gy, const void* x, void* gx, int64_t n,
+                     hipStream_t stream);
+void launch_colsum(const void* x, float* partials, float* out, int64_t B,
+                   int64_t M, int64_t N, int chunks, hipStream_t stream);
 void launch_attn_fwd(const void* qkv, void* ctx, void* Psave, int N,
                      int S, int H, float scale, float p,
                      unsigned long long seed, hipStream_t stream);

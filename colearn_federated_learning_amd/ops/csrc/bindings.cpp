@@ -222,6 +222,45 @@ torch::Tensor head_bwd(torch::Tensor probs, torch::Tensor pooled,
   return dx;
 }
 
+// ---- fast elementwise (BERT path) ----------------------------------------
+torch::Tensor gelu_fwd(torch::Tensor x) {
+  check_bf16(x, "x");
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
+  auto xc = x.contiguous();
+  auto y = torch::empty_like(xc);
+  launch_gelu_fwd(xc.data_ptr(), y.data_ptr(), xc.numel(), stream());
+  HIP_CHECK_LAST();
+  return y;
+}
+
+torch::Tensor gelu_bwd(torch::Tensor gy, torch::Tensor x) {
+  check_bf16(gy, "gy");
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
+  auto gc = gy.contiguous(), xc = x.contiguous();
+  auto gx = torch::empty_like(gc);
+  launch_gelu_bwd(gc.data_ptr(), xc.data_ptr(), gx.data_ptr(),
+                  gc.numel(), stream());
+  HIP_CHECK_LAST();
+  return gx;
+}
+
+// column sums over dim 1 of [B, M, N] bf16 -> fp32 written into ``out``
+// [B, N] (stacked bias grads, deterministic fixed-order chunks)
+void colsum_f32(torch::Tensor x, torch::Tensor out) {
+  check_bf16(x, "x");
+  check_f32(out, "out");
+  TORCH_CHECK(x.dim() == 3 && out.is_contiguous(), "colsum shapes");
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
+  auto xc = x.contiguous();
+  const int64_t B = x.size(0), M = x.size(1), N = x.size(2);
+  const int chunks = 16;
+  auto partials = torch::empty({B * chunks * N},
+                               out.options());
+  launch_colsum(xc.data_ptr(), partials.data_ptr<float>(),
+                out.data_ptr<float>(), B, M, N, chunks, stream());
+  HIP_CHECK_LAST();
+}
+
 // ---- fused attention (seq<=128, dh=64) -----------------------------------
 std::tuple<torch::Tensor, torch::Tensor> attn_fwd(
     torch::Tensor qkv, int64_t H, double scale, double p, int64_t seed) {
@@ -941,6 +980,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scale_rows_mean", &scale_rows_mean);
   m.def("philox_gaussian_add", &philox_gaussian_add);
   m.def("attn_fwd", &attn_fwd);
+  m.def("gelu_fwd", &gelu_fwd);
+  m.def("gelu_bwd", &gelu_bwd);
+  m.def("colsum_f32", &colsum_f32);
   m.def("attn_bwd", &attn_bwd);
   m.def("head_fwd", &head_fwd);
   m.def("head_bwd", &head_bwd);

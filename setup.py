@@ -32,6 +32,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "head.hip"),
         os.path.join(CSRC, "shadow.hip"),
         os.path.join(CSRC, "attn.hip"),
+        os.path.join(CSRC, "elemwise.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
