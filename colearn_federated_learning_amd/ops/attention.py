@@ -17,17 +17,17 @@ from . import native_enabled, require_ext
 class _FusedAttnFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, qkv, heads, scale, p, seed):
-        out, P = require_ext().attn_fwd(qkv, heads, scale, p, seed)
-        ctx.save_for_backward(qkv, P)
+        out, P, mask = require_ext().attn_fwd(qkv, heads, scale, p, seed)
+        ctx.save_for_backward(qkv, P, mask)
         ctx.args = (heads, scale, p, seed)
         return out
 
     @staticmethod
     def backward(ctx, dctx):
-        qkv, P = ctx.saved_tensors
+        qkv, P, mask = ctx.saved_tensors
         heads, scale, p, seed = ctx.args
-        dqkv = require_ext().attn_bwd(dctx.contiguous(), qkv, P, heads,
-                                      scale, p, seed)
+        dqkv = require_ext().attn_bwd(dctx.contiguous(), qkv, P, mask,
+                                      heads, scale, p, seed)
         return dqkv, None, None, None, None
 
 

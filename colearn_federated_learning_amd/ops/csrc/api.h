@@ -96,13 +96,14 @@ void launch_gelu_bwd(const void* gy, const void* x, void* gx, int64_t n,
                      hipStream_t stream);
 void launch_colsum(const void* x, float* partials, float* out, int64_t B,
                    int64_t M, int64_t N, int chunks, hipStream_t stream);
-void launch_attn_fwd(const void* qkv, void* ctx, void* Psave, int N,
-                     int S, int H, float scale, float p,
-                     unsigned long long seed, hipStream_t stream);
-void launch_attn_bwd(const void* dctx, const void* qkv, const void* Psave,
-                     void* dS, void* dqkv, int N, int S, int H,
+void launch_attn_fwd(const void* qkv, void* ctx, void* Psave,
+                     unsigned long long* mask, int N, int S, int H,
                      float scale, float p, unsigned long long seed,
                      hipStream_t stream);
+void launch_attn_bwd(const void* dctx, const void* qkv, const void* Psave,
+                     const unsigned long long* mask, void* dS, void* dqkv,
+                     int N, int S, int H, float scale, float p,
+                     unsigned long long seed, hipStream_t stream);
 void launch_philox_gaussian_add(float* out, int64_t n,
                                 unsigned long long seed, float std,
                                 hipStream_t stream);

@@ -154,7 +154,8 @@ using namespace attn;
 // [N*H, S, S] bf16 (softmax probs, pre-dropout)
 __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const short* __restrict__ qkv, short* __restrict__ ctx,
-    short* __restrict__ Psave, int N, int S, int H, float scale, float p,
+    short* __restrict__ Psave, unsigned long long* __restrict__ maskout,
+    int N, int S, int H, float scale, float p,
     unsigned long long seed) {
   // 48 KB total; after the score phase the P tile OVERLAYS q+k (their
   // last read is fenced by the barrier below)
@@ -225,8 +226,11 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       for (int n2 = 0; n2 < 8; ++n2) acc[m][n2][j] *= inv;
     }
   }
-  // save probs (bf16) to global + stage dropped probs into LDS
+  // save probs (bf16) to global + stage dropped probs into LDS; the
+  // dropout decisions are ALSO saved as one u64 bitfield per thread
+  // (fragment-ordered) so the backward never re-runs philox
   short* Pg = Psave + blk * S * S;
+  unsigned long long mbits = 0;
 #pragma unroll
   for (int m = 0; m < 2; ++m)
 #pragma unroll
@@ -237,12 +241,14 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         const int col = n2 * 16 + (lane & 15);
         const float pr = acc[m][n2][j];
         Pg[row * S + col] = f2bits(pr);
-        const float pd =
-            pr * drop_scale(seed, blk, row, col, p, inv_keep);
+        const float ds = drop_scale(seed, blk, row, col, p, inv_keep);
+        if (ds > 0.f)
+          mbits |= 1ull << ((m * 8 + n2) * 4 + j);
         *reinterpret_cast<short*>(
             reinterpret_cast<char*>(p_t) + tsq(row, col >> 3) +
-            (col & 7) * 2) = f2bits(pd);
+            (col & 7) * 2) = f2bits(pr * ds);
       }
+  maskout[blk * 256 + threadIdx.x] = mbits;
   __syncthreads();
   // C = Pd @ V : [32 x DH] per wave
   f32x4_t cacc[2][4];
@@ -303,7 +309,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 // dS is written to global (bf16) for kernel 2.
 __global__ __launch_bounds__(256) void attn_bwd1_kernel(
     const short* __restrict__ dctx, const short* __restrict__ qkv,
-    const short* __restrict__ Psave, short* __restrict__ dS_g,
+    const short* __restrict__ Psave,
+    const unsigned long long* __restrict__ maskin, short* __restrict__ dS_g,
     short* __restrict__ dqkv, int N, int S, int H, float scale, float p,
     unsigned long long seed) {
   __shared__ short smem[S_MAX * TSQ_SH + 2 * S_MAX * TDH_SH];  // 64 KB
@@ -328,6 +335,7 @@ __global__ __launch_bounds__(256) void attn_bwd1_kernel(
   const int m0 = wid * 32;
   const int rg = lane >> 4;
   const float inv_keep = p > 0.f ? 1.f / (1.f - p) : 1.f;
+  const unsigned long long mbits = maskin[blk * 256 + threadIdx.x];
 
   // dPd = dC @ V^T : [32 x S] per wave (A = dC rows, B = V rows)
   f32x4_t acc[2][8];
@@ -365,8 +373,9 @@ __global__ __launch_bounds__(256) void attn_bwd1_kernel(
         pr[n2] = bits2f(*reinterpret_cast<const short*>(
             reinterpret_cast<const char*>(p_t) + tsq(row, col >> 3) +
             (col & 7) * 2));
-        dp[n2] = acc[m][n2][j] *
-                 drop_scale(seed, blk, row, col, p, inv_keep);
+        const float ds =
+            (mbits >> ((m * 8 + n2) * 4 + j)) & 1 ? inv_keep : 0.f;
+        dp[n2] = acc[m][n2][j] * (p > 0.f ? ds : 1.f);
         rd += dp[n2] * pr[n2];
       }
       for (int off = 1; off < 16; off <<= 1) rd += __shfl_xor(rd, off);
@@ -379,21 +388,25 @@ __global__ __launch_bounds__(256) void attn_bwd1_kernel(
     }
   }
   __syncthreads();
-  // overwrite p_t in place with Pd = P * mask (for dV)
-  {
-    const int spr = S / 8;
-    for (int e = threadIdx.x; e < S * spr; e += 256) {
-      const int r = e / spr, slot = e % spr;
-      short* cell = reinterpret_cast<short*>(
-          reinterpret_cast<char*>(p_t) + tsq(r, slot));
+  // overwrite p_t in place with Pd = P * mask (for dV) — each thread
+  // scales exactly the elements its mask bitfield covers (fragment
+  // order), so no philox runs in the backward at all
+  if (p > 0.f) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int col = slot * 8 + j;
-        cell[j] = f2bits(
-            bits2f(cell[j]) *
-            drop_scale(seed, blk, r, col, p, inv_keep));
-      }
-    }
+    for (int m = 0; m < 2; ++m)
+#pragma unroll
+      for (int n2 = 0; n2 < 8; ++n2)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const int row = m0 + m * 16 + rg * 4 + j;
+          const int col = n2 * 16 + (lane & 15);
+          const float ds =
+              (mbits >> ((m * 8 + n2) * 4 + j)) & 1 ? inv_keep : 0.f;
+          short* cell = reinterpret_cast<short*>(
+              reinterpret_cast<char*>(p_t) + tsq(row, col >> 3) +
+              (col & 7) * 2);
+          *cell = f2bits(bits2f(*cell) * ds);
+        }
   }
   __syncthreads();
   // dV = Pd^T @ dC : [32 key rows x DH] per wave; A = Pd^T (transposed
@@ -530,21 +543,22 @@ __global__ __launch_bounds__(256) void attn_bwd2_kernel(
 }
 
 // ---- launchers ------------------------------------------------------------
-void launch_attn_fwd(const void* qkv, void* ctx, void* Psave, int N,
-                     int S, int H, float scale, float p,
-                     unsigned long long seed, hipStream_t stream) {
+void launch_attn_fwd(const void* qkv, void* ctx, void* Psave,
+                     unsigned long long* mask, int N, int S, int H,
+                     float scale, float p, unsigned long long seed,
+                     hipStream_t stream) {
   attn_fwd_kernel<<<N * H, 256, 0, stream>>>(
       static_cast<const short*>(qkv), static_cast<short*>(ctx),
-      static_cast<short*>(Psave), N, S, H, scale, p, seed);
+      static_cast<short*>(Psave), mask, N, S, H, scale, p, seed);
 }
 
 void launch_attn_bwd(const void* dctx, const void* qkv, const void* Psave,
-                     void* dS, void* dqkv, int N, int S, int H,
-                     float scale, float p, unsigned long long seed,
-                     hipStream_t stream) {
+                     const unsigned long long* mask, void* dS, void* dqkv,
+                     int N, int S, int H, float scale, float p,
+                     unsigned long long seed, hipStream_t stream) {
   attn_bwd1_kernel<<<N * H, 256, 0, stream>>>(
       static_cast<const short*>(dctx), static_cast<const short*>(qkv),
-      static_cast<const short*>(Psave), static_cast<short*>(dS),
+      static_cast<const short*>(Psave), mask, static_cast<short*>(dS),
       static_cast<short*>(dqkv), N, S, H, scale, p, seed);
   attn_bwd2_kernel<<<N * H, 256, 0, stream>>>(
       static_cast<const short*>(dS), static_cast<const short*>(qkv),

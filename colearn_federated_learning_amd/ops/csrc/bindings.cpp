@@ -262,7 +262,7 @@ void colsum_f32(torch::Tensor x, torch::Tensor out) {
 }
 
 // ---- fused attention (seq<=128, dh=64) -----------------------------------
-std::tuple<torch::Tensor, torch::Tensor> attn_fwd(
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> attn_fwd(
     torch::Tensor qkv, int64_t H, double scale, double p, int64_t seed) {
   check_bf16(qkv, "qkv");
   TORCH_CHECK(qkv.dim() == 5 && qkv.size(2) == 3 && qkv.size(4) == 64,
@@ -274,17 +274,21 @@ std::tuple<torch::Tensor, torch::Tensor> attn_fwd(
   auto ctx = torch::empty({N, S, H * 64}, qkv.options());
   auto P = torch::empty({static_cast<int64_t>(N) * H, S, S},
                         qkv.options());
-  launch_attn_fwd(qc.data_ptr(), ctx.data_ptr(), P.data_ptr(), N, S,
-                  static_cast<int>(H), static_cast<float>(scale),
+  auto mask = torch::empty({static_cast<int64_t>(N) * H * 256},
+                           qkv.options().dtype(torch::kInt64));
+  launch_attn_fwd(qc.data_ptr(), ctx.data_ptr(), P.data_ptr(),
+                  reinterpret_cast<unsigned long long*>(
+                      mask.data_ptr<int64_t>()),
+                  N, S, static_cast<int>(H), static_cast<float>(scale),
                   static_cast<float>(p),
                   static_cast<unsigned long long>(seed), stream());
   HIP_CHECK_LAST();
-  return {ctx, P};
+  return {ctx, P, mask};
 }
 
 torch::Tensor attn_bwd(torch::Tensor dctx, torch::Tensor qkv,
-                       torch::Tensor P, int64_t H, double scale, double p,
-                       int64_t seed) {
+                       torch::Tensor P, torch::Tensor mask, int64_t H,
+                       double scale, double p, int64_t seed) {
   check_bf16(dctx, "dctx");
   check_bf16(qkv, "qkv");
   const int N = qkv.size(0), S = qkv.size(1);
@@ -294,6 +298,8 @@ torch::Tensor attn_bwd(torch::Tensor dctx, torch::Tensor qkv,
   auto dS = torch::empty_like(P);
   auto dqkv = torch::empty_like(qc);
   launch_attn_bwd(dc.data_ptr(), qc.data_ptr(), P.data_ptr(),
+                  reinterpret_cast<const unsigned long long*>(
+                      mask.data_ptr<int64_t>()),
                   dS.data_ptr(), dqkv.data_ptr(), N, S,
                   static_cast<int>(H), static_cast<float>(scale),
                   static_cast<float>(p),
