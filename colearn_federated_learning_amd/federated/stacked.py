@@ -237,8 +237,18 @@ class StackedEngine:
             cfg.train.optimizer, self.sflat.flat_w, self.sflat.flat_g,
             cfg.train, flat_ref=self.flat_ref,
         )
+        # stacked DP: per-sample grads for ALL clients' samples in one
+        # backward (hook engine over the stacked modules), clip + per-
+        # client mean in one grouped kernel sweep per parameter
+        self.gs_engine = None
+        if cfg.dp.enabled:
+            from ..ops.grad_sample import GradSampleEngine
+
+            self.gs_engine = GradSampleEngine(self.model)
         self.runner = None
-        if cfg.model.name == "bert_base":
+        if cfg.dp.enabled:
+            pass  # eager per-sample path (no graphs yet)
+        elif cfg.model.name == "bert_base":
             # hipGraph replays with sort/embedding kernels intermittently
             # fault on ROCm 7.2 (see steprunner.py), and the stacked BERT
             # kernels are large enough that graphs buy ~nothing — eager.
@@ -257,6 +267,39 @@ class StackedEngine:
                 (self.G * bs,), torch.float32, torch.long, device,
                 cfg.train.dtype == "bfloat16",
             )
+
+    # ------------------------------------------------------------------
+    def _dp_step(self, x, y, use_bf16: bool):
+        """One per-sample-clipped DP step for ALL G clients at once.
+
+        Per-sample grads are weight-free quantities of (input, grad-out),
+        so the hook rules run on the stacked modules unchanged with
+        batch = G*B; the fused grouped reduction then clips each SAMPLE
+        to C and means each CLIENT's rows in one launch per parameter
+        (ops.dp.clip_and_reduce(groups=G))."""
+        from ..ops import dp as dp_ops
+
+        B = x.shape[0] // self.G
+        self.optimizer.zero_grad()
+        with self.gs_engine:
+            with torch.autocast("cuda", torch.bfloat16,
+                                enabled=use_bf16):
+                loss_vec = self.model.loss_terms(x, y)
+            loss_vec.sum().backward()
+            per_sample = self.gs_engine.grad_samples()
+            clipped = dp_ops.clip_and_reduce(
+                per_sample, self.cfg.dp.clip_norm,
+                pre_scale=float(B), groups=self.G,
+            )
+            trainable = [p for p in self.model.parameters()
+                         if p.requires_grad]
+            assert len(trainable) == len(clipped)
+            for p, g in zip(trainable, clipped):
+                p.grad.copy_(g.reshape(p.shape))
+            self.gs_engine.clear()
+        self.optimizer.step()
+        self.sflat.sync_shadow()
+        return loss_vec
 
     # ------------------------------------------------------------------
     def train_round(
@@ -314,6 +357,8 @@ class StackedEngine:
                                                      non_blocking=True)
                 if self.runner is not None and self.runner.matches(x):
                     loss_vec = self.runner.step(x, y)
+                elif self.gs_engine is not None:
+                    loss_vec = self._dp_step(x, y, use_bf16)
                 else:
                     self.optimizer.zero_grad()
                     with torch.autocast("cuda", torch.bfloat16,
@@ -388,8 +433,8 @@ def stackable(cfg: FederatedConfig, clients: List[FederatedClient]) -> bool:
     if cfg.model.name not in ("logreg", "resnet18", "resnet50",
                               "bert_base"):
         return False
-    if cfg.dp.enabled:  # per-sample path stays sequential
-        return False
+    if cfg.dp.enabled and cfg.model.name not in ("resnet18", "resnet50"):
+        return False  # stacked per-sample path exists for the conv models
     if len(clients) < 1:
         return False
     # G=1 is worth stacking too: the stacked engine carries the bf16

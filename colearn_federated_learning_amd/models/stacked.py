@@ -277,26 +277,37 @@ class SLinear(nn.Module):
         return out.reshape(-1, out.shape[-1])
 
 
+class SGroupNorm(nn.Module):
+    """Per-client GroupNorm (DP mode): normalize per SAMPLE (GroupNorm
+    semantics), affine per CLIENT.  A standalone module so the
+    per-sample-grad engine can hook its exact output (the residual/ReLU
+    epilogue lives in _SGNAct, outside the hook boundary)."""
+
+    def __init__(self, G: int, planes: int) -> None:
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(G, planes))
+        self.bias = nn.Parameter(torch.zeros(G, planes))
+        self.G = G
+        self.groups = min(32, planes)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        B = x.shape[0] // self.G
+        xn = F.group_norm(x, self.groups)
+        w = self.weight.repeat_interleave(B, dim=0)  # [G*B, C]
+        b = self.bias.repeat_interleave(B, dim=0)
+        return xn * w[:, :, None, None] + b[:, :, None, None]
+
+
 class _SGNAct(nn.Module):
-    """Stacked GroupNorm substitute (DP mode) — per-client loop."""
+    """Stacked GroupNorm substitute (DP mode) + residual/ReLU epilogue."""
 
     def __init__(self, G: int, planes: int, relu: bool) -> None:
         super().__init__()
-        self.G = G
-        self.weight = nn.Parameter(torch.ones(G, planes))
-        self.bias = nn.Parameter(torch.zeros(G, planes))
-        self.groups = min(32, planes)
+        self.norm = SGroupNorm(G, planes)
         self.do_relu = relu
 
     def forward(self, x: torch.Tensor, residual=None) -> torch.Tensor:
-        G = self.G
-        B = x.shape[0] // G
-        outs = [
-            F.group_norm(x[g * B:(g + 1) * B], self.groups, self.weight[g],
-                         self.bias[g])
-            for g in range(G)
-        ]
-        y = torch.cat(outs, dim=0)
+        y = self.norm(x)
         if residual is not None:
             y = y + residual
         return F.relu(y) if self.do_relu else y
@@ -367,7 +378,8 @@ class StackedResNet(ColearnModel):
                  small_input: bool = True, group_norm: bool = False) -> None:
         super().__init__()
         self.G = G
-        self.inplanes = 64
+        self.group_norm = group_norm  # DP mode: per-sample hooks need
+        self.inplanes = 64            # the SLinear module call (no fused head)
         if small_input:
             self.stem = nn.Sequential(
                 SConv2d(G, in_channels, 64, 3, 1, 1),
@@ -409,8 +421,8 @@ class StackedResNet(ColearnModel):
         from ..ops.head import fused_head_applicable, fused_head_loss
 
         feat = self.features(x)
-        if fused_head_applicable(feat, self.fc.weight,
-                                 self.fc.weight.shape[1]):
+        if not self.group_norm and fused_head_applicable(
+                feat, self.fc.weight, self.fc.weight.shape[1]):
             # fused avgpool+FC+CE (4 launches fwd+bwd, fp32 head math,
             # direct grad writes) — csrc/head.hip
             return fused_head_loss(feat, self.fc.weight, self.fc.bias,

@@ -73,7 +73,8 @@ void launch_sq_norm_per_row(const float* g, float* sq, float* partials,
                             int64_t rows, int64_t cols, int chunks,
                             hipStream_t stream);
 void launch_scale_rows_mean(const float* g, const float* scale, float* out,
-                            int64_t rows, int64_t cols, hipStream_t stream);
+                            int64_t rows, int64_t cols, int groups,
+                            hipStream_t stream);
 void launch_topk(const float* v, int64_t n, int64_t k, int* idx, float* val,
                  unsigned* workspace /* >= 1036 u32 */, int* cand,
                  int cand_cap, hipStream_t stream);

@@ -104,13 +104,14 @@ void sq_norm_per_row(torch::Tensor g, torch::Tensor sq) {
 }
 
 void scale_rows_mean(torch::Tensor g, torch::Tensor scale,
-                     torch::Tensor out) {
+                     torch::Tensor out, int64_t groups) {
   check_f32(g, "g");
+  TORCH_CHECK(g.size(0) % groups == 0, "rows must divide by groups");
   const c10::hip::HIPGuardMasqueradingAsCUDA guard(g.device());
   auto gc = g.contiguous();
   launch_scale_rows_mean(gc.data_ptr<float>(), scale.data_ptr<float>(),
                          out.data_ptr<float>(), gc.size(0), gc.size(1),
-                         stream());
+                         static_cast<int>(groups), stream());
   HIP_CHECK_LAST();
 }
 

@@ -54,13 +54,20 @@ __global__ void sq_norm_combine_kernel(const float* __restrict__ partials,
 }
 
 // ---- out[d] = (1/rows) * sum_b scale[b] * g[b,d] --------------------------
+// rows are split into gridDim.z equal GROUPS (stacked DP clients): each
+// z writes its group's mean into out[z * cols ..]
 __global__ void scale_rows_mean_kernel(const float* __restrict__ g,
                                        const float* __restrict__ scale,
                                        float* __restrict__ out, int64_t rows,
                                        int64_t cols) {
+  const int64_t rpg = rows / gridDim.z;
+  g += static_cast<int64_t>(blockIdx.z) * rpg * cols;
+  scale += static_cast<int64_t>(blockIdx.z) * rpg;
+  out += static_cast<int64_t>(blockIdx.z) * cols;
+  rows = rpg;
   int64_t d = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) * 4;
   const float inv = 1.f / static_cast<float>(rows);
-  const int64_t stride = grid_stride() * 4;
+  const int64_t stride = static_cast<int64_t>(blockDim.x) * gridDim.x * 4;
   for (; d + 3 < cols; d += stride) {
     float acc[4] = {0.f, 0.f, 0.f, 0.f};
     for (int64_t b = 0; b < rows; ++b) {
@@ -357,9 +364,11 @@ void launch_sq_norm_per_row(const float* g, float* sq, float* partials,
 }
 
 void launch_scale_rows_mean(const float* g, const float* scale, float* out,
-                            int64_t rows, int64_t cols, hipStream_t stream) {
-  scale_rows_mean_kernel<<<elemwise_grid((cols + 3) / 4), 256, 0, stream>>>(
-      g, scale, out, rows, cols);
+                            int64_t rows, int64_t cols, int groups,
+                            hipStream_t stream) {
+  dim3 grid(elemwise_grid((cols + 3) / 4), 1, groups);
+  scale_rows_mean_kernel<<<grid, 256, 0, stream>>>(g, scale, out, rows,
+                                                   cols);
 }
 
 void launch_topk(const float* v, int64_t n, int64_t k, int* idx, float* val,

@@ -34,13 +34,18 @@ def per_sample_norms(per_sample: Sequence[torch.Tensor]) -> torch.Tensor:
 
 def clip_and_reduce(
     per_sample: Sequence[torch.Tensor], clip_norm: float,
-    pre_scale: float = 1.0,
+    pre_scale: float = 1.0, groups: int = 1,
 ) -> List[torch.Tensor]:
     """Clip each sample's total grad to clip_norm; return per-param means.
 
     ``pre_scale``: the per-sample grads are treated as pre_scale * g (the
     hook engine passes unscaled grads of a MEAN loss with pre_scale = B)
     — the factor folds into the clip scale so no extra sweep is needed.
+
+    ``groups`` > 1 (stacked DP): the B rows are ``groups`` clients'
+    samples client-major; returns per-param per-CLIENT means
+    [groups, *shape] — the whole multi-client DP step reduces in one
+    launch per parameter.
     """
     b = per_sample[0].shape[0]
     norms = per_sample_norms(per_sample) * pre_scale
@@ -50,13 +55,20 @@ def clip_and_reduce(
     if per_sample[0].is_cuda and native_enabled():
         m = require_ext()
         for g in per_sample:
-            o = torch.empty(g.shape[1:], device=g.device, dtype=torch.float32)
-            m.scale_rows_mean(g.reshape(b, -1), scale, o.reshape(-1))
+            shape = ((groups,) + tuple(g.shape[1:]) if groups > 1
+                     else g.shape[1:])
+            o = torch.empty(shape, device=g.device, dtype=torch.float32)
+            m.scale_rows_mean(g.reshape(b, -1), scale, o.reshape(-1),
+                              groups)
             out.append(o)
         return out
     for g in per_sample:
-        flat = g.reshape(b, -1).float()
-        o = (flat * scale[:, None]).mean(dim=0).reshape(g.shape[1:])
+        flat = (g.reshape(b, -1).float() * scale[:, None])
+        if groups > 1:
+            o = flat.view(groups, b // groups, -1).mean(dim=1)
+            o = o.reshape((groups,) + tuple(g.shape[1:]))
+        else:
+            o = flat.mean(dim=0).reshape(g.shape[1:])
         out.append(o)
     return out
 

@@ -82,6 +82,56 @@ def _groupnorm_grad_sample(module, x: torch.Tensor,
     }
 
 
+# ---- stacked-module rules (per-sample grads are weight-free, so the
+# template rules apply with batch = G*B; shapes come out per-TEMPLATE) ----
+def _sconv_grad_sample(module, x: torch.Tensor,
+                       gy: torch.Tensor) -> Dict[str, torch.Tensor]:
+    b = x.shape[0]
+    if x.is_cuda and gy.dtype == torch.bfloat16 and x.dtype != torch.bfloat16:
+        x = x.bfloat16()
+    # stacked weight is [G, CO, KH, KW, CI]
+    kh, kw = module.weight.shape[2], module.weight.shape[3]
+    if (x.is_cuda and native_enabled() and x.dtype == torch.bfloat16
+            and gy.dtype == torch.bfloat16):
+        gw = require_ext().conv2d_wgrad_fp32_km(
+            gy.contiguous(), x.contiguous(), kh, kw,
+            module.stride, module.padding, b,
+        )
+        return {"weight": gw}
+    co, ci = module.weight.shape[1], module.weight.shape[4]
+    gws = [
+        torch.nn.grad.conv2d_weight(
+            x[i : i + 1].float(), (co, ci, kh, kw),
+            gy[i : i + 1].float(), module.stride, module.padding,
+        ).permute(0, 2, 3, 1)
+        for i in range(b)
+    ]
+    return {"weight": torch.stack(gws).float()}
+
+
+def _sgn_grad_sample(module, x: torch.Tensor,
+                     gy: torch.Tensor) -> Dict[str, torch.Tensor]:
+    xf, gf = x.float(), gy.float()
+    xhat = F.group_norm(xf, module.groups, eps=1e-5)
+    dims = tuple(range(2, x.dim()))
+    return {
+        "weight": (gf * xhat).sum(dim=dims),
+        "bias": gf.sum(dim=dims),
+    }
+
+
+def _slinear_grad_sample(module, x: torch.Tensor,
+                         gy: torch.Tensor) -> Dict[str, torch.Tensor]:
+    # per-sample grads vs the sample's OWN client slice = plain outer
+    # product (a sample only touches its client's weights)
+    x2 = x.reshape(x.shape[0], -1, x.shape[-1]).float()
+    g2 = gy.reshape(gy.shape[0], -1, gy.shape[-1]).float()
+    return {
+        "weight": torch.einsum("bsi,bsj->bij", g2, x2),
+        "bias": g2.sum(dim=1),
+    }
+
+
 class GradSampleEngine:
     """Attach to a model; after ``loss.backward()`` on a MEAN loss, every
     supported parameter has ``.grad_sample`` ([B, *shape], fp32, unscaled
@@ -91,6 +141,7 @@ class GradSampleEngine:
         from ..models.base import CLinear
         from ..models.bert import _Linear as BertLinear
         from ..models.resnet import _Conv2d
+        from ..models.stacked import SConv2d, SGroupNorm, SLinear
 
         self.model = model
         self._handles = []
@@ -102,6 +153,12 @@ class GradSampleEngine:
                 self._rules.append((mod, _linear_grad_sample))
             elif isinstance(mod, nn.GroupNorm):
                 self._rules.append((mod, _groupnorm_grad_sample))
+            elif isinstance(mod, SConv2d):
+                self._rules.append((mod, _sconv_grad_sample))
+            elif isinstance(mod, SLinear):
+                self._rules.append((mod, _slinear_grad_sample))
+            elif isinstance(mod, SGroupNorm):
+                self._rules.append((mod, _sgn_grad_sample))
         covered = {
             p for mod, _ in self._rules for p in mod.parameters(recurse=False)
         }
