@@ -40,3 +40,16 @@ sec = t(lambda: g.pow(2).sum(dim=1))
 print(f"torch sqsum  : {sec*1e3:8.2f} ms  {B*D*4/sec/1e12:6.2f} TB/s")
 sec = t(lambda: g.view(G, B // G, D).mean(dim=1))
 print(f"torch gmean  : {sec*1e3:8.2f} ms  {B*D*4/sec/1e12:6.2f} TB/s")
+
+# full config-5-like list: 20 conv tensors + small params, ~23 GB live
+print("\n-- full-list clip_and_reduce (23 GB live) --")
+from colearn_federated_learning_amd.ops.dp import clip_and_reduce
+
+shapes = ([2_359_296] * 4 + [1_179_648] * 3 + [589_824] * 4 +
+          [294_912] * 3 + [147_456] * 4 + [36864, 8192, 5120] +
+          [512] * 20 + [64] * 20)
+per = [torch.randn(B, d, device="cuda") for d in shapes]
+print(f"total {sum(s for s in shapes)*B*4/1e9:.1f} GB")
+sec = t(lambda: clip_and_reduce(per, 1.0, pre_scale=64.0, groups=G), n=5)
+tot = sum(shapes) * B * 4 * 2  # read twice (norms + scale)
+print(f"clip_and_reduce: {sec*1e3:8.2f} ms  {tot/sec/1e12:6.2f} TB/s eff")
