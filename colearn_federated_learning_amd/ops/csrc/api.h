@@ -69,12 +69,12 @@ void launch_fused_adam(float* w, const float* g, float* m, float* v,
                        int64_t ref_period, float lr, float b1, float b2,
                        float eps, float wd, float mu, float bc1, float bc2,
                        bool has_prox, hipStream_t stream);
-void launch_sq_norm_per_row(const float* g, float* sq, float* partials,
-                            int64_t rows, int64_t cols, int chunks,
-                            hipStream_t stream);
-void launch_scale_rows_mean(const float* g, const float* scale, float* out,
-                            int64_t rows, int64_t cols, int groups,
-                            hipStream_t stream);
+void launch_sq_norm_per_row(const void* g, bool bf16_in, float* sq,
+                            float* partials, int64_t rows, int64_t cols,
+                            int chunks, hipStream_t stream);
+void launch_scale_rows_mean(const void* g, bool bf16_in,
+                            const float* scale, float* out, int64_t rows,
+                            int64_t cols, int groups, hipStream_t stream);
 void launch_topk(const float* v, int64_t n, int64_t k, int* idx, float* val,
                  unsigned* workspace /* >= 1036 u32 */, int* cand,
                  int cand_cap, hipStream_t stream);
@@ -95,6 +95,9 @@ void launch_gelu_fwd(const void* x, void* y, int64_t n,
                      hipStream_t stream);
 void launch_gelu_bwd(const void* gy, const void* x, void* gx, int64_t n,
                      hipStream_t stream);
+void launch_gn_grad_sample(const void* x, const void* gy, float* gw,
+                           float* gb, int N, int C, int HW, int groups,
+                           float eps, hipStream_t stream);
 void launch_colsum(const void* x, float* partials, float* out, int64_t B,
                    int64_t M, int64_t N, int chunks, hipStream_t stream);
 void launch_attn_fwd(const void* qkv, void* ctx, void* Psave,

@@ -88,7 +88,9 @@ void fused_adam(torch::Tensor w, torch::Tensor g, torch::Tensor m,
 
 // ---- DP / top-k -----------------------------------------------------------
 void sq_norm_per_row(torch::Tensor g, torch::Tensor sq) {
-  check_f32(g, "g");
+  const bool bf16_in = g.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(bf16_in || g.scalar_type() == torch::kFloat32,
+              "g must be fp32 or bf16");
   check_f32(sq, "sq");
   const c10::hip::HIPGuardMasqueradingAsCUDA guard(g.device());
   auto gc = g.contiguous();
@@ -97,7 +99,7 @@ void sq_norm_per_row(torch::Tensor g, torch::Tensor sq) {
       std::min<int64_t>(std::max<int64_t>(2048 / rows, 1), 64));
   auto partials = torch::empty({rows * chunks},
                                g.options().dtype(torch::kFloat32));
-  launch_sq_norm_per_row(gc.data_ptr<float>(), sq.data_ptr<float>(),
+  launch_sq_norm_per_row(gc.data_ptr(), bf16_in, sq.data_ptr<float>(),
                          partials.data_ptr<float>(), rows, cols, chunks,
                          stream());
   HIP_CHECK_LAST();
@@ -105,11 +107,14 @@ void sq_norm_per_row(torch::Tensor g, torch::Tensor sq) {
 
 void scale_rows_mean(torch::Tensor g, torch::Tensor scale,
                      torch::Tensor out, int64_t groups) {
-  check_f32(g, "g");
+  const bool bf16_in = g.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(bf16_in || g.scalar_type() == torch::kFloat32,
+              "g must be fp32 or bf16");
   TORCH_CHECK(g.size(0) % groups == 0, "rows must divide by groups");
   const c10::hip::HIPGuardMasqueradingAsCUDA guard(g.device());
   auto gc = g.contiguous();
-  launch_scale_rows_mean(gc.data_ptr<float>(), scale.data_ptr<float>(),
+  launch_scale_rows_mean(gc.data_ptr(), bf16_in,
+                         scale.data_ptr<float>(),
                          out.data_ptr<float>(), gc.size(0), gc.size(1),
                          static_cast<int>(groups), stream());
   HIP_CHECK_LAST();
@@ -243,6 +248,31 @@ torch::Tensor gelu_bwd(torch::Tensor gy, torch::Tensor x) {
                   gc.numel(), stream());
   HIP_CHECK_LAST();
   return gx;
+}
+
+// GroupNorm per-sample param grads: x/gy channels-last bf16 [N,C,H,W]
+// -> (gw, gb) fp32 [N, C]
+std::tuple<torch::Tensor, torch::Tensor> gn_grad_sample(
+    torch::Tensor x, torch::Tensor gy, int64_t groups, double eps) {
+  check_bf16(x, "x");
+  check_bf16(gy, "gy");
+  TORCH_CHECK(x.dim() == 4, "x must be NCHW");
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
+  auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
+  auto gc = gy.contiguous(torch::MemoryFormat::ChannelsLast);
+  const int N = x.size(0), C = x.size(1);
+  const int HW = x.size(2) * x.size(3);
+  TORCH_CHECK(((C <= 256 && 256 % C == 0) || C % 256 == 0)
+                  && groups <= 64 && C % groups == 0,
+              "gn_grad_sample: C must divide/be divisible by 256");
+  auto gw = torch::empty({N, C}, x.options().dtype(torch::kFloat32));
+  auto gb = torch::empty_like(gw);
+  launch_gn_grad_sample(xc.data_ptr(), gc.data_ptr(),
+                        gw.data_ptr<float>(), gb.data_ptr<float>(), N, C,
+                        HW, static_cast<int>(groups),
+                        static_cast<float>(eps), stream());
+  HIP_CHECK_LAST();
+  return {gw, gb};
 }
 
 // column sums over dim 1 of [B, M, N] bf16 -> fp32 written into ``out``
@@ -990,6 +1020,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gelu_fwd", &gelu_fwd);
   m.def("gelu_bwd", &gelu_bwd);
   m.def("colsum_f32", &colsum_f32);
+  m.def("gn_grad_sample", &gn_grad_sample);
   m.def("attn_bwd", &attn_bwd);
   m.def("head_fwd", &head_fwd);
   m.def("head_bwd", &head_bwd);

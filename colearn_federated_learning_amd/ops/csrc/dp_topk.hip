@@ -17,23 +17,35 @@ namespace colearn {
 // ---- per-sample squared norms: sq[b] += sum_d g[b,d]^2 --------------------
 // deterministic AND parallel: per-(row, chunk) partials in fixed slots,
 // combined per row in chunk order.
-__global__ void sq_norm_partial_kernel(const float* __restrict__ g,
+template <typename T>
+DEV_INLINE float ld_f(const T* p) {
+  if constexpr (sizeof(T) == 2)
+    return b2f(*reinterpret_cast<const bf16*>(p));
+  else
+    return *reinterpret_cast<const float*>(p);
+}
+
+template <typename T>
+__global__ void sq_norm_partial_kernel(const T* __restrict__ g,
                                        float* __restrict__ partials,
                                        int64_t rows, int64_t cols,
                                        int chunks) {
   __shared__ float scratch[4];
   const int row = blockIdx.x;
   if (row >= rows) return;
-  const float* grow = g + row * cols;
+  const T* grow = g + row * cols;
   float acc = 0.f;
   int64_t i = (static_cast<int64_t>(blockIdx.y) * blockDim.x + threadIdx.x) * 4;
   const int64_t stride = static_cast<int64_t>(gridDim.y) * blockDim.x * 4;
   for (; i + 3 < cols; i += stride) {
-    const float4 v = *reinterpret_cast<const float4*>(grow + i);
-    acc += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const float v = ld_f(grow + i + j);
+      acc += v * v;
+    }
   }
   if (blockIdx.y == 0 && threadIdx.x < (cols & 3)) {
-    const float v = grow[(cols & ~int64_t(3)) + threadIdx.x];
+    const float v = ld_f(grow + (cols & ~int64_t(3)) + threadIdx.x);
     acc += v * v;
   }
   acc = block_allreduce_sum<256>(acc, scratch);
@@ -56,7 +68,8 @@ __global__ void sq_norm_combine_kernel(const float* __restrict__ partials,
 // ---- out[d] = (1/rows) * sum_b scale[b] * g[b,d] --------------------------
 // rows are split into gridDim.z equal GROUPS (stacked DP clients): each
 // z writes its group's mean into out[z * cols ..]
-__global__ void scale_rows_mean_kernel(const float* __restrict__ g,
+template <typename T>
+__global__ void scale_rows_mean_kernel(const T* __restrict__ g,
                                        const float* __restrict__ scale,
                                        float* __restrict__ out, int64_t rows,
                                        int64_t cols) {
@@ -72,11 +85,9 @@ __global__ void scale_rows_mean_kernel(const float* __restrict__ g,
     float acc[4] = {0.f, 0.f, 0.f, 0.f};
     for (int64_t b = 0; b < rows; ++b) {
       const float s = scale[b];
-      const float4 v = *reinterpret_cast<const float4*>(g + b * cols + d);
-      acc[0] += s * v.x;
-      acc[1] += s * v.y;
-      acc[2] += s * v.z;
-      acc[3] += s * v.w;
+      const T* p = g + b * cols + d;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[j] += s * ld_f(p + j);
     }
     *reinterpret_cast<float4*>(out + d) = make_float4(
         acc[0] * inv, acc[1] * inv, acc[2] * inv, acc[3] * inv);
@@ -84,7 +95,8 @@ __global__ void scale_rows_mean_kernel(const float* __restrict__ g,
   if (blockIdx.x == 0 && threadIdx.x < (cols & 3)) {
     const int64_t j = (cols & ~int64_t(3)) + threadIdx.x;
     float acc = 0.f;
-    for (int64_t b = 0; b < rows; ++b) acc += scale[b] * g[b * cols + j];
+    for (int64_t b = 0; b < rows; ++b)
+      acc += scale[b] * ld_f(g + b * cols + j);
     out[j] = acc * inv;
   }
 }
@@ -353,22 +365,30 @@ __global__ void philox_gaussian_add_kernel(float* __restrict__ out,
 }
 
 // ---- launchers ------------------------------------------------------------
-void launch_sq_norm_per_row(const float* g, float* sq, float* partials,
-                            int64_t rows, int64_t cols, int chunks,
-                            hipStream_t stream) {
+void launch_sq_norm_per_row(const void* g, bool bf16_in, float* sq,
+                            float* partials, int64_t rows, int64_t cols,
+                            int chunks, hipStream_t stream) {
   dim3 grid(static_cast<unsigned>(rows), chunks);
-  sq_norm_partial_kernel<<<grid, 256, 0, stream>>>(g, partials, rows, cols,
-                                                   chunks);
+  if (bf16_in)
+    sq_norm_partial_kernel<short><<<grid, 256, 0, stream>>>(
+        static_cast<const short*>(g), partials, rows, cols, chunks);
+  else
+    sq_norm_partial_kernel<float><<<grid, 256, 0, stream>>>(
+        static_cast<const float*>(g), partials, rows, cols, chunks);
   sq_norm_combine_kernel<<<(rows + 255) / 256, 256, 0, stream>>>(
       partials, sq, rows, chunks);
 }
 
-void launch_scale_rows_mean(const float* g, const float* scale, float* out,
-                            int64_t rows, int64_t cols, int groups,
-                            hipStream_t stream) {
+void launch_scale_rows_mean(const void* g, bool bf16_in,
+                            const float* scale, float* out, int64_t rows,
+                            int64_t cols, int groups, hipStream_t stream) {
   dim3 grid(elemwise_grid((cols + 3) / 4), 1, groups);
-  scale_rows_mean_kernel<<<grid, 256, 0, stream>>>(g, scale, out, rows,
-                                                   cols);
+  if (bf16_in)
+    scale_rows_mean_kernel<short><<<grid, 256, 0, stream>>>(
+        static_cast<const short*>(g), scale, out, rows, cols);
+  else
+    scale_rows_mean_kernel<float><<<grid, 256, 0, stream>>>(
+        static_cast<const float*>(g), scale, out, rows, cols);
 }
 
 void launch_topk(const float* v, int64_t n, int64_t k, int* idx, float* val,

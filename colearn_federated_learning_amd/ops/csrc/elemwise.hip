@@ -97,6 +97,98 @@ __global__ void colsum_combine_kernel(const float* __restrict__ partials,
   out[b * N + col] = acc;
 }
 
+// GroupNorm per-sample parameter gradients (stacked DP rule), fused:
+//   gw[n,c] = sum_hw gy * xhat ,  gb[n,c] = sum_hw gy
+// with xhat recomputed from per-(sample, group) stats — 3 bf16 passes
+// over the activation instead of the ~10 fp32 passes of the torch
+// composition (float casts + group_norm recompute + muls + reduces).
+// x/gy are channels-last [N, C, H, W] bf16; one block per sample;
+// deterministic fixed-order combines.
+template <int BLOCK>
+__global__ void gn_grad_sample_kernel(const bf16* __restrict__ x,
+                                      const bf16* __restrict__ gy,
+                                      float* __restrict__ gw,
+                                      float* __restrict__ gb, int C,
+                                      int HW, int groups, float eps) {
+  extern __shared__ float lds[];  // [2][max(spc*C, C)] partials
+  const int n = blockIdx.x;
+  const short* xs = reinterpret_cast<const short*>(x) +
+                    static_cast<int64_t>(n) * HW * C;
+  const short* gs = reinterpret_cast<const short*>(gy) +
+                    static_cast<int64_t>(n) * HW * C;
+  // C <= BLOCK: spc hw-splits per channel; C > BLOCK: one split, each
+  // thread owns C/BLOCK channels.
+  const int spc = C <= BLOCK ? BLOCK / C : 1;
+  const int cpt = C <= BLOCK ? 1 : C / BLOCK;
+  const int slots = spc * C;  // LDS floats per plane
+#pragma unroll 1
+  for (int t = 0; t < cpt; ++t) {
+    const int c = (threadIdx.x % C) + t * BLOCK;
+    const int sl = threadIdx.x / C;
+    float s = 0.f, ss = 0.f;
+    if (sl < spc && c < C) {
+      for (int p = sl; p < HW; p += spc) {
+        const float v = bits2f(xs[static_cast<int64_t>(p) * C + c]);
+        s += v;
+        ss += v * v;
+      }
+      lds[sl * C + c] = s;
+      lds[slots + sl * C + c] = ss;
+    }
+  }
+  __syncthreads();
+  __shared__ float gmean[64], ginv[64];
+  if (threadIdx.x < groups) {
+    const int cpg = C / groups;
+    float as = 0.f, assq = 0.f;
+    for (int cc = threadIdx.x * cpg; cc < (threadIdx.x + 1) * cpg; ++cc)
+      for (int k = 0; k < spc; ++k) {
+        as += lds[k * C + cc];
+        assq += lds[slots + k * C + cc];
+      }
+    const float cnt = static_cast<float>(cpg) * HW;
+    const float mu = as / cnt;
+    gmean[threadIdx.x] = mu;
+    ginv[threadIdx.x] = rsqrtf(fmaxf(assq / cnt - mu * mu, 0.f) + eps);
+  }
+  __syncthreads();
+#pragma unroll 1
+  for (int t = 0; t < cpt; ++t) {
+    const int c = (threadIdx.x % C) + t * BLOCK;
+    const int sl = threadIdx.x / C;
+    if (sl < spc && c < C) {
+      const float mu = gmean[c / (C / groups)];
+      const float inv = ginv[c / (C / groups)];
+      float aw = 0.f, ab = 0.f;
+      for (int p = sl; p < HW; p += spc) {
+        const int64_t i = static_cast<int64_t>(p) * C + c;
+        const float g = bits2f(gs[i]);
+        aw += g * (bits2f(xs[i]) - mu) * inv;
+        ab += g;
+      }
+      if (spc == 1) {  // no cross-split combine needed
+        gw[static_cast<int64_t>(n) * C + c] = aw;
+        gb[static_cast<int64_t>(n) * C + c] = ab;
+      } else {
+        lds[sl * C + c] = aw;
+        lds[slots + sl * C + c] = ab;
+      }
+    }
+  }
+  if (spc > 1) {
+    __syncthreads();
+    if (threadIdx.x < C) {
+      float w = 0.f, b = 0.f;
+      for (int k = 0; k < spc; ++k) {
+        w += lds[k * C + threadIdx.x];
+        b += lds[slots + k * C + threadIdx.x];
+      }
+      gw[static_cast<int64_t>(n) * C + threadIdx.x] = w;
+      gb[static_cast<int64_t>(n) * C + threadIdx.x] = b;
+    }
+  }
+}
+
 // ---- launchers ------------------------------------------------------------
 void launch_gelu_fwd(const void* x, void* y, int64_t n,
                      hipStream_t stream) {
@@ -109,6 +201,17 @@ void launch_gelu_bwd(const void* gy, const void* x, void* gx, int64_t n,
   gelu_bwd_kernel<<<elemwise_grid((n + 7) / 8), 256, 0, stream>>>(
       static_cast<const bf16*>(gy), static_cast<const bf16*>(x),
       static_cast<bf16*>(gx), n);
+}
+
+void launch_gn_grad_sample(const void* x, const void* gy, float* gw,
+                           float* gb, int N, int C, int HW, int groups,
+                           float eps, hipStream_t stream) {
+  const int spc = C <= 256 ? 256 / C : 1;
+  const size_t lds = 2 * static_cast<size_t>(spc * C > C ? spc * C : C) *
+                     sizeof(float);
+  gn_grad_sample_kernel<256><<<N, 256, lds, stream>>>(
+      static_cast<const bf16*>(x), static_cast<const bf16*>(gy), gw, gb,
+      C, HW, groups, eps);
 }
 
 void launch_colsum(const void* x, float* partials, float* out, int64_t B,

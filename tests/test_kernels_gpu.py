@@ -762,3 +762,40 @@ def test_fused_attention_dropout_stats_and_determinism():
     d0 = fused_attention(qkv, H, scale, 0.0, 0)
     assert (a.float() - d0.float()).abs().mean() < 0.15
     assert abs(a.float().mean() - d0.float().mean()) < 2e-2
+
+
+def test_gn_grad_sample_kernel_matches_torch():
+    m = _ext()
+    import torch.nn.functional as F
+
+    torch.manual_seed(19)
+    for C, HW_side in ((64, 8), (256, 4), (512, 4)):
+        N = 12
+        groups = min(32, C)
+        x = (torch.randn(N, C, HW_side, HW_side, device=DEV)
+             .bfloat16().contiguous(memory_format=torch.channels_last))
+        gy = (torch.randn(N, C, HW_side, HW_side, device=DEV)
+              .bfloat16().contiguous(memory_format=torch.channels_last))
+        gw, gb = m.gn_grad_sample(x, gy, groups, 1e-5)
+        xf, gf = x.float(), gy.float()
+        xhat = F.group_norm(xf, groups, eps=1e-5)
+        ref_w = (gf * xhat).sum(dim=(2, 3))
+        ref_b = gf.sum(dim=(2, 3))
+        torch.testing.assert_close(gw, ref_w, rtol=2e-2, atol=2e-2)
+        torch.testing.assert_close(gb, ref_b, rtol=2e-2, atol=2e-2)
+
+
+def test_clip_kernels_bf16_inputs():
+    from colearn_federated_learning_amd.ops.dp import clip_and_reduce
+
+    _ext()
+    torch.manual_seed(20)
+    G, B, D = 4, 8, 1000
+    g = (torch.randn(G * B, D, device=DEV) * 0.3).bfloat16()
+    out = clip_and_reduce([g], clip_norm=0.5, pre_scale=float(B),
+                          groups=G)[0]
+    gf = g.float()
+    norms = (gf * B).norm(dim=1)
+    scale = (0.5 / norms.clamp_min(1e-12)).clamp_max(1.0) * B
+    ref = (gf * scale[:, None]).view(G, B, D).mean(dim=1)
+    torch.testing.assert_close(out, ref, rtol=1e-2, atol=1e-3)
