@@ -242,9 +242,15 @@ class StackedEngine:
         # client mean in one grouped kernel sweep per parameter
         self.gs_engine = None
         if cfg.dp.enabled:
+            from ..models.stacked import SConv2d, SLinear
             from ..ops.grad_sample import GradSampleEngine
 
             self.gs_engine = GradSampleEngine(self.model)
+            # per-sample rules own the weight grads: autograd's mean
+            # wgrad / fc-grad kernels would be overwritten — skip them
+            for mod in self.model.modules():
+                if isinstance(mod, (SConv2d, SLinear)):
+                    mod.dp_mode = True
         self.runner = None
         if cfg.dp.enabled:
             pass  # eager per-sample path (no graphs yet)

@@ -96,15 +96,19 @@ class _SConv2dShadowFn(torch.autograd.Function):
 
     @staticmethod
     def forward(x, w, shadow, shadow_t, gslot, stride: int, padding: int,
-                G: int):
+                G: int, skip_wgrad: bool = False):
         return require_ext().conv2d_fwd_km(x, shadow, stride, padding)
 
     @staticmethod
     def setup_context(ctx, inputs, output):
-        x, w, shadow, shadow_t, gslot, stride, padding, G = inputs
+        (x, w, shadow, shadow_t, gslot, stride, padding, G,
+         *rest) = inputs
         ctx.save_for_backward(x, shadow, shadow_t)
         ctx.gslot = gslot
         ctx.stride, ctx.padding, ctx.G = stride, padding, G
+        # DP: the per-sample hook rule produces the weight grads; the
+        # autograd mean-wgrad would be thrown away — skip the kernel
+        ctx.skip_wgrad = rest[0] if rest else False
 
     @staticmethod
     def backward(ctx, gy):
@@ -123,7 +127,9 @@ class _SConv2dShadowFn(torch.autograd.Function):
         else:
             gx = m.conv2d_dgrad_km(gy, shadow, x.shape[2], x.shape[3],
                                    s, p)
-        if ctx.gslot is not None:
+        if ctx.skip_wgrad:
+            gw = None
+        elif ctx.gslot is not None:
             # write the fp32 grad straight into the flat-grad view; no
             # AccumulateGrad add kernel runs for this parameter
             m.conv2d_wgrad_km32(gy, x, shadow.shape[2], shadow.shape[3],
@@ -133,7 +139,7 @@ class _SConv2dShadowFn(torch.autograd.Function):
             e = torch.empty(0, device=x.device, dtype=torch.float32)
             gw = m.conv2d_wgrad_km32(gy, x, shadow.shape[2],
                                      shadow.shape[3], s, p, G, e)
-        return gx, gw, None, None, None, None, None, None
+        return gx, gw, None, None, None, None, None, None, None
 
 
 class SConv2d(nn.Module):
@@ -146,6 +152,7 @@ class SConv2d(nn.Module):
         self.wshadow = None    # bf16 flat-buffer view (StackedFlatParams)
         self.wshadow_t = None  # pre-transposed dgrad layout [G,CI,R,S,CO]
         self.wgslot = None     # fp32 flat-grad view (direct wgrad target)
+        self.dp_mode = False   # per-sample rule owns the weight grads
         self.G, self.stride, self.padding = G, stride, padding
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
@@ -155,7 +162,7 @@ class SConv2d(nn.Module):
             return _SConv2dShadowFn.apply(x, self.weight, self.wshadow,
                                           self.wshadow_t, self.wgslot,
                                           self.stride, self.padding,
-                                          self.G)
+                                          self.G, self.dp_mode)
         w = self.weight
         if torch.is_autocast_enabled("cuda") and x.is_cuda:
             dt = torch.get_autocast_dtype("cuda")
@@ -220,17 +227,18 @@ class _SLinearShadowFn(torch.autograd.Function):
     AccumulateGrad adds)."""
 
     @staticmethod
-    def forward(x3, w, b, wsh, gslot_w, gslot_b):
+    def forward(x3, w, b, wsh, gslot_w, gslot_b, skip_wgrad: bool = False):
         return require_ext().bmm_nt_bias(x3, wsh, b)
 
     @staticmethod
     def setup_context(ctx, inputs, output):
-        x3, w, b, wsh, gslot_w, gslot_b = inputs
+        x3, w, b, wsh, gslot_w, gslot_b, *rest = inputs
         ctx.save_for_backward(x3, wsh)
         # grad slots stash on ctx directly: they are written in place by
         # every layer's backward, so the saved-tensor version check would
         # (correctly for normal tensors, wrongly here) reject them
         ctx.gslot_w, ctx.gslot_b = gslot_w, gslot_b
+        ctx.skip_wgrad = rest[0] if rest else False
 
     @staticmethod
     def backward(ctx, gy):
@@ -239,9 +247,10 @@ class _SLinearShadowFn(torch.autograd.Function):
         m = require_ext()
         gy = gy.contiguous()
         gx = m.bmm_nn(gy, wsh)
-        m.bmm_tn_f32(gy, x3, gslot_w)
-        m.colsum_f32(gy, gslot_b)  # deterministic fixed-order bias grads
-        return gx, None, None, None, None, None
+        if not ctx.skip_wgrad:  # DP: per-sample rule owns these
+            m.bmm_tn_f32(gy, x3, gslot_w)
+            m.colsum_f32(gy, gslot_b)  # deterministic bias grads
+        return gx, None, None, None, None, None, None
 
 
 class SLinear(nn.Module):
@@ -251,6 +260,7 @@ class SLinear(nn.Module):
         self.bias = nn.Parameter(torch.zeros(G, cout))
         nn.init.normal_(self.weight, std=0.02)
         self.wshadow = None  # bf16 flat-buffer view (StackedFlatParams)
+        self.dp_mode = False
         self.G = G
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
@@ -266,7 +276,7 @@ class SLinear(nn.Module):
                 x3 = x3.bfloat16()
             out = _SLinearShadowFn.apply(
                 x3.contiguous(), self.weight, self.bias, self.wshadow,
-                self.weight.grad, self.bias.grad,
+                self.weight.grad, self.bias.grad, self.dp_mode,
             )
             return out.reshape(-1, out.shape[-1])
         w, b = self.weight, self.bias
@@ -293,8 +303,12 @@ class SGroupNorm(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         B = x.shape[0] // self.G
         xn = F.group_norm(x, self.groups)
-        w = self.weight.repeat_interleave(B, dim=0)  # [G*B, C]
-        b = self.bias.repeat_interleave(B, dim=0)
+        # DETACHED affine in the input dtype: the per-sample hook rule
+        # owns the weight/bias grads (autograd's would be overwritten),
+        # and matching dtype avoids promoting the whole activation to
+        # fp32 through the broadcast
+        w = self.weight.detach().to(x.dtype).repeat_interleave(B, dim=0)
+        b = self.bias.detach().to(x.dtype).repeat_interleave(B, dim=0)
         return xn * w[:, :, None, None] + b[:, :, None, None]
 
 

@@ -571,15 +571,38 @@ __global__ void wgrad_combine_kernel(const float* __restrict__ gwf,
   const int g = blockIdx.y;
   const float* base = gwf + static_cast<int64_t>(g) * splitk * n_per_g;
   OUT* out = gw + static_cast<int64_t>(g) * n_per_g;
-  int64_t i = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
-  const int64_t stride = static_cast<int64_t>(blockDim.x) * gridDim.x;
-  for (; i < n_per_g; i += stride) {
+  int64_t i =
+      (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) * 4;
+  const int64_t stride = static_cast<int64_t>(blockDim.x) * gridDim.x * 4;
+  for (; i + 3 < n_per_g; i += stride) {
+    float4 a = *reinterpret_cast<const float4*>(base + i);
+    for (int sk = 1; sk < splitk; ++sk) {
+      const float4 v =
+          *reinterpret_cast<const float4*>(base + sk * n_per_g + i);
+      a.x += v.x;
+      a.y += v.y;
+      a.z += v.z;
+      a.w += v.w;
+    }
+    if constexpr (sizeof(OUT) == 2) {
+      short4v o;
+      o.x[0] = f2bits(a.x);
+      o.x[1] = f2bits(a.y);
+      o.x[2] = f2bits(a.z);
+      o.x[3] = f2bits(a.w);
+      *reinterpret_cast<short4v*>(reinterpret_cast<short*>(out) + i) = o;
+    } else {
+      *reinterpret_cast<float4*>(reinterpret_cast<float*>(out) + i) = a;
+    }
+  }
+  if (blockIdx.x == 0 && threadIdx.x < (n_per_g & 3)) {
+    const int64_t j = (n_per_g & ~int64_t(3)) + threadIdx.x;
     float acc = 0.f;
-    for (int sk = 0; sk < splitk; ++sk) acc += base[sk * n_per_g + i];
+    for (int sk = 0; sk < splitk; ++sk) acc += base[sk * n_per_g + j];
     if constexpr (sizeof(OUT) == 2)
-      out[i] = f2b(acc);
+      out[j] = f2b(acc);
     else
-      out[i] = acc;
+      out[j] = acc;
   }
 }
 
@@ -676,7 +699,13 @@ void launch_conv_wgrad_out(const void* gy, const void* x, float* gwf,
                  static_cast<const short*>(x), gwf, cs, NPQ, RSC, G,
                  splitk, stream);
   const int64_t n_per_g = static_cast<int64_t>(cs.K) * RSC;
-  int cgrid_x = elemwise_grid(n_per_g, 256, G > 0 ? 2048 / G : 2048);
+  // target ~8-16k total blocks: the per-sample-grad path launches with
+  // G in the thousands (the old 2048/G cap collapsed to ONE block per g
+  // and ran the combine at 0.5 TB/s)
+  int cap = G > 0 ? 16384 / G : 2048;
+  if (cap < 1) cap = 1;
+  if (cap > 2048) cap = 2048;
+  int cgrid_x = elemwise_grid((n_per_g + 3) / 4, 256, cap);
   if (cgrid_x < 1) cgrid_x = 1;
   if (f32_out)
     wgrad_combine_kernel<<<dim3(cgrid_x, G), 256, 0, stream>>>(
@@ -895,7 +924,13 @@ void launch_conv_wgrad_variant(const void* gy, const void* x, float* gwf,
         static_cast<const short*>(gy), static_cast<const short*>(x), gwf,
         cs, NPQ, RSC, nbm, nbn, splitk);
   const int64_t n_per_g = static_cast<int64_t>(cs.K) * RSC;
-  int cgrid_x = elemwise_grid(n_per_g, 256, G > 0 ? 2048 / G : 2048);
+  // target ~8-16k total blocks: the per-sample-grad path launches with
+  // G in the thousands (the old 2048/G cap collapsed to ONE block per g
+  // and ran the combine at 0.5 TB/s)
+  int cap = G > 0 ? 16384 / G : 2048;
+  if (cap < 1) cap = 1;
+  if (cap > 2048) cap = 2048;
+  int cgrid_x = elemwise_grid((n_per_g + 3) / 4, 256, cap);
   if (cgrid_x < 1) cgrid_x = 1;
   wgrad_combine_kernel<<<dim3(cgrid_x, G), 256, 0, stream>>>(
       gwf, static_cast<bf16*>(gw), n_per_g, splitk);
