@@ -78,6 +78,7 @@ class StackedFlatParams:
         issue each step (~10% of step time — gpurun_out/eager_attrib.txt).
         """
         from ..models.stacked import SBatchNorm2d, SConv2d, SLinear
+        from ..models.stacked_bert import SLayerNorm
 
         self.flat_bf16 = torch.empty_like(self.flat_w,
                                           dtype=torch.bfloat16)
@@ -114,6 +115,9 @@ class StackedFlatParams:
                                  else "weight"]
                 mod.wshadow = (self.flat_bf16[off : off + self.G * n]
                                .view(mod.weight.shape))
+            elif isinstance(mod, SLayerNorm):
+                mod.gslot_w = mod.weight.grad
+                mod.gslot_b = mod.bias.grad
         dev = self.device
         self.flat_bf16_t = torch.empty(max(t_off, 1),
                                        dtype=torch.bfloat16, device=dev)

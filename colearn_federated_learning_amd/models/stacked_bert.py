@@ -42,11 +42,13 @@ class SLayerNorm(nn.Module):
         super().__init__()
         self.weight = nn.Parameter(torch.ones(G, hidden))
         self.bias = nn.Parameter(torch.zeros(G, hidden))
+        self.gslot_w = None  # flat-grad views (direct grad targets)
+        self.gslot_b = None
         self.G, self.eps = G, eps
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         y, _, _ = _LayerNormFn.apply(x, self.weight, self.bias, self.eps,
-                                     self.G)
+                                     self.G, self.gslot_w, self.gslot_b)
         return y
 
 

@@ -117,6 +117,9 @@ void launch_layernorm_fwd(const void* x, const float* w, const float* b,
                           void* y, float* mean, float* invstd, int64_t rows,
                           int H, float eps, int G, hipStream_t stream);
 int norms_reduce_grid(int64_t n8);
+void launch_ln_combine_slots(const float* partials, float* gw, float* gb,
+                             int H, int G, int nchunk,
+                             hipStream_t stream);
 void launch_layernorm_bwd(const void* gy, const void* x, const float* w,
                           const float* mean, const float* invstd, void* gx,
                           float* gw_gb, float* partials, int64_t rows, int H,

@@ -284,7 +284,10 @@ void colsum_f32(torch::Tensor x, torch::Tensor out) {
   const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   auto xc = x.contiguous();
   const int64_t B = x.size(0), M = x.size(1), N = x.size(2);
-  const int chunks = 16;
+  // enough chunks to fill the chip: [B, chunks, N/256] blocks
+  int chunks = static_cast<int>(
+      std::min<int64_t>(std::max<int64_t>(4096 / ((N / 256 + 1) * B), 16),
+                        128));
   auto partials = torch::empty({B * chunks * N},
                                out.options());
   launch_colsum(xc.data_ptr(), partials.data_ptr<float>(),
@@ -424,6 +427,40 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> layernorm_bwd(
   HIP_CHECK_LAST();
   return {gx, gw_gb.narrow(1, 0, H).clone(),
           gw_gb.narrow(1, H, H).clone()};
+}
+
+// direct-slot variant: the combined [G, H] grads land straight in the
+// flat-grad views (no clones, no AccumulateGrad adds)
+torch::Tensor layernorm_bwd_slots(torch::Tensor gy, torch::Tensor x,
+                                  torch::Tensor w, torch::Tensor mean,
+                                  torch::Tensor invstd, int64_t G,
+                                  torch::Tensor gw_slot,
+                                  torch::Tensor gb_slot) {
+  check_bf16(gy, "gy");
+  check_f32(gw_slot, "gw_slot");
+  check_f32(gb_slot, "gb_slot");
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
+  auto gc = gy.contiguous(), xc = x.contiguous();
+  auto mc = mean.contiguous(), ic = invstd.contiguous();
+  const int H = static_cast<int>(x.size(-1));
+  const int64_t rows = x.numel() / H;
+  auto gx = torch::empty_like(xc);
+  auto gw_gb = torch::empty({G, 2 * H},
+                            x.options().dtype(torch::kFloat32));
+  auto partials = torch::empty({G * 64 * 2 * H},
+                               x.options().dtype(torch::kFloat32));
+  launch_layernorm_bwd(
+      gc.data_ptr(), xc.data_ptr(), w.data_ptr<float>(),
+      mc.data_ptr<float>(), ic.data_ptr<float>(), gx.data_ptr(),
+      gw_gb.data_ptr<float>(), partials.data_ptr<float>(), rows, H,
+      static_cast<int>(G), stream());
+  // overwrite the interleaved combine with a direct-to-slot combine
+  launch_ln_combine_slots(partials.data_ptr<float>(),
+                          gw_slot.data_ptr<float>(),
+                          gb_slot.data_ptr<float>(), H,
+                          static_cast<int>(G), 64, stream());
+  HIP_CHECK_LAST();
+  return gx;
 }
 
 // ---- BatchNorm (channels-last) -------------------------------------------
@@ -1030,6 +1067,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scatter_add_pairs", &scatter_add_pairs);
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("layernorm_bwd_slots", &layernorm_bwd_slots);
   m.def("batchnorm_fwd", &batchnorm_fwd);
   m.def("batchnorm_bwd", &batchnorm_bwd);
   m.def("gemm_nt", &gemm_nt);

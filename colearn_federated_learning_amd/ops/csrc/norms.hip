@@ -547,6 +547,34 @@ void launch_layernorm_fwd(const void* x, const float* w, const float* b,
       invstd, rows, H, eps, rows / G);
 }
 
+// combine LN gw/gb partials [G][nchunk][2H] straight into the two flat
+// grad-slot tensors (no clone, no AccumulateGrad)
+__global__ void ln_combine_slots_kernel(const float* __restrict__ partials,
+                                        float* __restrict__ gw,
+                                        float* __restrict__ gb, int H,
+                                        int nblk) {
+  const int g = blockIdx.y;
+  const float* base =
+      partials + static_cast<int64_t>(g) * nblk * 2 * H;
+  const int h = blockIdx.x * blockDim.x + threadIdx.x;
+  if (h >= H) return;
+  float aw = 0.f, ab = 0.f;
+  for (int b = 0; b < nblk; ++b) {
+    aw += base[static_cast<int64_t>(b) * 2 * H + h];
+    ab += base[static_cast<int64_t>(b) * 2 * H + H + h];
+  }
+  gw[static_cast<int64_t>(g) * H + h] = aw;
+  gb[static_cast<int64_t>(g) * H + h] = ab;
+}
+
+void launch_ln_combine_slots(const float* partials, float* gw, float* gb,
+                             int H, int G, int nchunk,
+                             hipStream_t stream) {
+  dim3 g2((H + 255) / 256, G);
+  ln_combine_slots_kernel<<<g2, 256, 0, stream>>>(partials, gw, gb, H,
+                                                  nchunk);
+}
+
 void launch_layernorm_bwd(const void* gy_, const void* x_, const float* w,
                           const float* mean, const float* invstd, void* gx_,
                           float* gw_gb /*[G*2H] out*/, float* partials,

@@ -164,7 +164,8 @@ class _LayerNormFn(torch.autograd.Function):
     generate_vmap_rule = True
 
     @staticmethod
-    def forward(x, weight, bias, eps: float, groups: int = 1):
+    def forward(x, weight, bias, eps: float, groups: int = 1,
+                gslot_w=None, gslot_b=None):
         if _native(x):
             y, mean, invstd = require_ext().layernorm_fwd(
                 x.contiguous(), weight.view(-1), bias.view(-1), eps,
@@ -200,6 +201,8 @@ class _LayerNormFn(torch.autograd.Function):
         _, mean, invstd = output
         ctx.save_for_backward(x, weight, mean, invstd)
         ctx.groups = inputs[4] if len(inputs) > 4 else 1
+        ctx.gslot_w = inputs[5] if len(inputs) > 5 else None
+        ctx.gslot_b = inputs[6] if len(inputs) > 6 else None
         ctx.n_inputs = len(inputs)
         ctx.mark_non_differentiable(output[1], output[2])
         ctx.set_materialize_grads(False)  # see _BatchNormFn
@@ -209,7 +212,15 @@ class _LayerNormFn(torch.autograd.Function):
         x, weight, mean, invstd = ctx.saved_tensors
         G = ctx.groups
         gy = gy.contiguous()
-        if _native(x):
+        if _native(x) and ctx.gslot_w is not None:
+            # direct-slot path: gw/gb land straight in the flat-grad
+            # views (no clones, no AccumulateGrad adds)
+            gx = require_ext().layernorm_bwd_slots(
+                gy, x.contiguous(), weight.view(-1), mean, invstd, G,
+                ctx.gslot_w.view(-1), ctx.gslot_b.view(-1),
+            )
+            gw = gb = None
+        elif _native(x):
             gx, gw, gb = require_ext().layernorm_bwd(
                 gy, x.contiguous(), weight.view(-1), mean, invstd, G
             )
@@ -246,7 +257,7 @@ class _LayerNormFn(torch.autograd.Function):
                 - xhat * (wg * xhat).mean(dim=-1, keepdim=True)
             )
             gx = gx.to(x.dtype)
-        out = [gx, gw, gb, None, None]
+        out = [gx, gw, gb, None, None, None, None]
         return tuple(out[: ctx.n_inputs])
 
 
