@@ -118,6 +118,15 @@ class StackedFlatParams:
             elif isinstance(mod, SLayerNorm):
                 mod.gslot_w = mod.weight.grad
                 mod.gslot_b = mod.bias.grad
+            else:
+                from ..models.stacked_bert import SEmbedding
+
+                if isinstance(mod, SEmbedding):
+                    off, n = by_name[f"{mod_name}.weight" if mod_name
+                                     else "weight"]
+                    mod.wshadow = (
+                        self.flat_bf16[off : off + self.G * n]
+                        .view(mod.weight.shape))
         dev = self.device
         self.flat_bf16_t = torch.empty(max(t_off, 1),
                                        dtype=torch.bfloat16, device=dev)

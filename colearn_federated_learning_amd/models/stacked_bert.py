@@ -59,6 +59,7 @@ class SEmbedding(nn.Module):
         super().__init__()
         self.weight = nn.Parameter(torch.empty(G, n, hidden))
         nn.init.normal_(self.weight, std=0.02)
+        self.wshadow = None  # bf16 flat view (tied MLM decoder operand)
         self.G = G
 
     def forward(self, ids: torch.Tensor) -> torch.Tensor:
@@ -189,11 +190,28 @@ class StackedBert(ColearnModel):
         hk = torch.gather(h3, 1, order[..., None].expand(-1, -1, d))
         yk = torch.gather(y2, 1, order)  # [G, k]
         hk = self.head_ln(colearn_gelu(self.head_fc(hk.reshape(G * k, d))))
-        w = self.tok_emb.weight
-        if hk.dtype != w.dtype:
-            w = w.to(hk.dtype)
-        logits = (colearn_matmul_nt(hk.view(G, k, d), w)
-                  + self.head_bias[:, None, :])     # [G, k, V]
+        if (self.tok_emb.wshadow is not None
+                and hk.dtype == torch.bfloat16
+                and self.tok_emb.weight.grad is not None):
+            # tied decoder through the shadow-linear path: bf16 shadow
+            # operand, bias fused into the GEMM epilogue (no fp32 logits
+            # promotion), fp32 decoder wgrad straight into the embedding
+            # grad slot (the embedding-gather backward scatter-adds on
+            # top afterwards — backward runs decoder-first, so the
+            # overwrite-then-accumulate order is correct)
+            from .stacked import _SLinearShadowFn
+
+            logits = _SLinearShadowFn.apply(
+                hk.view(G, k, d).contiguous(), self.tok_emb.weight,
+                self.head_bias, self.tok_emb.wshadow,
+                self.tok_emb.weight.grad, self.head_bias.grad, False,
+            )
+        else:
+            w = self.tok_emb.weight
+            if hk.dtype != w.dtype:
+                w = w.to(hk.dtype)
+            logits = (colearn_matmul_nt(hk.view(G, k, d), w)
+                      + self.head_bias[:, None, :])     # [G, k, V]
         if logits.is_cuda and logits.dtype == torch.bfloat16:
             # CE straight on bf16 logits: torch's log_softmax accumulates
             # in fp32 internally; the explicit .float() materialized a
