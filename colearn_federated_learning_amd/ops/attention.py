@@ -32,11 +32,13 @@ class _FusedAttnFn(torch.autograd.Function):
 
 
 def fused_attention_applicable(qkv: torch.Tensor) -> bool:
-    """qkv [N, S, 3, H, dh]: native path needs bf16 CUDA, dh 64,
-    S <= 128 and a multiple of 32 (the BERT-base MLM config)."""
+    """qkv [N, S, 3, H, dh]: native path needs bf16 CUDA, dh 64 and
+    S == 128 exactly — the kernel's fragment layout is specialized to
+    the BERT-base MLM config (8 key-fragments per wave); other shapes
+    fall back to the batched-GEMM composition."""
     return (qkv.is_cuda and native_enabled()
             and qkv.dtype == torch.bfloat16 and qkv.shape[-1] == 64
-            and qkv.shape[1] <= 128 and qkv.shape[1] % 32 == 0)
+            and qkv.shape[1] == 128)
 
 
 def fused_attention(qkv: torch.Tensor, heads: int, scale: float,

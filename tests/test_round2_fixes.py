@@ -209,3 +209,23 @@ def test_grouped_clip_and_reduce_cpu_oracle():
     scale = (0.7 / norms.clamp_min(1e-12)).clamp_max(1.0) * B
     ref = (g * scale[:, None]).view(G, B, D).mean(dim=1)
     torch.testing.assert_close(out, ref, rtol=1e-5, atol=1e-6)
+
+
+def test_stacked_dp_subsampled():
+    """Stacked DP with clients_per_round < num_clients: the engine pool
+    builds a right-sized DP engine and matches sequential DP."""
+    cfg = driver_config(5).override(
+        rounds=1, num_clients=4, clients_per_round=2,
+        data__samples_per_client=16, train__batch_size=8,
+        train__dtype="float32", train__max_local_steps=1,
+        dp__noise_multiplier=0.0, num_evaluators=0,
+    )
+    s_stack = FederatedServer(cfg)
+    s_seq = FederatedServer(cfg.override(train__use_stacked=False))
+    ra = s_stack.run_round(0)
+    rb = s_seq.run_round(0)
+    assert set(ra["client_losses"]) == set(rb["client_losses"])
+    assert len(ra["client_losses"]) == 2
+    torch.testing.assert_close(s_stack.global_flat, s_seq.global_flat,
+                               rtol=1e-3, atol=5e-5)
+    assert s_stack.stacked._engines[2].gs_engine is not None
