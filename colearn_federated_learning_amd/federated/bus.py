@@ -14,7 +14,6 @@ from __future__ import annotations
 
 import fnmatch
 import threading
-from collections import defaultdict
 from dataclasses import dataclass, field
 from typing import Any, Callable, Dict, List, Tuple
 

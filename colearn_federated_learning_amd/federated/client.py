@@ -12,7 +12,6 @@ count + loss) is preserved.
 
 from __future__ import annotations
 
-import math
 from dataclasses import dataclass
 from typing import Dict, Optional
 
@@ -228,7 +227,7 @@ def _per_sample_clipped_backward(
     except NotImplementedError:
         pass
 
-    from torch.func import functional_call, grad, vmap
+    from torch.func import grad, vmap
 
     params = {k: v.detach() for k, v in model.named_parameters()}
     buffers = {k: v.detach() for k, v in model.named_buffers()}

@@ -20,7 +20,7 @@ layout, so slice extraction is a segment-wise reshape.
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
 import torch
 

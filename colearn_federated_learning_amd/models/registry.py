@@ -2,8 +2,6 @@
 
 from __future__ import annotations
 
-import torch
-
 from ..config import DataConfig, ModelConfig
 from .base import ColearnModel, LogisticRegression
 

@@ -18,7 +18,7 @@ stacked engine is fully testable without a GPU.
 
 from __future__ import annotations
 
-from typing import List, Optional, Type
+from typing import List, Optional
 
 import torch
 import torch.nn as nn

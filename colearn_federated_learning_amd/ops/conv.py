@@ -7,8 +7,6 @@ CPU path = F.conv2d (oracle).
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 import torch.nn.functional as F
 

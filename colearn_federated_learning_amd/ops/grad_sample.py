@@ -24,7 +24,7 @@ touches the [B, *param] tensors.
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 import torch
 import torch.nn as nn

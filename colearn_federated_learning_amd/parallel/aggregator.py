@@ -27,7 +27,7 @@ import torch
 import torch.distributed as dist
 
 from ..config import AggregationConfig
-from ..ops.topk import topk_compress, topk_decompress
+from ..ops.topk import topk_decompress
 
 
 class Aggregator:
