@@ -287,6 +287,34 @@ class SLinear(nn.Module):
         return out.reshape(-1, out.shape[-1])
 
 
+class _SGNFn(torch.autograd.Function):
+    """Native stacked GroupNorm (fwd + gx backward); the per-client
+    weight/bias grads come from the per-sample hook rule, so this
+    Function returns None for them."""
+
+    @staticmethod
+    def forward(x, w, b, groups: int, B_per_client: int):
+        y, mean, invstd = require_ext().sgn_fwd(
+            x, w, b, groups, B_per_client, 1e-5)
+        return y, mean, invstd
+
+    @staticmethod
+    def setup_context(ctx, inputs, output):
+        x, w = inputs[0], inputs[1]
+        _, mean, invstd = output
+        ctx.save_for_backward(x, w, mean, invstd)
+        ctx.groups, ctx.bpc = inputs[3], inputs[4]
+        ctx.mark_non_differentiable(output[1], output[2])
+        ctx.set_materialize_grads(False)
+
+    @staticmethod
+    def backward(ctx, gy, _gm, _gi):
+        x, w, mean, invstd = ctx.saved_tensors
+        gx = require_ext().sgn_bwd(gy.contiguous(), x.contiguous(), w,
+                                   mean, invstd, ctx.groups, ctx.bpc)
+        return gx, None, None, None, None
+
+
 class SGroupNorm(nn.Module):
     """Per-client GroupNorm (DP mode): normalize per SAMPLE (GroupNorm
     semantics), affine per CLIENT.  A standalone module so the
@@ -302,6 +330,10 @@ class SGroupNorm(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         B = x.shape[0] // self.G
+        if (_native(x) and self.weight.shape[1] % 8 == 0):
+            y, _, _ = _SGNFn.apply(x, self.weight, self.bias,
+                                   self.groups, B)
+            return y
         xn = F.group_norm(x, self.groups)
         # DETACHED affine in the input dtype: the per-sample hook rule
         # owns the weight/bias grads (autograd's would be overwritten),

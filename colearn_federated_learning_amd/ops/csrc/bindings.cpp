@@ -250,6 +250,55 @@ torch::Tensor gelu_bwd(torch::Tensor gy, torch::Tensor x) {
   return gx;
 }
 
+// Stacked GroupNorm fwd/bwd: per-sample stats, per-client affine
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> sgn_fwd(
+    torch::Tensor x, torch::Tensor w, torch::Tensor b, int64_t groups,
+    int64_t B_per_client, double eps) {
+  check_bf16(x, "x");
+  check_f32(w, "w");
+  check_f32(b, "b");
+  TORCH_CHECK(x.dim() == 4, "x must be NCHW");
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
+  auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
+  const int N = x.size(0), C = x.size(1);
+  const int HW = x.size(2) * x.size(3);
+  TORCH_CHECK(((C <= 256 && 256 % C == 0) || C % 256 == 0)
+                  && groups <= 64 && C % groups == 0,
+              "sgn: C must divide/be divisible by 256");
+  auto y = torch::empty_like(xc);
+  auto mean = torch::empty({N, groups},
+                           x.options().dtype(torch::kFloat32));
+  auto invstd = torch::empty_like(mean);
+  launch_sgn_fwd(xc.data_ptr(), w.contiguous().data_ptr<float>(),
+                 b.contiguous().data_ptr<float>(), y.data_ptr(),
+                 mean.data_ptr<float>(), invstd.data_ptr<float>(), N, C,
+                 HW, static_cast<int>(groups),
+                 static_cast<int>(B_per_client), static_cast<float>(eps),
+                 stream());
+  HIP_CHECK_LAST();
+  return {y, mean, invstd};
+}
+
+torch::Tensor sgn_bwd(torch::Tensor gy, torch::Tensor x, torch::Tensor w,
+                      torch::Tensor mean, torch::Tensor invstd,
+                      int64_t groups, int64_t B_per_client) {
+  check_bf16(gy, "gy");
+  check_bf16(x, "x");
+  const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
+  auto gc = gy.contiguous(torch::MemoryFormat::ChannelsLast);
+  auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
+  const int N = x.size(0), C = x.size(1);
+  const int HW = x.size(2) * x.size(3);
+  auto gx = torch::empty_like(xc);
+  launch_sgn_bwd(gc.data_ptr(), xc.data_ptr(),
+                 w.contiguous().data_ptr<float>(),
+                 mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                 gx.data_ptr(), N, C, HW, static_cast<int>(groups),
+                 static_cast<int>(B_per_client), stream());
+  HIP_CHECK_LAST();
+  return gx;
+}
+
 // GroupNorm per-sample param grads: x/gy channels-last bf16 [N,C,H,W]
 // -> (gw, gb) fp32 [N, C]
 std::tuple<torch::Tensor, torch::Tensor> gn_grad_sample(
@@ -1058,6 +1107,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gelu_bwd", &gelu_bwd);
   m.def("colsum_f32", &colsum_f32);
   m.def("gn_grad_sample", &gn_grad_sample);
+  m.def("sgn_fwd", &sgn_fwd);
+  m.def("sgn_bwd", &sgn_bwd);
   m.def("attn_bwd", &attn_bwd);
   m.def("head_fwd", &head_fwd);
   m.def("head_bwd", &head_bwd);
