@@ -287,52 +287,62 @@ class SLinear(nn.Module):
         return out.reshape(-1, out.shape[-1])
 
 
-class _SGNFn(torch.autograd.Function):
-    """Native stacked GroupNorm (fwd + gx backward); the per-client
-    weight/bias grads come from the per-sample hook rule, so this
-    Function returns None for them."""
+class _SGNActFn(torch.autograd.Function):
+    """Native stacked GroupNorm + per-client affine + optional residual
+    add + ReLU, fwd and gx/gres backward in two kernels (elemwise.hip).
+    The per-client weight/bias grads come from the per-sample hook rule
+    (DP mode), so this Function returns None for them."""
 
     @staticmethod
-    def forward(x, w, b, groups: int, B_per_client: int):
+    def forward(x, w, b, residual, groups: int, B_per_client: int,
+                relu: bool):
         y, mean, invstd = require_ext().sgn_fwd(
-            x, w, b, groups, B_per_client, 1e-5)
+            x, w, b,
+            residual if residual is not None
+            else torch.empty(0, device=x.device, dtype=x.dtype),
+            groups, B_per_client, 1e-5, relu)
         return y, mean, invstd
 
     @staticmethod
     def setup_context(ctx, inputs, output):
         x, w = inputs[0], inputs[1]
-        _, mean, invstd = output
-        ctx.save_for_backward(x, w, mean, invstd)
-        ctx.groups, ctx.bpc = inputs[3], inputs[4]
+        y, mean, invstd = output
+        ctx.save_for_backward(x, w, y, mean, invstd)
+        ctx.groups, ctx.bpc, ctx.relu = inputs[4], inputs[5], inputs[6]
+        ctx.has_res = inputs[3] is not None
         ctx.mark_non_differentiable(output[1], output[2])
         ctx.set_materialize_grads(False)
 
     @staticmethod
     def backward(ctx, gy, _gm, _gi):
-        x, w, mean, invstd = ctx.saved_tensors
-        gx = require_ext().sgn_bwd(gy.contiguous(), x.contiguous(), w,
-                                   mean, invstd, ctx.groups, ctx.bpc)
-        return gx, None, None, None, None
+        x, w, y, mean, invstd = ctx.saved_tensors
+        gx, gres = require_ext().sgn_bwd(
+            gy.contiguous(), x.contiguous(), y, w, mean, invstd,
+            ctx.groups, ctx.bpc, ctx.relu, ctx.has_res)
+        return (gx, None, None, gres if ctx.has_res else None, None,
+                None, None)
 
 
-class SGroupNorm(nn.Module):
-    """Per-client GroupNorm (DP mode): normalize per SAMPLE (GroupNorm
-    semantics), affine per CLIENT.  A standalone module so the
-    per-sample-grad engine can hook its exact output (the residual/ReLU
-    epilogue lives in _SGNAct, outside the hook boundary)."""
+class _SGNAct(nn.Module):
+    """Stacked GroupNorm substitute (DP mode): per-SAMPLE stats,
+    per-CLIENT affine, fused residual/ReLU epilogue on the native path.
+    Params live directly on this module (the per-sample grad engine
+    hooks it and masks grad_output by the saved ReLU output)."""
 
-    def __init__(self, G: int, planes: int) -> None:
+    def __init__(self, G: int, planes: int, relu: bool) -> None:
         super().__init__()
         self.weight = nn.Parameter(torch.ones(G, planes))
         self.bias = nn.Parameter(torch.zeros(G, planes))
         self.G = G
         self.groups = min(32, planes)
+        self.do_relu = relu
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, residual=None) -> torch.Tensor:
         B = x.shape[0] // self.G
-        if (_native(x) and self.weight.shape[1] % 8 == 0):
-            y, _, _ = _SGNFn.apply(x, self.weight, self.bias,
-                                   self.groups, B)
+        if _native(x) and self.weight.shape[1] % 8 == 0:
+            y, _, _ = _SGNActFn.apply(x, self.weight, self.bias,
+                                      residual, self.groups, B,
+                                      self.do_relu)
             return y
         xn = F.group_norm(x, self.groups)
         # DETACHED affine in the input dtype: the per-sample hook rule
@@ -341,19 +351,7 @@ class SGroupNorm(nn.Module):
         # fp32 through the broadcast
         w = self.weight.detach().to(x.dtype).repeat_interleave(B, dim=0)
         b = self.bias.detach().to(x.dtype).repeat_interleave(B, dim=0)
-        return xn * w[:, :, None, None] + b[:, :, None, None]
-
-
-class _SGNAct(nn.Module):
-    """Stacked GroupNorm substitute (DP mode) + residual/ReLU epilogue."""
-
-    def __init__(self, G: int, planes: int, relu: bool) -> None:
-        super().__init__()
-        self.norm = SGroupNorm(G, planes)
-        self.do_relu = relu
-
-    def forward(self, x: torch.Tensor, residual=None) -> torch.Tensor:
-        y = self.norm(x)
+        y = xn * w[:, :, None, None] + b[:, :, None, None]
         if residual is not None:
             y = y + residual
         return F.relu(y) if self.do_relu else y

@@ -96,13 +96,14 @@ void launch_gelu_fwd(const void* x, void* y, int64_t n,
 void launch_gelu_bwd(const void* gy, const void* x, void* gx, int64_t n,
                      hipStream_t stream);
 void launch_sgn_fwd(const void* x, const float* w, const float* b,
-                    void* y, float* mean, float* invstd, int N, int C,
-                    int HW, int groups, int B_per_client, float eps,
-                    hipStream_t stream);
-void launch_sgn_bwd(const void* gy, const void* x, const float* w,
-                    const float* mean, const float* invstd, void* gx,
+                    const void* res, void* y, float* mean, float* invstd,
                     int N, int C, int HW, int groups, int B_per_client,
-                    hipStream_t stream);
+                    float eps, bool relu, hipStream_t stream);
+void launch_sgn_bwd(const void* gy, const void* x, const void* yout,
+                    const float* w, const float* mean,
+                    const float* invstd, void* gx, void* gres, int N,
+                    int C, int HW, int groups, int B_per_client,
+                    bool relu, hipStream_t stream);
 void launch_gn_grad_sample(const void* x, const void* gy, float* gw,
                            float* gb, int N, int C, int HW, int groups,
                            float eps, hipStream_t stream);

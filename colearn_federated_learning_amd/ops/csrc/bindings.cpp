@@ -252,14 +252,18 @@ torch::Tensor gelu_bwd(torch::Tensor gy, torch::Tensor x) {
 
 // Stacked GroupNorm fwd/bwd: per-sample stats, per-client affine
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> sgn_fwd(
-    torch::Tensor x, torch::Tensor w, torch::Tensor b, int64_t groups,
-    int64_t B_per_client, double eps) {
+    torch::Tensor x, torch::Tensor w, torch::Tensor b,
+    torch::Tensor residual, int64_t groups, int64_t B_per_client,
+    double eps, bool relu) {
   check_bf16(x, "x");
   check_f32(w, "w");
   check_f32(b, "b");
   TORCH_CHECK(x.dim() == 4, "x must be NCHW");
   const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
+  const bool has_res = residual.numel() > 0;
+  torch::Tensor rc;
+  if (has_res) rc = residual.contiguous(torch::MemoryFormat::ChannelsLast);
   const int N = x.size(0), C = x.size(1);
   const int HW = x.size(2) * x.size(3);
   TORCH_CHECK(((C <= 256 && 256 % C == 0) || C % 256 == 0)
@@ -270,18 +274,20 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> sgn_fwd(
                            x.options().dtype(torch::kFloat32));
   auto invstd = torch::empty_like(mean);
   launch_sgn_fwd(xc.data_ptr(), w.contiguous().data_ptr<float>(),
-                 b.contiguous().data_ptr<float>(), y.data_ptr(),
+                 b.contiguous().data_ptr<float>(),
+                 has_res ? rc.data_ptr() : nullptr, y.data_ptr(),
                  mean.data_ptr<float>(), invstd.data_ptr<float>(), N, C,
                  HW, static_cast<int>(groups),
                  static_cast<int>(B_per_client), static_cast<float>(eps),
-                 stream());
+                 relu, stream());
   HIP_CHECK_LAST();
   return {y, mean, invstd};
 }
 
-torch::Tensor sgn_bwd(torch::Tensor gy, torch::Tensor x, torch::Tensor w,
-                      torch::Tensor mean, torch::Tensor invstd,
-                      int64_t groups, int64_t B_per_client) {
+std::tuple<torch::Tensor, torch::Tensor> sgn_bwd(
+    torch::Tensor gy, torch::Tensor x, torch::Tensor yout,
+    torch::Tensor w, torch::Tensor mean, torch::Tensor invstd,
+    int64_t groups, int64_t B_per_client, bool relu, bool need_gres) {
   check_bf16(gy, "gy");
   check_bf16(x, "x");
   const c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
@@ -290,13 +296,19 @@ torch::Tensor sgn_bwd(torch::Tensor gy, torch::Tensor x, torch::Tensor w,
   const int N = x.size(0), C = x.size(1);
   const int HW = x.size(2) * x.size(3);
   auto gx = torch::empty_like(xc);
-  launch_sgn_bwd(gc.data_ptr(), xc.data_ptr(),
+  torch::Tensor gres;
+  if (need_gres)
+    gres = torch::empty_like(gc);
+  else
+    gres = torch::empty({0}, gc.options());
+  launch_sgn_bwd(gc.data_ptr(), xc.data_ptr(), yout.data_ptr(),
                  w.contiguous().data_ptr<float>(),
                  mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                 gx.data_ptr(), N, C, HW, static_cast<int>(groups),
-                 static_cast<int>(B_per_client), stream());
+                 gx.data_ptr(), need_gres ? gres.data_ptr() : nullptr, N,
+                 C, HW, static_cast<int>(groups),
+                 static_cast<int>(B_per_client), relu, stream());
   HIP_CHECK_LAST();
-  return gx;
+  return {gx, gres};
 }
 
 // GroupNorm per-sample param grads: x/gy channels-last bf16 [N,C,H,W]

@@ -199,11 +199,12 @@ template <int BLOCK>
 __global__ void sgn_fwd_kernel(const bf16* __restrict__ x,
                                const float* __restrict__ w,
                                const float* __restrict__ b,
+                               const bf16* __restrict__ res,
                                bf16* __restrict__ y,
                                float* __restrict__ mean_out,
                                float* __restrict__ invstd_out, int C,
                                int HW, int groups, int B_per_client,
-                               float eps) {
+                               float eps, bool relu) {
   extern __shared__ float lds[];
   const int n = blockIdx.x;
   const int g = n / B_per_client;  // client
@@ -258,9 +259,16 @@ __global__ void sgn_fwd_kernel(const bf16* __restrict__ x,
       const int grp = c / (C / groups);
       const float mu = gmean[grp], inv = ginv[grp];
       const float wc = wg[c], bc = bg[c];
+      const short* rs =
+          res ? reinterpret_cast<const short*>(res) +
+                    static_cast<int64_t>(n) * HW * C
+              : nullptr;
       for (int p = sl; p < HW; p += spc) {
         const int64_t i = static_cast<int64_t>(p) * C + c;
-        ys[i] = f2bits((bits2f(xs[i]) - mu) * inv * wc + bc);
+        float v = (bits2f(xs[i]) - mu) * inv * wc + bc;
+        if (rs) v += bits2f(rs[i]);
+        if (relu) v = fmaxf(v, 0.f);
+        ys[i] = f2bits(v);
       }
     }
   }
@@ -271,11 +279,13 @@ __global__ void sgn_fwd_kernel(const bf16* __restrict__ x,
 template <int BLOCK>
 __global__ void sgn_bwd_kernel(const bf16* __restrict__ gy,
                                const bf16* __restrict__ x,
+                               const bf16* __restrict__ yout,
                                const float* __restrict__ w,
                                const float* __restrict__ mean,
                                const float* __restrict__ invstd,
-                               bf16* __restrict__ gx, int C, int HW,
-                               int groups, int B_per_client) {
+                               bf16* __restrict__ gx,
+                               bf16* __restrict__ gres, int C, int HW,
+                               int groups, int B_per_client, bool relu) {
   extern __shared__ float lds[];
   const int n = blockIdx.x;
   const int g = n / B_per_client;
@@ -303,10 +313,18 @@ __global__ void sgn_bwd_kernel(const bf16* __restrict__ gy,
     if (sl < spc && c < C) {
       const int grp = c / (C / groups);
       const float mu = gmu[grp], inv = ginv[grp], wc = wg[c];
+      const short* yo = reinterpret_cast<const short*>(yout) +
+                        static_cast<int64_t>(n) * HW * C;
+      short* gr = gres ? reinterpret_cast<short*>(gres) +
+                             static_cast<int64_t>(n) * HW * C
+                       : nullptr;
       float s1 = 0.f, s2 = 0.f;  // sum(du), sum(du * xhat)
       for (int p = sl; p < HW; p += spc) {
         const int64_t i = static_cast<int64_t>(p) * C + c;
-        const float du = bits2f(gs[i]) * wc;
+        float g = bits2f(gs[i]);
+        if (relu && bits2f(yo[i]) <= 0.f) g = 0.f;
+        if (gr) gr[i] = f2bits(g);
+        const float du = g * wc;
         s1 += du;
         s2 += du * (bits2f(xs[i]) - mu) * inv;
       }
@@ -336,9 +354,13 @@ __global__ void sgn_bwd_kernel(const bf16* __restrict__ gy,
       const int grp = c / (C / groups);
       const float mu = gmu[grp], inv = ginv[grp], wc = wg[c];
       const float m1 = g1[grp], m2 = g2[grp];
+      const short* yo = reinterpret_cast<const short*>(yout) +
+                        static_cast<int64_t>(n) * HW * C;
       for (int p = sl; p < HW; p += spc) {
         const int64_t i = static_cast<int64_t>(p) * C + c;
-        const float du = bits2f(gs[i]) * wc;
+        float g = bits2f(gs[i]);
+        if (relu && bits2f(yo[i]) <= 0.f) g = 0.f;
+        const float du = g * wc;
         const float xhat = (bits2f(xs[i]) - mu) * inv;
         os[i] = f2bits(inv * (du - m1 - xhat * m2));
       }
@@ -361,27 +383,31 @@ void launch_gelu_bwd(const void* gy, const void* x, void* gx, int64_t n,
 }
 
 void launch_sgn_fwd(const void* x, const float* w, const float* b,
-                    void* y, float* mean, float* invstd, int N, int C,
-                    int HW, int groups, int B_per_client, float eps,
-                    hipStream_t stream) {
+                    const void* res, void* y, float* mean, float* invstd,
+                    int N, int C, int HW, int groups, int B_per_client,
+                    float eps, bool relu, hipStream_t stream) {
   const int spc = C <= 256 ? 256 / C : 1;
   const size_t lds = 2 * static_cast<size_t>(spc * C > C ? spc * C : C) *
                      sizeof(float);
   sgn_fwd_kernel<256><<<N, 256, lds, stream>>>(
-      static_cast<const bf16*>(x), w, b, static_cast<bf16*>(y), mean,
-      invstd, C, HW, groups, B_per_client, eps);
+      static_cast<const bf16*>(x), w, b, static_cast<const bf16*>(res),
+      static_cast<bf16*>(y), mean, invstd, C, HW, groups, B_per_client,
+      eps, relu);
 }
 
-void launch_sgn_bwd(const void* gy, const void* x, const float* w,
-                    const float* mean, const float* invstd, void* gx,
-                    int N, int C, int HW, int groups, int B_per_client,
-                    hipStream_t stream) {
+void launch_sgn_bwd(const void* gy, const void* x, const void* yout,
+                    const float* w, const float* mean,
+                    const float* invstd, void* gx, void* gres, int N,
+                    int C, int HW, int groups, int B_per_client,
+                    bool relu, hipStream_t stream) {
   const int spc = C <= 256 ? 256 / C : 1;
   const size_t lds = 2 * static_cast<size_t>(spc * C > C ? spc * C : C) *
                      sizeof(float);
   sgn_bwd_kernel<256><<<N, 256, lds, stream>>>(
-      static_cast<const bf16*>(gy), static_cast<const bf16*>(x), w, mean,
-      invstd, static_cast<bf16*>(gx), C, HW, groups, B_per_client);
+      static_cast<const bf16*>(gy), static_cast<const bf16*>(x),
+      static_cast<const bf16*>(yout), w, mean, invstd,
+      static_cast<bf16*>(gx), static_cast<bf16*>(gres), C, HW, groups,
+      B_per_client, relu);
 }
 
 void launch_gn_grad_sample(const void* x, const void* gy, float* gw,

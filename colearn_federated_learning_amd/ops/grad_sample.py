@@ -114,12 +114,18 @@ def _sconv_grad_sample(module, x: torch.Tensor,
     return {"weight": torch.stack(gws).float()}
 
 
-def _sgn_grad_sample(module, x: torch.Tensor,
-                     gy: torch.Tensor) -> Dict[str, torch.Tensor]:
+def _sgnact_grad_sample(module, x: torch.Tensor,
+                        gy: torch.Tensor) -> Dict[str, torch.Tensor]:
+    # grad_output is wrt the POST-ReLU output: mask by the saved output
+    # (saved by the engine's forward hook) before the affine-grad sums
+    if module.do_relu:
+        y = module._gs_output
+        gy = gy * (y > 0).to(gy.dtype)
     if (x.is_cuda and native_enabled() and x.dtype == torch.bfloat16
             and gy.dtype == torch.bfloat16):
         # fused kernel: 3 bf16 passes instead of ~10 fp32 passes
-        gw, gb = require_ext().gn_grad_sample(x, gy, module.groups, 1e-5)
+        gw, gb = require_ext().gn_grad_sample(x, gy.contiguous(),
+                                              module.groups, 1e-5)
         return {"weight": gw, "bias": gb}
     xf, gf = x.float(), gy.float()
     xhat = F.group_norm(xf, module.groups, eps=1e-5)
@@ -151,7 +157,7 @@ class GradSampleEngine:
         from ..models.base import CLinear
         from ..models.bert import _Linear as BertLinear
         from ..models.resnet import _Conv2d
-        from ..models.stacked import SConv2d, SGroupNorm, SLinear
+        from ..models.stacked import SConv2d, SLinear, _SGNAct
 
         self.model = model
         self._handles = []
@@ -167,8 +173,8 @@ class GradSampleEngine:
                 self._rules.append((mod, _sconv_grad_sample))
             elif isinstance(mod, SLinear):
                 self._rules.append((mod, _slinear_grad_sample))
-            elif isinstance(mod, SGroupNorm):
-                self._rules.append((mod, _sgn_grad_sample))
+            elif isinstance(mod, _SGNAct):
+                self._rules.append((mod, _sgnact_grad_sample))
         covered = {
             p for mod, _ in self._rules for p in mod.parameters(recurse=False)
         }
@@ -199,10 +205,15 @@ class GradSampleEngine:
         for mod, _ in self._rules:
             if hasattr(mod, "_gs_input"):
                 del mod._gs_input
+            if hasattr(mod, "_gs_output"):
+                del mod._gs_output
 
     @staticmethod
-    def _save_input(mod, inputs, _output) -> None:
+    def _save_input(mod, inputs, output) -> None:
         mod._gs_input = inputs[0].detach()
+        # the GN rule masks grad_output by the post-ReLU output
+        mod._gs_output = (output.detach()
+                          if isinstance(output, torch.Tensor) else None)
 
     def _make_bwd(self, rule):
         def hook(mod, _grad_input, grad_output):
