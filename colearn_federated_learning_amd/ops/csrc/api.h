@@ -104,9 +104,10 @@ void launch_sgn_bwd(const void* gy, const void* x, const void* yout,
                     const float* invstd, void* gx, void* gres, int N,
                     int C, int HW, int groups, int B_per_client,
                     bool relu, hipStream_t stream);
-void launch_gn_grad_sample(const void* x, const void* gy, float* gw,
-                           float* gb, int N, int C, int HW, int groups,
-                           float eps, hipStream_t stream);
+void launch_gn_grad_sample(const void* x, const void* gy,
+                           const void* yout, float* gw, float* gb, int N,
+                           int C, int HW, int groups, float eps,
+                           bool relu, hipStream_t stream);
 void launch_colsum(const void* x, float* partials, float* out, int64_t B,
                    int64_t M, int64_t N, int chunks, hipStream_t stream);
 void launch_attn_fwd(const void* qkv, void* ctx, void* Psave,

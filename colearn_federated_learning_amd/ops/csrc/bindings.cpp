@@ -314,7 +314,8 @@ std::tuple<torch::Tensor, torch::Tensor> sgn_bwd(
 // GroupNorm per-sample param grads: x/gy channels-last bf16 [N,C,H,W]
 // -> (gw, gb) fp32 [N, C]
 std::tuple<torch::Tensor, torch::Tensor> gn_grad_sample(
-    torch::Tensor x, torch::Tensor gy, int64_t groups, double eps) {
+    torch::Tensor x, torch::Tensor gy, torch::Tensor yout,
+    int64_t groups, double eps, bool relu) {
   check_bf16(x, "x");
   check_bf16(gy, "gy");
   TORCH_CHECK(x.dim() == 4, "x must be NCHW");
@@ -326,12 +327,15 @@ std::tuple<torch::Tensor, torch::Tensor> gn_grad_sample(
   TORCH_CHECK(((C <= 256 && 256 % C == 0) || C % 256 == 0)
                   && groups <= 64 && C % groups == 0,
               "gn_grad_sample: C must divide/be divisible by 256");
+  torch::Tensor yc;
+  if (relu) yc = yout.contiguous(torch::MemoryFormat::ChannelsLast);
   auto gw = torch::empty({N, C}, x.options().dtype(torch::kFloat32));
   auto gb = torch::empty_like(gw);
   launch_gn_grad_sample(xc.data_ptr(), gc.data_ptr(),
+                        relu ? yc.data_ptr() : nullptr,
                         gw.data_ptr<float>(), gb.data_ptr<float>(), N, C,
                         HW, static_cast<int>(groups),
-                        static_cast<float>(eps), stream());
+                        static_cast<float>(eps), relu, stream());
   HIP_CHECK_LAST();
   return {gw, gb};
 }

@@ -107,15 +107,20 @@ __global__ void colsum_combine_kernel(const float* __restrict__ partials,
 template <int BLOCK>
 __global__ void gn_grad_sample_kernel(const bf16* __restrict__ x,
                                       const bf16* __restrict__ gy,
+                                      const bf16* __restrict__ yout,
                                       float* __restrict__ gw,
                                       float* __restrict__ gb, int C,
-                                      int HW, int groups, float eps) {
+                                      int HW, int groups, float eps,
+                                      bool relu) {
   extern __shared__ float lds[];  // [2][max(spc*C, C)] partials
   const int n = blockIdx.x;
   const short* xs = reinterpret_cast<const short*>(x) +
                     static_cast<int64_t>(n) * HW * C;
   const short* gs = reinterpret_cast<const short*>(gy) +
                     static_cast<int64_t>(n) * HW * C;
+  const short* yo = yout ? reinterpret_cast<const short*>(yout) +
+                               static_cast<int64_t>(n) * HW * C
+                         : nullptr;
   // C <= BLOCK: spc hw-splits per channel; C > BLOCK: one split, each
   // thread owns C/BLOCK channels.
   const int spc = C <= BLOCK ? BLOCK / C : 1;
@@ -162,7 +167,8 @@ __global__ void gn_grad_sample_kernel(const bf16* __restrict__ x,
       float aw = 0.f, ab = 0.f;
       for (int p = sl; p < HW; p += spc) {
         const int64_t i = static_cast<int64_t>(p) * C + c;
-        const float g = bits2f(gs[i]);
+        float g = bits2f(gs[i]);
+        if (relu && bits2f(yo[i]) <= 0.f) g = 0.f;
         aw += g * (bits2f(xs[i]) - mu) * inv;
         ab += g;
       }
@@ -410,15 +416,16 @@ void launch_sgn_bwd(const void* gy, const void* x, const void* yout,
       B_per_client, relu);
 }
 
-void launch_gn_grad_sample(const void* x, const void* gy, float* gw,
-                           float* gb, int N, int C, int HW, int groups,
-                           float eps, hipStream_t stream) {
+void launch_gn_grad_sample(const void* x, const void* gy,
+                           const void* yout, float* gw, float* gb, int N,
+                           int C, int HW, int groups, float eps,
+                           bool relu, hipStream_t stream) {
   const int spc = C <= 256 ? 256 / C : 1;
   const size_t lds = 2 * static_cast<size_t>(spc * C > C ? spc * C : C) *
                      sizeof(float);
   gn_grad_sample_kernel<256><<<N, 256, lds, stream>>>(
-      static_cast<const bf16*>(x), static_cast<const bf16*>(gy), gw, gb,
-      C, HW, groups, eps);
+      static_cast<const bf16*>(x), static_cast<const bf16*>(gy),
+      static_cast<const bf16*>(yout), gw, gb, C, HW, groups, eps, relu);
 }
 
 void launch_colsum(const void* x, float* partials, float* out, int64_t B,

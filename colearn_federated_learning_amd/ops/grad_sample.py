@@ -116,17 +116,17 @@ def _sconv_grad_sample(module, x: torch.Tensor,
 
 def _sgnact_grad_sample(module, x: torch.Tensor,
                         gy: torch.Tensor) -> Dict[str, torch.Tensor]:
-    # grad_output is wrt the POST-ReLU output: mask by the saved output
-    # (saved by the engine's forward hook) before the affine-grad sums
-    if module.do_relu:
-        y = module._gs_output
-        gy = gy * (y > 0).to(gy.dtype)
+    # grad_output is wrt the POST-ReLU output: the kernel masks by the
+    # saved output in-flight (no gy*(y>0) temporary)
     if (x.is_cuda and native_enabled() and x.dtype == torch.bfloat16
             and gy.dtype == torch.bfloat16):
-        # fused kernel: 3 bf16 passes instead of ~10 fp32 passes
-        gw, gb = require_ext().gn_grad_sample(x, gy.contiguous(),
-                                              module.groups, 1e-5)
+        yout = (module._gs_output if module.do_relu
+                else torch.empty(0, device=x.device, dtype=x.dtype))
+        gw, gb = require_ext().gn_grad_sample(
+            x, gy, yout, module.groups, 1e-5, module.do_relu)
         return {"weight": gw, "bias": gb}
+    if module.do_relu:
+        gy = gy * (module._gs_output > 0).to(gy.dtype)
     xf, gf = x.float(), gy.float()
     xhat = F.group_norm(xf, module.groups, eps=1e-5)
     dims = tuple(range(2, x.dim()))

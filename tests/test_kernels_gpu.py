@@ -776,7 +776,9 @@ def test_gn_grad_sample_kernel_matches_torch():
              .bfloat16().contiguous(memory_format=torch.channels_last))
         gy = (torch.randn(N, C, HW_side, HW_side, device=DEV)
               .bfloat16().contiguous(memory_format=torch.channels_last))
-        gw, gb = m.gn_grad_sample(x, gy, groups, 1e-5)
+        gw, gb = m.gn_grad_sample(
+            x, gy, torch.empty(0, device=DEV, dtype=torch.bfloat16),
+            groups, 1e-5, False)
         xf, gf = x.float(), gy.float()
         xhat = F.group_norm(xf, groups, eps=1e-5)
         ref_w = (gf * xhat).sum(dim=(2, 3))
@@ -838,7 +840,8 @@ def test_sgn_fwd_bwd_matches_torch():
          .contiguous(memory_format=torch.channels_last))
     w = torch.rand(G, C, device=DEV) + 0.5
     b = torch.randn(G, C, device=DEV) * 0.1
-    y, mean, invstd = m.sgn_fwd(x, w, b, groups, B, 1e-5)
+    empty = torch.empty(0, device=DEV, dtype=torch.bfloat16)
+    y, mean, invstd = m.sgn_fwd(x, w, b, empty, groups, B, 1e-5, False)
     xf = x.float()
     ref = F.group_norm(xf, groups, eps=1e-5)
     wf = w.repeat_interleave(B, dim=0)
@@ -847,7 +850,8 @@ def test_sgn_fwd_bwd_matches_torch():
     torch.testing.assert_close(y.float(), ref, rtol=2e-2, atol=2e-2)
     # backward gx vs autograd (params detached — matches the Function)
     gy = (torch.randn_like(x))
-    gx = m.sgn_bwd(gy, x, w, mean, invstd, groups, B)
+    gx, _ = m.sgn_bwd(gy, x, y, w, mean, invstd, groups, B, False,
+                      False)
     xo = x.float().clone().requires_grad_(True)
     yo = F.group_norm(xo, groups, eps=1e-5)
     yo = yo * wf[:, :, None, None] + bf[:, :, None, None]
