@@ -896,8 +896,11 @@ torch::Tensor conv2d_wgrad_km(torch::Tensor gy, torch::Tensor x, int64_t R,
   auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
   const int64_t rsc = static_cast<int64_t>(cs.R) * cs.S * cs.C;
   const int splitk = conv_wgrad_splitk(cs, static_cast<int>(G));
-  auto gwf = torch::empty({G * splitk, cs.K, rsc},
-                          x.options().dtype(torch::kFloat32));
+  // splitk == 1 writes the final dtype straight from the epilogue —
+  // don't reserve the (up to ~180 GB at G=4096) slab
+  auto gwf = torch::empty(
+      {splitk > 1 ? G * splitk : 0, cs.K, rsc},
+      x.options().dtype(torch::kFloat32));
   auto gw = torch::empty({G, cs.K, cs.R, cs.S, cs.C}, x.options());
   launch_conv_wgrad(gc.data_ptr(), xc.data_ptr(), gwf.data_ptr<float>(),
                     gw.data_ptr(), cs, static_cast<int>(G), stream());
@@ -934,8 +937,11 @@ torch::Tensor conv2d_wgrad_km32(torch::Tensor gy, torch::Tensor x,
   auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
   const int64_t rsc = static_cast<int64_t>(cs.R) * cs.S * cs.C;
   const int splitk = conv_wgrad_splitk(cs, static_cast<int>(G));
-  auto gwf = torch::empty({G * splitk, cs.K, rsc},
-                          x.options().dtype(torch::kFloat32));
+  // splitk == 1 writes the final dtype straight from the epilogue —
+  // don't reserve the (up to ~180 GB at G=4096) slab
+  auto gwf = torch::empty(
+      {splitk > 1 ? G * splitk : 0, cs.K, rsc},
+      x.options().dtype(torch::kFloat32));
   const bool direct = out.numel() > 0 && ci == cs.C;
   torch::Tensor gw;
   if (direct) {
@@ -1083,8 +1089,11 @@ torch::Tensor conv2d_wgrad(torch::Tensor gy, torch::Tensor x, int64_t R,
   auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
   const int64_t rsc = static_cast<int64_t>(cs.R) * cs.S * cs.C;
   const int splitk = conv_wgrad_splitk(cs, static_cast<int>(G));
-  auto gwf = torch::empty({G * splitk, cs.K, rsc},
-                          x.options().dtype(torch::kFloat32));
+  // splitk == 1 writes the final dtype straight from the epilogue —
+  // don't reserve the (up to ~180 GB at G=4096) slab
+  auto gwf = torch::empty(
+      {splitk > 1 ? G * splitk : 0, cs.K, rsc},
+      x.options().dtype(torch::kFloat32));
   auto gw_rsci = torch::empty({G, cs.K, cs.R, cs.S, cs.C}, x.options());
   const char* env = getenv("COLEARN_WGRAD_VARIANT");
   const int variant = env ? atoi(env) : 0;

@@ -381,10 +381,10 @@ static void conv_nt_dispatch(const short* A, const short* B, bf16* C,
 // variants restore occupancy.  Waves always cover 64 output rows; the 4
 // waves split the other way: TM=128 -> 2x2 waves of 64x(TN/2),
 // TM=64 -> 1x4 waves of 64x(TN/4).
-template <int TM, int TN>
+template <int TM, int TN, typename OUT = float>
 __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
     const short* __restrict__ gy, const short* __restrict__ x,
-    float* __restrict__ gwf, ConvShape cs, int64_t NPQ, int64_t RSC,
+    OUT* __restrict__ gwf, ConvShape cs, int64_t NPQ, int64_t RSC,
     int nbm, int nbn, int splitk) {
   constexpr int WAVES_M = TM / 64;
   constexpr int WAVES_N = 4 / WAVES_M;
@@ -415,8 +415,8 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
   const short* gyg = gy + static_cast<int64_t>(g) * NPQ * cs.K;
   const short* xg =
       x + static_cast<int64_t>(g) * cs.N * cs.H * cs.W * cs.C;
-  float* slab = gwf + (static_cast<int64_t>(g) * splitk + blockIdx.y) *
-                          cs.K * RSC;
+  OUT* slab = gwf + (static_cast<int64_t>(g) * splitk + blockIdx.y) *
+                        cs.K * RSC;
   const int tile = blockIdx.x;
   const int co0 = (tile / nbn) * TM;
   const int rc0 = (tile % nbn) * TN;
@@ -556,8 +556,12 @@ __global__ __launch_bounds__(kBlockThreads) void conv_wgrad_kernel(
       for (int j = 0; j < 4; ++j) {
         const int co = co0 + wm + m * 16 + rg * 4 + j;
         const int rc = rc0 + wn + n * 16 + cif;
-        if (co < cs.K && rc < RSC)
-          slab[static_cast<int64_t>(co) * RSC + rc] = acc[m][n][j];
+        if (co < cs.K && rc < RSC) {
+          if constexpr (sizeof(OUT) == 2)
+            slab[static_cast<int64_t>(co) * RSC + rc] = f2b(acc[m][n][j]);
+          else
+            slab[static_cast<int64_t>(co) * RSC + rc] = acc[m][n][j];
+        }
       }
 }
 
@@ -655,23 +659,24 @@ int conv_wgrad_splitk(ConvShape cs, int G) {
   return static_cast<int>(want < cap ? want : cap);
 }
 
-static void wgrad_dispatch(const short* gy, const short* x, float* gwf,
+template <typename OUT>
+static void wgrad_dispatch(const short* gy, const short* x, OUT* gwf,
                            ConvShape cs, int64_t NPQ, int64_t RSC, int G,
                            int splitk, hipStream_t stream) {
   int tm, tn, nbm, nbn;
   wgrad_tiles(cs, RSC, tm, tn, nbm, nbn);
   dim3 grid(nbm * nbn, splitk, G);
   if (tm == 64 && tn == 64)
-    conv_wgrad_kernel<64, 64><<<grid, kBlockThreads, 0, stream>>>(
+    conv_wgrad_kernel<64, 64, OUT><<<grid, kBlockThreads, 0, stream>>>(
         gy, x, gwf, cs, NPQ, RSC, nbm, nbn, splitk);
   else if (tm == 64)
-    conv_wgrad_kernel<64, 128><<<grid, kBlockThreads, 0, stream>>>(
+    conv_wgrad_kernel<64, 128, OUT><<<grid, kBlockThreads, 0, stream>>>(
         gy, x, gwf, cs, NPQ, RSC, nbm, nbn, splitk);
   else if (tn == 64)
-    conv_wgrad_kernel<128, 64><<<grid, kBlockThreads, 0, stream>>>(
+    conv_wgrad_kernel<128, 64, OUT><<<grid, kBlockThreads, 0, stream>>>(
         gy, x, gwf, cs, NPQ, RSC, nbm, nbn, splitk);
   else
-    conv_wgrad_kernel<128, 128><<<grid, kBlockThreads, 0, stream>>>(
+    conv_wgrad_kernel<128, 128, OUT><<<grid, kBlockThreads, 0, stream>>>(
         gy, x, gwf, cs, NPQ, RSC, nbm, nbn, splitk);
 }
 
@@ -679,9 +684,9 @@ void launch_conv_wgrad_nosum(const void* gy, const void* x, float* gwf,
                              ConvShape cs, int G, hipStream_t stream) {
   const int64_t NPQ = static_cast<int64_t>(cs.N) * cs.P * cs.Q;
   const int64_t RSC = static_cast<int64_t>(cs.R) * cs.S * cs.C;
-  wgrad_dispatch(static_cast<const short*>(gy),
-                 static_cast<const short*>(x), gwf, cs, NPQ, RSC, G, 1,
-                 stream);
+  wgrad_dispatch<float>(static_cast<const short*>(gy),
+                        static_cast<const short*>(x), gwf, cs, NPQ, RSC,
+                        G, 1, stream);
 }
 
 void launch_conv_wgrad(const void* gy, const void* x, float* gwf, void* gw,
@@ -695,9 +700,25 @@ void launch_conv_wgrad_out(const void* gy, const void* x, float* gwf,
   const int64_t NPQ = static_cast<int64_t>(cs.N) * cs.P * cs.Q;
   const int64_t RSC = static_cast<int64_t>(cs.R) * cs.S * cs.C;
   const int splitk = conv_wgrad_splitk(cs, G);
-  wgrad_dispatch(static_cast<const short*>(gy),
-                 static_cast<const short*>(x), gwf, cs, NPQ, RSC, G,
-                 splitk, stream);
+  if (splitk == 1) {
+    // no K-split: the epilogue writes the FINAL output dtype straight
+    // into gw — no fp32 slab, no combine pass (the per-sample DP path
+    // at G=4096 wrote ~180 GB/step of slabs just to cast them)
+    if (f32_out)
+      wgrad_dispatch<float>(static_cast<const short*>(gy),
+                            static_cast<const short*>(x),
+                            static_cast<float*>(gw), cs, NPQ, RSC, G, 1,
+                            stream);
+    else
+      wgrad_dispatch<bf16>(static_cast<const short*>(gy),
+                           static_cast<const short*>(x),
+                           static_cast<bf16*>(gw), cs, NPQ, RSC, G, 1,
+                           stream);
+    return;
+  }
+  wgrad_dispatch<float>(static_cast<const short*>(gy),
+                        static_cast<const short*>(x), gwf, cs, NPQ, RSC,
+                        G, splitk, stream);
   const int64_t n_per_g = static_cast<int64_t>(cs.K) * RSC;
   // target ~8-16k total blocks: the per-sample-grad path launches with
   // G in the thousands (the old 2048/G cap collapsed to ONE block per g
