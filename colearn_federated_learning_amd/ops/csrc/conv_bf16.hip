@@ -371,9 +371,13 @@ static void conv_nt_dispatch(const short* A, const short* B, bf16* C,
 // ==== wgrad ===============================================================
 // gw[g, co, rsci]: 128x128 tile per block (4 waves, 4x4 fragments each,
 // 128 FLOP per staged LDS element — 2x the 64x64 v1), BK=32 over npq;
-// deterministic split-K slabs per client.  grid = (tiles, splitk, G).
-// Both operands transposed-scatter into K-major LDS (b16 writes; the
-// tr_b16 hardware-transpose read is the planned next step).
+// deterministic split-K slabs per client; splitk==1 writes the FINAL
+// dtype from the epilogue (no slab/combine).  grid = (tiles, splitk, G).
+// Both operands transposed-scatter into the XOR-swizzled slot LDS
+// (round 2: fragment reads are single conflict-free ds_read_b128s and
+// the physical-row remap spreads the scatter-write banks — MFMA-busy
+// ratio 0.76 -> 0.96, see profiles/wgrad_r2_evidence.md; the earlier
+// ds_read_b64_tr_b16 hardware-transpose plan is superseded).
 //
 // Tile size is a template param: shapes with K <= 64 (ResNet stem/layer1)
 // or RSC <= 64 (1x1 downsamples) were burning >half their MFMA issue on
