@@ -17,13 +17,13 @@ from colearn_federated_learning_amd import driver_config
 
 
 def _make_cfg(tmpdir: str, **over):
-    cfg = driver_config(1).override(
+    base = dict(
         rounds=3, num_clients=4, num_evaluators=1,
         data__samples_per_client=64,
         checkpoint_dir=os.path.join(tmpdir, "ck"), checkpoint_every=0,
-        **over,
     )
-    return cfg
+    base.update(over)  # tests may override the harness defaults too
+    return driver_config(1).override(**base)
 
 
 def _worker(rank: int, world: int, tmpdir: str, port: int, over_json: str):
@@ -111,6 +111,30 @@ def test_world2_equals_world1(tmp_path, over):
         assert float(diff.mean()) < 1e-5 and float(diff.max()) < 5e-2
     else:
         torch.testing.assert_close(w1, w2, rtol=1e-4, atol=1e-5)
+
+
+def test_world2_resnet_stacked(tmp_path):
+    """ResNet layout-invariance: stacked engines of different G (world1
+    trains its clients in one G-sized stack, world2 in two smaller stacks)
+    plus BN running-buffer aggregation across ranks must agree.
+
+    Tolerance is aggregate, not elementwise: torch's CPU conv is not
+    bitwise reproducible across *batch sizes* (oneDNN picks blocking by
+    shape, so the same sample through a G*B=32 stack vs a G*B=16 stack
+    differs at ~1e-7), and a ReLU boundary flip on the second local step
+    turns that into isolated ~1e-2 spikes in one client's delta.  Measured
+    baseline: mean ~6e-6, max ~4e-3.  A real layout bug (wrong client
+    set, wrong weighting, BN buffers folded to the wrong slot) moves the
+    MEAN by orders of magnitude."""
+    over = {"model__name": "resnet18", "model__image_size": 32,
+            "data__samples_per_client": 16, "train__batch_size": 8,
+            "train__max_local_steps": 2, "rounds": 1}
+    d1, d2 = str(tmp_path / "w1"), str(tmp_path / "w2")
+    os.makedirs(d1), os.makedirs(d2)
+    _run_world(1, d1, over=over)
+    _run_world(2, d2, over=over)
+    diff = (_final_weights(d1) - _final_weights(d2)).abs()
+    assert float(diff.mean()) < 1e-4 and float(diff.max()) < 5e-2
 
 
 def test_sparse_transport_matches_dense(tmp_path):
