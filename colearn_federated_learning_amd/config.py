@@ -39,6 +39,13 @@ class ModelConfig:
     in_channels: int = 3
     image_size: int = 32
     dropout: float = 0.1  # bert_base only (0 => deterministic paths)
+    # bert_base architecture overrides (None => BertConfig defaults,
+    # i.e. the real 110M bert-base).  Small values give a CPU-testable
+    # transformer with the full stacked/MLM machinery.
+    hidden_size: Optional[int] = None
+    num_layers: Optional[int] = None
+    num_heads: Optional[int] = None
+    intermediate_size: Optional[int] = None
 
 
 @dataclass

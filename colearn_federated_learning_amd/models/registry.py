@@ -25,11 +25,25 @@ def build_model(model_cfg: ModelConfig, data_cfg: DataConfig,
             group_norm=dp_mode,  # DP-SGD is incompatible with batch stats
         )
     if name == "bert_base":
-        from .bert import BertConfig, BertForMaskedLM
+        from .bert import BertForMaskedLM
 
-        return BertForMaskedLM(BertConfig(
-            vocab_size=data_cfg.vocab_size,
-            max_position_embeddings=max(data_cfg.seq_len, 512),
-            dropout=model_cfg.dropout,
-        ))
+        return BertForMaskedLM(bert_config(model_cfg, data_cfg))
     raise ValueError(f"unknown model {name!r}")
+
+
+def bert_config(model_cfg: ModelConfig, data_cfg: DataConfig):
+    """One BertConfig for BOTH the sequential and stacked builders, so the
+    two engines always agree on architecture (equivalence tests depend on
+    this).  ModelConfig's optional dims override the bert-base defaults."""
+    from .bert import BertConfig
+
+    kw = dict(
+        vocab_size=data_cfg.vocab_size,
+        max_position_embeddings=max(data_cfg.seq_len, 512),
+        dropout=model_cfg.dropout,
+    )
+    for f in ("hidden_size", "num_layers", "num_heads", "intermediate_size"):
+        v = getattr(model_cfg, f)
+        if v is not None:
+            kw[f] = v
+    return BertConfig(**kw)

@@ -515,12 +515,8 @@ def build_stacked_model(model_cfg, data_cfg, G: int,
             small_input=model_cfg.image_size <= 64, group_norm=dp_mode,
         )
     if name == "bert_base":
-        from .bert import BertConfig
+        from .registry import bert_config
         from .stacked_bert import StackedBert
 
-        return StackedBert(G, BertConfig(
-            vocab_size=data_cfg.vocab_size,
-            max_position_embeddings=max(data_cfg.seq_len, 512),
-            dropout=model_cfg.dropout,
-        ))
+        return StackedBert(G, bert_config(model_cfg, data_cfg))
     return None

@@ -224,11 +224,11 @@ class _LayerNormFn(torch.autograd.Function):
             gx, gw, gb = require_ext().layernorm_bwd(
                 gy, x.contiguous(), weight.view(-1), mean, invstd, G
             )
-            if G > 1:
-                gw = gw.view(weight.shape)
-                gb = gb.view(weight.shape)
-            else:
-                gw, gb = gw.view(-1), gb.view(-1)
+            # weight.shape covers both plain [H] and stacked [G, H] —
+            # including the G==1 stack ([1, H]), whose grad must keep the
+            # leading slot dim (8-rank runs stack ONE client per rank)
+            gw = gw.view(weight.shape)
+            gb = gb.view(weight.shape)
         elif G > 1:
             h = x.shape[-1]
             rows = x.numel() // h
@@ -249,8 +249,8 @@ class _LayerNormFn(torch.autograd.Function):
             xf, gyf = x.float(), gy.float()
             xhat = (xf - mean[..., None]) * invstd[..., None]
             h = x.shape[-1]
-            gw = (gyf * xhat).reshape(-1, h).sum(dim=0)
-            gb = gyf.reshape(-1, h).sum(dim=0)
+            gw = (gyf * xhat).reshape(-1, h).sum(dim=0).view(weight.shape)
+            gb = gyf.reshape(-1, h).sum(dim=0).view(weight.shape)
             wg = gyf * weight
             gx = invstd[..., None] * (
                 wg - wg.mean(dim=-1, keepdim=True)

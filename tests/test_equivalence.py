@@ -23,7 +23,8 @@ def _make_cfg(tmpdir: str, **over):
         checkpoint_dir=os.path.join(tmpdir, "ck"), checkpoint_every=0,
     )
     base.update(over)  # tests may override the harness defaults too
-    return driver_config(1).override(**base)
+    idx = base.pop("_base", 1)  # which driver config to start from
+    return driver_config(idx).override(**base)
 
 
 def _worker(rank: int, world: int, tmpdir: str, port: int, over_json: str):
@@ -135,6 +136,26 @@ def test_world2_resnet_stacked(tmp_path):
     _run_world(2, d2, over=over)
     diff = (_final_weights(d1) - _final_weights(d2)).abs()
     assert float(diff.mean()) < 1e-4 and float(diff.max()) < 5e-2
+
+
+def test_world2_bert_stacked(tmp_path):
+    """Tiny-BERT layout-invariance, incl. the G=1 stack (8-rank layouts
+    put ONE client per rank; a [1,H] grouped-LN weight once lost its slot
+    dim in backward there).  Dropout 0 => deterministic; the transformer
+    CPU path is reproducible across batch sizes, so tolerance is tight."""
+    over = {"_base": 4, "model__dropout": 0.0,
+            "model__hidden_size": 64, "model__num_layers": 2,
+            "model__num_heads": 2, "model__intermediate_size": 128,
+            "data__vocab_size": 512, "data__samples_per_client": 16,
+            "train__batch_size": 8, "train__max_local_steps": 2,
+            "train__dtype": "float32", "rounds": 1,
+            "num_clients": 4, "num_evaluators": 1}
+    d1, d2 = str(tmp_path / "w1"), str(tmp_path / "w2")
+    os.makedirs(d1), os.makedirs(d2)
+    _run_world(1, d1, over=over)
+    _run_world(2, d2, over=over)
+    torch.testing.assert_close(_final_weights(d1), _final_weights(d2),
+                               rtol=1e-4, atol=1e-5)
 
 
 def test_sparse_transport_matches_dense(tmp_path):

@@ -1,5 +1,6 @@
 """Norm layers vs stock PyTorch (the CPU oracle used for GPU kernels too)."""
 
+import pytest
 import torch
 
 from colearn_federated_learning_amd.ops.norms import (
@@ -57,3 +58,29 @@ def test_layernorm_matches_torch():
                                rtol=1e-4, atol=1e-4)
     torch.testing.assert_close(ours.bias.grad, ref.bias.grad,
                                rtol=1e-4, atol=1e-4)
+
+
+@pytest.mark.parametrize("G", [1, 3])
+def test_grouped_layernorm_slot_dim(G):
+    """Grouped (stacked-client) LN: weight [G,H] must get a [G,H] grad —
+    including G==1, where the slot dim once got squeezed away (the 8-rank
+    layout stacks one client per rank)."""
+    from colearn_federated_learning_amd.ops.norms import _LayerNormFn
+
+    torch.manual_seed(0)
+    B, S, H = 2, 5, 16
+    x = torch.randn(G * B, S, H, requires_grad=True)
+    w = torch.randn(G, H, requires_grad=True)
+    b = torch.randn(G, H, requires_grad=True)
+    y, _, _ = _LayerNormFn.apply(x, w, b, 1e-5, G)
+    y.backward(torch.randn_like(y))
+    assert w.grad.shape == (G, H) and b.grad.shape == (G, H)
+    # per-slot oracle
+    x3 = x.detach().reshape(G, B, S, H)
+    for g in range(G):
+        xr = x3[g].clone().requires_grad_(True)
+        wr = w.detach()[g].clone().requires_grad_(True)
+        br = b.detach()[g].clone().requires_grad_(True)
+        yr = torch.nn.functional.layer_norm(xr, (H,), wr, br, 1e-5)
+        torch.testing.assert_close(
+            yr, y.detach().reshape(G, B, S, H)[g], rtol=1e-4, atol=1e-5)
