@@ -30,3 +30,29 @@ def test_bench_json_contract():
     assert d["data"] == "synthetic"
     assert "ResNet-18" in d["metric"]
     assert d["config"]["model"] == "resnet18"
+
+
+def test_bench_world2_torchrun():
+    """The driver's exact N>1 launch shape (torch.distributed.run with 2
+    ranks over gloo on CPU): rank 0 must emit one JSON line with
+    n_gpus=2 and the whole-job aggregate value."""
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port), "bench.py", "--gpus", "2",
+         "--steps", "1", "--warmup", "0", "--samples-per-client", "32",
+         "--set", "num_clients=2", "--set", "train__batch_size=16"],
+        cwd=ROOT, capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [ln for ln in out.stdout.strip().splitlines()
+             if ln.startswith("{")]
+    assert len(lines) == 1, out.stdout[-2000:]  # ONE line, rank 0 only
+    d = json.loads(lines[0])
+    assert d["n_gpus"] == 2 and d["value"] > 0
+    assert "dp2" in d["config"]["parallelism"]
