@@ -82,3 +82,22 @@ def test_config_override_unknown_key_raises():
         driver_config(2).override(samples_per_client=8)  # missing data__
     with pytest.raises(KeyError):
         driver_config(2).override(train__learning_rate=0.1)
+
+
+def test_load_yaml_file(tmp_path):
+    """load_yaml: a config written as YAML to disk loads back identically
+    (the CLI's --config file path)."""
+    import yaml
+
+    from colearn_federated_learning_amd.config import (
+        driver_config,
+        load_yaml,
+    )
+
+    cfg = driver_config(4).override(rounds=7, train__lr=0.125,
+                                    model__num_layers=2)
+    p = tmp_path / "cfg.yaml"
+    p.write_text(yaml.safe_dump(cfg.to_dict()))
+    back = load_yaml(str(p))
+    assert back.to_dict() == cfg.to_dict()
+    assert back.rounds == 7 and back.model.num_layers == 2
