@@ -12,7 +12,6 @@ so a networked MQTT transport could be slotted in behind the same API.
 
 from __future__ import annotations
 
-import fnmatch
 import threading
 from dataclasses import dataclass, field
 from typing import Any, Callable, Dict, List, Tuple
@@ -49,15 +48,17 @@ class MessageBus:
 
     @staticmethod
     def _match(pattern: str, topic: str) -> bool:
-        # MQTT wildcards: '+' one level, '#' remaining levels.
-        glob = pattern.replace("+", "*").replace("#", "*")
+        # MQTT wildcards, level-wise: '+' matches exactly ONE level,
+        # '#' (last level only) matches all remaining levels.
         if pattern == topic:
             return True
-        if "#" in pattern:
-            prefix = pattern.split("#", 1)[0]
-            return topic.startswith(prefix) or fnmatch.fnmatch(topic, glob)
         pp, tp = pattern.split("/"), topic.split("/")
-        if len(pp) != len(tp):
+        if pp and pp[-1] == "#":
+            pp = pp[:-1]
+            if len(tp) < len(pp):
+                return False
+            tp = tp[: len(pp)]
+        elif len(pp) != len(tp):
             return False
         return all(p == "+" or p == t for p, t in zip(pp, tp))
 

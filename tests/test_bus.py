@@ -43,3 +43,17 @@ def test_collect():
     bus.publish("colearn/metrics/0", {"x": 1}, retain=True)
     avail = bus.collect("colearn/availability")
     assert list(avail.values()) == [{"client_id": 0}]
+
+
+def test_mqtt_wildcard_semantics():
+    """'+' is exactly one level; '#' covers all remaining levels and may
+    follow '+' levels (real MQTT semantics, not glob)."""
+    m = MessageBus._match
+    assert m("a/+/b", "a/x/b")
+    assert not m("a/+/b", "a/x/y/b")      # '+' must not cross levels
+    assert m("a/#", "a/b/c/d")
+    assert m("a/+/b/#", "a/x/b/c/d")
+    assert not m("a/+/b/#", "a/x/y/b/c")  # '+' crossed a level
+    assert m("a/+/#", "a/x/b")
+    assert not m("a/+/#", "a")            # '+' needs its level present
+    assert m("#", "anything/at/all")
