@@ -38,6 +38,10 @@ COMBOS = [
          train__local_epochs=2),
     # bert + topk + fedprox (stacked transformer with sparsified deltas)
     dict(TINY_BERT, agg__topk_ratio=0.1, train__fedprox_mu=0.01),
+    # DP + bert: no stacked per-sample path for transformers, so this
+    # exercises the sequential vmap per-sample fallback end-to-end
+    dict(TINY_BERT, dp__enabled=True, dp__noise_multiplier=0.3,
+         num_clients=2, train__max_local_steps=1, rounds=1),
     # resnet + evaluators + drops + sequential path (stacking off)
     {"_base": 2, "model__image_size": 32, "num_evaluators": 2,
      "drop_client_prob": 0.3, "train__use_stacked": False,
