@@ -77,6 +77,10 @@ def make_client_dataset(
     if data_cfg.dataset == "text_synth":
         s = data_cfg.seq_len
         v = data_cfg.vocab_size
+        if v < 104:
+            raise ValueError(
+                f"text_synth needs vocab_size >= 104 (the [MASK]-style "
+                f"token id is 103), got {v}")
         lo = min(1000, v // 2)  # reserve low ids for special tokens
         ids = torch.randint(lo, v, (n, s), generator=gen)
         # MLM masking: select ~mlm_probability positions; label = original id

@@ -70,3 +70,20 @@ def test_text_dataset_mlm():
     assert 0.05 < frac < 0.3
     # masked positions carry the mask token in inputs
     assert (ds.x[masked] == 103).all()
+
+
+def test_text_synth_vocab_floor():
+    """vocab_size must cover the [MASK] token id (103) — loud error, not
+    out-of-range embedding indices downstream."""
+    import pytest
+
+    from colearn_federated_learning_amd.config import driver_config
+    from colearn_federated_learning_amd.data.synthetic import (
+        make_client_dataset,
+    )
+
+    cfg = driver_config(4).override(data__vocab_size=50,
+                                    data__samples_per_client=4)
+    with pytest.raises(ValueError, match="vocab_size"):
+        make_client_dataset(cfg.data, cfg.model, 0,
+                            torch.zeros(4, dtype=torch.long))
