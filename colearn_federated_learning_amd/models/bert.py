@@ -35,6 +35,12 @@ class BertConfig:
     dropout: float = 0.1
     layer_norm_eps: float = 1e-12
 
+    def __post_init__(self) -> None:
+        if self.hidden_size % self.num_heads:
+            raise ValueError(
+                f"hidden_size {self.hidden_size} must be divisible by "
+                f"num_heads {self.num_heads}")
+
 
 class _Linear(nn.Module):
     def __init__(self, cin: int, cout: int) -> None:
