@@ -6,8 +6,10 @@ The .so lands next to the Python wrappers
 (colearn_federated_learning_amd/ops/_colearn_hip*.so) so repo snapshots
 carry it to GPU boxes.  Sources are native HIP written for CDNA4; torch's
 build machinery mechanically copies them to ``*_hip.hip`` (its hipify
-pass), but there is no CUDA in them to translate — the copies are
-byte-identical modulo the rename and are gitignored.
+pass).  There is no CUDA in them to translate — the pass only adds the
+``hip_runtime.h`` include and rewrites the ``<<<>>>`` launches to the
+equivalent ``hipLaunchKernelGGL`` spelling — and the generated copies
+are gitignored.
 """
 
 import os
