@@ -213,16 +213,23 @@ class GradSampleEngine:
 
     def _make_fwd(self, rule):
         def fwd_hook(mod, inputs, output):
-            mod._gs_input = inputs[0].detach()
-            # the GN rule masks grad_output by the post-ReLU output
-            mod._gs_output = (output.detach()
-                              if isinstance(output, torch.Tensor)
-                              else None)
+            # per-CALL saves live in this closure (not on the module), so
+            # a weight-shared module invoked twice in one forward gets the
+            # right input paired with each call's grad; per-call grads
+            # ACCUMULATE into grad_sample (cleared each step).
+            x_saved = inputs[0].detach()
+            out_saved = (output.detach()
+                         if isinstance(output, torch.Tensor) else None)
             if isinstance(output, torch.Tensor) and output.requires_grad:
                 def grad_cb(gy, mod=mod):
-                    gs = rule(mod, mod._gs_input, gy.detach())
+                    # the GN rule masks grad_output by the post-ReLU output
+                    mod._gs_input = x_saved
+                    mod._gs_output = out_saved
+                    gs = rule(mod, x_saved, gy.detach())
                     for name, g in gs.items():
-                        getattr(mod, name).grad_sample = g
+                        p = getattr(mod, name)
+                        prev = getattr(p, "grad_sample", None)
+                        p.grad_sample = g if prev is None else prev + g
 
                 output.register_hook(grad_cb)
 

@@ -68,3 +68,38 @@ def test_gelu_vmap_rule():
     xf = x.clone().requires_grad_(True)
     colearn_gelu(xf).sum().backward()
     torch.testing.assert_close(g, xf.grad)
+
+
+def test_grad_sample_shared_module():
+    """A weight-shared module applied twice must produce the SUM of both
+    calls' per-sample grads (per-call closure saves + accumulation)."""
+    import torch.nn as nn
+
+    from colearn_federated_learning_amd.ops.grad_sample import (
+        GradSampleEngine,
+    )
+
+    torch.manual_seed(0)
+    lin = nn.Linear(4, 4, bias=False)
+
+    class Twice(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.lin = lin
+
+        def forward(self, x):
+            return self.lin(torch.relu(self.lin(x)))
+
+    m = Twice()
+    x = torch.randn(3, 4)
+    eng = GradSampleEngine(m)
+    with eng:
+        m(x).sum().backward()
+        gs = lin.weight.grad_sample.clone()
+        eng.clear()
+    # oracle: per-sample autograd
+    for i in range(3):
+        m.zero_grad()
+        m(x[i : i + 1]).sum().backward()
+        torch.testing.assert_close(gs[i], lin.weight.grad, rtol=1e-5,
+                                   atol=1e-6)
