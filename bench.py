@@ -125,6 +125,9 @@ def main() -> None:
             "warmup": args.warmup,
             "ms_per_step": elapsed / args.steps * 1000.0,
             "higher_is_better": True,
+            # the driver configs fix TOTAL work (num_clients and shard
+            # sizes don't grow with N): adding GPUs splits the same
+            # clients across ranks => strong scaling
             "scaling": "strong",
             "vs_baseline": None,
             "rounds_to_target": rounds_to_target,
