@@ -41,9 +41,13 @@ def save_checkpoint(
     tmp = path + ".tmp"
     torch.save(payload, tmp)
     os.replace(tmp, path)
+    # LATEST is a byte copy of the round file (serialize once), written
+    # with the same tmp+rename dance so readers never see a torn file
+    import shutil
+
     latest = os.path.join(path_dir, LATEST)
     tmp = latest + ".tmp"
-    torch.save(payload, tmp)
+    shutil.copyfile(path, tmp)
     os.replace(tmp, latest)
     return path
 
