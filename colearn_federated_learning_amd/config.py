@@ -34,7 +34,7 @@ class DataConfig:
 
 @dataclass
 class ModelConfig:
-    name: str = "logreg"  # logreg | resnet18 | resnet50 | bert_base
+    name: str = "logreg"  # logreg | mnist_cnn | resnet18 | resnet50 | bert_base
     num_classes: int = 10
     in_channels: int = 3
     image_size: int = 32

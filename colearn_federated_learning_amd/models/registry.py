@@ -14,6 +14,14 @@ def build_model(model_cfg: ModelConfig, data_cfg: DataConfig,
             model_cfg.in_channels * model_cfg.image_size * model_cfg.image_size
         )
         return LogisticRegression(in_features, model_cfg.num_classes)
+    if name == "mnist_cnn":
+        from .cnn import MnistCNN
+
+        return MnistCNN(
+            in_channels=model_cfg.in_channels,
+            num_classes=model_cfg.num_classes,
+            image_size=model_cfg.image_size,
+        )
     if name in ("resnet18", "resnet50"):
         from .resnet import resnet18, resnet50
 
