@@ -38,6 +38,11 @@ COMBOS = [
          train__local_epochs=2),
     # bert + topk + fedprox (stacked transformer with sparsified deltas)
     dict(TINY_BERT, agg__topk_ratio=0.1, train__fedprox_mu=0.01),
+    # DP on the reference's MNIST CNN: conv+linear per-sample hook rules
+    {"_base": 1, "model__name": "mnist_cnn", "model__in_channels": 1,
+     "model__image_size": 28, "dp__enabled": True,
+     "dp__noise_multiplier": 0.3, "rounds": 1, "num_clients": 2,
+     "train__max_local_steps": 1},
     # DP + bert: no stacked per-sample path for transformers, so this
     # exercises the sequential vmap per-sample fallback end-to-end
     dict(TINY_BERT, dp__enabled=True, dp__noise_multiplier=0.3,
