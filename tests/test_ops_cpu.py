@@ -103,3 +103,30 @@ def test_grad_sample_shared_module():
         m(x[i : i + 1]).sum().backward()
         torch.testing.assert_close(gs[i], lin.weight.grad, rtol=1e-5,
                                    atol=1e-6)
+
+
+def test_linear_and_matmul_vmap_rules():
+    from torch.func import grad, vmap
+
+    from colearn_federated_learning_amd.ops.gemm import (
+        colearn_linear,
+        colearn_matmul,
+    )
+
+    torch.manual_seed(1)
+    w = torch.randn(3, 5)
+    xs = torch.randn(4, 5)
+
+    def f(xi, wi):
+        return colearn_linear(xi.unsqueeze(0), wi).sum()
+
+    gw = vmap(grad(f, argnums=1), in_dims=(0, None))(xs, w)
+    for i in range(4):
+        wi = w.clone().requires_grad_(True)
+        colearn_linear(xs[i : i + 1], wi).sum().backward()
+        torch.testing.assert_close(gw[i], wi.grad)
+
+    a = torch.randn(4, 2, 6)
+    b = torch.randn(4, 6, 3)
+    out = vmap(colearn_matmul)(a, b)
+    torch.testing.assert_close(out, a @ b)
