@@ -19,7 +19,10 @@ from typing import Any, Dict, List, Optional
 class DataConfig:
     """Synthetic dataset shape + federated sharding."""
 
-    dataset: str = "mnist_synth"  # mnist_synth | cifar10_synth | text_synth
+    dataset: str = "mnist_synth"  # mnist_synth | cifar10_synth | text_synth | file
+    # dataset="file": torch.save'd {"x": [N,C,H,W] float, "y": [N] int64}
+    # partitioned across clients by the same IID/Dirichlet axes (data/files.py)
+    data_path: Optional[str] = None
     num_classes: int = 10
     samples_per_client: int = 1024
     # sharding: "iid" or "dirichlet" (non-IID label skew, alpha controls skew)

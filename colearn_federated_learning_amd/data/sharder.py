@@ -56,6 +56,12 @@ def dirichlet_shards(cfg: DataConfig, num_clients: int) -> List[torch.Tensor]:
 
 
 def make_shards(cfg: DataConfig, num_clients: int) -> List[torch.Tensor]:
+    if cfg.dataset == "file":
+        # real-dataset path: "shards" are per-client sample INDICES into
+        # the file (data/files.py); make_client_dataset slices by them
+        from .files import partition_file_indices
+
+        return partition_file_indices(cfg, num_clients)
     if cfg.dataset == "text_synth":
         # text shards have no label skew axis; each client gets a seeded id
         # sequence of the right length (labels unused, generator keys on cid)

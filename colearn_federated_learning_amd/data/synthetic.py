@@ -62,6 +62,12 @@ def make_client_dataset(
     across clients, exactly the non-IID axis the reference's partitioner
     controls.
     """
+    if data_cfg.dataset == "file":
+        from .files import load_file_dataset
+
+        x, y = load_file_dataset(data_cfg.data_path)
+        idx = labels  # the sharder returned sample indices for "file"
+        return ClientShard(x[idx].clone(), y[idx].clone(), client_id)
     n = labels.shape[0]
     gen = torch.Generator().manual_seed(
         stable_hash(data_cfg.seed, "inputs", client_id)
