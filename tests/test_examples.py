@@ -11,6 +11,7 @@ ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 @pytest.mark.parametrize("script", [
     "run_federated.py", "dp_federated.py", "bert_mlm.py",
+    "file_dataset.py",
 ])
 def test_example_runs(script):
     out = subprocess.run(
