@@ -75,7 +75,7 @@ class StackedFlatParams:
         dgrad weight in tiled-transpose passes — replacing both the
         per-layer autocast casts (~5% of step time, round 1) and the
         ~20 per-call permute+contiguous copies the dgrad path used to
-        issue each step (~10% of step time — gpurun_out/eager_attrib.txt).
+        issue each step (~10% of step time — tools/gpu_debug/eager_attrib.py).
         """
         from ..models.stacked import SBatchNorm2d, SConv2d, SLinear
         from ..models.stacked_bert import SLayerNorm

@@ -2,7 +2,7 @@
 
 VERDICT r1 next #4: BERT was the least-native path — 8 clients trained
 SEQUENTIALLY, 222 GEMM launches per client step, 65% of step time in
-small torch kernels (gpurun_out/bert_prof.txt).  Stacking trains all G
+small torch kernels (tools/gpu_debug attribution runs).  Stacking trains all G
 clients at once with per-client weights:
 
 - every projection is ONE batched GEMM over G clients

@@ -3,7 +3,7 @@
 //
 // Replaces the torch chain (score bmm + scale + fp32 softmax + dropout +
 // context bmm + 3 reshape copies, fwd AND bwd — measured ~65% of BERT
-// step time as small kernels, gpurun_out/bert_prof2.txt): the whole
+// step time as small kernels, tools/gpu_debug profiling run): the whole
 // head-attention runs out of LDS with MFMA tiles, reading q/k/v straight
 // from the fused qkv projection layout [N, S, 3, H, dh] and writing ctx
 // in [N, S, H*dh] — no transpose materialization anywhere.

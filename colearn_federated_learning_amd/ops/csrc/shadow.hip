@@ -3,7 +3,7 @@
 // One launch per train step replaces (a) the monolithic fp32->bf16 flat
 // cast AND (b) ~20 per-conv weight permute+contiguous copies that the
 // dgrad path issued every step (measured 1.1 ms/step of elementwise /
-// copy launches — gpurun_out/eager_attrib.txt).
+// copy launches — tools/gpu_debug/eager_attrib.py run).
 //
 // Block ranges:
 //   [0, cast_blocks)   : flat fp32 -> bf16 cast, 8 elems/thread
