@@ -508,6 +508,11 @@ def stackable(cfg: FederatedConfig, clients: List[FederatedClient]) -> bool:
     shard size (lock-step batches) and the model family is implemented."""
     if not cfg.train.use_stacked:
         return False
+    if cfg.straggler_timeout_s is not None:
+        # the per-client mid-fit deadline only exists on the sequential
+        # path; in a lock-step stack no client can straggle relative to
+        # its rank-mates, so honor the knob by falling back
+        return False
     if cfg.model.name not in ("logreg", "resnet18", "resnet50",
                               "bert_base"):
         return False

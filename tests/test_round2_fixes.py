@@ -229,3 +229,22 @@ def test_stacked_dp_subsampled():
     torch.testing.assert_close(s_stack.global_flat, s_seq.global_flat,
                                rtol=1e-3, atol=5e-5)
     assert s_stack.stacked._engines[2].gs_engine is not None
+
+
+def test_straggler_timeout_disables_stacking():
+    """straggler_timeout_s must be HONORED: the per-client mid-fit
+    deadline only exists on the sequential path, so setting it falls the
+    engine back from the lock-step stack."""
+    from colearn_federated_learning_amd import driver_config
+    from colearn_federated_learning_amd.federated.stacked import stackable
+
+    cfg = driver_config(1).override(num_clients=2)
+
+    class _FakeClient:
+        def __init__(self):
+            self.shard = [0] * 8
+
+    clients = [_FakeClient(), _FakeClient()]
+    assert stackable(cfg, clients)
+    cfg2 = cfg.override(straggler_timeout_s=5.0)
+    assert not stackable(cfg2, clients)
