@@ -74,3 +74,24 @@ def test_end_to_end_federated_run_on_file(tmp_path):
     assert len(hist) == 2
     losses = [r["mean_client_loss"] for r in hist]
     assert all(l == l for l in losses) and losses[-1] < losses[0]
+
+
+def test_file_dataset_world2_equals_world1(tmp_path):
+    """Layout invariance holds for file-loaded data too: the partition is
+    global-seeded, each rank slices only its own clients' samples."""
+    import os
+
+    import tests.test_equivalence as te
+
+    p = tmp_path / "d.pt"
+    _write_dataset(p, n=400)
+    over = {"data__dataset": "file", "data__data_path": str(p),
+            "data__samples_per_client": 40, "model__in_channels": 1,
+            "model__image_size": 28, "rounds": 2}
+    d1, d2 = str(tmp_path / "w1"), str(tmp_path / "w2")
+    os.makedirs(d1), os.makedirs(d2)
+    te._run_world(1, d1, over=over)
+    te._run_world(2, d2, over=over)
+    torch.testing.assert_close(te._final_weights(d1),
+                               te._final_weights(d2),
+                               rtol=1e-4, atol=1e-5)
