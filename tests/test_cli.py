@@ -61,3 +61,16 @@ def test_server_plus_client_processes(tmp_path):
     ck = torch.load(tmp_path / "ck" / "colearn_latest.pt",
                     map_location="cpu", weights_only=False)
     assert ck["round"] == 1
+
+
+def test_module_entry_point(tmp_path):
+    """python -m colearn_federated_learning_amd (no .cli suffix) works."""
+    p = _run_proc([
+        sys.executable, "-m", "colearn_federated_learning_amd",
+        "server", "--driver-config", "1",
+        "rounds=1", "data.samples_per_client=32",
+        f"checkpoint_dir={tmp_path / 'ck'}",
+    ])
+    out, _ = p.communicate(timeout=300)
+    assert p.returncode == 0, out
+    assert "session done" in out
