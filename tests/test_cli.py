@@ -74,3 +74,19 @@ def test_module_entry_point(tmp_path):
     out, _ = p.communicate(timeout=300)
     assert p.returncode == 0, out
     assert "session done" in out
+
+
+def test_gpu_tools_importable():
+    """The multi-GPU tools must at least parse/import on CPU so they
+    don't rot before the day an 8-GPU node runs them."""
+    import importlib.util as ilu
+
+    for rel in ("tools/comm_sweep.py", "tools/rccl_validate.py",
+                "tools/gpu_debug/g1_check.py"):
+        path = os.path.join(os.path.dirname(os.path.dirname(
+            os.path.abspath(__file__))), rel)
+        spec = ilu.spec_from_file_location("_tool_mod", path)
+        mod = ilu.module_from_spec(spec)
+        # import-time side effects only; main() guarded by __main__
+        spec.loader.exec_module(mod)
+        assert hasattr(mod, "main")
