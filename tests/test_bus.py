@@ -57,3 +57,37 @@ def test_mqtt_wildcard_semantics():
     assert m("a/+/#", "a/x/b")
     assert not m("a/+/#", "a")            # '+' needs its level present
     assert m("#", "anything/at/all")
+
+
+def test_mqtt_match_property():
+    """_match vs an independent recursive MQTT matcher on random
+    topic/pattern pairs."""
+    import itertools
+    import random
+
+    def oracle(pp, tp):
+        # recursive MQTT semantics: '+' one level, trailing '#' any tail
+        if not pp:
+            return not tp
+        if pp[0] == "#":
+            return len(pp) == 1
+        if not tp:
+            return False
+        if pp[0] == "+" or pp[0] == tp[0]:
+            return oracle(pp[1:], tp[1:])
+        return False
+
+    rng = random.Random(0)
+    atoms = ["a", "b", "colearn", "availability", "3"]
+    wild = atoms + ["+"]
+    for _ in range(2000):
+        tlen = rng.randint(1, 4)
+        plen = rng.randint(1, 4)
+        topic = "/".join(rng.choice(atoms) for _ in range(tlen))
+        parts = [rng.choice(wild) for _ in range(plen)]
+        if rng.random() < 0.4:
+            parts[-1] = "#"
+        pattern = "/".join(parts)
+        want = oracle(pattern.split("/"), topic.split("/"))
+        got = MessageBus._match(pattern, topic)
+        assert got == want, (pattern, topic, got, want)
