@@ -430,6 +430,12 @@ class StackedEnginePool:
     a sampling/drop pattern needs them and cached (graph capture and
     stacked-weight allocation are one-time costs)."""
 
+    # lazy engines kept beyond the full one: each holds G x weights +
+    # optimizer state (+ DP graph pools with per-sample slabs), so an
+    # unbounded cache under drop-injection (a new active count most
+    # rounds) could exhaust HBM at 64-client scale
+    MAX_LAZY = 2
+
     def __init__(self, cfg: FederatedConfig,
                  clients: List[FederatedClient],
                  device: torch.device) -> None:
@@ -439,12 +445,22 @@ class StackedEnginePool:
         self._engines: Dict[int, StackedEngine] = {
             len(clients): StackedEngine(cfg, len(clients), device)
         }
+        self._lazy_lru: List[int] = []  # non-full engine keys, LRU order
 
     def get(self, n_active: int) -> StackedEngine:
         eng = self._engines.get(n_active)
         if eng is None:
             eng = StackedEngine(self.cfg, n_active, self.device)
             self._engines[n_active] = eng
+        if n_active != len(self.all_clients):
+            if n_active in self._lazy_lru:
+                self._lazy_lru.remove(n_active)
+            self._lazy_lru.append(n_active)
+            while len(self._lazy_lru) > self.MAX_LAZY:
+                evict = self._lazy_lru.pop(0)
+                del self._engines[evict]
+                if self.device.type == "cuda":
+                    torch.cuda.empty_cache()
         return eng
 
     @property

@@ -139,3 +139,39 @@ def test_stacked_bert_param_order_matches_template():
         tnames, snames, tmpl.named_parameters(), stk.named_parameters()
     ):
         assert sp.shape == (3,) + tuple(tp.shape), (tn, sn)
+
+
+def test_engine_pool_evicts_lazy_engines():
+    """Drop-injection creates a new active count most rounds; the pool
+    must bound cached lazy engines (each holds G x weights + optimizer
+    state and, under DP, graph pools) while keeping the full engine."""
+    from colearn_federated_learning_amd import driver_config
+    from colearn_federated_learning_amd.federated.client import (
+        FederatedClient,
+    )
+    from colearn_federated_learning_amd.data.sharder import make_shards
+    from colearn_federated_learning_amd.data.synthetic import (
+        make_client_dataset,
+    )
+    from colearn_federated_learning_amd.federated.stacked import (
+        StackedEnginePool,
+    )
+
+    cfg = driver_config(1).override(num_clients=6,
+                                    data__samples_per_client=8)
+    dev = torch.device("cpu")
+    shards = make_shards(cfg.data, 6)
+    clients = [
+        FederatedClient(i, cfg, make_client_dataset(cfg.data, cfg.model,
+                                                    i, shards[i]), dev)
+        for i in range(6)
+    ]
+    pool = StackedEnginePool(cfg, clients, dev)
+    for n in (5, 4, 3, 2):
+        pool.get(n)
+    keys = set(pool._engines)
+    assert 6 in keys  # full engine never evicted
+    assert len(keys - {6}) == pool.MAX_LAZY == 2
+    assert keys - {6} == {3, 2}  # LRU: oldest lazy engines gone
+    pool.get(3)  # re-request: rebuilt and cached again
+    assert 3 in pool._engines
