@@ -29,7 +29,7 @@ import torch
 import torch.distributed as dist
 
 # repo root on sys.path when launched by file path (torchrun)
-sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(__import__("os").path.abspath(__file__))))
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
 def bench_point(flat: torch.Tensor, bucket_bytes: int, streams: int,
