@@ -95,3 +95,14 @@ def test_file_dataset_world2_equals_world1(tmp_path):
     torch.testing.assert_close(te._final_weights(d1),
                                te._final_weights(d2),
                                rtol=1e-4, atol=1e-5)
+
+
+def test_whole_dataset_split(tmp_path):
+    """samples_per_client=0 => split the whole file evenly."""
+    p = tmp_path / "d.pt"
+    _write_dataset(p, n=100)
+    cfg = _cfg(p, data__samples_per_client=0)
+    parts = partition_file_indices(cfg.data, 4)
+    assert all(len(q) == 25 for q in parts)
+    allidx = torch.cat(parts)
+    assert torch.unique(allidx).numel() == 100
