@@ -20,12 +20,14 @@ def _write_dataset(path, n=200, classes=10):
 
 
 def _cfg(path, **over):
-    return driver_config(1).override(
+    base = dict(
         data__dataset="file", data__data_path=str(path),
         data__samples_per_client=40, model__in_channels=1,
         model__image_size=28, rounds=2, num_clients=4,
-        train__batch_size=8, **over,
+        train__batch_size=8,
     )
+    base.update(over)
+    return driver_config(1).override(**base)
 
 
 def test_iid_partition_disjoint_and_deterministic(tmp_path):
