@@ -7,10 +7,12 @@ the global model, evaluate, checkpoint, log — repeat.
 
 MI355X re-architecture: clients are virtual and sharded across RCCL ranks
 (client cid lives on rank cid % world).  Each rank trains its sampled
-clients sequentially on its GPU out of ONE flat parameter buffer, folds
-each client's delta into a flat fp32 accumulator (optionally top-k
-compressed per client), and a single bucketed weighted all-reduce replaces
-the reference's per-client websocket model shipping.  Everything that feeds
+clients SIMULTANEOUSLY in a stacked engine (per-client weight slices,
+grid.z = client — federated/stacked.py; sequential fallback for
+unstackable configs), folds each client's delta into a flat fp32
+accumulator (optionally top-k compressed per client), and a single
+bucketed weighted all-reduce replaces the reference's per-client
+websocket model shipping.  Everything that feeds
 cross-rank state (sampling, DP noise, drop injection) is computed from
 rank-consistent RNG so a run is equivalent on any GPU count
 (SURVEY.md §4b.3 — tests/test_equivalence.py).
